@@ -1,0 +1,348 @@
+"""Core runtime: Estimator/Model orchestration for SPMD fit/transform.
+
+Reference equivalent: core.py (1,968 lines) — `_CumlCaller._call_cuml_fit_func`
+turns a Spark DataFrame into a barrier-mode mapInPandas job with one task per
+GPU, runs a fit closure under a NCCL context, and collects model attributes
+(reference core.py:742-1013). The MI355X-native redesign removes the
+driver/executor split entirely: the framework is SPMD — every rank (one
+process per GPU, launched by torchrun or run single-process) executes
+`Estimator.fit(df)` on its local shard, the solvers reduce over RCCL, and
+every rank ends up holding the full (replicated, small) model. That keeps the
+reference's algorithmic structure (PartitionDescriptor allGather, replicated
+model state, rank-0 result authority) while dropping Spark's barrier-stage /
+Arrow-UDF machinery, which has no MI355X equivalent worth recreating.
+
+Hooks per algorithm (mirroring reference core.py:1283/1695's
+`_get_cuml_fit_func` / `_get_cuml_transform_func` contract):
+- `Estimator._fit_array(X, y, pdesc, params) -> dict of model attributes`
+- `Model._transform_array(X) -> column(s) to append`
+"""
+
+from __future__ import annotations
+
+import json
+import os
+import pickle
+from abc import abstractmethod
+from typing import Any, Callable, Dict, Iterator, List, Optional, Sequence, Tuple, Union
+
+import numpy as np
+import torch
+
+from .data import DataFrame, extract_features, _is_sparse
+from .params import (
+    HasFeaturesCol,
+    HasFeaturesCols,
+    HasLabelCol,
+    HasPredictionCol,
+    HasVerbose,
+    Param,
+    Params,
+    _NativeParams,
+)
+from .parallel.context import Comm, PartitionDescriptor, get_comm
+from .utils import get_logger
+
+
+class _FitContext:
+    """Everything the per-algorithm fit hook needs (the reference passes the
+    same via the params dict into the fit closure, core.py:983-994)."""
+
+    def __init__(self, comm: Comm, pdesc: PartitionDescriptor, device: torch.device):
+        self.comm = comm
+        self.pdesc = pdesc
+        self.device = device
+        # shared across the param maps of one fit call so a multi-model sweep
+        # reuses one data pass (reference single-pass fitMultiple,
+        # core.py:1177-1228 / regression.py:657-674)
+        self.cache: Dict[str, Any] = {}
+
+
+class Estimator(_NativeParams, HasVerbose):
+    """Base of all estimators (reference `_CumlEstimator`, core.py:1155)."""
+
+    def __init__(self) -> None:
+        super().__init__()
+
+    # -- hooks ------------------------------------------------------------
+    @abstractmethod
+    def _fit_array(
+        self,
+        X: Any,
+        y: Optional[Any],
+        ctx: _FitContext,
+        params: Dict[str, Any],
+    ) -> Dict[str, Any]:
+        """SPMD fit of the local shard; must return the model attribute dict
+        (identical on every rank after the final broadcast)."""
+
+    @abstractmethod
+    def _create_model(self, attrs: Dict[str, Any]) -> "Model":
+        """Build the Model object from fitted attributes (reference
+        `_create_pyspark_model`, core.py:1267)."""
+
+    def _is_supervised(self) -> bool:
+        return isinstance(self, HasLabelCol)
+
+    def _use_sparse(self, X: Any) -> bool:
+        return _is_sparse(X)
+
+    # -- fit orchestration -------------------------------------------------
+    def _extract_xy(self, df: DataFrame) -> Tuple[Any, Optional[np.ndarray]]:
+        features_col, features_cols = self._get_input_columns()
+        X = extract_features(df, features_col, features_cols, self._float32_inputs)
+        y = None
+        if self._is_supervised():
+            label_col = self.getOrDefault("labelCol")
+            if label_col not in df.columns:
+                raise ValueError(f"label column {label_col!r} not in dataframe")
+            y = np.asarray(df[label_col]).astype(
+                np.float32 if self._float32_inputs else np.float64
+            )
+        return X, y
+
+    def fit(self, df: DataFrame, params: Optional[Dict[Any, Any]] = None) -> "Model":
+        if params:
+            est = self.copy(params)
+            return est.fit(df)
+        return self._fit(df)
+
+    def _fit(self, df: DataFrame) -> "Model":
+        models = self._fit_internal(df, [{}])
+        return models[0]
+
+    def fitMultiple(
+        self, df: DataFrame, paramMaps: Sequence[Dict[Any, Any]]
+    ) -> Iterator[Tuple[int, "Model"]]:
+        """Fit all param maps against one data load (reference single-pass
+        fitMultiple, core.py:1177-1228); yields (index, model)."""
+        native_maps = [self._resolve_param_map(pm) for pm in paramMaps]
+        models = self._fit_internal(df, native_maps)
+
+        def _iter() -> Iterator[Tuple[int, Model]]:
+            for i, m in enumerate(models):
+                yield i, m
+
+        return _iter()
+
+    def _resolve_param_map(self, pm: Dict[Any, Any]) -> Dict[str, Any]:
+        """Translate a {Param: value} map into native param overrides."""
+        out: Dict[str, Any] = {}
+        mapping = self._param_mapping()
+        vmap = self._param_value_mapping()
+        for p, v in pm.items():
+            name = p.name if isinstance(p, Param) else str(p)
+            native = mapping.get(name, name)
+            if native in (None, ""):
+                continue
+            if native in vmap:
+                mv = vmap[native](v)
+                if mv is None:
+                    raise ValueError(f"Value {v!r} unsupported for {name}")
+                v = mv
+            out[native] = v
+        return out
+
+    def _fit_internal(self, df: DataFrame, param_maps: List[Dict[str, Any]]) -> List["Model"]:
+        comm = get_comm()
+        X, y = self._extract_xy(df)
+        nnz = X.nnz if _is_sparse(X) else None
+        pdesc = PartitionDescriptor.build(comm, X.shape[0], X.shape[1], nnz=nnz)
+        if pdesc.m == 0:
+            raise RuntimeError("Dataset is empty across all ranks.")
+        ctx = _FitContext(comm, pdesc, comm.device)
+        logger = get_logger(self.__class__)
+
+        models: List[Model] = []
+        for pm in param_maps:
+            params = dict(self._native_params)
+            params.update(pm)
+            attrs = self._fit_array(X, y, ctx, params)
+            model = self._create_model(attrs)
+            self._copyValues(model)
+            model._native_params = dict(params)
+            model._float32_inputs = self._float32_inputs
+            models.append(model)
+        if comm.device.type == "cuda":
+            torch.cuda.synchronize(comm.device)
+        return models
+
+    # -- persistence -------------------------------------------------------
+    def write(self) -> "_Writer":
+        return _Writer(self)
+
+    def save(self, path: str) -> None:
+        self.write().save(path)
+
+    @classmethod
+    def read(cls) -> "_Reader":
+        return _Reader(cls)
+
+    @classmethod
+    def load(cls, path: str):
+        return cls.read().load(path)
+
+
+class Model(_NativeParams, HasVerbose):
+    """Base of fitted models (reference `_CumlModel`, core.py:1379)."""
+
+    def __init__(self, **model_attributes: Any) -> None:
+        super().__init__()
+        self._model_attributes = model_attributes
+
+    def _get_model_attributes(self) -> Dict[str, Any]:
+        return self._model_attributes
+
+    # -- hooks ------------------------------------------------------------
+    @abstractmethod
+    def _transform_array(self, X: Any) -> Union[np.ndarray, Dict[str, np.ndarray]]:
+        """Compute the output column(s) for local rows; return either one
+        array (named by predictionCol/outputCol) or a dict name->array."""
+
+    def _out_col_name(self) -> str:
+        if isinstance(self, HasPredictionCol) or self.hasParam("predictionCol"):
+            return self.getOrDefault("predictionCol")
+        return self.getOrDefault("outputCol")
+
+    # -- transform orchestration ------------------------------------------
+    def transform(self, df: DataFrame) -> DataFrame:
+        """Append output columns to the local shard (reference
+        `_CumlModelWithColumns._transform`, core.py:1797: a pandas_udf over
+        the feature struct; here a direct device-batched call)."""
+        features_col, features_cols = self._get_input_columns()
+        X = extract_features(df, features_col, features_cols, self._float32_inputs)
+        out = self._transform_array(X)
+        if isinstance(out, dict):
+            res = df
+            for name, col in out.items():
+                res = res.with_column(name, col)
+            return res
+        return df.with_column(self._out_col_name(), out)
+
+    # -- persistence -------------------------------------------------------
+    def write(self) -> "_Writer":
+        return _Writer(self)
+
+    def save(self, path: str) -> None:
+        self.write().save(path)
+
+    @classmethod
+    def read(cls) -> "_Reader":
+        return _Reader(cls)
+
+    @classmethod
+    def load(cls, path: str):
+        return cls.read().load(path)
+
+
+# ---------------------------------------------------------------------------
+# Persistence: JSON metadata + npz arrays
+# (reference core.py:268-355 writes Spark param metadata JSON + a
+# model_attributes JSON text file; big arrays via parquet for UMAP
+# umap.py:1569-1727. Here: metadata.json + attributes.npz, rank 0 writes.)
+# ---------------------------------------------------------------------------
+
+
+class _Writer:
+    def __init__(self, instance: Union[Estimator, Model]):
+        self.instance = instance
+
+    def save(self, path: str) -> None:
+        self._save(path, overwrite=False)
+
+    def overwrite(self) -> "_OverwriteWriter":
+        return _OverwriteWriter(self.instance)
+
+    def _save(self, path: str, overwrite: bool) -> None:
+        comm = get_comm()
+        inst = self.instance
+        if comm.rank == 0:
+            if os.path.exists(path):
+                if not overwrite:
+                    raise FileExistsError(
+                        f"Path {path} already exists (use .write().overwrite())."
+                    )
+            os.makedirs(path, exist_ok=True)
+            meta = {
+                "class": f"{type(inst).__module__}.{type(inst).__qualname__}",
+                "uid": inst.uid,
+                "paramMap": {p.name: inst._paramMap[p] for p in inst._paramMap},
+                "defaultParamMap": {
+                    p.name: inst._defaultParamMap[p] for p in inst._defaultParamMap
+                },
+                "native_params": _jsonable(inst._native_params),
+                "float32_inputs": inst._float32_inputs,
+                "sparkRapidsMlAmdVersion": __import__("spark_rapids_ml_amd").__version__,
+            }
+            with open(os.path.join(path, "metadata.json"), "w") as f:
+                json.dump(meta, f, default=_json_default)
+            if isinstance(inst, Model):
+                arrays = {}
+                scalars = {}
+                for k, v in inst._get_model_attributes().items():
+                    if isinstance(v, np.ndarray):
+                        arrays[k] = v
+                    elif isinstance(v, torch.Tensor):
+                        arrays[k] = v.cpu().numpy()
+                    else:
+                        scalars[k] = v
+                np.savez(os.path.join(path, "attributes.npz"), **arrays)
+                with open(os.path.join(path, "attributes.json"), "w") as f:
+                    json.dump(_jsonable(scalars), f, default=_json_default)
+        comm.barrier()
+
+
+class _OverwriteWriter(_Writer):
+    def save(self, path: str) -> None:
+        self._save(path, overwrite=True)
+
+
+class _Reader:
+    def __init__(self, cls: type):
+        self.cls = cls
+
+    def load(self, path: str):
+        with open(os.path.join(path, "metadata.json")) as f:
+            meta = json.load(f)
+        cls = self.cls
+        attrs: Dict[str, Any] = {}
+        attr_json = os.path.join(path, "attributes.json")
+        attr_npz = os.path.join(path, "attributes.npz")
+        is_model = issubclass(cls, Model)
+        if is_model:
+            if os.path.exists(attr_json):
+                with open(attr_json) as f:
+                    attrs.update(json.load(f))
+            if os.path.exists(attr_npz):
+                with np.load(attr_npz, allow_pickle=False) as z:
+                    for k in z.files:
+                        attrs[k] = z[k]
+            inst = cls(**attrs)
+        else:
+            inst = cls()
+        for name, v in meta.get("defaultParamMap", {}).items():
+            if inst.hasParam(name):
+                inst._defaultParamMap[inst.getParam(name)] = v
+        for name, v in meta.get("paramMap", {}).items():
+            if inst.hasParam(name):
+                inst._paramMap[inst.getParam(name)] = v
+        inst._native_params.update(meta.get("native_params", {}))
+        inst._float32_inputs = meta.get("float32_inputs", True)
+        return inst
+
+
+def _jsonable(d: Dict[str, Any]) -> Dict[str, Any]:
+    out = {}
+    for k, v in d.items():
+        if isinstance(v, np.generic):
+            v = v.item()
+        out[k] = v
+    return out
+
+
+def _json_default(o: Any) -> Any:
+    if isinstance(o, np.generic):
+        return o.item()
+    if isinstance(o, np.ndarray):
+        return o.tolist()
+    raise TypeError(f"not JSON serializable: {type(o)}")

@@ -1,0 +1,142 @@
+"""Evaluators (pyspark.ml.evaluation equivalents over the SPMD runtime).
+
+The reference plugs Spark's evaluators into its single-pass
+transform+evaluate (reference core.py:1572-1693, metrics/*); pyspark is not
+a dependency here, so the evaluator classes themselves are provided with the
+same names/params. Each evaluator computes per-rank sufficient statistics
+and merges them with one allgather — no raw predictions cross ranks.
+"""
+
+from __future__ import annotations
+
+from typing import Any, Dict, Optional
+
+import numpy as np
+
+from .data import DataFrame
+from .metrics import MulticlassMetrics, RegressionMetrics
+from .params import HasLabelCol, HasPredictionCol, Param, Params, TypeConverters
+from .parallel.context import get_comm
+
+
+class Evaluator(Params):
+    def evaluate(self, df: DataFrame) -> float:
+        raise NotImplementedError
+
+    def isLargerBetter(self) -> bool:
+        return True
+
+
+class RegressionEvaluator(Evaluator, HasLabelCol, HasPredictionCol):
+    """rmse|mse|r2|mae|var (Spark RegressionEvaluator parity)."""
+
+    metricName = Param("eval", "metricName", "rmse|mse|r2|mae|var.", TypeConverters.toString)
+
+    def __init__(self, **kwargs: Any) -> None:
+        super().__init__()
+        self._setDefault(metricName="rmse")
+        self._set(**kwargs)
+
+    def setMetricName(self, value: str) -> "RegressionEvaluator":
+        return self._set(metricName=value)  # type: ignore[return-value]
+
+    def setLabelCol(self, value: str) -> "RegressionEvaluator":
+        return self._set(labelCol=value)  # type: ignore[return-value]
+
+    def setPredictionCol(self, value: str) -> "RegressionEvaluator":
+        return self._set(predictionCol=value)  # type: ignore[return-value]
+
+    def isLargerBetter(self) -> bool:
+        return self.getOrDefault("metricName") in ("r2", "var")
+
+    def evaluate(self, df: DataFrame) -> float:
+        comm = get_comm()
+        lab = np.asarray(df[self.getOrDefault("labelCol")])
+        pred = np.asarray(df[self.getOrDefault("predictionCol")])
+        local = RegressionMetrics.from_predictions(lab, pred)
+        bufs = comm.allgather_obj(local._buf.to_list())
+        from .metrics.RegressionMetrics import _SummarizerBuffer
+
+        merged = local
+        first = True
+        for b in bufs:
+            rm = RegressionMetrics(_SummarizerBuffer.from_list(b))
+            merged = rm if first else merged.merge(rm)
+            first = False
+        return merged.evaluate(self.getOrDefault("metricName"))
+
+
+class MulticlassClassificationEvaluator(Evaluator, HasLabelCol, HasPredictionCol):
+    """f1|accuracy|weighted*|hammingLoss|logLoss (Spark parity)."""
+
+    metricName = Param("eval", "metricName", "metric.", TypeConverters.toString)
+    probabilityCol = Param("eval", "probabilityCol", "for logLoss.", TypeConverters.toString)
+
+    def __init__(self, **kwargs: Any) -> None:
+        super().__init__()
+        self._setDefault(metricName="f1", probabilityCol="probability")
+        self._set(**kwargs)
+
+    def setMetricName(self, value: str) -> "MulticlassClassificationEvaluator":
+        return self._set(metricName=value)  # type: ignore[return-value]
+
+    def setLabelCol(self, value: str) -> "MulticlassClassificationEvaluator":
+        return self._set(labelCol=value)  # type: ignore[return-value]
+
+    def setPredictionCol(self, value: str) -> "MulticlassClassificationEvaluator":
+        return self._set(predictionCol=value)  # type: ignore[return-value]
+
+    def isLargerBetter(self) -> bool:
+        return self.getOrDefault("metricName") not in ("hammingLoss", "logLoss")
+
+    def evaluate(self, df: DataFrame) -> float:
+        comm = get_comm()
+        lab = np.asarray(df[self.getOrDefault("labelCol")], dtype=np.float64)
+        pred = np.asarray(df[self.getOrDefault("predictionCol")], dtype=np.float64)
+        conf: Dict = {}
+        for l, p in zip(lab, pred):
+            key = (float(l), float(p))
+            conf[key] = conf.get(key, 0.0) + 1.0
+        log_loss = None
+        if self.getOrDefault("metricName") == "logLoss":
+            probs = np.asarray(df[self.getOrDefault("probabilityCol")], dtype=np.float64)
+            eps = 1e-15
+            pl = np.clip(probs[np.arange(len(lab)), lab.astype(int)], eps, 1.0)
+            log_loss = float(-np.log(pl).sum())
+        all_stats = comm.allgather_obj((list(conf.items()), log_loss))
+        merged = MulticlassMetrics()
+        for items, ll in all_stats:
+            m = MulticlassMetrics.from_confusion({tuple(k): v for k, v in items}, ll)
+            merged = merged.merge(m)
+        return merged.evaluate(self.getOrDefault("metricName"))
+
+
+class BinaryClassificationEvaluator(Evaluator, HasLabelCol):
+    """areaUnderROC via a global rank-sum (Mann-Whitney) over gathered
+    scores."""
+
+    metricName = Param("eval", "metricName", "areaUnderROC.", TypeConverters.toString)
+    rawPredictionCol = Param("eval", "rawPredictionCol", "scores col.", TypeConverters.toString)
+
+    def __init__(self, **kwargs: Any) -> None:
+        super().__init__()
+        self._setDefault(metricName="areaUnderROC", rawPredictionCol="rawPrediction")
+        self._set(**kwargs)
+
+    def evaluate(self, df: DataFrame) -> float:
+        comm = get_comm()
+        lab = np.asarray(df[self.getOrDefault("labelCol")], dtype=np.float64)
+        raw = np.asarray(df[self.getOrDefault("rawPredictionCol")])
+        score = raw[:, 1] if raw.ndim == 2 else raw
+        parts = comm.allgather_obj((score, lab))
+        s = np.concatenate([p[0] for p in parts])
+        l = np.concatenate([p[1] for p in parts])
+        pos = s[l == 1.0]
+        neg = s[l != 1.0]
+        if len(pos) == 0 or len(neg) == 0:
+            return 0.5
+        from scipy.stats import rankdata
+
+        ranks = rankdata(np.concatenate([pos, neg]))
+        auc = (ranks[: len(pos)].sum() - len(pos) * (len(pos) + 1) / 2) / (len(pos) * len(neg))
+        return float(auc)

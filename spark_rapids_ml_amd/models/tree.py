@@ -1,0 +1,665 @@
+"""Random forest: histogram-based GPU trainer + ensemble-parallel fit.
+
+Reference design (reference tree.py, 744 lines): embarrassingly parallel —
+trees are divided across workers (`_estimators_per_worker`, tree.py:330-341),
+each worker fits its sub-forest on its LOCAL partition only (single-GPU cuML
+RF, histogram split finding with n_bins≤128), the serialized sub-forests are
+allGather'ed and concatenated (tree.py:424-460). The trainer here is a
+from-scratch histogram forest in torch (CPU/ROCm; the histogram build is the
+HIP-kernel candidate): depth-wise frontier, per-(node,feature,bin) class
+counts / moment stats via scatter-add, vectorized gain scan, per-node
+feature subsampling.
+
+Forest serialization is an own format (arrays per tree: feature, threshold,
+children, leaf values) playing the role of treelite bytes + JSON dump
+(reference tree.py:424-460): pickled for the allGather merge, stored in the
+model attributes, and dumpable as JSON for Spark-tree translation parity.
+"""
+
+from __future__ import annotations
+
+import math
+import pickle
+from typing import Any, Dict, List, Optional, Tuple
+
+import numpy as np
+import torch
+
+from ..core import Estimator, Model, _FitContext
+from ..data import to_device_tensor
+from ..params import (
+    HasFeaturesCol,
+    HasFeaturesCols,
+    HasLabelCol,
+    HasPredictionCol,
+    HasProbabilityCol,
+    HasRawPredictionCol,
+    Param,
+    TypeConverters,
+)
+from ..utils import as_numpy, get_logger
+
+
+class _RandomForestParams(HasFeaturesCol, HasFeaturesCols, HasLabelCol, HasPredictionCol):
+    numTrees = Param("rf", "numTrees", "number of trees.", TypeConverters.toInt)
+    maxDepth = Param("rf", "maxDepth", "max tree depth.", TypeConverters.toInt)
+    maxBins = Param("rf", "maxBins", "max histogram bins (<=256).", TypeConverters.toInt)
+    minInstancesPerNode = Param(
+        "rf", "minInstancesPerNode", "min rows per child.", TypeConverters.toInt
+    )
+    minInfoGain = Param("rf", "minInfoGain", "min split gain.", TypeConverters.toFloat)
+    subsamplingRate = Param("rf", "subsamplingRate", "bootstrap fraction.", TypeConverters.toFloat)
+    featureSubsetStrategy = Param(
+        "rf",
+        "featureSubsetStrategy",
+        "auto|all|sqrt|log2|onethird|n|fraction.",
+        TypeConverters.toString,
+    )
+    seed = Param("rf", "seed", "random seed.", TypeConverters.toInt)
+    bootstrap = Param("rf", "bootstrap", "bootstrap rows.", TypeConverters.toBoolean)
+    impurity = Param("rf", "impurity", "gini|entropy|variance.", TypeConverters.toString)
+
+    def __init__(self, **kwargs: Any) -> None:
+        super().__init__(**kwargs)
+        self._setDefault(
+            numTrees=20,
+            maxDepth=5,
+            maxBins=32,
+            minInstancesPerNode=1,
+            minInfoGain=0.0,
+            subsamplingRate=1.0,
+            featureSubsetStrategy="auto",
+            seed=0,
+            bootstrap=True,
+        )
+
+    @classmethod
+    def _param_mapping(cls) -> Dict[str, Optional[str]]:
+        # reference tree.py:93-135
+        return {
+            "numTrees": "n_estimators",
+            "maxDepth": "max_depth",
+            "maxBins": "n_bins",
+            "minInstancesPerNode": "min_samples_leaf",
+            "minInfoGain": "min_impurity_decrease",
+            "subsamplingRate": "max_samples",
+            "featureSubsetStrategy": "max_features",
+            "seed": "random_state",
+            "bootstrap": "bootstrap",
+            "impurity": "split_criterion",
+        }
+
+    @classmethod
+    def _get_native_params_default(cls) -> Dict[str, Any]:
+        return {
+            "n_estimators": 20,
+            "max_depth": 5,
+            "n_bins": 32,
+            "min_samples_leaf": 1,
+            "min_impurity_decrease": 0.0,
+            "max_samples": 1.0,
+            "max_features": "auto",
+            "random_state": 0,
+            "bootstrap": True,
+            "split_criterion": None,
+            "max_batch_size": 4096,
+            "verbose": False,
+        }
+
+    def getNumTrees(self) -> int:
+        return self.getOrDefault("numTrees")
+
+    def getMaxDepth(self) -> int:
+        return self.getOrDefault("maxDepth")
+
+
+def _estimators_per_worker(n_estimators: int, world: int) -> List[int]:
+    """Tree counts per rank (reference tree.py:330-341): floor division with
+    the remainder spread over the first ranks."""
+    base = n_estimators // world
+    out = [base] * world
+    for i in range(n_estimators % world):
+        out[i] += 1
+    return out
+
+
+def _resolve_max_features(strategy: str, d: int, task: str) -> int:
+    s = str(strategy).lower()
+    if s == "auto":
+        s = "sqrt" if task == "classification" else "onethird"
+    if s == "all":
+        return d
+    if s == "sqrt":
+        return max(1, int(math.sqrt(d)))
+    if s == "log2":
+        return max(1, int(math.log2(max(2, d))))
+    if s == "onethird":
+        return max(1, int(d / 3.0))
+    try:
+        v = float(s)
+        if v.is_integer() and v >= 1:
+            return min(d, int(v))
+        if 0 < v <= 1:
+            return max(1, int(v * d))
+    except ValueError:
+        pass
+    raise ValueError(f"unsupported featureSubsetStrategy {strategy!r}")
+
+
+# ---------------------------------------------------------------------------
+# Binning
+# ---------------------------------------------------------------------------
+
+
+def _compute_bin_edges(
+    X: torch.Tensor, n_bins: int, comm, sample_cap: int = 65536, seed: int = 0
+) -> torch.Tensor:
+    """Approximate per-feature quantile edges from a global row sample
+    (reference cuML RF quantile binning). Returns [d, n_bins-1] f32 edges."""
+    n, d = X.shape
+    rng = np.random.default_rng(seed)
+    cap_local = max(1, sample_cap // max(1, comm.world_size))
+    if n > cap_local:
+        idx = torch.from_numpy(rng.choice(n, cap_local, replace=False)).to(X.device)
+        sample = X[idx]
+    else:
+        sample = X
+    parts = comm.allgather_obj(as_numpy(sample))
+    full = np.concatenate([p for p in parts if len(p)], axis=0)
+    qs = np.linspace(0, 1, n_bins + 1)[1:-1]
+    edges = np.quantile(full.astype(np.float64), qs, axis=0).T.astype(np.float32)  # [d, nb-1]
+    # strictly increasing edges per feature (degenerate features collapse)
+    edges = np.maximum.accumulate(edges + np.arange(edges.shape[1]) * 0.0, axis=1)
+    return torch.from_numpy(np.ascontiguousarray(edges)).to(X.device)
+
+
+def _bin_data(X: torch.Tensor, edges: torch.Tensor) -> torch.Tensor:
+    """bin(x) = #edges <= x, i.e. searchsorted right; uint8 [n,d].
+    Inference rule derived from this: bin(x) <= b  <=>  x < edges[b]."""
+    b = torch.searchsorted(edges, X.T.contiguous(), right=True)  # [d, n]
+    return b.T.to(torch.uint8).contiguous()
+
+
+# ---------------------------------------------------------------------------
+# Tree growing
+# ---------------------------------------------------------------------------
+
+
+class _Tree:
+    """Flat-array binary tree. Arrays grow as nodes are added."""
+
+    __slots__ = ("feature", "threshold", "left", "right", "is_leaf", "value", "n_nodes")
+
+    def __init__(self, value_width: int):
+        self.feature: List[int] = []
+        self.threshold: List[float] = []
+        self.left: List[int] = []
+        self.right: List[int] = []
+        self.is_leaf: List[bool] = []
+        self.value: List[np.ndarray] = []
+        self.n_nodes = 0
+
+    def add_node(self, value: np.ndarray) -> int:
+        i = self.n_nodes
+        self.n_nodes += 1
+        self.feature.append(-1)
+        self.threshold.append(0.0)
+        self.left.append(-1)
+        self.right.append(-1)
+        self.is_leaf.append(True)
+        self.value.append(value)
+        return i
+
+    def to_arrays(self) -> Dict[str, np.ndarray]:
+        return {
+            "feature": np.asarray(self.feature, dtype=np.int32),
+            "threshold": np.asarray(self.threshold, dtype=np.float32),
+            "left": np.asarray(self.left, dtype=np.int32),
+            "right": np.asarray(self.right, dtype=np.int32),
+            "is_leaf": np.asarray(self.is_leaf, dtype=bool),
+            "value": np.stack(self.value).astype(np.float32),
+        }
+
+
+def _grow_tree(
+    Xb: torch.Tensor,  # [n,d] uint8 binned
+    y: torch.Tensor,  # [n] int64 class idx (classification) or f32 target
+    edges: torch.Tensor,  # [d, nb-1]
+    task: str,
+    n_classes: int,
+    n_bins: int,
+    max_depth: int,
+    min_leaf: int,
+    min_gain: float,
+    max_features: int,
+    gen: torch.Generator,
+    node_batch: int = 4096,
+    feat_chunk: int = 256,
+) -> Dict[str, np.ndarray]:
+    n, d = Xb.shape
+    dev = Xb.device
+    value_width = n_classes if task == "classification" else 2  # (mean, count)
+    tree = _Tree(value_width)
+
+    if task == "classification":
+        y_idx = y.to(torch.int64)
+        root_counts = torch.bincount(y_idx, minlength=n_classes).to(torch.float32)
+        root_val = as_numpy(root_counts)
+    else:
+        yf = y.to(torch.float32)
+        root_val = np.array([float(yf.mean().item()) if n else 0.0, float(n)], dtype=np.float32)
+
+    root = tree.add_node(root_val)
+    node_of_row = torch.zeros(n, dtype=torch.int64, device=dev)
+    frontier = [root]  # node ids whose rows are still splittable
+
+    for depth in range(max_depth):
+        if not frontier:
+            break
+        new_frontier: List[int] = []
+        for bstart in range(0, len(frontier), node_batch):
+            batch = frontier[bstart : bstart + node_batch]
+            B = len(batch)
+            batch_t = torch.tensor(batch, dtype=torch.int64, device=dev)
+            # map node id -> local 0..B-1 (rows not in batch -> -1)
+            lut = torch.full((tree.n_nodes,), -1, dtype=torch.int64, device=dev)
+            lut[batch_t] = torch.arange(B, dtype=torch.int64, device=dev)
+            local = lut[node_of_row]
+            in_batch = local >= 0
+            rows = torch.nonzero(in_batch).flatten()
+            if rows.numel() == 0:
+                continue
+            loc = local[rows]
+            if task == "classification":
+                yb = y_idx[rows]
+            else:
+                yb = yf[rows]
+
+            best_gain = torch.full((B,), -1.0, dtype=torch.float32, device=dev)
+            best_feat = torch.full((B,), -1, dtype=torch.int64, device=dev)
+            best_bin = torch.full((B,), -1, dtype=torch.int64, device=dev)
+            best_lcnt = torch.zeros((B,), dtype=torch.float32, device=dev)
+            if task == "classification":
+                best_lval = torch.zeros((B, n_classes), dtype=torch.float32, device=dev)
+                best_rval = torch.zeros((B, n_classes), dtype=torch.float32, device=dev)
+            else:
+                best_lval = torch.zeros((B, 2), dtype=torch.float32, device=dev)
+                best_rval = torch.zeros((B, 2), dtype=torch.float32, device=dev)
+
+            # per-node feature subset mask
+            if max_features < d:
+                scores = torch.rand((B, d), generator=gen, device=dev)
+                sel = scores.argsort(dim=1)[:, :max_features]
+                fmask = torch.zeros((B, d), dtype=torch.bool, device=dev)
+                fmask.scatter_(1, sel, True)
+            else:
+                fmask = torch.ones((B, d), dtype=torch.bool, device=dev)
+
+            for f0 in range(0, d, feat_chunk):
+                f1 = min(d, f0 + feat_chunk)
+                F = f1 - f0
+                bins = Xb[rows][:, f0:f1].to(torch.int64)  # [m, F]
+                base = (loc[:, None] * F + torch.arange(F, device=dev)[None, :]) * n_bins + bins
+                if task == "classification":
+                    hist = torch.zeros(B * F * n_bins * n_classes, dtype=torch.float32, device=dev)
+                    idx = base * n_classes + yb[:, None]
+                    hist.index_add_(0, idx.flatten(), torch.ones(idx.numel(), device=dev))
+                    H = hist.view(B, F, n_bins, n_classes)
+                    gain, sbin, lval, rval, lcnt = _best_split_class(H, min_leaf)
+                else:
+                    cnt = torch.zeros(B * F * n_bins, dtype=torch.float32, device=dev)
+                    sm = torch.zeros_like(cnt)
+                    s2 = torch.zeros_like(cnt)
+                    flat = base.flatten()
+                    ones = torch.ones(flat.numel(), device=dev)
+                    cnt.index_add_(0, flat, ones)
+                    sm.index_add_(0, flat, yb[:, None].expand(-1, F).flatten())
+                    s2.index_add_(0, flat, (yb * yb)[:, None].expand(-1, F).flatten())
+                    H = torch.stack(
+                        [cnt.view(B, F, n_bins), sm.view(B, F, n_bins), s2.view(B, F, n_bins)],
+                        dim=-1,
+                    )
+                    gain, sbin, lval, rval, lcnt = _best_split_reg(H, min_leaf)
+                gain = torch.where(fmask[:, f0:f1], gain, torch.full_like(gain, -1.0))
+                # best feature within chunk
+                g, fidx = gain.max(dim=1)
+                upd = g > best_gain
+                best_gain = torch.where(upd, g, best_gain)
+                ar = torch.arange(B, device=dev)
+                best_feat = torch.where(upd, fidx + f0, best_feat)
+                best_bin = torch.where(upd, sbin[ar, fidx], best_bin)
+                best_lcnt = torch.where(upd, lcnt[ar, fidx], best_lcnt)
+                best_lval = torch.where(upd[:, None], lval[ar, fidx], best_lval)
+                best_rval = torch.where(upd[:, None], rval[ar, fidx], best_rval)
+
+            # materialize splits
+            bg = as_numpy(best_gain)
+            bf = as_numpy(best_feat)
+            bb = as_numpy(best_bin)
+            lv = as_numpy(best_lval)
+            rv = as_numpy(best_rval)
+            edges_np = as_numpy(edges)
+            split_nodes = []
+            for i, nid in enumerate(batch):
+                if bg[i] <= min_gain or bf[i] < 0:
+                    continue
+                f, b = int(bf[i]), int(bb[i])
+                thr = float(edges_np[f, b])
+                lchild = tree.add_node(lv[i])
+                rchild = tree.add_node(rv[i])
+                tree.feature[nid] = f
+                tree.threshold[nid] = thr
+                tree.left[nid] = lchild
+                tree.right[nid] = rchild
+                tree.is_leaf[nid] = False
+                split_nodes.append((nid, f, b, lchild, rchild))
+                if depth + 1 < max_depth:
+                    lcount = lv[i][-1] if task == "regression" else lv[i].sum()
+                    rcount = rv[i][-1] if task == "regression" else rv[i].sum()
+                    if lcount >= 2 * min_leaf:
+                        new_frontier.append(lchild)
+                    if rcount >= 2 * min_leaf:
+                        new_frontier.append(rchild)
+            if split_nodes:
+                nid_t = torch.tensor([s[0] for s in split_nodes], device=dev)
+                f_t = torch.tensor([s[1] for s in split_nodes], device=dev)
+                b_t = torch.tensor([s[2] for s in split_nodes], device=dev)
+                l_t = torch.tensor([s[3] for s in split_nodes], device=dev)
+                r_t = torch.tensor([s[4] for s in split_nodes], device=dev)
+                lut2 = torch.full((tree.n_nodes,), -1, dtype=torch.int64, device=dev)
+                lut2[nid_t] = torch.arange(len(split_nodes), dtype=torch.int64, device=dev)
+                sl = lut2[node_of_row]
+                moved = sl >= 0
+                mrows = torch.nonzero(moved).flatten()
+                srel = sl[mrows]
+                go_left = Xb[mrows].gather(1, f_t[srel].view(-1, 1)).flatten().to(
+                    torch.int64
+                ) <= b_t[srel]
+                node_of_row[mrows] = torch.where(go_left, l_t[srel], r_t[srel])
+        frontier = new_frontier
+
+    return tree.to_arrays()
+
+
+def _best_split_class(H: torch.Tensor, min_leaf: int):
+    """H: [B,F,nb,C] class counts. Returns per (B,F): best gain (impurity
+    decrease weighted by node fraction, Spark/CART gini), split bin, left/right
+    class-count values, left count."""
+    csum = H.cumsum(dim=2)  # left counts for split at bin b (bins <= b)
+    total = csum[:, :, -1:, :]  # [B,F,1,C]
+    left = csum[:, :, :-1, :]  # split bins 0..nb-2
+    right = total - left
+    lc = left.sum(dim=3)
+    rc = right.sum(dim=3)
+    nt = total.sum(dim=3)  # [B,F,1]
+    gini = lambda cnt, tot: 1.0 - ((cnt / torch.clamp(tot.unsqueeze(-1), min=1e-12)) ** 2).sum(
+        dim=-1
+    )
+    g_parent = gini(total, nt)  # [B,F,1]
+    g_left = gini(left, lc)
+    g_right = gini(right, rc)
+    ntc = torch.clamp(nt, min=1e-12)
+    gain = g_parent - (lc / ntc) * g_left - (rc / ntc) * g_right  # [B,F,nb-1]
+    valid = (lc >= min_leaf) & (rc >= min_leaf)
+    gain = torch.where(valid, gain, torch.full_like(gain, -1.0))
+    best_gain, best_bin = gain.max(dim=2)  # [B,F]
+    B, F = best_gain.shape
+    ar_b = torch.arange(B, device=H.device)[:, None].expand(B, F)
+    ar_f = torch.arange(F, device=H.device)[None, :].expand(B, F)
+    lval = left[ar_b, ar_f, best_bin]  # [B,F,C]
+    rval = right[ar_b, ar_f, best_bin]
+    lcnt = lc[ar_b, ar_f, best_bin]
+    return best_gain, best_bin, lval, rval, lcnt
+
+
+def _best_split_reg(H: torch.Tensor, min_leaf: int):
+    """H: [B,F,nb,3] = (count,sum,sum2). Variance-reduction gain."""
+    cs = H.cumsum(dim=2)
+    total = cs[:, :, -1:, :]
+    left = cs[:, :, :-1, :]
+    right = total - left
+    lc, ls, ls2 = left[..., 0], left[..., 1], left[..., 2]
+    rc, rs, rs2 = right[..., 0], right[..., 1], right[..., 2]
+    tc, ts, ts2 = total[..., 0], total[..., 1], total[..., 2]
+    var = lambda c, s, s2: torch.clamp(s2 / torch.clamp(c, min=1e-12) - (s / torch.clamp(c, min=1e-12)) ** 2, min=0.0)
+    tcc = torch.clamp(tc, min=1e-12)
+    gain = var(tc, ts, ts2) - (lc / tcc) * var(lc, ls, ls2) - (rc / tcc) * var(rc, rs, rs2)
+    valid = (lc >= min_leaf) & (rc >= min_leaf)
+    gain = torch.where(valid, gain, torch.full_like(gain, -1.0))
+    best_gain, best_bin = gain.max(dim=2)
+    B, F = best_gain.shape
+    ar_b = torch.arange(B, device=H.device)[:, None].expand(B, F)
+    ar_f = torch.arange(F, device=H.device)[None, :].expand(B, F)
+    lcb = lc[ar_b, ar_f, best_bin]
+    lsb = ls[ar_b, ar_f, best_bin]
+    rcb = rc[ar_b, ar_f, best_bin]
+    rsb = rs[ar_b, ar_f, best_bin]
+    lval = torch.stack([lsb / torch.clamp(lcb, min=1e-12), lcb], dim=-1)
+    rval = torch.stack([rsb / torch.clamp(rcb, min=1e-12), rcb], dim=-1)
+    return best_gain, best_bin, lval, rval, lcb
+
+
+# ---------------------------------------------------------------------------
+# Estimator / Model
+# ---------------------------------------------------------------------------
+
+
+class _RandomForestEstimator(_RandomForestParams, Estimator):
+    _task = "regression"
+
+    def setNumTrees(self, value: int):
+        return self._set_params(numTrees=value)
+
+    def setMaxDepth(self, value: int):
+        return self._set_params(maxDepth=value)
+
+    def setMaxBins(self, value: int):
+        return self._set_params(maxBins=value)
+
+    def setSeed(self, value: int):
+        return self._set_params(seed=value)
+
+    def setImpurity(self, value: str):
+        return self._set_params(impurity=value)
+
+    def setFeaturesCol(self, value):
+        if isinstance(value, (list, tuple)):
+            return self._set_params(featuresCols=list(value))
+        return self._set_params(featuresCol=value)
+
+    def setLabelCol(self, value: str):
+        return self._set_params(labelCol=value)
+
+    def setPredictionCol(self, value: str):
+        return self._set_params(predictionCol=value)
+
+    def setFeatureSubsetStrategy(self, value: str):
+        return self._set_params(featureSubsetStrategy=value)
+
+    def _fit_array(
+        self, X: Any, y: Optional[Any], ctx: _FitContext, params: Dict[str, Any]
+    ) -> Dict[str, Any]:
+        comm, pdesc = ctx.comm, ctx.pdesc
+        n_estimators = int(params["n_estimators"])
+        max_depth = int(params["max_depth"])
+        n_bins = min(256, int(params["n_bins"]))
+        min_leaf = int(params["min_samples_leaf"])
+        min_gain = float(params["min_impurity_decrease"])
+        seed = int(params["random_state"])
+        bootstrap = bool(params["bootstrap"])
+        max_samples = float(params["max_samples"])
+
+        Xt = to_device_tensor(np.ascontiguousarray(X, dtype=np.float32), ctx.device)
+        n, d = Xt.shape
+
+        if self._task == "classification":
+            y_local = np.asarray(y)
+            bad = (y_local < 0) | (y_local != np.floor(y_local))
+            if bad.any():
+                raise ValueError("RandomForestClassifier labels must be non-negative integers")
+            local_max = float(y_local.max()) if len(y_local) else -1.0
+            n_classes = int(comm.allreduce_scalar(local_max, "max")) + 1
+            # reference requires labels to cover [0, numClasses) (tree.py:415-421)
+            yt = to_device_tensor(y_local.astype(np.int64), ctx.device)
+        else:
+            n_classes = 0
+            yt = to_device_tensor(np.asarray(y, dtype=np.float32), ctx.device)
+
+        edges = _compute_bin_edges(Xt, n_bins, comm, seed=seed)
+        Xb = _bin_data(Xt, edges)
+        max_features = _resolve_max_features(str(params["max_features"]), d, self._task)
+
+        counts = _estimators_per_worker(n_estimators, comm.world_size)
+        my_trees = counts[comm.rank]
+        tree_id0 = sum(counts[: comm.rank])
+
+        trees: List[Dict[str, np.ndarray]] = []
+        for t in range(my_trees):
+            tseed = seed + 1315423911 * (tree_id0 + t + 1)
+            gen = torch.Generator(device=ctx.device)
+            gen.manual_seed(tseed & 0x7FFFFFFFFFFF)
+            if bootstrap and n > 0:
+                n_draw = max(1, int(round(max_samples * n)))
+                rows = torch.randint(0, n, (n_draw,), generator=gen, device=ctx.device)
+                Xb_t, y_t = Xb[rows], yt[rows]
+            else:
+                Xb_t, y_t = Xb, yt
+            trees.append(
+                _grow_tree(
+                    Xb_t,
+                    y_t,
+                    edges,
+                    self._task,
+                    n_classes,
+                    n_bins,
+                    max_depth,
+                    min_leaf,
+                    min_gain,
+                    max_features,
+                    gen,
+                    node_batch=int(params.get("max_batch_size", 4096)),
+                )
+            )
+
+        # merge sub-forests (reference allGathers treelite bytes, tree.py:424-447)
+        blobs = comm.allgather_obj(pickle.dumps(trees, protocol=4))
+        forest: List[Dict[str, np.ndarray]] = []
+        for b in blobs:
+            forest.extend(pickle.loads(b))
+
+        return {
+            "forest_": np.frombuffer(pickle.dumps(forest, protocol=4), dtype=np.uint8).copy(),
+            "n_classes_": n_classes,
+            "n_features_": d,
+            "n_trees_": len(forest),
+        }
+
+
+class _RandomForestModel(_RandomForestParams, Model):
+    _task = "regression"
+
+    def __init__(
+        self,
+        forest_: np.ndarray,
+        n_classes_: int = 0,
+        n_features_: int = 0,
+        n_trees_: int = 0,
+        **kwargs: Any,
+    ) -> None:
+        super().__init__(
+            forest_=np.asarray(forest_, dtype=np.uint8),
+            n_classes_=int(n_classes_),
+            n_features_=int(n_features_),
+            n_trees_=int(n_trees_),
+        )
+        self._trees: Optional[List[Dict[str, np.ndarray]]] = None
+
+    @property
+    def trees(self) -> List[Dict[str, np.ndarray]]:
+        if self._trees is None:
+            self._trees = pickle.loads(self._model_attributes["forest_"].tobytes())
+        return self._trees
+
+    @property
+    def numTrees(self) -> int:
+        return self._model_attributes["n_trees_"]
+
+    @property
+    def numFeatures(self) -> int:
+        return self._model_attributes["n_features_"]
+
+    @property
+    def numClasses(self) -> int:
+        return self._model_attributes["n_classes_"]
+
+    def dump_as_json(self) -> str:
+        """Forest dump for Spark-tree translation parity (reference
+        treelite dump_as_json, tree.py:449-460)."""
+        import json
+
+        return json.dumps(
+            [
+                {k: v.tolist() for k, v in t.items()}
+                for t in self.trees
+            ]
+        )
+
+    def setFeaturesCol(self, value):
+        if isinstance(value, (list, tuple)):
+            return self._set_params(featuresCols=list(value))
+        return self._set_params(featuresCol=value)
+
+    def setPredictionCol(self, value: str):
+        return self._set_params(predictionCol=value)
+
+    def _predict_raw(self, X: Any) -> torch.Tensor:
+        """Vectorized forest traversal (the FIL-analog inference path,
+        reference tree.py:682-721): iterate depth levels with gathers.
+        Returns [n, C] vote/probability sums (classification) or [n] mean."""
+        from ..parallel.context import get_comm
+
+        device = get_comm().device
+        Xt = to_device_tensor(np.ascontiguousarray(X, dtype=np.float32), device)
+        n = Xt.shape[0]
+        if self._task == "classification":
+            acc = torch.zeros((n, self.numClasses), dtype=torch.float32, device=device)
+        else:
+            acc = torch.zeros(n, dtype=torch.float32, device=device)
+        for t in self.trees:
+            feature = torch.from_numpy(t["feature"]).to(device, torch.int64)
+            thr = torch.from_numpy(t["threshold"]).to(device)
+            left = torch.from_numpy(t["left"]).to(device, torch.int64)
+            right = torch.from_numpy(t["right"]).to(device, torch.int64)
+            leaf = torch.from_numpy(t["is_leaf"]).to(device)
+            value = torch.from_numpy(t["value"]).to(device)
+            node = torch.zeros(n, dtype=torch.int64, device=device)
+            while True:
+                at_leaf = leaf[node]
+                if bool(at_leaf.all()):
+                    break
+                f = feature[node].clamp(min=0)
+                xv = Xt.gather(1, f.view(-1, 1)).flatten()
+                go_left = xv <= thr[node]
+                nxt = torch.where(go_left, left[node], right[node])
+                node = torch.where(at_leaf, node, nxt)
+            v = value[node]
+            if self._task == "classification":
+                acc += v / torch.clamp(v.sum(dim=1, keepdim=True), min=1e-12)
+            else:
+                acc += v[:, 0]
+        return acc
+
+    def _transform_array(self, X: Any):
+        acc = self._predict_raw(X)
+        if self._task == "classification":
+            probs = acc / self.numTrees
+            pred = probs.argmax(dim=1)
+            out = {
+                self.getOrDefault("predictionCol"): as_numpy(pred).astype(np.float64),
+            }
+            if self.hasParam("probabilityCol"):
+                out[self.getOrDefault("probabilityCol")] = as_numpy(probs)
+            if self.hasParam("rawPredictionCol"):
+                out[self.getOrDefault("rawPredictionCol")] = as_numpy(acc)
+            return out
+        return as_numpy(acc / self.numTrees).astype(np.float64)

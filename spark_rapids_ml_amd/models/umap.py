@@ -1,0 +1,408 @@
+"""UMAP: single-GPU fit + broadcast-replicated transform (reference umap.py).
+
+Reference design (umap.py:678,1346): fit runs on ONE GPU over (optionally
+`sample_fraction`-subsampled) data — the reference coalesces to a single
+partition (umap.py:923-945); the model holds the embedding + the raw training
+data; transform broadcasts both and embeds each rank's rows locally
+(umap.py:1404-1551). Here: rank 0 fits (kNN graph -> fuzzy simplicial set ->
+SGD embedding, all torch on-device), result broadcast over RCCL.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Any, Dict, Optional, Tuple
+
+import numpy as np
+import torch
+
+from ..core import Estimator, Model, _FitContext
+from ..data import DataFrame, extract_features, to_device_tensor
+from ..params import (
+    HasFeaturesCol,
+    HasFeaturesCols,
+    HasLabelCol,
+    HasOutputCol,
+    Param,
+    TypeConverters,
+)
+from ..ops.torch_ref import knn_topk
+from ..utils import as_numpy, get_logger
+
+
+def find_ab_params(spread: float = 1.0, min_dist: float = 0.1) -> Tuple[float, float]:
+    """Fit the differentiable curve 1/(1+a x^{2b}) to the desired fuzzy
+    membership shape (reference calls cuML's find_ab_params, umap.py:1518-1524)."""
+    from scipy.optimize import curve_fit
+
+    def curve(x, a, b):
+        return 1.0 / (1.0 + a * x ** (2 * b))
+
+    xv = np.linspace(0, spread * 3, 300)
+    yv = np.zeros_like(xv)
+    yv[xv < min_dist] = 1.0
+    yv[xv >= min_dist] = np.exp(-(xv[xv >= min_dist] - min_dist) / spread)
+    params, _ = curve_fit(curve, xv, yv)
+    return float(params[0]), float(params[1])
+
+
+def _smooth_knn(dists: np.ndarray, local_connectivity: float = 1.0) -> Tuple[np.ndarray, np.ndarray]:
+    """Per-row (sigma, rho) by binary search: sum_j exp(-(d_ij-rho)/sigma) =
+    log2(k) (umap-learn smooth_knn_dist semantics)."""
+    n, k = dists.shape
+    target = math.log2(k)
+    rho = np.zeros(n)
+    sigma = np.zeros(n)
+    for i in range(n):
+        row = dists[i]
+        nz = row[row > 0]
+        if len(nz) >= local_connectivity:
+            rho[i] = nz[int(local_connectivity) - 1] if int(local_connectivity) >= 1 else nz[0]
+        elif len(nz) > 0:
+            rho[i] = nz[-1]
+        lo, hi, mid = 0.0, np.inf, 1.0
+        for _ in range(64):
+            val = np.exp(-np.maximum(row - rho[i], 0.0) / mid).sum()
+            if abs(val - target) < 1e-5:
+                break
+            if val > target:
+                hi = mid
+                mid = (lo + hi) / 2.0
+            else:
+                lo = mid
+                mid = mid * 2 if hi == np.inf else (lo + hi) / 2.0
+        sigma[i] = max(mid, 1e-3 * (row.mean() if row.mean() > 0 else 1.0))
+    return sigma, rho
+
+
+def _fuzzy_simplicial_set(
+    knn_d: np.ndarray, knn_i: np.ndarray, set_op_mix_ratio: float, local_connectivity: float
+) -> Tuple[np.ndarray, np.ndarray, np.ndarray]:
+    """Returns COO (rows, cols, vals) of the symmetrized membership graph."""
+    import scipy.sparse as sp
+
+    n, k = knn_d.shape
+    sigma, rho = _smooth_knn(knn_d, local_connectivity)
+    vals = np.exp(-np.maximum(knn_d - rho[:, None], 0.0) / sigma[:, None])
+    vals[knn_i == np.arange(n)[:, None]] = 0.0  # no self loops
+    rows = np.repeat(np.arange(n), k)
+    W = sp.coo_matrix((vals.ravel(), (rows, knn_i.ravel())), shape=(n, n)).tocsr()
+    Wt = W.T.tocsr()
+    prod = W.multiply(Wt)
+    sym = set_op_mix_ratio * (W + Wt - prod) + (1 - set_op_mix_ratio) * prod
+    sym = sym.tocoo()
+    return sym.row.astype(np.int64), sym.col.astype(np.int64), sym.data.astype(np.float32)
+
+
+def _optimize_embedding(
+    emb: torch.Tensor,
+    heads: torch.Tensor,
+    tails: torch.Tensor,
+    weights: torch.Tensor,
+    n_epochs: int,
+    a: float,
+    b: float,
+    lr: float,
+    neg_rate: int,
+    repulsion: float,
+    gen: torch.Generator,
+    move_tail: bool = True,
+    tail_emb: Optional[torch.Tensor] = None,
+) -> torch.Tensor:
+    """Edge-sampled SGD (umap-learn schedule, vectorized): each edge gets
+    epochs_per_sample = max(w)/w; per epoch, due edges attract and spawn
+    negative repulsions. index_add makes the update Hogwild-equivalent."""
+    device = emb.device
+    n_emb = tail_emb if tail_emb is not None else emb
+    eps = weights.max() / torch.clamp(weights, min=1e-12)  # epochs per sample
+    next_due = eps.clone()
+    n_vertices = n_emb.shape[0]
+    clip = 4.0
+    for epoch in range(1, n_epochs + 1):
+        alpha = lr * (1.0 - epoch / n_epochs)
+        due = next_due <= epoch
+        if not bool(due.any()):
+            continue
+        h = heads[due]
+        t = tails[due]
+        next_due[due] += eps[due]
+        eh = emb[h]
+        et = n_emb[t]
+        diff = eh - et
+        d2 = (diff * diff).sum(dim=1, keepdim=True)
+        # attraction: grad coef = -2ab d^{2(b-1)} / (1 + a d^{2b})
+        gcoef = (-2.0 * a * b * d2.clamp(min=1e-12) ** (b - 1.0)) / (
+            1.0 + a * d2.clamp(min=1e-12) ** b
+        )
+        g = torch.clamp(gcoef * diff, -clip, clip)
+        emb.index_add_(0, h, alpha * g)
+        if move_tail:
+            emb.index_add_(0, t, -alpha * g)
+        # negative sampling
+        m = h.shape[0]
+        neg = torch.randint(0, n_vertices, (m * neg_rate,), generator=gen, device=device)
+        hrep = h.repeat_interleave(neg_rate)
+        diff = emb[hrep] - n_emb[neg]
+        d2 = (diff * diff).sum(dim=1, keepdim=True)
+        gcoef = (2.0 * repulsion * b) / (
+            (0.001 + d2) * (1.0 + a * d2.clamp(min=1e-12) ** b)
+        )
+        g = torch.clamp(gcoef * diff, -clip, clip)
+        emb.index_add_(0, hrep, alpha * g)
+    return emb
+
+
+class _UMAPParams(HasFeaturesCol, HasFeaturesCols, HasLabelCol, HasOutputCol):
+    n_neighbors = Param("umap", "n_neighbors", "kNN graph size.", TypeConverters.toFloat)
+    n_components = Param("umap", "n_components", "embedding dim.", TypeConverters.toInt)
+    n_epochs = Param("umap", "n_epochs", "SGD epochs (0=auto).", TypeConverters.toInt)
+    learning_rate = Param("umap", "learning_rate", "initial SGD step.", TypeConverters.toFloat)
+    init = Param("umap", "init", "spectral|random.", TypeConverters.toString)
+    min_dist = Param("umap", "min_dist", "min embedding distance.", TypeConverters.toFloat)
+    spread = Param("umap", "spread", "embedding spread.", TypeConverters.toFloat)
+    set_op_mix_ratio = Param("umap", "set_op_mix_ratio", "union/intersection mix.", TypeConverters.toFloat)
+    local_connectivity = Param("umap", "local_connectivity", "min local links.", TypeConverters.toFloat)
+    repulsion_strength = Param("umap", "repulsion_strength", "negative weight.", TypeConverters.toFloat)
+    negative_sample_rate = Param("umap", "negative_sample_rate", "negatives per edge.", TypeConverters.toInt)
+    a = Param("umap", "a", "curve param a.", TypeConverters.toFloat)
+    b = Param("umap", "b", "curve param b.", TypeConverters.toFloat)
+    random_state = Param("umap", "random_state", "seed.", TypeConverters.toInt)
+    sample_fraction = Param(
+        "umap", "sample_fraction", "fit subsample fraction (reference umap.py:923-945).", TypeConverters.toFloat
+    )
+
+    def __init__(self, **kwargs: Any) -> None:
+        super().__init__(**kwargs)
+        self._setDefault(
+            n_neighbors=15.0,
+            n_components=2,
+            n_epochs=0,
+            learning_rate=1.0,
+            init="spectral",
+            min_dist=0.1,
+            spread=1.0,
+            set_op_mix_ratio=1.0,
+            local_connectivity=1.0,
+            repulsion_strength=1.0,
+            negative_sample_rate=5,
+            a=None,
+            b=None,
+            random_state=42,
+            sample_fraction=1.0,
+            outputCol="embedding",
+        )
+
+    @classmethod
+    def _param_mapping(cls) -> Dict[str, Optional[str]]:
+        return {p: p for p in [
+            "n_neighbors", "n_components", "n_epochs", "learning_rate", "init",
+            "min_dist", "spread", "set_op_mix_ratio", "local_connectivity",
+            "repulsion_strength", "negative_sample_rate", "a", "b", "random_state",
+        ]}
+
+    @classmethod
+    def _get_native_params_default(cls) -> Dict[str, Any]:
+        return {"n_neighbors": 15, "n_components": 2, "verbose": False}
+
+
+class UMAP(_UMAPParams, Estimator):
+    """UMAP estimator (reference UMAP, umap.py:678): single-rank fit on
+    (sampled) gathered data, model replicated by broadcast."""
+
+    def __init__(self, **kwargs: Any) -> None:
+        super().__init__()
+        self._set_params(**kwargs)
+
+    def setFeaturesCol(self, value) -> "UMAP":
+        if isinstance(value, (list, tuple)):
+            return self._set_params(featuresCols=list(value))
+        return self._set_params(featuresCol=value)
+
+    def setOutputCol(self, value: str) -> "UMAP":
+        return self._set_params(outputCol=value)
+
+    def setK(self, value: int) -> "UMAP":
+        return self._set_params(n_neighbors=value)
+
+    def _is_supervised(self) -> bool:
+        return False
+
+    def _fit_array(
+        self, X: Any, y: Optional[Any], ctx: _FitContext, params: Dict[str, Any]
+    ) -> Dict[str, Any]:
+        comm = ctx.comm
+        frac = float(self.getOrDefault("sample_fraction"))
+        seed = int(self.getOrDefault("random_state"))
+        rng = np.random.default_rng(seed)
+        X_local = np.ascontiguousarray(X, dtype=np.float32)
+        if frac < 1.0 and X_local.shape[0] > 0:
+            m = max(1, int(frac * X_local.shape[0]))
+            X_local = X_local[rng.choice(X_local.shape[0], m, replace=False)]
+        parts = comm.allgather_obj(X_local)  # gather to all; rank0 fits
+        X_fit = np.concatenate([p for p in parts if len(p)], axis=0)
+
+        if comm.rank == 0:
+            emb = self._fit_single(X_fit, ctx.device)
+        else:
+            emb = np.zeros((X_fit.shape[0], int(self.getOrDefault("n_components"))), dtype=np.float32)
+        emb_t = torch.from_numpy(emb).to(comm._coll_device())
+        comm.broadcast(emb_t, src=0)
+        emb = as_numpy(emb_t)
+
+        return {
+            "embedding_": emb.astype(np.float32),
+            "raw_data_": X_fit.astype(np.float32),
+            "a_": self._ab()[0],
+            "b_": self._ab()[1],
+        }
+
+    def _ab(self) -> Tuple[float, float]:
+        a = self.getOrDefault("a")
+        b = self.getOrDefault("b")
+        if a is None or b is None:
+            return find_ab_params(
+                float(self.getOrDefault("spread")), float(self.getOrDefault("min_dist"))
+            )
+        return float(a), float(b)
+
+    def _fit_single(self, X: np.ndarray, device: torch.device) -> np.ndarray:
+        n = X.shape[0]
+        k = int(float(self.getOrDefault("n_neighbors")))
+        k = min(k, max(2, n - 1))
+        dim = int(self.getOrDefault("n_components"))
+        seed = int(self.getOrDefault("random_state"))
+        Xt = to_device_tensor(X, device)
+        d, i = knn_topk(Xt, Xt, k + 1)
+        knn_d, knn_i = as_numpy(d)[:, 1:], as_numpy(i)[:, 1:]  # drop self
+
+        rows, cols, vals = _fuzzy_simplicial_set(
+            knn_d,
+            knn_i,
+            float(self.getOrDefault("set_op_mix_ratio")),
+            float(self.getOrDefault("local_connectivity")),
+        )
+        a, b = self._ab()
+
+        n_epochs = int(self.getOrDefault("n_epochs")) or (500 if n <= 10000 else 200)
+        emb = self._init_embedding(X, rows, cols, vals, dim, seed)
+        emb_t = torch.from_numpy(emb).to(device)
+        gen = torch.Generator(device=device)
+        gen.manual_seed(seed)
+        emb_t = _optimize_embedding(
+            emb_t,
+            torch.from_numpy(rows).to(device),
+            torch.from_numpy(cols).to(device),
+            torch.from_numpy(vals).to(device),
+            n_epochs,
+            a,
+            b,
+            float(self.getOrDefault("learning_rate")),
+            int(self.getOrDefault("negative_sample_rate")),
+            float(self.getOrDefault("repulsion_strength")),
+            gen,
+        )
+        return as_numpy(emb_t)
+
+    def _init_embedding(self, X, rows, cols, vals, dim, seed) -> np.ndarray:
+        n = X.shape[0]
+        init = self.getOrDefault("init")
+        if init == "spectral" and n > dim + 1:
+            try:
+                import scipy.sparse as sp
+                from scipy.sparse.linalg import eigsh
+
+                W = sp.coo_matrix((vals, (rows, cols)), shape=(n, n)).tocsr()
+                deg = np.asarray(W.sum(axis=1)).flatten()
+                dinv = 1.0 / np.sqrt(np.maximum(deg, 1e-12))
+                L = sp.eye(n) - sp.diags(dinv) @ W @ sp.diags(dinv)
+                k_eig = min(dim + 1, n - 1)
+                w, v = eigsh(L, k=k_eig, sigma=0.0, which="LM", maxiter=2000)
+                emb = v[:, 1 : dim + 1]
+                scale = 10.0 / max(1e-12, np.abs(emb).max())
+                return (emb * scale).astype(np.float32)
+            except Exception:
+                pass
+        rng = np.random.default_rng(seed)
+        return (rng.uniform(-10, 10, size=(n, dim))).astype(np.float32)
+
+    def _create_model(self, attrs: Dict[str, Any]) -> "UMAPModel":
+        return UMAPModel(**attrs)
+
+
+class UMAPModel(_UMAPParams, Model):
+    """Fitted UMAP model (reference UMAPModel, umap.py:1346): holds the
+    embedding and the raw training data (both replicated — the reference
+    broadcasts them in <=8GB chunks, umap.py:1404-1447)."""
+
+    def __init__(
+        self,
+        embedding_: np.ndarray,
+        raw_data_: np.ndarray,
+        a_: float = 1.577,
+        b_: float = 0.895,
+        **kwargs: Any,
+    ) -> None:
+        super().__init__(
+            embedding_=np.asarray(embedding_),
+            raw_data_=np.asarray(raw_data_),
+            a_=float(a_),
+            b_=float(b_),
+        )
+
+    @property
+    def embedding(self) -> np.ndarray:
+        return self._model_attributes["embedding_"]
+
+    @property
+    def rawData(self) -> np.ndarray:
+        return self._model_attributes["raw_data_"]
+
+    def _out_col_name(self) -> str:
+        return self.getOrDefault("outputCol")
+
+    def _transform_array(self, X: Any) -> np.ndarray:
+        """Embed new points: weighted average of their training-kNN's
+        embeddings + a few attract-only refinement epochs (reference
+        umap.py:1528-1549 runs cuML UMAP.transform per partition)."""
+        from ..parallel.context import get_comm
+
+        device = get_comm().device
+        Xt = to_device_tensor(np.ascontiguousarray(X, dtype=np.float32), device)
+        n_new = Xt.shape[0]
+        if n_new == 0:
+            return np.zeros((0, self.embedding.shape[1]), dtype=np.float32)
+        raw = to_device_tensor(self.rawData, device)
+        emb = to_device_tensor(self.embedding, device)
+        k = int(float(self.getOrDefault("n_neighbors")))
+        k = min(k, raw.shape[0])
+        d, i = knn_topk(Xt, raw, k)
+        knn_d, knn_i = as_numpy(d), as_numpy(i)
+        sigma, rho = _smooth_knn(knn_d, float(self.getOrDefault("local_connectivity")))
+        w = np.exp(-np.maximum(knn_d - rho[:, None], 0.0) / sigma[:, None]).astype(np.float32)
+        w = w / np.clip(w.sum(axis=1, keepdims=True), 1e-12, None)
+        wt = torch.from_numpy(w).to(device)
+        new_emb = (wt[:, :, None] * emb[i]).sum(dim=1)
+
+        # few refinement epochs against the training embedding
+        heads = torch.arange(n_new, device=device).repeat_interleave(k)
+        tails = i.flatten() if isinstance(i, np.ndarray) else i.flatten()
+        tails = torch.from_numpy(knn_i.flatten()).to(device)
+        weights = torch.from_numpy(w.flatten()).to(device)
+        gen = torch.Generator(device=device)
+        gen.manual_seed(int(self.getOrDefault("random_state")))
+        new_emb = _optimize_embedding(
+            new_emb,
+            heads,
+            tails,
+            weights,
+            n_epochs=30,
+            a=self._model_attributes["a_"],
+            b=self._model_attributes["b_"],
+            lr=float(self.getOrDefault("learning_rate")),
+            neg_rate=int(self.getOrDefault("negative_sample_rate")),
+            repulsion=float(self.getOrDefault("repulsion_strength")),
+            gen=gen,
+            move_tail=False,
+            tail_emb=emb,
+        )
+        return as_numpy(new_emb)

@@ -1,0 +1,153 @@
+"""CrossValidator: k-fold CV with single-pass multi-model fit
+(reference tuning.py, 186 LoC).
+
+The reference fits ALL param maps in one data pass per fold via fitMultiple
+and evaluates them in one transform job (reference tuning.py:123-130);
+folds run on a driver thread pool. Here the SPMD runtime makes thread-level
+fold parallelism a collectives hazard (concurrent all-reduces would
+interleave), so folds run sequentially — the single-pass property per fold
+is preserved, which is where the reference's speedup comes from.
+"""
+
+from __future__ import annotations
+
+import numpy as np
+from typing import Any, Dict, List, Optional, Sequence, Tuple
+
+from .core import Estimator, Model, _Reader, _Writer
+from .data import DataFrame
+from .evaluation import Evaluator
+from .params import Param, Params, TypeConverters
+from .parallel.context import get_comm
+from .utils import get_logger
+
+
+class ParamGridBuilder:
+    """pyspark.ml.tuning.ParamGridBuilder equivalent."""
+
+    def __init__(self) -> None:
+        self._grid: Dict[Any, List[Any]] = {}
+
+    def addGrid(self, param: Any, values: List[Any]) -> "ParamGridBuilder":
+        self._grid[param] = list(values)
+        return self
+
+    def baseOn(self, *args: Tuple[Any, Any]) -> "ParamGridBuilder":
+        for p, v in args:
+            self._grid[p] = [v]
+        return self
+
+    def build(self) -> List[Dict[Any, Any]]:
+        import itertools
+
+        keys = list(self._grid.keys())
+        maps = []
+        for combo in itertools.product(*[self._grid[k] for k in keys]):
+            maps.append(dict(zip(keys, combo)))
+        return maps or [{}]
+
+
+class CrossValidator(Params):
+    """k-fold cross validation (reference CrossValidator, tuning.py:56+)."""
+
+    numFolds = Param("cv", "numFolds", "number of folds.", TypeConverters.toInt)
+    seed = Param("cv", "seed", "fold-split seed.", TypeConverters.toInt)
+    parallelism = Param("cv", "parallelism", "accepted for API parity; folds run sequentially (SPMD collectives).", TypeConverters.toInt)
+    collectSubModels = Param("cv", "collectSubModels", "keep all sub-models.", TypeConverters.toBoolean)
+
+    def __init__(
+        self,
+        estimator: Optional[Estimator] = None,
+        estimatorParamMaps: Optional[List[Dict[Any, Any]]] = None,
+        evaluator: Optional[Evaluator] = None,
+        numFolds: int = 3,
+        seed: int = 42,
+        parallelism: int = 1,
+        collectSubModels: bool = False,
+    ) -> None:
+        super().__init__()
+        self._setDefault(numFolds=3, seed=42, parallelism=1, collectSubModels=False)
+        self._set(numFolds=numFolds, seed=seed, parallelism=parallelism)
+        self._estimator = estimator
+        self._est_param_maps = estimatorParamMaps or [{}]
+        self._evaluator = evaluator
+
+    def setEstimator(self, est: Estimator) -> "CrossValidator":
+        self._estimator = est
+        return self
+
+    def setEstimatorParamMaps(self, maps: List[Dict[Any, Any]]) -> "CrossValidator":
+        self._est_param_maps = maps
+        return self
+
+    def setEvaluator(self, ev: Evaluator) -> "CrossValidator":
+        self._evaluator = ev
+        return self
+
+    def setNumFolds(self, value: int) -> "CrossValidator":
+        return self._set(numFolds=value)  # type: ignore[return-value]
+
+    def getEstimator(self) -> Estimator:
+        return self._estimator
+
+    def getEvaluator(self) -> Evaluator:
+        return self._evaluator
+
+    def getEstimatorParamMaps(self) -> List[Dict[Any, Any]]:
+        return self._est_param_maps
+
+    def _kfold(self, df: DataFrame) -> List[Tuple[DataFrame, DataFrame]]:
+        """Deterministic per-row fold assignment on each rank's local shard
+        (Spark assigns by rand(seed) — same semantics, shard-local)."""
+        comm = get_comm()
+        n_folds = self.getOrDefault("numFolds")
+        seed = self.getOrDefault("seed")
+        rng = np.random.default_rng(seed + 1000003 * comm.rank)
+        assign = rng.integers(0, n_folds, size=df.num_rows)
+        folds = []
+        for f in range(n_folds):
+            test_idx = np.nonzero(assign == f)[0]
+            train_idx = np.nonzero(assign != f)[0]
+            folds.append((df.take_local(train_idx), df.take_local(test_idx)))
+        return folds
+
+    def fit(self, df: DataFrame) -> "CrossValidatorModel":
+        assert self._estimator is not None and self._evaluator is not None
+        logger = get_logger(self.__class__)
+        epm = self._est_param_maps
+        n_folds = self.getOrDefault("numFolds")
+        metrics = np.zeros((len(epm), n_folds))
+        for fold, (train, test) in enumerate(self._kfold(df)):
+            # single data pass over all param maps (reference tuning.py:123-130)
+            for idx, model in self._estimator.fitMultiple(train, epm):
+                out = model.transform(test)
+                metrics[idx, fold] = self._evaluator.evaluate(out)
+        avg = metrics.mean(axis=1)
+        std = metrics.std(axis=1)
+        best = int(np.argmax(avg) if self._evaluator.isLargerBetter() else np.argmin(avg))
+        logger.info(f"CV best param map index {best}: avg metric {avg[best]:.6f}")
+        best_model = self._estimator.fit(df, epm[best])
+        cvm = CrossValidatorModel(
+            bestModel=best_model, avgMetrics=avg.tolist(), stdMetrics=std.tolist()
+        )
+        self._copyValues(cvm)
+        return cvm
+
+
+class CrossValidatorModel(Params):
+    """Holds the best model + per-param-map metrics (reference
+    tuning.py:141-157)."""
+
+    def __init__(
+        self,
+        bestModel: Optional[Model] = None,
+        avgMetrics: Optional[List[float]] = None,
+        stdMetrics: Optional[List[float]] = None,
+    ) -> None:
+        super().__init__()
+        self.bestModel = bestModel
+        self.avgMetrics = avgMetrics or []
+        self.stdMetrics = stdMetrics or []
+
+    def transform(self, df: DataFrame) -> DataFrame:
+        return self.bestModel.transform(df)

@@ -1,0 +1,112 @@
+"""KMeans correctness vs sklearn (pattern: reference tests/test_kmeans.py —
+generate blobs, fit, compare centers within tolerance)."""
+
+import numpy as np
+import pytest
+from sklearn.datasets import make_blobs
+
+from spark_rapids_ml_amd import KMeans, KMeansModel
+from spark_rapids_ml_amd.data import DataFrame
+
+from .dist_utils import run_distributed
+
+
+def _sorted_centers(C):
+    C = np.asarray(C)
+    order = np.lexsort(C.T[::-1])
+    return C[order]
+
+
+def _make(n=1000, d=8, k=4, seed=0):
+    X, y = make_blobs(n_samples=n, n_features=d, centers=k, cluster_std=0.5, random_state=seed)
+    return X.astype(np.float32), y
+
+
+def test_kmeans_matches_sklearn_blobs():
+    from sklearn.cluster import KMeans as SkKMeans
+
+    X, _ = _make()
+    df = DataFrame.from_numpy(X)
+    model = KMeans(k=4, maxIter=50, seed=3, tol=1e-6).fit(df)
+    sk = SkKMeans(n_clusters=4, n_init=10, random_state=0).fit(X)
+    ours = _sorted_centers(model.cluster_centers_)
+    theirs = _sorted_centers(sk.cluster_centers_)
+    assert np.allclose(ours, theirs, atol=0.05), f"{ours}\n!=\n{theirs}"
+    # inertia parity
+    assert model.trainingCost <= sk.inertia_ * 1.05
+
+
+def test_kmeans_random_init():
+    X, _ = _make()
+    df = DataFrame.from_numpy(X)
+    model = KMeans(k=4, maxIter=100, seed=5, initMode="random").fit(df)
+    assert model.cluster_centers_.shape == (4, 8)
+    assert model.trainingCost > 0
+
+
+def test_kmeans_transform_labels_consistent():
+    X, _ = _make()
+    df = DataFrame.from_numpy(X)
+    model = KMeans(k=4, maxIter=30, seed=1).fit(df)
+    out = model.transform(df)
+    labels = np.asarray(out["prediction"])
+    # assignment must equal nearest-center assignment
+    d = ((X[:, None, :] - model.cluster_centers_[None]) ** 2).sum(axis=2)
+    assert np.array_equal(labels, d.argmin(axis=1).astype(labels.dtype))
+
+
+def test_kmeans_predict_single_vector():
+    X, _ = _make()
+    model = KMeans(k=4, maxIter=20, seed=1).fit(DataFrame.from_numpy(X))
+    lab = model.predict(X[0])
+    assert 0 <= lab < 4
+
+
+def test_kmeans_persistence(tmp_model_path):
+    X, _ = _make(n=200)
+    model = KMeans(k=3, maxIter=10, seed=2).fit(DataFrame.from_numpy(X))
+    model.save(tmp_model_path)
+    loaded = KMeansModel.load(tmp_model_path)
+    assert np.allclose(loaded.cluster_centers_, model.cluster_centers_)
+    assert loaded.getOrDefault("k") == 3
+    out = loaded.transform(DataFrame.from_numpy(X))
+    assert "prediction" in out.columns
+
+
+def test_kmeans_estimator_persistence(tmp_model_path):
+    est = KMeans(k=5, maxIter=7)
+    est.save(tmp_model_path)
+    loaded = KMeans.load(tmp_model_path)
+    assert loaded.getOrDefault("k") == 5
+    assert loaded.native_params["max_iter"] == 7
+
+
+def test_kmeans_k_greater_than_rows_raises():
+    X = np.random.rand(3, 2).astype(np.float32)
+    with pytest.raises(ValueError):
+        KMeans(k=10, maxIter=5).fit(DataFrame.from_numpy(X))
+
+
+# -- distributed (gloo world_size=2) ----------------------------------------
+
+
+def _dist_kmeans_fit(seed: int):
+    from spark_rapids_ml_amd.parallel.context import get_comm
+
+    comm = get_comm()
+    X, _ = _make(n=1000, seed=seed)
+    shard = X[comm.rank :: comm.world_size]
+    model = KMeans(k=4, maxIter=50, seed=3, tol=1e-6).fit(DataFrame.from_numpy(shard))
+    return model.cluster_centers_, model.trainingCost
+
+
+def test_kmeans_distributed_matches_single():
+    results = run_distributed(_dist_kmeans_fit, world_size=2, args=(0,))
+    c0, cost0 = results[0]
+    c1, cost1 = results[1]
+    assert np.allclose(c0, c1)  # replicated model identical across ranks
+    X, _ = _make(n=1000, seed=0)
+    single = KMeans(k=4, maxIter=50, seed=3, tol=1e-6).fit(DataFrame.from_numpy(X))
+    assert np.allclose(
+        _sorted_centers(c0), _sorted_centers(single.cluster_centers_), atol=0.05
+    )

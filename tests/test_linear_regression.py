@@ -1,0 +1,142 @@
+"""LinearRegression vs sklearn / closed forms (pattern: reference
+tests/test_linear_model.py)."""
+
+import numpy as np
+import pytest
+from sklearn.linear_model import ElasticNet, LinearRegression as SkOLS, Ridge
+
+from spark_rapids_ml_amd import LinearRegression, LinearRegressionModel
+from spark_rapids_ml_amd.data import DataFrame
+
+from .dist_utils import run_distributed
+
+
+def _data(n=400, d=10, seed=0, noise=0.1):
+    rng = np.random.default_rng(seed)
+    X = rng.normal(size=(n, d)) * rng.uniform(0.5, 2.0, size=d)
+    w = rng.normal(size=d)
+    y = X @ w + 3.0 + noise * rng.normal(size=n)
+    return X.astype(np.float64), y.astype(np.float64), w
+
+
+def test_ols_matches_sklearn():
+    X, y, _ = _data()
+    df = DataFrame.from_numpy(X, y)
+    model = LinearRegression(regParam=0.0).fit(df)
+    sk = SkOLS().fit(X, y)
+    assert np.allclose(model.coefficients, sk.coef_, atol=1e-6)
+    assert np.isclose(model.intercept, sk.intercept_, atol=1e-6)
+
+
+def test_ols_no_intercept():
+    X, y, _ = _data()
+    model = LinearRegression(regParam=0.0, fitIntercept=False).fit(DataFrame.from_numpy(X, y))
+    sk = SkOLS(fit_intercept=False).fit(X, y)
+    assert np.allclose(model.coefficients, sk.coef_, atol=1e-6)
+    assert model.intercept == 0.0
+
+
+def test_ridge_matches_glmnet_objective():
+    """Our ridge solves min 1/(2n)||ỹ-X̃w̃||² + λ/2||w̃||² in standardized
+    space (Spark semantics). Verify against the closed form computed in
+    numpy directly."""
+    X, y, _ = _data(n=300, d=6, seed=1)
+    lam = 0.3
+    model = LinearRegression(regParam=lam, elasticNetParam=0.0).fit(
+        DataFrame.from_numpy(X, y)
+    )
+    n = len(y)
+    xbar, ybar = X.mean(0), y.mean()
+    sx = X.std(0, ddof=1)
+    sy = y.std(ddof=1)
+    Xs = (X - xbar) / sx
+    ys = (y - ybar) / sy
+    wt = np.linalg.solve(Xs.T @ Xs / n + lam * np.eye(X.shape[1]), Xs.T @ ys / n)
+    w = wt * sy / sx
+    b = ybar - w @ xbar
+    assert np.allclose(model.coefficients, w, atol=1e-6)
+    assert np.isclose(model.intercept, b, atol=1e-6)
+
+
+def test_elasticnet_matches_closed_objective():
+    """CD result must minimize the stated objective: compare objective value
+    against scikit-learn ElasticNet on the standardized problem."""
+    X, y, _ = _data(n=300, d=8, seed=2)
+    lam, alpha = 0.1, 0.5
+    model = LinearRegression(
+        regParam=lam, elasticNetParam=alpha, maxIter=500, tol=1e-9
+    ).fit(DataFrame.from_numpy(X, y))
+    n = len(y)
+    xbar, ybar = X.mean(0), y.mean()
+    sx = X.std(0, ddof=1)
+    sy = y.std(ddof=1)
+    Xs = (X - xbar) / sx
+    ys = (y - ybar) / sy
+    # sklearn ElasticNet: 1/(2n)||y-Xw||² + a*l1||w||₁ + a(1-l1)/2||w||²
+    sk = ElasticNet(alpha=lam, l1_ratio=alpha, fit_intercept=False, tol=1e-10, max_iter=100000).fit(
+        Xs, ys
+    )
+    w_expect = sk.coef_ * sy / sx
+    assert np.allclose(model.coefficients, w_expect, atol=1e-4)
+
+
+def test_sparsity_under_l1():
+    X, y, _ = _data(n=200, d=20, seed=3)
+    model = LinearRegression(regParam=0.5, elasticNetParam=1.0, maxIter=300).fit(
+        DataFrame.from_numpy(X, y)
+    )
+    assert (np.abs(model.coefficients) < 1e-10).sum() > 0
+
+
+def test_transform_prediction():
+    X, y, _ = _data()
+    model = LinearRegression().fit(DataFrame.from_numpy(X, y))
+    out = model.transform(DataFrame.from_numpy(X))
+    pred = np.asarray(out["prediction"])
+    expect = X @ model.coefficients + model.intercept
+    assert np.allclose(pred, expect, atol=1e-3)
+    # r2 close to 1 on low-noise data
+    ss_res = ((pred - y) ** 2).sum()
+    ss_tot = ((y - y.mean()) ** 2).sum()
+    assert 1 - ss_res / ss_tot > 0.98
+
+
+def test_persistence(tmp_model_path):
+    X, y, _ = _data(n=100)
+    model = LinearRegression(regParam=0.1).fit(DataFrame.from_numpy(X, y))
+    model.save(tmp_model_path)
+    loaded = LinearRegressionModel.load(tmp_model_path)
+    assert np.allclose(loaded.coefficients, model.coefficients)
+    assert np.isclose(loaded.intercept, model.intercept)
+
+
+def test_fit_multiple_single_pass():
+    X, y, _ = _data(n=200, d=5)
+    df = DataFrame.from_numpy(X, y)
+    est = LinearRegression()
+    maps = [
+        {est.getParam("regParam"): 0.0},
+        {est.getParam("regParam"): 0.5},
+    ]
+    models = dict(est.fitMultiple(df, maps))
+    assert len(models) == 2
+    assert not np.allclose(models[0].coefficients, models[1].coefficients)
+
+
+def _dist_ols(seed: int):
+    from spark_rapids_ml_amd.parallel.context import get_comm
+
+    comm = get_comm()
+    X, y, _ = _data(n=400, seed=seed)
+    sl = slice(comm.rank, None, comm.world_size)
+    model = LinearRegression().fit(DataFrame.from_numpy(X[sl], y[sl]))
+    return np.asarray(model.coefficients), model.intercept
+
+
+def test_ols_distributed_matches_single():
+    results = run_distributed(_dist_ols, world_size=2, args=(0,))
+    X, y, _ = _data(n=400, seed=0)
+    sk = SkOLS().fit(X, y)
+    for coef, icpt in results:
+        assert np.allclose(coef, sk.coef_, atol=1e-6)
+        assert np.isclose(icpt, sk.intercept_, atol=1e-6)

@@ -1,0 +1,169 @@
+"""LogisticRegression vs sklearn (pattern: reference
+tests/test_logistic_regression.py)."""
+
+import numpy as np
+import pytest
+from sklearn.datasets import make_classification
+from sklearn.linear_model import LogisticRegression as SkLR
+
+from spark_rapids_ml_amd import LogisticRegression, LogisticRegressionModel
+from spark_rapids_ml_amd.data import DataFrame
+
+from .dist_utils import run_distributed
+
+
+def _binary(n=500, d=8, seed=0):
+    X, y = make_classification(
+        n_samples=n, n_features=d, n_informative=d // 2, random_state=seed
+    )
+    return X.astype(np.float64), y.astype(np.float64)
+
+
+def _multi(n=600, d=10, k=3, seed=0):
+    X, y = make_classification(
+        n_samples=n,
+        n_features=d,
+        n_informative=d // 2,
+        n_classes=k,
+        n_clusters_per_class=1,
+        random_state=seed,
+    )
+    return X.astype(np.float64), y.astype(np.float64)
+
+
+def test_binary_no_reg_matches_sklearn():
+    X, y = _binary()
+    model = LogisticRegression(regParam=0.0, maxIter=200, tol=1e-10).fit(
+        DataFrame.from_numpy(X, y)
+    )
+    sk = SkLR(penalty=None, max_iter=2000, tol=1e-10).fit(X, y)
+
+    def mean_logloss(w, b):
+        z = X @ w + b
+        t = 2 * y - 1
+        return np.mean(np.logaddexp(0, -t * z))
+
+    ours = mean_logloss(model.coefficients, model.intercept)
+    theirs = mean_logloss(sk.coef_[0], sk.intercept_[0])
+    # unregularized optimum is flat: compare objective values, not coefs
+    assert ours <= theirs * (1 + 1e-3) + 1e-6
+    # agreement on predictions is the robust check
+    out = model.transform(DataFrame.from_numpy(X))
+    pred = np.asarray(out["prediction"])
+    assert (pred == sk.predict(X)).mean() > 0.99
+
+
+def test_binary_l2_matches_sklearn():
+    X, y = _binary(seed=1)
+    n = len(y)
+    lam = 0.1
+    # sklearn: min 0.5 w'w + C Σloss  <=>  ours(std=False): 1/n Σloss + λ/2 w'w
+    # => C = 1/(λ n)
+    model = LogisticRegression(
+        regParam=lam, standardization=False, maxIter=300, tol=1e-12
+    ).fit(DataFrame.from_numpy(X, y))
+    sk = SkLR(penalty="l2", C=1.0 / (lam * n), max_iter=5000, tol=1e-12).fit(X, y)
+    assert np.allclose(model.coefficients, sk.coef_[0], rtol=1e-3, atol=1e-4)
+    assert np.isclose(model.intercept, sk.intercept_[0], rtol=1e-3, atol=1e-4)
+
+
+def test_binary_l1_owlqn():
+    X, y = _binary(seed=2)
+    n = len(y)
+    lam = 0.05
+    model = LogisticRegression(
+        regParam=lam, elasticNetParam=1.0, standardization=False, maxIter=500, tol=1e-12
+    ).fit(DataFrame.from_numpy(X, y))
+    sk = SkLR(
+        penalty="l1", C=1.0 / (lam * n), solver="saga", max_iter=20000, tol=1e-10
+    ).fit(X, y)
+
+    def obj(w, b):
+        z = X @ w + b
+        t = 2 * y - 1
+        return np.mean(np.logaddexp(0, -t * z)) + lam * np.abs(w).sum()
+
+    # OWL-QN must reach at least saga's objective and produce sparsity
+    assert obj(model.coefficients, model.intercept) <= obj(sk.coef_[0], sk.intercept_[0]) * (
+        1 + 1e-3
+    )
+    assert (np.abs(model.coefficients) < 1e-8).sum() > 0
+    out = model.transform(DataFrame.from_numpy(X))
+    assert (np.asarray(out["prediction"]) == y).mean() > 0.8
+
+
+def test_multinomial_matches_sklearn():
+    X, y = _multi()
+    model = LogisticRegression(regParam=0.01, standardization=False, maxIter=300, tol=1e-10).fit(
+        DataFrame.from_numpy(X, y)
+    )
+    n = len(y)
+    sk = SkLR(penalty="l2", C=1.0 / (0.01 * n), max_iter=5000, tol=1e-10).fit(X, y)
+    assert model.numClasses == 3
+    out = model.transform(DataFrame.from_numpy(X))
+    agree = (np.asarray(out["prediction"]) == sk.predict(X)).mean()
+    assert agree > 0.98
+    # intercepts centered (reference classification.py:1135-1147)
+    assert abs(model.interceptVector.mean()) < 1e-8
+
+
+def test_probability_and_raw_outputs():
+    X, y = _binary()
+    model = LogisticRegression(maxIter=100).fit(DataFrame.from_numpy(X, y))
+    out = model.transform(DataFrame.from_numpy(X))
+    probs = np.asarray(out["probability"])
+    raw = np.asarray(out["rawPrediction"])
+    assert probs.shape == (len(y), 2)
+    assert np.allclose(probs.sum(axis=1), 1.0, atol=1e-6)
+    assert np.allclose(raw[:, 1], -raw[:, 0])
+
+
+def test_single_class_degenerate():
+    X = np.random.rand(50, 4)
+    y = np.ones(50)
+    model = LogisticRegression().fit(DataFrame.from_numpy(X, y))
+    assert np.isinf(model.interceptVector).any()
+    out = model.transform(DataFrame.from_numpy(X))
+    assert (np.asarray(out["prediction"]) == 1.0).all()
+
+
+def test_classes_sorted():
+    X, y = _binary()
+    y = y + 3  # labels {3,4}
+    model = LogisticRegression().fit(DataFrame.from_numpy(X, y))
+    assert np.array_equal(model.classes_, [3.0, 4.0])
+    out = model.transform(DataFrame.from_numpy(X))
+    assert set(np.unique(np.asarray(out["prediction"]))) <= {3.0, 4.0}
+
+
+def test_persistence(tmp_model_path):
+    X, y = _binary(n=200)
+    model = LogisticRegression(regParam=0.01).fit(DataFrame.from_numpy(X, y))
+    model.save(tmp_model_path)
+    loaded = LogisticRegressionModel.load(tmp_model_path)
+    assert np.allclose(loaded.coefficientMatrix, model.coefficientMatrix)
+    assert np.array_equal(loaded.classes_, model.classes_)
+
+
+def _dist_logreg(seed: int):
+    from spark_rapids_ml_amd.parallel.context import get_comm
+
+    comm = get_comm()
+    X, y = _binary(n=500, seed=seed)
+    sl = slice(comm.rank, None, comm.world_size)
+    model = LogisticRegression(regParam=0.01, maxIter=200, tol=1e-10).fit(
+        DataFrame.from_numpy(X[sl], y[sl])
+    )
+    return np.asarray(model.coefficients), model.intercept
+
+
+def test_logreg_distributed_matches_single():
+    results = run_distributed(_dist_logreg, world_size=2, args=(0,))
+    X, y = _binary(n=500, seed=0)
+    single = LogisticRegression(regParam=0.01, maxIter=200, tol=1e-10).fit(
+        DataFrame.from_numpy(X, y)
+    )
+    for coef, icpt in results:
+        # fp summation order differs between world sizes; optimizer-level agreement
+        assert np.allclose(coef, single.coefficients, rtol=1e-2, atol=1e-3)
+        assert np.isclose(icpt, single.intercept, rtol=1e-2, atol=1e-3)
