@@ -145,25 +145,33 @@ class Params:
         self._params: Optional[List[Param]] = None
         self._copy_params()
 
+    @classmethod
+    def _declared_params(cls) -> Dict[str, Param]:
+        """All Param declarations in the MRO, including ones shadowed by a
+        property on a more-derived class (e.g. a model exposing a same-named
+        attribute property)."""
+        found: Dict[str, Param] = {}
+        for klass in reversed(cls.__mro__):
+            for name, attr in vars(klass).items():
+                if isinstance(attr, Param):
+                    found[name] = attr
+        return found
+
     def _copy_params(self) -> None:
         """Bind class-level Param declarations to this instance."""
-        cls = type(self)
-        for name in dir(cls):
-            attr = getattr(cls, name, None)
-            if isinstance(attr, Param):
-                setattr(self, name, attr._copy_new_parent(self))
+        self._bound_params: Dict[str, Param] = {}
+        for name, attr in self._declared_params().items():
+            bound = attr._copy_new_parent(self)
+            self._bound_params[name] = bound
+            try:
+                setattr(self, name, bound)
+            except AttributeError:
+                pass  # shadowed by a read-only property; getParam still works
 
     @property
     def params(self) -> List[Param]:
         if self._params is None:
-            self._params = sorted(
-                [
-                    getattr(self, x)
-                    for x in dir(self)
-                    if x != "params" and isinstance(getattr(type(self), x, None), Param)
-                ],
-                key=lambda p: p.name,
-            )
+            self._params = sorted(self._bound_params.values(), key=lambda p: p.name)
         return self._params
 
     def hasParam(self, paramName: str) -> bool:

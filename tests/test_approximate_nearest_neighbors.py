@@ -1,0 +1,61 @@
+"""ANN recall tests (pattern: reference
+tests/test_approximate_nearest_neighbors.py)."""
+
+import numpy as np
+import pytest
+from sklearn.neighbors import NearestNeighbors as SkNN
+
+from spark_rapids_ml_amd import ApproximateNearestNeighbors
+from spark_rapids_ml_amd.data import DataFrame
+
+
+def _data(n=2000, d=16, seed=0):
+    rng = np.random.default_rng(seed)
+    return rng.normal(size=(n, d)).astype(np.float32)
+
+
+def _recall(idx, sk_idx):
+    hits = 0
+    for a, b in zip(idx, sk_idx):
+        hits += len(set(a.tolist()) & set(b.tolist()))
+    return hits / sk_idx.size
+
+
+@pytest.mark.parametrize("algo,params", [
+    ("ivfflat", {"nlist": 32, "nprobe": 8}),
+    ("ivfpq", {"nlist": 32, "nprobe": 8, "refine_ratio": 2.0}),
+])
+def test_ann_recall(algo, params):
+    X = _data()
+    Q = X[:100]
+    model = ApproximateNearestNeighbors(k=10, algorithm=algo, algoParams=params).fit(
+        DataFrame.from_numpy(X)
+    )
+    _, _, knn_df = model.kneighbors(DataFrame.from_numpy(Q))
+    idx = np.asarray(knn_df["indices"])
+    sk = SkNN(n_neighbors=10).fit(X)
+    _, sk_idx = sk.kneighbors(Q)
+    rec = _recall(idx, sk_idx)
+    assert rec > 0.7, f"{algo} recall {rec}"
+
+
+def test_ann_full_probe_is_exact():
+    X = _data(n=500)
+    Q = X[:50]
+    model = ApproximateNearestNeighbors(
+        k=5, algorithm="ivfflat", algoParams={"nlist": 8, "nprobe": 8}
+    ).fit(DataFrame.from_numpy(X))
+    _, _, knn_df = model.kneighbors(DataFrame.from_numpy(Q))
+    idx = np.asarray(knn_df["indices"])
+    sk = SkNN(n_neighbors=5).fit(X)
+    _, sk_idx = sk.kneighbors(Q)
+    assert _recall(idx, sk_idx) == 1.0
+
+
+def test_ann_unsupported_algo_raises():
+    X = _data(n=100)
+    model = ApproximateNearestNeighbors(k=3, algorithm="bogus").fit(
+        DataFrame.from_numpy(X)
+    )
+    with pytest.raises(ValueError):
+        model.kneighbors(DataFrame.from_numpy(X[:5]))

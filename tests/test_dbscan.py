@@ -1,0 +1,59 @@
+"""DBSCAN vs sklearn (pattern: reference tests/test_dbscan.py)."""
+
+import numpy as np
+import pytest
+from sklearn.cluster import DBSCAN as SkDBSCAN
+from sklearn.datasets import make_blobs
+from sklearn.metrics import adjusted_rand_score
+
+from spark_rapids_ml_amd import DBSCAN
+from spark_rapids_ml_amd.data import DataFrame
+
+from .dist_utils import run_distributed
+
+
+def _data(n=300, seed=0):
+    X, y = make_blobs(n_samples=n, n_features=4, centers=3, cluster_std=0.4, random_state=seed)
+    return X.astype(np.float32), y
+
+
+def test_dbscan_matches_sklearn():
+    X, _ = _data()
+    model = DBSCAN(eps=1.0, min_samples=5).fit(DataFrame.from_numpy(X))
+    out = model.transform(DataFrame.from_numpy(X))
+    labels = np.asarray(out["prediction"])
+    sk = SkDBSCAN(eps=1.0, min_samples=5).fit(X)
+    assert adjusted_rand_score(labels, sk.labels_) == 1.0
+    # same noise set
+    assert np.array_equal(labels == -1, sk.labels_ == -1)
+
+
+def test_dbscan_noise_detection():
+    X, _ = _data(n=200)
+    X = np.vstack([X, np.full((3, 4), 50.0, dtype=np.float32)])  # outliers
+    model = DBSCAN(eps=1.0, min_samples=5).fit(DataFrame.from_numpy(X))
+    out = model.transform(DataFrame.from_numpy(X))
+    labels = np.asarray(out["prediction"])
+    assert (labels[-3:] == -1).all()
+
+
+def _dist_dbscan(seed: int):
+    from spark_rapids_ml_amd.parallel.context import get_comm
+
+    comm = get_comm()
+    X, _ = _data(n=300, seed=seed)
+    shard = X[comm.rank :: comm.world_size]
+    model = DBSCAN(eps=1.0, min_samples=5).fit(DataFrame.from_numpy(shard))
+    out = model.transform(DataFrame.from_numpy(shard))
+    return np.asarray(out["prediction"])
+
+
+def test_dbscan_distributed_matches_sklearn():
+    results = run_distributed(_dist_dbscan, world_size=2, args=(0,))
+    X, _ = _data(n=300, seed=0)
+    sk = SkDBSCAN(eps=1.0, min_samples=5).fit(X)
+    # reassemble interleaved shards
+    labels = np.empty(len(X))
+    labels[0::2] = results[0]
+    labels[1::2] = results[1]
+    assert adjusted_rand_score(labels, sk.labels_) == 1.0

@@ -1,0 +1,84 @@
+"""Metrics parity vs sklearn (pattern: reference tests/test_metrics.py)."""
+
+import numpy as np
+import pytest
+from sklearn import metrics as skm
+
+from spark_rapids_ml_amd.metrics import MulticlassMetrics, RegressionMetrics
+from spark_rapids_ml_amd.metrics.RegressionMetrics import _SummarizerBuffer
+
+
+def _conf(y_true, y_pred):
+    conf = {}
+    for l, p in zip(y_true, y_pred):
+        conf[(float(l), float(p))] = conf.get((float(l), float(p)), 0.0) + 1.0
+    return conf
+
+
+def test_multiclass_metrics_match_sklearn():
+    rng = np.random.default_rng(0)
+    y_true = rng.integers(0, 3, 500).astype(float)
+    y_pred = np.where(rng.random(500) < 0.7, y_true, rng.integers(0, 3, 500)).astype(float)
+    m = MulticlassMetrics.from_confusion(_conf(y_true, y_pred))
+    assert np.isclose(m.evaluate("accuracy"), skm.accuracy_score(y_true, y_pred))
+    assert np.isclose(m.evaluate("f1"), skm.f1_score(y_true, y_pred, average="weighted"))
+    assert np.isclose(
+        m.evaluate("weightedPrecision"),
+        skm.precision_score(y_true, y_pred, average="weighted"),
+    )
+    assert np.isclose(
+        m.evaluate("weightedRecall"), skm.recall_score(y_true, y_pred, average="weighted")
+    )
+    assert np.isclose(m.evaluate("hammingLoss"), skm.hamming_loss(y_true, y_pred))
+
+
+def test_multiclass_merge_associative():
+    rng = np.random.default_rng(1)
+    y_true = rng.integers(0, 4, 300).astype(float)
+    y_pred = rng.integers(0, 4, 300).astype(float)
+    whole = MulticlassMetrics.from_confusion(_conf(y_true, y_pred))
+    a = MulticlassMetrics.from_confusion(_conf(y_true[:100], y_pred[:100]))
+    b = MulticlassMetrics.from_confusion(_conf(y_true[100:], y_pred[100:]))
+    merged = a.merge(b)
+    assert np.isclose(merged.evaluate("f1"), whole.evaluate("f1"))
+    assert np.isclose(merged.evaluate("accuracy"), whole.evaluate("accuracy"))
+
+
+def test_log_loss():
+    rng = np.random.default_rng(2)
+    y = rng.integers(0, 2, 200).astype(float)
+    probs = rng.random((200, 2))
+    probs = probs / probs.sum(axis=1, keepdims=True)
+    ll_local = float(-np.log(np.clip(probs[np.arange(200), y.astype(int)], 1e-15, 1)).sum())
+    m = MulticlassMetrics.from_confusion(_conf(y, probs.argmax(1).astype(float)), ll_local)
+    assert np.isclose(m.evaluate("logLoss"), skm.log_loss(y, probs))
+
+
+def test_regression_metrics_match_sklearn():
+    rng = np.random.default_rng(3)
+    y = rng.normal(size=400)
+    pred = y + 0.3 * rng.normal(size=400)
+    m = RegressionMetrics.from_predictions(y, pred)
+    assert np.isclose(m.evaluate("mse"), skm.mean_squared_error(y, pred))
+    assert np.isclose(m.evaluate("rmse"), np.sqrt(skm.mean_squared_error(y, pred)))
+    assert np.isclose(m.evaluate("mae"), skm.mean_absolute_error(y, pred))
+    assert np.isclose(m.evaluate("r2"), skm.r2_score(y, pred))
+
+
+def test_regression_merge_matches_whole():
+    rng = np.random.default_rng(4)
+    y = rng.normal(size=300)
+    pred = y + 0.2 * rng.normal(size=300)
+    whole = RegressionMetrics.from_predictions(y, pred)
+    a = RegressionMetrics.from_predictions(y[:120], pred[:120])
+    b = RegressionMetrics.from_predictions(y[120:], pred[120:])
+    merged = a.merge(b)
+    for name in ("rmse", "mse", "mae", "r2"):
+        assert np.isclose(merged.evaluate(name), whole.evaluate(name)), name
+
+
+def test_summarizer_buffer_roundtrip():
+    buf = _SummarizerBuffer([1.0, 2.0], [0.1, 0.2], [1.1, 2.2], [3.0, 4.0], 10)
+    back = _SummarizerBuffer.from_list(buf.to_list())
+    assert back.total_count == 10
+    assert back._curr_mean == [1.0, 2.0]

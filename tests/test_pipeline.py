@@ -1,0 +1,47 @@
+"""Pipeline + VectorAssembler bypass (pattern: reference
+tests_no_import_change + pipeline tests)."""
+
+import numpy as np
+import pytest
+from sklearn.datasets import make_classification
+
+from spark_rapids_ml_amd import LogisticRegression
+from spark_rapids_ml_amd.data import DataFrame
+from spark_rapids_ml_amd.pipeline import NoOpTransformer, Pipeline, VectorAssembler
+
+
+def _scalar_df(n=300, d=4, seed=0):
+    X, y = make_classification(n_samples=n, n_features=d, n_informative=3, n_redundant=0, random_state=seed)
+    cols = {f"c{i}": X[:, i].astype(np.float64) for i in range(d)}
+    cols["label"] = y.astype(np.float64)
+    return DataFrame(cols), X, y
+
+
+def test_vector_assembler():
+    df, X, _ = _scalar_df()
+    va = VectorAssembler(inputCols=["c0", "c1", "c2", "c3"], outputCol="features")
+    out = va.transform(df)
+    assert np.allclose(np.asarray(out["features"]), X.astype(np.float32), atol=1e-5)
+
+
+def test_pipeline_bypasses_assembler():
+    df, X, y = _scalar_df()
+    va = VectorAssembler(inputCols=["c0", "c1", "c2", "c3"], outputCol="features")
+    lr = LogisticRegression(maxIter=50)
+    pipe = Pipeline(stages=[va, lr])
+    model = pipe.fit(df)
+    # assembler replaced by NoOp; estimator consumed featuresCols directly
+    assert isinstance(model.stages[0], NoOpTransformer)
+    assert lr.getFeaturesCols() == ["c0", "c1", "c2", "c3"]
+    out = model.transform(df)
+    acc = (np.asarray(out["prediction"]) == y).mean()
+    assert acc > 0.65
+
+
+def test_pipeline_without_assembler():
+    df, X, y = _scalar_df()
+    df = df.with_column("features", X.astype(np.float32))
+    lr = LogisticRegression(maxIter=50)
+    model = Pipeline(stages=[lr]).fit(df)
+    out = model.transform(df)
+    assert (np.asarray(out["prediction"]) == y).mean() > 0.65

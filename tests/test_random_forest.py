@@ -1,0 +1,145 @@
+"""RandomForest classifier/regressor (pattern: reference
+tests/test_random_forest.py — accuracy/r2 parity vs sklearn)."""
+
+import numpy as np
+import pytest
+from sklearn.datasets import make_classification, make_regression
+from sklearn.ensemble import (
+    RandomForestClassifier as SkRFC,
+    RandomForestRegressor as SkRFR,
+)
+
+from spark_rapids_ml_amd import (
+    RandomForestClassifier,
+    RandomForestClassificationModel,
+    RandomForestRegressor,
+    RandomForestRegressionModel,
+)
+from spark_rapids_ml_amd.data import DataFrame
+
+from .dist_utils import run_distributed
+
+
+def _cls_data(n=600, d=10, seed=0):
+    X, y = make_classification(
+        n_samples=n, n_features=d, n_informative=6, random_state=seed
+    )
+    return X.astype(np.float32), y.astype(np.float64)
+
+
+def _reg_data(n=600, d=10, seed=0):
+    X, y = make_regression(n_samples=n, n_features=d, n_informative=6, noise=5.0, random_state=seed)
+    return X.astype(np.float32), y.astype(np.float64)
+
+
+def test_rfc_accuracy_close_to_sklearn():
+    X, y = _cls_data()
+    model = RandomForestClassifier(numTrees=30, maxDepth=8, maxBins=64, seed=0).fit(
+        DataFrame.from_numpy(X, y)
+    )
+    out = model.transform(DataFrame.from_numpy(X))
+    acc = (np.asarray(out["prediction"]) == y).mean()
+    sk = SkRFC(n_estimators=30, max_depth=8, random_state=0).fit(X, y)
+    sk_acc = (sk.predict(X) == y).mean()
+    assert acc > sk_acc - 0.05, f"ours {acc} vs sklearn {sk_acc}"
+    probs = np.asarray(out["probability"])
+    assert probs.shape == (len(y), 2)
+    assert np.allclose(probs.sum(axis=1), 1.0, atol=1e-5)
+
+
+def test_rfr_r2_close_to_sklearn():
+    X, y = _reg_data()
+    model = RandomForestRegressor(numTrees=30, maxDepth=8, maxBins=64, seed=0).fit(
+        DataFrame.from_numpy(X, y)
+    )
+    out = model.transform(DataFrame.from_numpy(X))
+    pred = np.asarray(out["prediction"])
+    r2 = 1 - ((pred - y) ** 2).sum() / ((y - y.mean()) ** 2).sum()
+    sk = SkRFR(n_estimators=30, max_depth=8, random_state=0).fit(X, y)
+    sk_r2 = sk.score(X, y)
+    assert r2 > sk_r2 - 0.1, f"ours {r2} vs sklearn {sk_r2}"
+
+
+def test_rfc_multiclass():
+    X, y = make_classification(
+        n_samples=500,
+        n_features=8,
+        n_informative=6,
+        n_classes=3,
+        n_clusters_per_class=1,
+        random_state=1,
+    )
+    X = X.astype(np.float32)
+    y = y.astype(np.float64)
+    model = RandomForestClassifier(numTrees=20, maxDepth=8, seed=1).fit(
+        DataFrame.from_numpy(X, y)
+    )
+    assert model.numClasses == 3
+    out = model.transform(DataFrame.from_numpy(X))
+    acc = (np.asarray(out["prediction"]) == y).mean()
+    assert acc > 0.85
+
+
+def test_rfc_bad_labels_raise():
+    X = np.random.rand(50, 4).astype(np.float32)
+    y = np.array([0.5] * 50)
+    with pytest.raises(ValueError):
+        RandomForestClassifier(numTrees=2).fit(DataFrame.from_numpy(X, y))
+
+
+def test_rf_num_trees_and_depth():
+    X, y = _cls_data(n=200)
+    model = RandomForestClassifier(numTrees=7, maxDepth=3, seed=0).fit(
+        DataFrame.from_numpy(X, y)
+    )
+    assert model.numTrees == 7
+    assert model.numFeatures == 10
+    for t in model.trees:
+        assert t["feature"].shape == t["left"].shape
+        # depth-3 tree has at most 2^4-1 nodes
+        assert t["feature"].shape[0] <= 15
+
+
+def test_rf_persistence(tmp_model_path):
+    X, y = _cls_data(n=200)
+    model = RandomForestClassifier(numTrees=5, maxDepth=4, seed=0).fit(
+        DataFrame.from_numpy(X, y)
+    )
+    model.save(tmp_model_path)
+    loaded = RandomForestClassificationModel.load(tmp_model_path)
+    assert loaded.numTrees == 5
+    o1 = model.transform(DataFrame.from_numpy(X))
+    o2 = loaded.transform(DataFrame.from_numpy(X))
+    assert np.array_equal(np.asarray(o1["prediction"]), np.asarray(o2["prediction"]))
+
+
+def test_rf_dump_as_json():
+    X, y = _cls_data(n=100)
+    model = RandomForestClassifier(numTrees=2, maxDepth=3, seed=0).fit(
+        DataFrame.from_numpy(X, y)
+    )
+    import json
+
+    dumped = json.loads(model.dump_as_json())
+    assert len(dumped) == 2
+    assert "feature" in dumped[0]
+
+
+def _dist_rf(seed: int):
+    from spark_rapids_ml_amd.parallel.context import get_comm
+
+    comm = get_comm()
+    X, y = _cls_data(n=600, seed=seed)
+    sl = slice(comm.rank, None, comm.world_size)
+    model = RandomForestClassifier(numTrees=10, maxDepth=6, seed=0).fit(
+        DataFrame.from_numpy(X[sl], y[sl])
+    )
+    out = model.transform(DataFrame.from_numpy(X, y))
+    return model.numTrees, (np.asarray(out["prediction"]) == y).mean()
+
+
+def test_rf_distributed_trees_split_across_ranks():
+    results = run_distributed(_dist_rf, world_size=2, args=(0,))
+    for n_trees, acc in results:
+        assert n_trees == 10  # merged forest has all trees
+        assert acc > 0.85

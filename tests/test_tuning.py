@@ -1,0 +1,78 @@
+"""CrossValidator + evaluators (pattern: reference tests/test_tuning.py)."""
+
+import numpy as np
+import pytest
+from sklearn.datasets import make_classification, make_regression
+
+from spark_rapids_ml_amd import LinearRegression, LogisticRegression
+from spark_rapids_ml_amd.data import DataFrame
+from spark_rapids_ml_amd.evaluation import (
+    BinaryClassificationEvaluator,
+    MulticlassClassificationEvaluator,
+    RegressionEvaluator,
+)
+from spark_rapids_ml_amd.tuning import CrossValidator, CrossValidatorModel, ParamGridBuilder
+
+
+def test_param_grid_builder():
+    lr = LinearRegression()
+    grid = (
+        ParamGridBuilder()
+        .addGrid(lr.getParam("regParam"), [0.0, 0.1])
+        .addGrid(lr.getParam("elasticNetParam"), [0.0, 0.5])
+        .build()
+    )
+    assert len(grid) == 4
+
+
+def test_cv_regression_selects_best():
+    X, y = make_regression(n_samples=400, n_features=8, noise=1.0, random_state=0)
+    df = DataFrame.from_numpy(X.astype(np.float64), y.astype(np.float64))
+    lr = LinearRegression()
+    grid = ParamGridBuilder().addGrid(lr.getParam("regParam"), [0.0, 100.0]).build()
+    cv = CrossValidator(
+        estimator=lr,
+        estimatorParamMaps=grid,
+        evaluator=RegressionEvaluator(metricName="rmse"),
+        numFolds=3,
+        seed=7,
+    )
+    model = cv.fit(df)
+    assert len(model.avgMetrics) == 2
+    # unregularized fits clean linear data far better than λ=100
+    assert model.avgMetrics[0] < model.avgMetrics[1]
+    assert model.bestModel.getOrDefault("regParam") == 0.0
+    out = model.transform(df)
+    assert "prediction" in out.columns
+
+
+def test_cv_classification():
+    X, y = make_classification(n_samples=400, n_features=8, random_state=0)
+    df = DataFrame.from_numpy(X.astype(np.float64), y.astype(np.float64))
+    lr = LogisticRegression(maxIter=50)
+    grid = ParamGridBuilder().addGrid(lr.getParam("regParam"), [0.01, 10.0]).build()
+    cv = CrossValidator(
+        estimator=lr,
+        estimatorParamMaps=grid,
+        evaluator=MulticlassClassificationEvaluator(metricName="f1"),
+        numFolds=3,
+    )
+    model = cv.fit(df)
+    assert model.bestModel.getOrDefault("regParam") == 0.01
+
+
+def test_evaluators_basic():
+    rng = np.random.default_rng(0)
+    y = rng.integers(0, 2, 200).astype(np.float64)
+    scores = np.where(y == 1, 0.8, 0.2) + 0.1 * rng.random(200)
+    df = DataFrame(
+        {
+            "label": y,
+            "prediction": (scores > 0.5).astype(np.float64),
+            "rawPrediction": np.stack([-scores, scores], axis=1),
+        }
+    )
+    auc = BinaryClassificationEvaluator().evaluate(df)
+    assert auc > 0.9
+    acc = MulticlassClassificationEvaluator(metricName="accuracy").evaluate(df)
+    assert acc > 0.9

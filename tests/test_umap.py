@@ -1,0 +1,62 @@
+"""UMAP embedding quality (pattern: reference tests/test_umap.py uses
+trustworthiness)."""
+
+import numpy as np
+import pytest
+from sklearn.datasets import make_blobs
+from sklearn.manifold import trustworthiness
+
+from spark_rapids_ml_amd import UMAP, UMAPModel
+from spark_rapids_ml_amd.data import DataFrame
+
+
+def _data(n=400, d=10, seed=0):
+    X, y = make_blobs(n_samples=n, n_features=d, centers=5, cluster_std=0.5, random_state=seed)
+    return X.astype(np.float32), y
+
+
+def test_umap_trustworthiness():
+    X, y = _data()
+    model = UMAP(n_neighbors=15, n_epochs=200, random_state=42).fit(DataFrame.from_numpy(X))
+    emb = model.embedding
+    assert emb.shape == (len(X), 2)
+    t = trustworthiness(X, emb, n_neighbors=15)
+    assert t > 0.8, f"trustworthiness {t}"
+
+
+def test_umap_transform_places_near_train():
+    X, y = _data(n=300)
+    model = UMAP(n_neighbors=10, n_epochs=150, random_state=1).fit(DataFrame.from_numpy(X))
+    out = model.transform(DataFrame.from_numpy(X[:50]))
+    emb_new = np.asarray(out["embedding"])
+    assert emb_new.shape == (50, 2)
+    # transformed points should land near their own training embedding
+    d = np.linalg.norm(emb_new - model.embedding[:50], axis=1)
+    spread = np.linalg.norm(model.embedding.max(0) - model.embedding.min(0))
+    assert np.median(d) < spread * 0.25
+
+
+def test_umap_sample_fraction():
+    X, y = _data(n=400)
+    model = UMAP(n_neighbors=10, n_epochs=100, sample_fraction=0.5, random_state=3).fit(
+        DataFrame.from_numpy(X)
+    )
+    assert model.rawData.shape[0] == 200
+
+
+def test_umap_persistence(tmp_model_path):
+    X, _ = _data(n=150)
+    model = UMAP(n_neighbors=8, n_epochs=80, random_state=5).fit(DataFrame.from_numpy(X))
+    model.save(tmp_model_path)
+    loaded = UMAPModel.load(tmp_model_path)
+    assert np.allclose(loaded.embedding, model.embedding)
+    assert np.allclose(loaded.rawData, model.rawData)
+
+
+def test_find_ab_params():
+    from spark_rapids_ml_amd.models.umap import find_ab_params
+
+    a, b = find_ab_params(1.0, 0.1)
+    # umap-learn reference values for spread=1, min_dist=0.1
+    assert abs(a - 1.577) < 0.05
+    assert abs(b - 0.895) < 0.05
