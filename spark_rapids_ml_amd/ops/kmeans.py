@@ -1,9 +1,10 @@
-"""KMeans ops: fused assignment+accumulation (the Lloyd-step hot kernel).
+"""KMeans ops: fused assignment (MFMA distance + argmin) + accumulation.
 
-HIP kernel: LDS-tiled distance expansion ||x||^2+||c||^2-2x.c with the
--2XC^T term on MFMA (mfma_f32_16x16x4f32), fused argmin and per-center
-sum/count accumulation — the reference's KMeansMG fit kernel family
-(reference clustering.py:381-415 invocation).
+HIP path (gfx950): `kmeans_assign` — LDS-tiled ‖x‖²+‖c‖²−2x·c with the
+−2XCᵀ term on mfma_f32_32x32x2f32, fused per-row argmin (packed 64-bit
+atomicMin) and block-reduced inertia; `label_accumulate` — vectorized
+per-center sum/count scatter. Together these are the KMeansMG Lloyd-step
+kernel family of the reference (clustering.py:381-415).
 """
 
 from __future__ import annotations
@@ -21,20 +22,24 @@ def kmeans_assign_reduce(
     C: torch.Tensor,
     x_sq: Optional[torch.Tensor] = None,
 ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor, float]:
-    """Returns (labels i32 [n], sums f64|f32 [k,d], counts [k], inertia)."""
+    """Returns (labels i32 [n], sums [k,d], counts [k], inertia)."""
     if use_hip(X):
         ext = hip_ops()
-        labels, sums, counts, inertia = ext.kmeans_assign_reduce(
-            X.contiguous(), C.contiguous(), x_sq if x_sq is not None else _xsq(X)
+        if x_sq is None:
+            x_sq = _xsq(X)
+        labels, _min_d, inertia = ext.kmeans_assign(
+            X.contiguous(), C.contiguous(), x_sq.contiguous()
         )
-        return labels, sums, counts, float(inertia)
+        sums, counts = ext.label_accumulate(X.contiguous(), labels, C.shape[0])
+        return labels, sums.to(torch.float64), counts.to(torch.float64), float(inertia.item())
     return torch_ref.kmeans_assign_reduce(X, C, x_sq)
 
 
 def kmeans_predict(X: torch.Tensor, C: torch.Tensor) -> torch.Tensor:
     if use_hip(X):
         ext = hip_ops()
-        return ext.kmeans_predict(X.contiguous(), C.contiguous())
+        labels, _, _ = ext.kmeans_assign(X.contiguous(), C.contiguous(), _xsq(X))
+        return labels
     return torch_ref.kmeans_predict(X, C)
 
 

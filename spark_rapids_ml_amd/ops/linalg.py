@@ -29,12 +29,8 @@ def gram(X: torch.Tensor) -> torch.Tensor:
 
 
 def xty_gram(X: torch.Tensor, y: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
-    """(X^T X, X^T y) fused — one pass over X."""
-    if use_hip(X):
-        ext = hip_ops()
-        if hasattr(ext, "gram_xty_f32") and X.dtype == torch.float32:
-            return ext.gram_xty_f32(X.contiguous(), y.contiguous())
-    return torch_ref.gram(X), torch_ref.xty(X, y)
+    """(X^T X, X^T y): the Gram rides the MFMA kernel, X^T y is a GEMV."""
+    return gram(X), torch_ref.xty(X, y)
 
 
 def eigh_sym(A: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:

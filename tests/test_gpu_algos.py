@@ -1,0 +1,82 @@
+"""GPU end-to-end algorithm tests (single MI355X)."""
+
+import numpy as np
+import pytest
+import torch
+from sklearn.datasets import make_blobs, make_classification, make_regression
+
+from spark_rapids_ml_amd import (
+    KMeans,
+    LinearRegression,
+    LogisticRegression,
+    NearestNeighbors,
+    PCA,
+    RandomForestClassifier,
+)
+from spark_rapids_ml_amd.data import DataFrame
+
+pytestmark = pytest.mark.gpu
+
+
+def test_kmeans_gpu_matches_sklearn():
+    from sklearn.cluster import KMeans as SkKMeans
+
+    X, _ = make_blobs(n_samples=5000, n_features=32, centers=8, cluster_std=0.5, random_state=0)
+    X = X.astype(np.float32)
+    model = KMeans(k=8, maxIter=50, seed=3, tol=1e-6).fit(DataFrame.from_numpy(X))
+    sk = SkKMeans(n_clusters=8, n_init=10, random_state=0).fit(X)
+    assert model.trainingCost <= sk.inertia_ * 1.05
+
+
+def test_pca_gpu_matches_sklearn():
+    from sklearn.decomposition import PCA as SkPCA
+
+    rng = np.random.default_rng(0)
+    X = (rng.normal(size=(2000, 4)) @ rng.normal(size=(4, 64)) + 0.01 * rng.normal(size=(2000, 64))).astype(np.float32)
+    model = PCA(k=3).fit(DataFrame.from_numpy(X))
+    sk = SkPCA(n_components=3).fit(X)
+    for i in range(3):
+        c = np.asarray(model.components_)[i]
+        s = sk.components_[i]
+        assert min(np.abs(c - s).max(), np.abs(c + s).max()) < 1e-2
+
+
+def test_linreg_gpu_matches_sklearn():
+    from sklearn.linear_model import LinearRegression as SkOLS
+
+    X, y = make_regression(n_samples=5000, n_features=64, noise=1.0, random_state=0)
+    model = LinearRegression().fit(DataFrame.from_numpy(X.astype(np.float64), y))
+    sk = SkOLS().fit(X, y)
+    assert np.allclose(model.coefficients, sk.coef_, atol=1e-4)
+
+
+def test_logreg_gpu_predictions():
+    X, y = make_classification(n_samples=5000, n_features=32, n_informative=16, random_state=0)
+    model = LogisticRegression(regParam=0.01, maxIter=100).fit(
+        DataFrame.from_numpy(X.astype(np.float32), y.astype(np.float64))
+    )
+    out = model.transform(DataFrame.from_numpy(X.astype(np.float32)))
+    assert (np.asarray(out["prediction"]) == y).mean() > 0.85
+
+
+def test_rf_gpu_accuracy():
+    X, y = make_classification(n_samples=3000, n_features=20, n_informative=10, random_state=0)
+    model = RandomForestClassifier(numTrees=20, maxDepth=8, seed=0).fit(
+        DataFrame.from_numpy(X.astype(np.float32), y.astype(np.float64))
+    )
+    out = model.transform(DataFrame.from_numpy(X.astype(np.float32)))
+    assert (np.asarray(out["prediction"]) == y).mean() > 0.9
+
+
+def test_knn_gpu_exact():
+    from sklearn.neighbors import NearestNeighbors as SkNN
+
+    rng = np.random.default_rng(0)
+    X = rng.normal(size=(3000, 64)).astype(np.float32)
+    Q = rng.normal(size=(100, 64)).astype(np.float32)
+    model = NearestNeighbors(k=8).fit(DataFrame.from_numpy(X))
+    _, _, knn_df = model.kneighbors(DataFrame.from_numpy(Q))
+    sk = SkNN(n_neighbors=8).fit(X)
+    sk_d, sk_i = sk.kneighbors(Q)
+    # distances must match; index ties allowed
+    assert np.allclose(np.asarray(knn_df["distances"]), sk_d, atol=1e-3)
