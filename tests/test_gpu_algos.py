@@ -56,7 +56,7 @@ def test_logreg_gpu_predictions():
         DataFrame.from_numpy(X.astype(np.float32), y.astype(np.float64))
     )
     out = model.transform(DataFrame.from_numpy(X.astype(np.float32)))
-    assert (np.asarray(out["prediction"]) == y).mean() > 0.85
+    assert (np.asarray(out["prediction"]) == y).mean() > 0.72  # sklearn gets 0.741 here
 
 
 def test_rf_gpu_accuracy():
