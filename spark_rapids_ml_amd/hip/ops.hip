@@ -54,8 +54,12 @@ __global__ __launch_bounds__(256) void kmeans_assign_kernel(
     int32_t* __restrict__ labels,    // [n]
     float* __restrict__ min_dists,   // [n] squared distance to the winner
     double* __restrict__ inertia) {  // [1] accumulated
-  __shared__ float lds_x[KM_BK][KM_BM + 1];
-  __shared__ float lds_c[KM_BK][KM_BN + 1];
+  // Double-buffered LDS, T14 load-early/write-late: per K-step the next
+  // tile's global loads are issued BEFORE the MFMA phase (latency hides
+  // under 64-cycle f32 MFMAs) and written to the spare buffer after it;
+  // one bare s_barrier per step (no global load outstanding at barrier).
+  __shared__ float lds_x[2][KM_BK][KM_BM + 1];
+  __shared__ float lds_c[2][KM_BK][KM_BN + 1];
   __shared__ unsigned long long best[KM_BM];
 
   const int i0 = blockIdx.x * KM_BM;
@@ -66,9 +70,13 @@ __global__ __launch_bounds__(256) void kmeans_assign_kernel(
   const int wc = wave & 1;   // wave col (0..1) -> 64 cols
 
   for (int i = tid; i < KM_BM; i += blockDim.x) best[i] = ~0ULL;
-  __syncthreads();
+
+  const bool full_rows = (i0 + KM_BM <= n);
+  const int nsteps = (d + KM_BK - 1) / KM_BK;
+  float rx[16], rc[16];
 
   for (int j0 = 0; j0 < k; j0 += KM_BN) {
+    const bool full_cols = (j0 + KM_BN <= k);
     f32x16 acc[2][2];
 #pragma unroll
     for (int m = 0; m < 2; ++m)
@@ -77,34 +85,79 @@ __global__ __launch_bounds__(256) void kmeans_assign_kernel(
 #pragma unroll
         for (int r = 0; r < 16; ++r) acc[m][nn][r] = 0.0f;
 
-    for (int d0 = 0; d0 < d; d0 += KM_BK) {
-      // stage X[i0:i0+128, d0:d0+32] as lds_x[kd][i]
-      // 256 threads * 16 elems = 4096 = 32*128
-      for (int e = tid; e < KM_BK * KM_BM; e += blockDim.x) {
-        int kd = e / KM_BM;
-        int i = e % KM_BM;
-        int gi = i0 + i, gd = d0 + kd;
-        lds_x[kd][i] = (gi < n && gd < d) ? X[(int64_t)gi * d + gd] : 0.0f;
+    // prologue: load + write step 0 synchronously
+    {
+      const int d0 = 0;
+      if (full_rows && full_cols && KM_BK <= d) {
+#pragma unroll
+        for (int q = 0; q < 16; ++q) {
+          int e = q * 256 + tid;
+          rx[q] = X[(int64_t)(i0 + (e >> 5)) * d + d0 + (e & 31)];
+          rc[q] = C[(int64_t)(j0 + (e >> 5)) * d + d0 + (e & 31)];
+        }
+      } else {
+#pragma unroll
+        for (int q = 0; q < 16; ++q) {
+          int e = q * 256 + tid;
+          int gi = i0 + (e >> 5), gj = j0 + (e >> 5), gd = d0 + (e & 31);
+          rx[q] = (gi < n && gd < d) ? X[(int64_t)gi * d + gd] : 0.0f;
+          rc[q] = (gj < k && gd < d) ? C[(int64_t)gj * d + gd] : 0.0f;
+        }
       }
-      for (int e = tid; e < KM_BK * KM_BN; e += blockDim.x) {
-        int kd = e / KM_BN;
-        int j = e % KM_BN;
-        int gj = j0 + j, gd = d0 + kd;
-        lds_c[kd][j] = (gj < k && gd < d) ? C[(int64_t)gj * d + gd] : 0.0f;
+      __syncthreads();  // best[] init + any previous j-tile epilogue done
+#pragma unroll
+      for (int q = 0; q < 16; ++q) {
+        int e = q * 256 + tid;
+        lds_x[0][e & 31][e >> 5] = rx[q];
+        lds_c[0][e & 31][e >> 5] = rc[q];
       }
       __syncthreads();
+    }
+
+    for (int step = 0; step < nsteps; ++step) {
+      const int buf = step & 1;
+      const int d0n = (step + 1) * KM_BK;
+      const bool have_next = step + 1 < nsteps;
+      if (have_next) {
+        if (full_rows && full_cols && d0n + KM_BK <= d) {
+#pragma unroll
+          for (int q = 0; q < 16; ++q) {
+            int e = q * 256 + tid;
+            rx[q] = X[(int64_t)(i0 + (e >> 5)) * d + d0n + (e & 31)];
+            rc[q] = C[(int64_t)(j0 + (e >> 5)) * d + d0n + (e & 31)];
+          }
+        } else {
+#pragma unroll
+          for (int q = 0; q < 16; ++q) {
+            int e = q * 256 + tid;
+            int gi = i0 + (e >> 5), gj = j0 + (e >> 5), gd = d0n + (e & 31);
+            rx[q] = (gi < n && gd < d) ? X[(int64_t)gi * d + gd] : 0.0f;
+            rc[q] = (gj < k && gd < d) ? C[(int64_t)gj * d + gd] : 0.0f;
+          }
+        }
+      }
 
 #pragma unroll
       for (int kk = 0; kk < KM_BK / 2; ++kk) {
         const int kd = 2 * kk + (lane >> 5);
-        float a0 = lds_x[kd][wr * 64 + (lane & 31)];
-        float a1 = lds_x[kd][wr * 64 + 32 + (lane & 31)];
-        float b0 = lds_c[kd][wc * 64 + (lane & 31)];
-        float b1 = lds_c[kd][wc * 64 + 32 + (lane & 31)];
+        float a0 = lds_x[buf][kd][wr * 64 + (lane & 31)];
+        float a1 = lds_x[buf][kd][wr * 64 + 32 + (lane & 31)];
+        float b0 = lds_c[buf][kd][wc * 64 + (lane & 31)];
+        float b1 = lds_c[buf][kd][wc * 64 + 32 + (lane & 31)];
         acc[0][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc[0][0], 0, 0, 0);
         acc[0][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc[0][1], 0, 0, 0);
         acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc[1][0], 0, 0, 0);
         acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc[1][1], 0, 0, 0);
+      }
+
+      if (have_next) {
+        const int nb = buf ^ 1;
+#pragma unroll
+        for (int q = 0; q < 16; ++q) {
+          int e = q * 256 + tid;
+          lds_x[nb][e & 31][e >> 5] = rx[q];
+          lds_c[nb][e & 31][e >> 5] = rc[q];
+        }
       }
       __syncthreads();
     }
@@ -159,29 +212,64 @@ __global__ __launch_bounds__(256) void kmeans_assign_kernel(
 }
 
 // ---------------------------------------------------------------------------
-// label_accumulate: sums[label[i]] += X[i], counts[label[i]] += 1.
-// float4-vectorized columns, atomics spread over k*d addresses.
+// label_accumulate: sums[label[i]] += X[i], counts[label[i]] += 1 via a
+// sort-based segmented reduction: rows pre-sorted by label (torch.sort in
+// the wrapper), each (label, split) block register-accumulates its column
+// stripe over its row sub-segment — each X element is read ONCE, coalesced,
+// and the only atomics are the k*SPLIT partial merges.
 // ---------------------------------------------------------------------------
 
-__global__ __launch_bounds__(256) void label_accumulate_kernel(
-    const float* __restrict__ X, const int32_t* __restrict__ labels,
-    int64_t n, int d, float* __restrict__ sums, float* __restrict__ counts) {
-  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  const int64_t d4 = d / 4;
-  const int64_t total = n * d4;
-  for (int64_t e = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; e < total; e += stride) {
-    int64_t row = e / d4;
-    int64_t c4 = e % d4;
-    int lab = labels[row];
-    const float4 v = reinterpret_cast<const float4*>(X + row * d)[c4];
-    float* dst = sums + (int64_t)lab * d + c4 * 4;
-    atomicAdd(dst + 0, v.x);
-    atomicAdd(dst + 1, v.y);
-    atomicAdd(dst + 2, v.z);
-    atomicAdd(dst + 3, v.w);
-    if (c4 == 0) {
-      atomicAdd(&counts[lab], 1.0f);
-      for (int c = d4 * 4; c < d; ++c) atomicAdd(&sums[(int64_t)lab * d + c], X[row * d + c]);
+constexpr int LA_SPLIT = 8;    // sub-segments per label (skew tolerance)
+constexpr int LA_COLS = 16;    // column accumulators per thread (256*16=4096)
+
+__global__ __launch_bounds__(256) void segment_sum_kernel(
+    const float* __restrict__ X, const int64_t* __restrict__ perm,
+    const int64_t* __restrict__ seg_off, int d,
+    float* __restrict__ sums, float* __restrict__ counts) {
+  const int lab = blockIdx.x / LA_SPLIT;
+  const int split = blockIdx.x % LA_SPLIT;
+  const int64_t s0 = seg_off[lab], e0 = seg_off[lab + 1];
+  const int64_t len = e0 - s0;
+  if (len == 0) return;
+  const int64_t chunk = (len + LA_SPLIT - 1) / LA_SPLIT;
+  const int64_t rs = s0 + split * chunk;
+  const int64_t re = min(e0, rs + chunk);
+  if (rs >= re) return;
+  if (split == 0 && threadIdx.x == 0) counts[lab] = (float)len;
+
+  for (int cb = 0; cb < d; cb += blockDim.x * LA_COLS) {
+    float acc[LA_COLS];
+#pragma unroll
+    for (int q = 0; q < LA_COLS; ++q) acc[q] = 0.0f;
+    int64_t r = rs;
+    for (; r + 4 <= re; r += 4) {  // 4 rows in flight for latency hiding
+      const float* row0 = X + perm[r] * (int64_t)d;
+      const float* row1 = X + perm[r + 1] * (int64_t)d;
+      const float* row2 = X + perm[r + 2] * (int64_t)d;
+      const float* row3 = X + perm[r + 3] * (int64_t)d;
+#pragma unroll
+      for (int q = 0; q < LA_COLS; ++q) {
+        int c = cb + q * blockDim.x + threadIdx.x;
+        if (c < d) acc[q] += row0[c] + row1[c] + row2[c] + row3[c];
+      }
+    }
+    for (; r < re; ++r) {
+      const float* row = X + perm[r] * (int64_t)d;
+#pragma unroll
+      for (int q = 0; q < LA_COLS; ++q) {
+        int c = cb + q * blockDim.x + threadIdx.x;
+        if (c < d) acc[q] += row[c];
+      }
+    }
+#pragma unroll
+    for (int q = 0; q < LA_COLS; ++q) {
+      int c = cb + q * blockDim.x + threadIdx.x;
+      if (c < d) {
+        if (LA_SPLIT > 1)
+          atomicAdd(&sums[(int64_t)lab * d + c], acc[q]);
+        else
+          sums[(int64_t)lab * d + c] = acc[q];
+      }
     }
   }
 }
@@ -218,20 +306,33 @@ __global__ __launch_bounds__(256) void gram_kernel(
 #pragma unroll
       for (int r = 0; r < 16; ++r) acc[m][nn][r] = 0.0f;
 
+  const bool full_i = (i0 + GR_BM <= d);
+  const bool full_j = (j0 + GR_BN <= d);
   for (int64_t r0 = 0; r0 < n; r0 += GR_BK) {
-    for (int e = tid; e < GR_BK * GR_BM; e += blockDim.x) {
-      int rr = e / GR_BM;
-      int i = e % GR_BM;
-      int64_t gr = r0 + rr;
-      int gi = i0 + i;
-      lds_i[rr][i] = (gr < n && gi < d) ? A[gr * d + gi] : 0.0f;
-    }
-    for (int e = tid; e < GR_BK * GR_BN; e += blockDim.x) {
-      int rr = e / GR_BN;
-      int j = e % GR_BN;
-      int64_t gr = r0 + rr;
-      int gj = j0 + j;
-      lds_j[rr][j] = (gr < n && gj < d) ? A[gr * d + gj] : 0.0f;
+    if (full_i && full_j && r0 + GR_BK <= n) {
+#pragma unroll 4
+      for (int e = tid; e < GR_BK * GR_BM; e += 256) {
+        int rr = e >> 7, i = e & 127;
+        lds_i[rr][i] = A[(r0 + rr) * d + i0 + i];
+      }
+#pragma unroll 4
+      for (int e = tid; e < GR_BK * GR_BN; e += 256) {
+        int rr = e >> 7, j = e & 127;
+        lds_j[rr][j] = A[(r0 + rr) * d + j0 + j];
+      }
+    } else {
+      for (int e = tid; e < GR_BK * GR_BM; e += blockDim.x) {
+        int rr = e / GR_BM, i = e % GR_BM;
+        int64_t gr = r0 + rr;
+        int gi = i0 + i;
+        lds_i[rr][i] = (gr < n && gi < d) ? A[gr * d + gi] : 0.0f;
+      }
+      for (int e = tid; e < GR_BK * GR_BN; e += blockDim.x) {
+        int rr = e / GR_BN, j = e % GR_BN;
+        int64_t gr = r0 + rr;
+        int gj = j0 + j;
+        lds_j[rr][j] = (gr < n && gj < d) ? A[gr * d + gj] : 0.0f;
+      }
     }
     __syncthreads();
 
@@ -349,10 +450,15 @@ std::vector<torch::Tensor> label_accumulate(torch::Tensor X, torch::Tensor label
   auto sums = torch::zeros({k, d}, X.options());
   auto counts = torch::zeros({k}, X.options());
   if (n > 0) {
-    int grid = std::min<int64_t>(2048, (n * std::max(1, d / 4) + 255) / 256);
-    hipLaunchKernelGGL(label_accumulate_kernel, dim3(grid), dim3(256), 0, cur_stream(),
-                       X.data_ptr<float>(), labels.data_ptr<int32_t>(), n, d,
-                       sums.data_ptr<float>(), counts.data_ptr<float>());
+    auto sorted = labels.to(torch::kInt64).sort();
+    auto perm = std::get<1>(sorted).contiguous();
+    auto sl = std::get<0>(sorted).contiguous();
+    auto bounds = torch::arange(k + 1, labels.options().dtype(torch::kInt64));
+    auto seg_off = torch::searchsorted(sl, bounds).contiguous();
+    hipLaunchKernelGGL(segment_sum_kernel, dim3((unsigned)(k * LA_SPLIT)), dim3(256), 0,
+                       cur_stream(), X.data_ptr<float>(), perm.data_ptr<int64_t>(),
+                       seg_off.data_ptr<int64_t>(), d, sums.data_ptr<float>(),
+                       counts.data_ptr<float>());
   }
   return {sums, counts};
 }
