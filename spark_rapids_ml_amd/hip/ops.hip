@@ -43,7 +43,8 @@ using f32x16 = __attribute__((ext_vector_type(16))) float;
 
 constexpr int KM_BM = 128;
 constexpr int KM_BN = 128;
-constexpr int KM_BK = 32;
+constexpr int KM_BK = 64;
+constexpr int KM_LD = KM_BK * KM_BM / 256;  // staging elems per thread (32)
 
 __global__ __launch_bounds__(256) void kmeans_assign_kernel(
     const float* __restrict__ X,     // [n,d] row-major
@@ -54,12 +55,13 @@ __global__ __launch_bounds__(256) void kmeans_assign_kernel(
     int32_t* __restrict__ labels,    // [n]
     float* __restrict__ min_dists,   // [n] squared distance to the winner
     double* __restrict__ inertia) {  // [1] accumulated
-  // Double-buffered LDS, T14 load-early/write-late: per K-step the next
-  // tile's global loads are issued BEFORE the MFMA phase (latency hides
-  // under 64-cycle f32 MFMAs) and written to the spare buffer after it;
-  // one bare s_barrier per step (no global load outstanding at barrier).
-  __shared__ float lds_x[2][KM_BK][KM_BM + 1];
-  __shared__ float lds_c[2][KM_BK][KM_BN + 1];
+  // Single-buffer BK=64 with write-after-barrier register pipelining (T14):
+  // per K-step 128 MFMAs (8192 issue cycles/wave) run between two barriers;
+  // the next tile's registers are written right after the first barrier and
+  // the tile after that's global loads are issued immediately, so HBM
+  // latency hides under the MFMA phase. 2 blocks/CU (67 KB LDS).
+  __shared__ float lds_x[KM_BK][KM_BM + 1];
+  __shared__ float lds_c[KM_BK][KM_BN + 1];
   __shared__ unsigned long long best[KM_BM];
 
   const int i0 = blockIdx.x * KM_BM;
@@ -73,7 +75,7 @@ __global__ __launch_bounds__(256) void kmeans_assign_kernel(
 
   const bool full_rows = (i0 + KM_BM <= n);
   const int nsteps = (d + KM_BK - 1) / KM_BK;
-  float rx[16], rc[16];
+  float rx[KM_LD], rc[KM_LD];
 
   for (int j0 = 0; j0 < k; j0 += KM_BN) {
     const bool full_cols = (j0 + KM_BN <= k);
@@ -85,82 +87,70 @@ __global__ __launch_bounds__(256) void kmeans_assign_kernel(
 #pragma unroll
         for (int r = 0; r < 16; ++r) acc[m][nn][r] = 0.0f;
 
-    // prologue: load + write step 0 synchronously
-    {
-      const int d0 = 0;
-      if (full_rows && full_cols && KM_BK <= d) {
-#pragma unroll
-        for (int q = 0; q < 16; ++q) {
-          int e = q * 256 + tid;
-          rx[q] = X[(int64_t)(i0 + (e >> 5)) * d + d0 + (e & 31)];
-          rc[q] = C[(int64_t)(j0 + (e >> 5)) * d + d0 + (e & 31)];
-        }
-      } else {
-#pragma unroll
-        for (int q = 0; q < 16; ++q) {
-          int e = q * 256 + tid;
-          int gi = i0 + (e >> 5), gj = j0 + (e >> 5), gd = d0 + (e & 31);
-          rx[q] = (gi < n && gd < d) ? X[(int64_t)gi * d + gd] : 0.0f;
-          rc[q] = (gj < k && gd < d) ? C[(int64_t)gj * d + gd] : 0.0f;
-        }
-      }
-      __syncthreads();  // best[] init + any previous j-tile epilogue done
-#pragma unroll
-      for (int q = 0; q < 16; ++q) {
-        int e = q * 256 + tid;
-        lds_x[0][e & 31][e >> 5] = rx[q];
-        lds_c[0][e & 31][e >> 5] = rc[q];
-      }
-      __syncthreads();
+    // loads: KM_LD=32 elems/thread; e = q*256+tid over 8192 = 128 rows x 64 kd
+#define KM_LOAD_T(dst, SRC, base_row, d0)                                      \
+  do {                                                                         \
+    if (full_rows && full_cols && (d0) + KM_BK <= d) {                         \
+      _Pragma("unroll") for (int q = 0; q < KM_LD; ++q) {                      \
+        int e = q * 256 + tid;                                                 \
+        dst[q] = SRC[(int64_t)((base_row) + (e >> 6)) * d + (d0) + (e & 63)];  \
+      }                                                                        \
+    } else {                                                                   \
+      _Pragma("unroll") for (int q = 0; q < KM_LD; ++q) {                      \
+        int e = q * 256 + tid;                                                 \
+        int gr = (base_row) + (e >> 6);                                        \
+        int gd = (d0) + (e & 63);                                              \
+        int lim = (&dst[0] == &rx[0]) ? n : k;                                 \
+        dst[q] = (gr < lim && gd < d) ? SRC[(int64_t)gr * d + gd] : 0.0f;      \
+      }                                                                        \
+    }                                                                          \
+  } while (0)
+
+#define KM_WRITE()                                                             \
+  do {                                                                         \
+    _Pragma("unroll") for (int q = 0; q < KM_LD; ++q) {                        \
+      int e = q * 256 + tid;                                                   \
+      lds_x[e & 63][e >> 6] = rx[q];                                           \
+      lds_c[e & 63][e >> 6] = rc[q];                                           \
+    }                                                                          \
+  } while (0)
+
+    // prologue: tile 0 into LDS, tile 1 into regs
+    KM_LOAD_T(rx, X, i0, 0);
+    KM_LOAD_T(rc, C, j0, 0);
+    __syncthreads();  // best[] init / previous epilogue complete
+    KM_WRITE();
+    if (nsteps > 1) {
+      KM_LOAD_T(rx, X, i0, KM_BK);
+      KM_LOAD_T(rc, C, j0, KM_BK);
     }
+    __syncthreads();
 
     for (int step = 0; step < nsteps; ++step) {
-      const int buf = step & 1;
-      const int d0n = (step + 1) * KM_BK;
-      const bool have_next = step + 1 < nsteps;
-      if (have_next) {
-        if (full_rows && full_cols && d0n + KM_BK <= d) {
-#pragma unroll
-          for (int q = 0; q < 16; ++q) {
-            int e = q * 256 + tid;
-            rx[q] = X[(int64_t)(i0 + (e >> 5)) * d + d0n + (e & 31)];
-            rc[q] = C[(int64_t)(j0 + (e >> 5)) * d + d0n + (e & 31)];
-          }
-        } else {
-#pragma unroll
-          for (int q = 0; q < 16; ++q) {
-            int e = q * 256 + tid;
-            int gi = i0 + (e >> 5), gj = j0 + (e >> 5), gd = d0n + (e & 31);
-            rx[q] = (gi < n && gd < d) ? X[(int64_t)gi * d + gd] : 0.0f;
-            rc[q] = (gj < k && gd < d) ? C[(int64_t)gj * d + gd] : 0.0f;
-          }
-        }
-      }
-
-#pragma unroll
+#pragma unroll 8
       for (int kk = 0; kk < KM_BK / 2; ++kk) {
         const int kd = 2 * kk + (lane >> 5);
-        float a0 = lds_x[buf][kd][wr * 64 + (lane & 31)];
-        float a1 = lds_x[buf][kd][wr * 64 + 32 + (lane & 31)];
-        float b0 = lds_c[buf][kd][wc * 64 + (lane & 31)];
-        float b1 = lds_c[buf][kd][wc * 64 + 32 + (lane & 31)];
+        float a0 = lds_x[kd][wr * 64 + (lane & 31)];
+        float a1 = lds_x[kd][wr * 64 + 32 + (lane & 31)];
+        float b0 = lds_c[kd][wc * 64 + (lane & 31)];
+        float b1 = lds_c[kd][wc * 64 + 32 + (lane & 31)];
         acc[0][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc[0][0], 0, 0, 0);
         acc[0][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc[0][1], 0, 0, 0);
         acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc[1][0], 0, 0, 0);
         acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc[1][1], 0, 0, 0);
       }
-
-      if (have_next) {
-        const int nb = buf ^ 1;
-#pragma unroll
-        for (int q = 0; q < 16; ++q) {
-          int e = q * 256 + tid;
-          lds_x[nb][e & 31][e >> 5] = rx[q];
-          lds_c[nb][e & 31][e >> 5] = rc[q];
+      __syncthreads();  // all waves done reading this tile
+      if (step + 1 < nsteps) {
+        KM_WRITE();  // vmcnt-waits for loads issued a full MFMA phase ago
+        if (step + 2 < nsteps) {
+          KM_LOAD_T(rx, X, i0, (step + 2) * KM_BK);
+          KM_LOAD_T(rc, C, j0, (step + 2) * KM_BK);
         }
+        __syncthreads();
       }
-      __syncthreads();
     }
+#undef KM_LOAD_T
+#undef KM_WRITE
 
     // epilogue: distances + packed argmin into LDS
 #pragma unroll
