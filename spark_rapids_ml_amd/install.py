@@ -1,0 +1,96 @@
+"""No-import-change acceleration (reference install.py:22-81).
+
+The reference installs proxy modules over `pyspark.ml.*` so unmodified
+pyspark scripts pick up the GPU classes. The analog here: scripts written
+against the REFERENCE package (`import spark_rapids_ml` / its submodules
+feature, clustering, classification, regression, knn, umap, tuning,
+pipeline) run unmodified on this framework — `install_aliases()` registers
+`spark_rapids_ml` (and, when pyspark is absent, `pyspark.ml`-shaped alias
+modules) in sys.modules pointing at the spark_rapids_ml_amd classes.
+"""
+
+from __future__ import annotations
+
+import sys
+import types
+from typing import Dict, List
+
+
+_SUBMODULE_EXPORTS: Dict[str, List[str]] = {
+    "feature": ["PCA", "PCAModel"],
+    "clustering": ["KMeans", "KMeansModel", "DBSCAN", "DBSCANModel"],
+    "classification": [
+        "LogisticRegression",
+        "LogisticRegressionModel",
+        "RandomForestClassifier",
+        "RandomForestClassificationModel",
+    ],
+    "regression": [
+        "LinearRegression",
+        "LinearRegressionModel",
+        "RandomForestRegressor",
+        "RandomForestRegressionModel",
+    ],
+    "knn": [
+        "NearestNeighbors",
+        "NearestNeighborsModel",
+        "ApproximateNearestNeighbors",
+        "ApproximateNearestNeighborsModel",
+    ],
+    "umap": ["UMAP", "UMAPModel"],
+    "tuning": ["CrossValidator", "CrossValidatorModel"],
+    "pipeline": ["Pipeline", "PipelineModel", "NoOpTransformer"],
+    "evaluation": [
+        "RegressionEvaluator",
+        "MulticlassClassificationEvaluator",
+        "BinaryClassificationEvaluator",
+    ],
+}
+
+
+def _make_module(name: str, exports: Dict[str, object]) -> types.ModuleType:
+    mod = types.ModuleType(name)
+    for k, v in exports.items():
+        setattr(mod, k, v)
+    mod.__all__ = list(exports)
+    return mod
+
+
+def install_aliases(alias_root: str = "spark_rapids_ml") -> None:
+    """Register `spark_rapids_ml` alias modules over this package."""
+    import spark_rapids_ml_amd as root
+
+    from .tuning import ParamGridBuilder
+
+    root_exports = {name: getattr(root, name) for name in root.__all__}
+    root_mod = _make_module(alias_root, root_exports)
+    sys.modules[alias_root] = root_mod
+
+    for sub, names in _SUBMODULE_EXPORTS.items():
+        exports = {}
+        for n in names:
+            if hasattr(root, n):
+                exports[n] = getattr(root, n)
+        if sub == "tuning":
+            exports["ParamGridBuilder"] = ParamGridBuilder
+        if sub == "pipeline":
+            from .pipeline import VectorAssembler
+
+            exports["VectorAssembler"] = VectorAssembler
+        if sub == "evaluation":
+            from . import evaluation
+
+            for n in names:
+                exports[n] = getattr(evaluation, n)
+        m = _make_module(f"{alias_root}.{sub}", exports)
+        sys.modules[f"{alias_root}.{sub}"] = m
+        setattr(root_mod, sub, m)
+
+
+def main() -> None:  # console-script parity with reference install.py
+    install_aliases()
+    print("spark_rapids_ml -> spark_rapids_ml_amd aliases installed")
+
+
+if __name__ == "__main__":
+    main()

@@ -173,7 +173,23 @@ class LogisticRegression(_LogisticRegressionParams, Estimator):
 
         n = pdesc.m
         d = pdesc.n
-        Xt = to_device_tensor(np.ascontiguousarray(X), ctx.device)
+        from ..data import _is_sparse
+
+        sparse = _is_sparse(X)
+        if sparse:
+            # CSR path (reference classification.py:960-966: int64 index
+            # escape only above 1e9 nnz; torch CSR uses int64 throughout)
+            import scipy.sparse as sp
+
+            Xcsr = X.tocsr()
+            Xt = torch.sparse_csr_tensor(
+                torch.from_numpy(Xcsr.indptr.astype(np.int64)),
+                torch.from_numpy(Xcsr.indices.astype(np.int64)),
+                torch.from_numpy(Xcsr.data.astype(np.float32)),
+                size=Xcsr.shape,
+            ).to(ctx.device)
+        else:
+            Xt = to_device_tensor(np.ascontiguousarray(X), ctx.device)
 
         # classes: global sorted unique labels (reference allgathers classes_)
         local_classes = np.unique(np.asarray(y)) if len(np.asarray(y)) else np.array([])
@@ -208,7 +224,17 @@ class LogisticRegression(_LogisticRegressionParams, Estimator):
         # feature scale (no centering: Spark's sparsity-preserving scaling)
         sigma = self._column_std(Xt, comm, n)
         sig_safe = torch.where(sigma > 0, sigma, torch.ones_like(sigma))
-        Xs = Xt / sig_safe[None, :].to(Xt.dtype)
+        XsT = None
+        if sparse:
+            inv = (1.0 / sig_safe).to(torch.float32)
+            vals = Xt.values() * inv[Xt.col_indices()]
+            Xs = torch.sparse_csr_tensor(
+                Xt.crow_indices(), Xt.col_indices(), vals, size=Xt.shape
+            )
+            # pre-transpose once for the per-iteration XᵀR SpMM
+            XsT = Xs.t().to_sparse_csr()
+        else:
+            Xs = Xt / sig_safe[None, :].to(Xt.dtype)
 
         ncol = d + (1 if fit_intercept else 0)
         w0 = torch.zeros(C_out * ncol, dtype=torch.float64, device=ctx.device)
@@ -221,8 +247,8 @@ class LogisticRegression(_LogisticRegressionParams, Estimator):
         l1 = lam * l1r
 
         def closure(wv: torch.Tensor) -> Tuple[float, torch.Tensor]:
-            W = wv.view(C_out, ncol).to(Xs.dtype)
-            grad, loss = logistic_grad_loss(Xs, y_t, W, fit_intercept)
+            W = wv.view(C_out, ncol).to(torch.float32 if sparse else Xs.dtype)
+            grad, loss = logistic_grad_loss(Xs, y_t, W, fit_intercept, XT=XsT)
             buf = torch.zeros(C_out * ncol + 1, dtype=torch.float64, device=wv.device)
             buf[:-1] = grad.to(torch.float64).flatten()
             buf[-1] = loss.to(torch.float64)
@@ -282,9 +308,15 @@ class LogisticRegression(_LogisticRegressionParams, Estimator):
     def _column_std(self, Xt: torch.Tensor, comm, n: int) -> torch.Tensor:
         buf = torch.zeros((2, Xt.shape[1]), dtype=torch.float64, device=Xt.device)
         if Xt.shape[0] > 0:
-            Xd = Xt.to(torch.float64)
-            buf[0] = Xd.sum(dim=0)
-            buf[1] = (Xd * Xd).sum(dim=0)
+            if Xt.layout == torch.sparse_csr:
+                vals = Xt.values().to(torch.float64)
+                cols = Xt.col_indices()
+                buf[0].index_add_(0, cols, vals)
+                buf[1].index_add_(0, cols, vals * vals)
+            else:
+                Xd = Xt.to(torch.float64)
+                buf[0] = Xd.sum(dim=0)
+                buf[1] = (Xd * Xd).sum(dim=0)
         comm.allreduce(comm.to_coll(buf))
         buf = buf.to(Xt.device)
         mean = buf[0] / n
@@ -362,10 +394,20 @@ class LogisticRegressionModel(_LogisticRegressionParams, Model):
         return self._set_params(threshold=value)
 
     def _transform_array(self, X: Any) -> Dict[str, np.ndarray]:
+        from ..data import _is_sparse
         from ..parallel.context import get_comm
 
         device = get_comm().device
-        Xt = to_device_tensor(np.ascontiguousarray(X), device)
+        if _is_sparse(X):
+            Xc = X.tocsr()
+            Xt = torch.sparse_csr_tensor(
+                torch.from_numpy(Xc.indptr.astype(np.int64)),
+                torch.from_numpy(Xc.indices.astype(np.int64)),
+                torch.from_numpy(Xc.data.astype(np.float32)),
+                size=Xc.shape,
+            ).to(device)
+        else:
+            Xt = to_device_tensor(np.ascontiguousarray(X), device)
         coef = torch.from_numpy(
             np.ascontiguousarray(self._model_attributes["coef_"])
         ).to(device, Xt.dtype)

@@ -21,26 +21,48 @@ def logistic_grad_loss(
     y_idx: torch.Tensor,
     W: torch.Tensor,
     fit_intercept: bool,
+    XT: Optional[torch.Tensor] = None,
 ) -> Tuple[torch.Tensor, torch.Tensor]:
     """Local-shard gradient (sum, unscaled) and loss (sum) for binary
-    (W: [1, d(+1)], sigmoid) or multinomial (W: [C, d(+1)], softmax)."""
-    if use_hip(X):
+    (W: [1, d(+1)], sigmoid) or multinomial (W: [C, d(+1)], softmax).
+
+    Dense X runs the GEMMs on MFMA (hipBLASLt) with the fused HIP
+    softmax-residual kernel between them. Sparse CSR X (reference sparse
+    path, classification.py:960-966) runs scores = X@Wᵀ and grad = (XᵀR)ᵀ
+    through rocSPARSE SpMM; pass XT (the pre-transposed CSR) to avoid a
+    transpose per iteration.
+    """
+    sparse = X.layout == torch.sparse_csr
+    n = X.shape[0]
+    d = X.shape[1]
+    C = W.shape[0]
+    coef = W[:, :d]
+    if sparse:
+        scores = torch.sparse.mm(X, coef.T.contiguous())
+    else:
+        scores = X @ coef.T
+    if fit_intercept:
+        scores = scores + W[:, d][None, :]
+
+    if use_hip(scores):
         ext = hip_ops()
-        if hasattr(ext, "softmax_residual_loss"):
-            n, d = X.shape
-            C = W.shape[0]
-            coef = W[:, :d]
-            scores = X @ coef.T
-            if fit_intercept:
-                scores = scores + W[:, d][None, :]
-            resid, loss = ext.softmax_residual_loss(scores.contiguous(), y_idx.contiguous())
-            grad_coef = resid.T @ X
-            if fit_intercept:
-                grad = torch.cat([grad_coef, resid.sum(dim=0)[:, None]], dim=1)
-            else:
-                grad = grad_coef
-            return grad, loss
-    return torch_ref.logistic_forward_grad(X, y_idx, W, fit_intercept)
+        resid, loss = ext.softmax_residual_loss(
+            scores.contiguous().to(torch.float32), y_idx.contiguous()
+        )
+        resid = resid.to(scores.dtype)
+    else:
+        resid, loss = torch_ref.softmax_residual(scores, y_idx)
+
+    if sparse:
+        assert XT is not None, "sparse path requires pre-transposed CSR"
+        grad_coef = torch.sparse.mm(XT, resid).T
+    else:
+        grad_coef = resid.T @ X
+    if fit_intercept:
+        grad = torch.cat([grad_coef, resid.sum(dim=0)[:, None]], dim=1)
+    else:
+        grad = grad_coef
+    return grad, loss
 
 
 def linear_grad_loss(

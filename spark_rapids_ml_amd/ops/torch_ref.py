@@ -104,6 +104,28 @@ def knn_topk(
     return dists, idx
 
 
+def softmax_residual(
+    scores: torch.Tensor, y_idx: torch.Tensor
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    """(resid [n,C], loss sum) — sigmoid when C==1 else softmax."""
+    n, C = scores.shape[0], scores.shape[1] if scores.dim() > 1 else 1
+    if C == 1:
+        z = scores[:, 0]
+        t = y_idx.to(z.dtype) * 2.0 - 1.0
+        loss = torch.nn.functional.softplus(-t * z).sum()
+        resid = (torch.sigmoid(z) - y_idx.to(z.dtype))[:, None]
+    else:
+        logp = torch.log_softmax(scores, dim=1)
+        loss = -logp.gather(1, y_idx.view(-1, 1).to(torch.int64)).sum()
+        resid = torch.exp(logp)
+        resid.scatter_add_(
+            1,
+            y_idx.view(-1, 1).to(torch.int64),
+            -torch.ones_like(y_idx, dtype=resid.dtype).view(-1, 1),
+        )
+    return resid, loss.reshape(())
+
+
 def logistic_forward_grad(
     X: torch.Tensor,
     y_idx: torch.Tensor,

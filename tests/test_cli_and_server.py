@@ -1,0 +1,104 @@
+"""No-import-change CLI + connect server tests (reference
+tests_no_import_change/ + jvm plugin suite analogs)."""
+
+import json
+import os
+import subprocess
+import sys
+
+import numpy as np
+import pytest
+
+
+def test_install_aliases():
+    from spark_rapids_ml_amd.install import install_aliases
+
+    install_aliases()
+    import spark_rapids_ml  # noqa: F401 - the alias
+
+    from spark_rapids_ml.clustering import KMeans
+    from spark_rapids_ml.classification import LogisticRegression
+    from spark_rapids_ml.tuning import CrossValidator, ParamGridBuilder
+
+    import spark_rapids_ml_amd
+
+    assert KMeans is spark_rapids_ml_amd.KMeans
+    assert LogisticRegression is spark_rapids_ml_amd.LogisticRegression
+
+
+def test_runner_executes_reference_style_script(tmp_path):
+    script = tmp_path / "app.py"
+    script.write_text(
+        """
+import numpy as np
+from spark_rapids_ml.clustering import KMeans
+from spark_rapids_ml_amd.data import DataFrame
+
+X = np.random.default_rng(0).normal(size=(200, 8)).astype("float32")
+model = KMeans(k=3, maxIter=5, seed=1).fit(DataFrame.from_numpy(X))
+print("CENTERS", model.cluster_centers_.shape)
+"""
+    )
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, "-m", "spark_rapids_ml_amd", str(script)],
+        capture_output=True,
+        text=True,
+        cwd=repo,
+        timeout=300,
+    )
+    assert out.returncode == 0, out.stderr
+    assert "CENTERS (3, 8)" in out.stdout
+
+
+@pytest.fixture
+def client(tmp_path):
+    from starlette.testclient import TestClient
+
+    from spark_rapids_ml_amd.connect_server import create_app
+
+    return TestClient(create_app())
+
+
+def test_server_fit_transform_roundtrip(client, tmp_path):
+    from spark_rapids_ml_amd.data import DataFrame
+
+    rng = np.random.default_rng(0)
+    X = rng.normal(size=(300, 8)).astype(np.float32)
+    data_path = str(tmp_path / "data")
+    DataFrame.from_numpy(X).write_parquet(data_path)
+
+    r = client.post(
+        "/fit",
+        json={"estimator": "KMeans", "params": {"k": 4, "maxIter": 5}, "data_path": data_path},
+    )
+    assert r.status_code == 200, r.text
+    body = r.json()
+    mid = body["model_id"]
+    assert np.asarray(body["attributes"]["cluster_centers_"]).shape == (4, 8)
+
+    out_path = str(tmp_path / "out")
+    r2 = client.post(
+        "/transform",
+        json={"model_id": mid, "data_path": data_path, "output_path": out_path},
+    )
+    assert r2.status_code == 200, r2.text
+    assert "prediction" in r2.json()["columns"]
+    back = DataFrame.read_parquet(out_path)
+    assert back.num_rows == 300
+
+    r3 = client.get("/models")
+    assert mid in r3.json()
+
+    save_path = str(tmp_path / "saved")
+    r4 = client.post(f"/models/{mid}/save", json={"path": save_path})
+    assert r4.status_code == 200
+    from spark_rapids_ml_amd import KMeansModel
+
+    loaded = KMeansModel.load(save_path)
+    assert loaded.cluster_centers_.shape == (4, 8)
+
+
+def test_server_bad_estimator(client):
+    r = client.post("/fit", json={"estimator": "Nope", "params": {}, "data_path": "/none"})
+    assert r.status_code == 400

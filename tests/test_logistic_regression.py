@@ -167,3 +167,27 @@ def test_logreg_distributed_matches_single():
         # fp summation order differs between world sizes; optimizer-level agreement
         assert np.allclose(coef, single.coefficients, rtol=1e-2, atol=1e-3)
         assert np.isclose(icpt, single.intercept, rtol=1e-2, atol=1e-3)
+
+
+def test_sparse_csr_matches_dense():
+    import scipy.sparse as sp
+
+    rng = np.random.default_rng(0)
+    X = rng.normal(size=(400, 30))
+    X[rng.random(X.shape) < 0.8] = 0.0  # 80% sparse
+    y = (X @ rng.normal(size=30) + 0.1 * rng.normal(size=400) > 0).astype(np.float64)
+    Xs = sp.csr_matrix(X.astype(np.float32))
+    dense_model = LogisticRegression(regParam=0.01, maxIter=200, tol=1e-10).fit(
+        DataFrame.from_numpy(X.astype(np.float32), y)
+    )
+    sparse_model = LogisticRegression(regParam=0.01, maxIter=200, tol=1e-10).fit(
+        DataFrame.from_numpy(Xs, y)
+    )
+    assert np.allclose(
+        sparse_model.coefficients, dense_model.coefficients, rtol=1e-3, atol=1e-4
+    )
+    out = sparse_model.transform(DataFrame.from_numpy(Xs))
+    out_d = dense_model.transform(DataFrame.from_numpy(X.astype(np.float32)))
+    assert (
+        np.asarray(out["prediction"]) == np.asarray(out_d["prediction"])
+    ).mean() > 0.99
