@@ -81,12 +81,18 @@ def gen_classification(
     comm = get_comm()
     rng = np.random.default_rng(seed)
     ninf = n_informative or max(2, num_cols // 10)
-    centroids = rng.normal(scale=2.0, size=(n_classes, ninf))
+    # centroid separation chosen so classes overlap (Bayes error ~10-20%):
+    # per-dim signal shrinks with sqrt(ninf) so the problem stays hard as
+    # dimensionality grows (a solver must actually iterate).
+    centroids = rng.normal(scale=2.0 / np.sqrt(ninf), size=(n_classes, ninf))
     rng_local = np.random.default_rng(seed + 104729 * (comm.rank + 1))
     n_local = _shard(num_rows, comm.rank, comm.world_size)
     y = rng_local.integers(0, n_classes, n_local)
     X = rng_local.normal(size=(n_local, num_cols))
     X[:, :ninf] += centroids[y]
+    # 1% label noise
+    flip = rng_local.random(n_local) < 0.01
+    y[flip] = rng_local.integers(0, n_classes, int(flip.sum()))
     return X.astype(dtype), y.astype(np.float64)
 
 

@@ -272,21 +272,42 @@ __global__ __launch_bounds__(256) void segment_sum_kernel(
 
 constexpr int GR_BM = 128;
 constexpr int GR_BN = 128;
-constexpr int GR_BK = 32;
+constexpr int GR_BK = 64;
+constexpr int GR_LD = GR_BK * GR_BM / 256;  // 32 staging elems per thread
 
+// gram_kernel: C[d,d] = A^T A, upper-triangle block enumeration (1D grid of
+// nb*(nb+1)/2 tiles x SPLIT row-slices; partials atomicAdd'ed) with the
+// same BK=64 write-after-barrier pipeline as the kmeans assign kernel.
 __global__ __launch_bounds__(256) void gram_kernel(
-    const float* __restrict__ A, int64_t n, int d, float* __restrict__ out) {
+    const float* __restrict__ A, int64_t n, int d, int nb, int split,
+    float* __restrict__ out) {
   __shared__ float lds_i[GR_BK][GR_BM + 1];
   __shared__ float lds_j[GR_BK][GR_BN + 1];
 
-  const int i0 = blockIdx.x * GR_BM;
-  const int j0 = blockIdx.y * GR_BN;
-  if (j0 + GR_BN <= i0) return;  // strictly-lower blocks mirrored at the end
+  const int tri = blockIdx.x / split;
+  const int slice = blockIdx.x % split;
+  // tri -> (bi, bj) upper triangle: bi = row of the triangle
+  int bi = (int)((sqrtf(8.0f * tri + 1.0f) - 1.0f) * 0.5f);
+  while ((bi + 1) * (bi + 2) / 2 <= tri) ++bi;
+  while (bi * (bi + 1) / 2 > tri) --bi;
+  const int bj_off = tri - bi * (bi + 1) / 2;
+  const int j0 = bi * GR_BN;          // bi-th diagonal stripe
+  const int i0 = bj_off * GR_BM;      // column tile within the stripe (i0<=j0)
+
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
   const int wr = wave >> 1;
   const int wc = wave & 1;
+
+  const int nsteps_total = (int)((n + GR_BK - 1) / GR_BK);
+  const int per = (nsteps_total + split - 1) / split;
+  const int s_begin = slice * per;
+  const int s_end = min(nsteps_total, s_begin + per);
+  if (s_begin >= s_end) return;
+
+  const bool full_i = (i0 + GR_BM <= d);
+  const bool full_j = (j0 + GR_BN <= d);
 
   f32x16 acc[2][2];
 #pragma unroll
@@ -296,37 +317,44 @@ __global__ __launch_bounds__(256) void gram_kernel(
 #pragma unroll
       for (int r = 0; r < 16; ++r) acc[m][nn][r] = 0.0f;
 
-  const bool full_i = (i0 + GR_BM <= d);
-  const bool full_j = (j0 + GR_BN <= d);
-  for (int64_t r0 = 0; r0 < n; r0 += GR_BK) {
-    if (full_i && full_j && r0 + GR_BK <= n) {
-#pragma unroll 4
-      for (int e = tid; e < GR_BK * GR_BM; e += 256) {
-        int rr = e >> 7, i = e & 127;
-        lds_i[rr][i] = A[(r0 + rr) * d + i0 + i];
-      }
-#pragma unroll 4
-      for (int e = tid; e < GR_BK * GR_BN; e += 256) {
-        int rr = e >> 7, j = e & 127;
-        lds_j[rr][j] = A[(r0 + rr) * d + j0 + j];
-      }
-    } else {
-      for (int e = tid; e < GR_BK * GR_BM; e += blockDim.x) {
-        int rr = e / GR_BM, i = e % GR_BM;
-        int64_t gr = r0 + rr;
-        int gi = i0 + i;
-        lds_i[rr][i] = (gr < n && gi < d) ? A[gr * d + gi] : 0.0f;
-      }
-      for (int e = tid; e < GR_BK * GR_BN; e += blockDim.x) {
-        int rr = e / GR_BN, j = e % GR_BN;
-        int64_t gr = r0 + rr;
-        int gj = j0 + j;
-        lds_j[rr][j] = (gr < n && gj < d) ? A[gr * d + gj] : 0.0f;
-      }
-    }
-    __syncthreads();
+  float rg[GR_LD];
 
-#pragma unroll
+#define GR_LOAD(dst_unused, col0, step, full)                                  \
+  do {                                                                         \
+    int64_t r0 = (int64_t)(step)*GR_BK;                                        \
+    if ((full) && r0 + GR_BK <= n) {                                           \
+      _Pragma("unroll") for (int q = 0; q < GR_LD; ++q) {                      \
+        int e = q * 256 + tid;                                                 \
+        rg[q] = A[(r0 + (e >> 7)) * d + (col0) + (e & 127)];                   \
+      }                                                                        \
+    } else {                                                                   \
+      _Pragma("unroll") for (int q = 0; q < GR_LD; ++q) {                      \
+        int e = q * 256 + tid;                                                 \
+        int64_t gr = r0 + (e >> 7);                                            \
+        int gc = (col0) + (e & 127);                                           \
+        rg[q] = (gr < n && gc < d) ? A[gr * d + gc] : 0.0f;                    \
+      }                                                                        \
+    }                                                                          \
+  } while (0)
+
+#define GR_WRITE(dstlds)                                                       \
+  do {                                                                         \
+    _Pragma("unroll") for (int q = 0; q < GR_LD; ++q) {                        \
+      int e = q * 256 + tid;                                                   \
+      dstlds[e >> 7][e & 127] = rg[q];                                         \
+    }                                                                          \
+  } while (0)
+
+  // Staging: consecutive lanes take consecutive COLUMNS of one data row
+  // (row-major A -> coalesced 512B reads); LDS write [rr][i] is contiguous.
+  GR_LOAD(rg, i0, s_begin, full_i);
+  GR_WRITE(lds_i);
+  GR_LOAD(rg, j0, s_begin, full_j);
+  GR_WRITE(lds_j);
+  __syncthreads();
+
+  for (int step = s_begin; step < s_end; ++step) {
+#pragma unroll 8
     for (int kk = 0; kk < GR_BK / 2; ++kk) {
       const int rr = 2 * kk + (lane >> 5);
       float a0 = lds_i[rr][wr * 64 + (lane & 31)];
@@ -339,7 +367,16 @@ __global__ __launch_bounds__(256) void gram_kernel(
       acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc[1][1], 0, 0, 0);
     }
     __syncthreads();
+    if (step + 1 < s_end) {
+      GR_LOAD(rg, i0, step + 1, full_i);
+      GR_WRITE(lds_i);
+      GR_LOAD(rg, j0, step + 1, full_j);
+      GR_WRITE(lds_j);
+      __syncthreads();
+    }
   }
+#undef GR_LOAD
+#undef GR_WRITE
 
 #pragma unroll
   for (int m = 0; m < 2; ++m)
@@ -350,10 +387,24 @@ __global__ __launch_bounds__(256) void gram_kernel(
         int row = i0 + wr * 64 + m * 32 + (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
         int col = j0 + wc * 64 + nn * 32 + (lane & 31);
         if (row < d && col < d && col >= row) {
-          out[(int64_t)row * d + col] = acc[m][nn][r];
-          if (col != row) out[(int64_t)col * d + row] = acc[m][nn][r];
+          float v = acc[m][nn][r];
+          if (split > 1) {
+            atomicAdd(&out[(int64_t)row * d + col], v);
+          } else {
+            out[(int64_t)row * d + col] = v;
+          }
         }
       }
+}
+
+// mirror upper triangle to lower
+__global__ void gram_mirror_kernel(float* __restrict__ out, int d) {
+  int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t total = (int64_t)d * d;
+  for (; idx < total; idx += (int64_t)gridDim.x * blockDim.x) {
+    int row = (int)(idx / d), col = (int)(idx % d);
+    if (col < row) out[idx] = out[(int64_t)col * d + row];
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -459,9 +510,15 @@ torch::Tensor gram_f32(torch::Tensor A) {
   const int d = (int)A.size(1);
   auto out = torch::zeros({d, d}, A.options());
   if (n > 0 && d > 0) {
-    dim3 grid((d + GR_BM - 1) / GR_BM, (d + GR_BN - 1) / GR_BN);
+    const int nb = (d + GR_BN - 1) / GR_BN;
+    const int tri = nb * (nb + 1) / 2;
+    int split = std::max(1, std::min(64, 768 / tri));
+    dim3 grid((unsigned)(tri * split));
     hipLaunchKernelGGL(gram_kernel, grid, dim3(256), 0, cur_stream(),
-                       A.data_ptr<float>(), n, d, out.data_ptr<float>());
+                       A.data_ptr<float>(), n, d, nb, split, out.data_ptr<float>());
+    int mgrid = (int)std::min<int64_t>(2048, ((int64_t)d * d + 255) / 256);
+    hipLaunchKernelGGL(gram_mirror_kernel, dim3(mgrid), dim3(256), 0, cur_stream(),
+                       out.data_ptr<float>(), d);
   }
   return out;
 }

@@ -34,12 +34,18 @@ def xty_gram(X: torch.Tensor, y: torch.Tensor) -> Tuple[torch.Tensor, torch.Tens
 
 
 def eigh_sym(A: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
-    """Eigendecomposition of a small symmetric d×d matrix, ascending
-    eigenvalues (torch.linalg.eigh; d is at most a few thousand so this is
-    never the hot path — reference used raft::linalg::eigDC, rapidsml_jni.cu:215-269)."""
-    # eigh of small matrices is faster + deterministic on CPU in f64
-    Ad = A.detach().to("cpu", torch.float64)
-    w, v = torch.linalg.eigh(Ad)
+    """Eigendecomposition of a symmetric d×d matrix, ascending eigenvalues
+    (reference used raft::linalg::eigDC, rapidsml_jni.cu:215-269). On GPU
+    this is rocSOLVER syevd via torch.linalg.eigh — a vendor solver, not a
+    CUDA port; CPU fallback for determinism in CPU tests."""
+    Ad = A.detach().to(torch.float64)
+    if A.is_cuda:
+        try:
+            w, v = torch.linalg.eigh(Ad)
+            return w, v
+        except Exception:
+            pass
+    w, v = torch.linalg.eigh(Ad.cpu())
     return w.to(A.device), v.to(A.device)
 
 
