@@ -314,9 +314,11 @@ class LogisticRegression(_LogisticRegressionParams, Estimator):
                 buf[0].index_add_(0, cols, vals)
                 buf[1].index_add_(0, cols, vals * vals)
             else:
-                Xd = Xt.to(torch.float64)
-                buf[0] = Xd.sum(dim=0)
-                buf[1] = (Xd * Xd).sum(dim=0)
+                # chunked so no full f64 copy of X is materialized
+                for s0 in range(0, Xt.shape[0], 1 << 18):
+                    xb = Xt[s0 : s0 + (1 << 18)]
+                    buf[0] += xb.sum(dim=0, dtype=torch.float64)
+                    buf[1] += (xb * xb).sum(dim=0, dtype=torch.float64)
         comm.allreduce(comm.to_coll(buf))
         buf = buf.to(Xt.device)
         mean = buf[0] / n
