@@ -237,43 +237,46 @@ class LogisticRegression(_LogisticRegressionParams, Estimator):
             Xs = Xt / sig_safe[None, :].to(Xt.dtype)
 
         ncol = d + (1 if fit_intercept else 0)
-        w0 = torch.zeros(C_out * ncol, dtype=torch.float64, device=ctx.device)
+        # optimizer state lives on CPU: the weight vector is C*(d+1) floats,
+        # so the L-BFGS two-loop recursion is hundreds of tiny ops that cost
+        # more as device launches + .item() syncs than as CPU math. Only the
+        # data-sized work (scores/residual/gradient GEMMs) runs on device.
+        w0 = torch.zeros(C_out * ncol, dtype=torch.float64)
 
-        pen_scale = torch.ones(d, dtype=torch.float64, device=ctx.device)
+        pen_scale = torch.ones(d, dtype=torch.float64)
         if not standardization:
-            pen_scale = 1.0 / (sig_safe.to(torch.float64) ** 2)
+            pen_scale = 1.0 / (sig_safe.to(torch.float64).cpu() ** 2)
 
         l2 = lam * (1.0 - l1r)
         l1 = lam * l1r
 
         def closure(wv: torch.Tensor) -> Tuple[float, torch.Tensor]:
-            W = wv.view(C_out, ncol).to(torch.float32 if sparse else Xs.dtype)
+            W = wv.to(ctx.device).view(C_out, ncol).to(
+                torch.float32 if sparse else Xs.dtype
+            )
             grad, loss = logistic_grad_loss(Xs, y_t, W, fit_intercept, XT=XsT)
-            buf = torch.zeros(C_out * ncol + 1, dtype=torch.float64, device=wv.device)
+            buf = torch.zeros(C_out * ncol + 1, dtype=torch.float64, device=grad.device)
             buf[:-1] = grad.to(torch.float64).flatten()
             buf[-1] = loss.to(torch.float64)
             comm.allreduce(comm.to_coll(buf))
-            buf = buf.to(wv.device)
+            buf = buf.cpu()
             g = buf[:-1] / n
-            total_loss = float(buf[-1].item()) / n
+            total_loss = float(buf[-1]) / n
             if l2 > 0:
-                Wd = wv.view(C_out, ncol)
-                coef = Wd[:, :d]
-                g = g.view(C_out, ncol).clone()
+                coef = wv.view(C_out, ncol)[:, :d]
+                g = g.view(C_out, ncol)
                 g[:, :d] += l2 * coef * pen_scale[None, :]
-                total_loss += 0.5 * l2 * float(
-                    ((coef**2) * pen_scale[None, :]).sum().item()
-                )
-                g = g.flatten()
+                total_loss += 0.5 * l2 * float(((coef**2) * pen_scale[None, :]).sum())
+                g = g.reshape(-1)
             return total_loss, g
 
         l1_vec = None
         if l1 > 0:
-            l1_vec = torch.zeros(C_out, ncol, dtype=torch.float64, device=ctx.device)
+            l1_vec = torch.zeros(C_out, ncol, dtype=torch.float64)
             if standardization:
                 l1_vec[:, :d] = l1
             else:
-                l1_vec[:, :d] = l1 / sig_safe.to(torch.float64)[None, :]
+                l1_vec[:, :d] = l1 / sig_safe.to(torch.float64).cpu()[None, :]
             l1_vec = l1_vec.flatten()
 
         w_opt, obj, n_iter = lbfgs(
@@ -288,7 +291,7 @@ class LogisticRegression(_LogisticRegressionParams, Estimator):
 
         W = w_opt.view(C_out, ncol)
         coef_s = W[:, :d]
-        coef = (coef_s / sig_safe.to(torch.float64)[None, :]).cpu().numpy()
+        coef = (coef_s / sig_safe.to(torch.float64).cpu()[None, :]).numpy()
         if fit_intercept:
             intercept = W[:, d].cpu().numpy()
         else:

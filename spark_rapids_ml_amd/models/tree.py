@@ -113,6 +113,17 @@ class _RandomForestParams(HasFeaturesCol, HasFeaturesCols, HasLabelCol, HasPredi
         return self.getOrDefault("maxDepth")
 
 
+def _max_lastdim(t: torch.Tensor):
+    """(values, indices) over the last dim. torch-CPU's indexed max/argmax is
+    ~30x slower than numpy in this build; route CPU through numpy."""
+    if t.is_cuda:
+        return t.max(dim=-1)
+    arr = t.detach().numpy()
+    idx = np.argmax(arr, axis=-1)
+    vals = np.take_along_axis(arr, np.expand_dims(idx, -1), axis=-1).squeeze(-1)
+    return torch.from_numpy(np.ascontiguousarray(vals)), torch.from_numpy(idx.astype(np.int64))
+
+
 def _estimators_per_worker(n_estimators: int, world: int) -> List[int]:
     """Tree counts per rank (reference tree.py:330-341): floor division with
     the remainder spread over the first ranks."""
@@ -179,9 +190,19 @@ def _compute_bin_edges(
 
 def _bin_data(X: torch.Tensor, edges: torch.Tensor) -> torch.Tensor:
     """bin(x) = #edges <= x, i.e. searchsorted right; uint8 [n,d].
-    Inference rule derived from this: bin(x) <= b  <=>  x < edges[b]."""
-    b = torch.searchsorted(edges, X.T.contiguous(), right=True)  # [d, n]
-    return b.T.to(torch.uint8).contiguous()
+    Inference rule derived from this: bin(x) <= b  <=>  x < edges[b].
+    Feature-chunked: searchsorted's kernel caps at 2^31 elements and the
+    transposed copy stays small."""
+    n, d = X.shape
+    out = torch.empty((n, d), dtype=torch.uint8, device=X.device)
+    fc = max(1, min(d, (1 << 30) // max(1, n)))
+    for f0 in range(0, d, fc):
+        f1 = min(d, f0 + fc)
+        b = torch.searchsorted(
+            edges[f0:f1], X[:, f0:f1].T.contiguous(), right=True
+        )  # [fc, n]
+        out[:, f0:f1] = b.T.to(torch.uint8)
+    return out
 
 
 # ---------------------------------------------------------------------------
@@ -333,7 +354,7 @@ def _grow_tree(
                     )
                     gain, sbin, lval, rval, lcnt = _best_split_reg(H, min_leaf)
                 # best feature within chunk
-                g, fidx = gain.max(dim=1)
+                g, fidx = _max_lastdim(gain)
                 upd = g > best_gain
                 best_gain = torch.where(upd, g, best_gain)
                 ar = torch.arange(B, device=dev)
@@ -417,7 +438,7 @@ def _best_split_class(H: torch.Tensor, min_leaf: int):
     gain = g_parent - (lc / ntc) * g_left - (rc / ntc) * g_right  # [B,F,nb-1]
     valid = (lc >= min_leaf) & (rc >= min_leaf)
     gain = torch.where(valid, gain, torch.full_like(gain, -1.0))
-    best_gain, best_bin = gain.max(dim=2)  # [B,F]
+    best_gain, best_bin = _max_lastdim(gain)  # [B,F]
     B, F = best_gain.shape
     ar_b = torch.arange(B, device=H.device)[:, None].expand(B, F)
     ar_f = torch.arange(F, device=H.device)[None, :].expand(B, F)
@@ -441,7 +462,7 @@ def _best_split_reg(H: torch.Tensor, min_leaf: int):
     gain = var(tc, ts, ts2) - (lc / tcc) * var(lc, ls, ls2) - (rc / tcc) * var(rc, rs, rs2)
     valid = (lc >= min_leaf) & (rc >= min_leaf)
     gain = torch.where(valid, gain, torch.full_like(gain, -1.0))
-    best_gain, best_bin = gain.max(dim=2)
+    best_gain, best_bin = _max_lastdim(gain)
     B, F = best_gain.shape
     ar_b = torch.arange(B, device=H.device)[:, None].expand(B, F)
     ar_f = torch.arange(F, device=H.device)[None, :].expand(B, F)

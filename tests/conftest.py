@@ -3,6 +3,15 @@ import sys
 
 import pytest
 
+os.environ.setdefault("OMP_NUM_THREADS", "8")
+
+def _cap_torch_threads():
+    import torch
+
+    torch.set_num_threads(min(8, os.cpu_count() or 8))
+
+_cap_torch_threads()
+
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 
