@@ -202,6 +202,372 @@ __global__ __launch_bounds__(256) void kmeans_assign_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// kmeans_assign_glds: same contract as kmeans_assign_kernel, staging via
+// async global_load_lds DMA (16B) into a 16B-XOR-swizzled linear LDS image.
+// 2 buffers, raw s_barrier + counted vmcnt (a plain __syncthreads would
+// drain the in-flight next tile, guide §5 pipelining note). Requires d%4==0
+// (16B-aligned rows); host falls back to the register-staged kernel else.
+// All LDS lives in ONE __shared__ object (guide §5.4 trap (a)).
+// ---------------------------------------------------------------------------
+
+constexpr int KG_BM = 128;
+constexpr int KG_BN = 128;
+constexpr int KG_BK = 32;
+// float offsets into the single LDS arena
+constexpr int KG_X0 = 0;                    // lds_x buf0 [128*32]
+constexpr int KG_X1 = 4096;                 // lds_x buf1
+constexpr int KG_C0 = 8192;                 // lds_c buf0
+constexpr int KG_C1 = 12288;                // lds_c buf1
+constexpr int KG_BEST = 16384;              // u64 best[128] = 256 floats
+constexpr int KG_SCRATCH = 16640;           // 8 doubles = 16 floats
+constexpr int KG_TOTAL = 16656;
+
+__device__ __forceinline__ int kg_swz(int i, int kd) {
+  // element (row i, col kd) -> float offset within a 128x32 tile image:
+  // 16B groups XOR-swizzled by (i&7) so column-strided b32 fragment reads
+  // spread over 8 bank quads (<=4-way conflict) while glds stays lane-linear
+  return i * KG_BK + ((((kd >> 2) ^ (i & 7)) << 2) | (kd & 3));
+}
+
+__global__ __launch_bounds__(256) void kmeans_assign_glds_kernel(
+    const float* __restrict__ X, const float* __restrict__ C,
+    const float* __restrict__ x_sq, const float* __restrict__ c_sq,
+    int n, int d, int k,
+    int32_t* __restrict__ labels, float* __restrict__ min_dists,
+    double* __restrict__ inertia) {
+  __shared__ __attribute__((aligned(16))) float smem[KG_TOTAL];
+
+  const int i0 = blockIdx.x * KG_BM;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = wave >> 1;
+  const int wc = wave & 1;
+
+  unsigned long long* best = reinterpret_cast<unsigned long long*>(&smem[KG_BEST]);
+  for (int i = tid; i < KG_BM; i += blockDim.x) best[i] = ~0ULL;
+
+  const bool full_rows = (i0 + KG_BM <= n);
+  const int nsteps = (d + KG_BK - 1) / KG_BK;
+  bool full_rows_cols = full_rows;  // && full_cols, set per j-tile
+
+  // Tile "pair fullness": both operands of a d-step go glds (8 instructions
+  // per wave) or both scalar — keeps the vmcnt immediates branch-constant.
+  auto pair_full = [&](int step) {
+    return full_rows_cols && (step + 1) * KG_BK <= d;
+  };
+
+  // per-wave glds issue: 4 instructions per operand per tile
+  // instruction q (0..15 across 4 waves) covers rows [q*8, q*8+8)
+  auto issue_glds = [&](const float* __restrict__ src, int base_row, int d0,
+                        int lds_off) {
+#pragma unroll
+    for (int qq = 0; qq < 4; ++qq) {
+      const int q = wave * 4 + qq;
+      const int i = q * 8 + (lane >> 3);
+      const int g = (lane & 7) ^ (i & 7);
+      const float* gp = src + (int64_t)(base_row + i) * d + d0 + (g << 2);
+      auto lp = (__attribute__((address_space(3))) uint32_t*)(&smem[lds_off + q * 256]);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)gp, lp, 16, 0, 0);
+    }
+  };
+  auto issue_scalar = [&](const float* __restrict__ src, int base_row, int lim,
+                          int d0, int lds_off) {
+    for (int e = tid; e < KG_BM * KG_BK; e += 256) {
+      int i = e >> 5, kd = e & 31;
+      int gr = base_row + i, gd = d0 + kd;
+      smem[lds_off + kg_swz(i, kd)] =
+          (gr < lim && gd < d) ? src[(int64_t)gr * d + gd] : 0.0f;
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  };
+  auto issue_tile = [&](int step, int xoff, int coff, int j0) {
+    const int d0 = step * KG_BK;
+    if (pair_full(step)) {
+      issue_glds(X, i0, d0, xoff);
+      issue_glds(C, j0, d0, coff);
+    } else {
+      issue_scalar(X, i0, n, d0, xoff);
+      issue_scalar(C, j0, k, d0, coff);
+    }
+  };
+
+  for (int j0 = 0; j0 < k; j0 += KG_BN) {
+    const bool full_cols = (j0 + KG_BN <= k);
+    full_rows_cols = full_rows && full_cols;
+    f32x16 acc[2][2];
+#pragma unroll
+    for (int m = 0; m < 2; ++m)
+#pragma unroll
+      for (int nn = 0; nn < 2; ++nn)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) acc[m][nn][r] = 0.0f;
+
+    // prologue: tile0 + tile1 in flight; wait until only tile1's glds remain
+    issue_tile(0, KG_X0, KG_C0, j0);
+    if (nsteps > 1) issue_tile(1, KG_X1, KG_C1, j0);
+    if (pair_full(0)) {
+      if (nsteps > 1 && pair_full(1)) {
+        asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+      } else {
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      }
+    }
+    __builtin_amdgcn_s_barrier();
+
+    for (int step = 0; step < nsteps; ++step) {
+      const int xb = (step & 1) ? KG_X1 : KG_X0;
+      const int cb = (step & 1) ? KG_C1 : KG_C0;
+      const int xi = wr * 64 + (lane & 31);
+      const int ci = wc * 64 + (lane & 31);
+#pragma unroll 8
+      for (int kk = 0; kk < KG_BK / 2; ++kk) {
+        const int kd = 2 * kk + (lane >> 5);
+        float a0 = smem[xb + kg_swz(xi, kd)];
+        float a1 = smem[xb + kg_swz(xi + 32, kd)];
+        float b0 = smem[cb + kg_swz(ci, kd)];
+        float b1 = smem[cb + kg_swz(ci + 32, kd)];
+        acc[0][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc[0][0], 0, 0, 0);
+        acc[0][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc[0][1], 0, 0, 0);
+        acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc[1][0], 0, 0, 0);
+        acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc[1][1], 0, 0, 0);
+      }
+      __builtin_amdgcn_s_barrier();  // everyone done reading buf[step&1]
+      if (step + 2 < nsteps) {
+        issue_tile(step + 2, xb, cb, j0);
+        // wait until only tile (step+2)'s glds remain -> (step+1) landed
+        if (pair_full(step + 1)) {
+          if (pair_full(step + 2)) {
+            asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+          } else {
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+          }
+        }
+        __builtin_amdgcn_s_barrier();
+      } else if (step + 1 < nsteps) {
+        if (pair_full(step + 1)) {
+          asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        }
+        __builtin_amdgcn_s_barrier();
+      }
+    }
+
+    // epilogue: distances + packed argmin (no glds outstanding)
+#pragma unroll
+    for (int m = 0; m < 2; ++m) {
+#pragma unroll
+      for (int nn = 0; nn < 2; ++nn) {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          int row = wr * 64 + m * 32 + (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
+          int col = wc * 64 + nn * 32 + (lane & 31);
+          int gi = i0 + row, gj = j0 + col;
+          if (gi < n && gj < k) {
+            float dist = x_sq[gi] + c_sq[gj] - 2.0f * acc[m][nn][r];
+            dist = dist < 0.0f ? 0.0f : dist;
+            unsigned long long packed =
+                ((unsigned long long)__float_as_uint(dist) << 32) |
+                (unsigned int)gj;
+            atomicMin(&best[row], packed);
+          }
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  double* block_inertia = reinterpret_cast<double*>(&smem[KG_SCRATCH]);
+  double partial = 0.0;
+  for (int i = tid; i < KG_BM; i += blockDim.x) {
+    int gi = i0 + i;
+    if (gi < n) {
+      unsigned long long p = best[i];
+      float dist = __uint_as_float((unsigned int)(p >> 32));
+      labels[gi] = (int32_t)(p & 0xffffffffu);
+      min_dists[gi] = dist;
+      partial += (double)dist;
+    }
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1)
+    partial += __shfl_down(partial, off, 64);
+  if (lane == 0) block_inertia[wave] = partial;
+  __syncthreads();
+  if (tid == 0)
+    atomicAdd(inertia, block_inertia[0] + block_inertia[1] + block_inertia[2] +
+                           block_inertia[3]);
+}
+
+// ---------------------------------------------------------------------------
+// kmeans_assign_glds3: 3-buffer glds span (guide: "+83%" row) — ONE block
+// per CU (99 KB LDS), 2 tile-pairs in flight across each raw barrier with
+// counted vmcnt. Requires d%4==0.
+// ---------------------------------------------------------------------------
+
+constexpr int K3_BUF = 4096;  // floats per operand buffer
+
+__global__ __launch_bounds__(256) void kmeans_assign_glds3_kernel(
+    const float* __restrict__ X, const float* __restrict__ C,
+    const float* __restrict__ x_sq, const float* __restrict__ c_sq,
+    int n, int d, int k,
+    int32_t* __restrict__ labels, float* __restrict__ min_dists,
+    double* __restrict__ inertia) {
+  // arena: x bufs [0,3*4096), c bufs [3*4096, 6*4096), best, scratch
+  __shared__ __attribute__((aligned(16))) float smem[6 * K3_BUF + 256 + 16];
+  const int BEST = 6 * K3_BUF;
+  const int SCR = BEST + 256;
+
+  const int i0 = blockIdx.x * KG_BM;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = wave >> 1;
+  const int wc = wave & 1;
+
+  unsigned long long* best = reinterpret_cast<unsigned long long*>(&smem[BEST]);
+  for (int i = tid; i < KG_BM; i += blockDim.x) best[i] = ~0ULL;
+
+  const bool full_rows = (i0 + KG_BM <= n);
+  const int nsteps = (d + KG_BK - 1) / KG_BK;
+  bool frc = full_rows;
+
+  auto pf = [&](int step) { return frc && (step + 1) * KG_BK <= d; };
+
+  auto issue_glds = [&](const float* __restrict__ src, int base_row, int d0,
+                        int lds_off) {
+#pragma unroll
+    for (int qq = 0; qq < 4; ++qq) {
+      const int q = wave * 4 + qq;
+      const int i = q * 8 + (lane >> 3);
+      const int g = (lane & 7) ^ (i & 7);
+      const float* gp = src + (int64_t)(base_row + i) * d + d0 + (g << 2);
+      auto lp = (__attribute__((address_space(3))) uint32_t*)(&smem[lds_off + q * 256]);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)gp, lp, 16, 0, 0);
+    }
+  };
+  auto issue_scalar = [&](const float* __restrict__ src, int base_row, int lim,
+                          int d0, int lds_off) {
+    for (int e = tid; e < KG_BM * KG_BK; e += 256) {
+      int i = e >> 5, kd = e & 31;
+      int gr = base_row + i, gd = d0 + kd;
+      smem[lds_off + kg_swz(i, kd)] =
+          (gr < lim && gd < d) ? src[(int64_t)gr * d + gd] : 0.0f;
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  };
+  auto issue_tile = [&](int step, int j0) {
+    const int d0 = step * KG_BK;
+    const int b = step % 3;
+    if (pf(step)) {
+      issue_glds(X, i0, d0, b * K3_BUF);
+      issue_glds(C, j0, d0, (3 + b) * K3_BUF);
+    } else {
+      issue_scalar(X, i0, n, d0, b * K3_BUF);
+      issue_scalar(C, j0, k, d0, (3 + b) * K3_BUF);
+    }
+  };
+  // wait until only tiles {a, b} (8 glds each when pair-full) remain
+  auto wait_leaving = [&](int ta, int tb) {
+    const bool fa = ta >= 0 && ta < nsteps && pf(ta);
+    const bool fb = tb >= 0 && tb < nsteps && pf(tb);
+    if (fa && fb) {
+      asm volatile("s_waitcnt vmcnt(16)" ::: "memory");
+    } else if (fa || fb) {
+      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+  };
+
+  for (int j0 = 0; j0 < k; j0 += KG_BN) {
+    frc = full_rows && (j0 + KG_BN <= k);
+    f32x16 acc[2][2];
+#pragma unroll
+    for (int m = 0; m < 2; ++m)
+#pragma unroll
+      for (int nn = 0; nn < 2; ++nn)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) acc[m][nn][r] = 0.0f;
+
+    issue_tile(0, j0);
+    if (nsteps > 1) issue_tile(1, j0);
+    if (nsteps > 2) issue_tile(2, j0);
+    wait_leaving(nsteps > 1 ? 1 : -1, nsteps > 2 ? 2 : -1);  // tile0 landed
+    __builtin_amdgcn_s_barrier();
+
+    for (int step = 0; step < nsteps; ++step) {
+      const int xb = (step % 3) * K3_BUF;
+      const int cb = (3 + step % 3) * K3_BUF;
+      const int xi = wr * 64 + (lane & 31);
+      const int ci = wc * 64 + (lane & 31);
+#pragma unroll 8
+      for (int kk = 0; kk < KG_BK / 2; ++kk) {
+        const int kd = 2 * kk + (lane >> 5);
+        float a0 = smem[xb + kg_swz(xi, kd)];
+        float a1 = smem[xb + kg_swz(xi + 32, kd)];
+        float b0 = smem[cb + kg_swz(ci, kd)];
+        float b1 = smem[cb + kg_swz(ci + 32, kd)];
+        acc[0][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc[0][0], 0, 0, 0);
+        acc[0][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc[0][1], 0, 0, 0);
+        acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc[1][0], 0, 0, 0);
+        acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc[1][1], 0, 0, 0);
+      }
+      __builtin_amdgcn_s_barrier();  // all waves done reading buf[step%3]
+      if (step + 3 < nsteps) issue_tile(step + 3, j0);
+      if (step + 1 < nsteps) {
+        wait_leaving(step + 2, step + 3);  // (step+1) landed
+        __builtin_amdgcn_s_barrier();
+      }
+    }
+
+#pragma unroll
+    for (int m = 0; m < 2; ++m) {
+#pragma unroll
+      for (int nn = 0; nn < 2; ++nn) {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          int row = wr * 64 + m * 32 + (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
+          int col = wc * 64 + nn * 32 + (lane & 31);
+          int gi = i0 + row, gj = j0 + col;
+          if (gi < n && gj < k) {
+            float dist = x_sq[gi] + c_sq[gj] - 2.0f * acc[m][nn][r];
+            dist = dist < 0.0f ? 0.0f : dist;
+            unsigned long long packed =
+                ((unsigned long long)__float_as_uint(dist) << 32) |
+                (unsigned int)gj;
+            atomicMin(&best[row], packed);
+          }
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  double* block_inertia = reinterpret_cast<double*>(&smem[SCR]);
+  double partial = 0.0;
+  for (int i = tid; i < KG_BM; i += blockDim.x) {
+    int gi = i0 + i;
+    if (gi < n) {
+      unsigned long long p = best[i];
+      float dist = __uint_as_float((unsigned int)(p >> 32));
+      labels[gi] = (int32_t)(p & 0xffffffffu);
+      min_dists[gi] = dist;
+      partial += (double)dist;
+    }
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1)
+    partial += __shfl_down(partial, off, 64);
+  if (lane == 0) block_inertia[wave] = partial;
+  __syncthreads();
+  if (tid == 0)
+    atomicAdd(inertia, block_inertia[0] + block_inertia[1] + block_inertia[2] +
+                           block_inertia[3]);
+}
+
+// ---------------------------------------------------------------------------
 // label_accumulate: sums[label[i]] += X[i], counts[label[i]] += 1 via a
 // sort-based segmented reduction: rows pre-sorted by label (torch.sort in
 // the wrapper), each (label, split) block register-accumulates its column
@@ -475,12 +841,31 @@ std::vector<torch::Tensor> kmeans_assign(torch::Tensor X, torch::Tensor C,
   auto min_dists = torch::empty({n}, X.options());
   auto inertia = torch::zeros({1}, X.options().dtype(torch::kFloat64));
   const int grid = (int)((n + KM_BM - 1) / KM_BM);
-  if (n > 0)
-    hipLaunchKernelGGL(kmeans_assign_kernel, dim3(grid), dim3(256), 0, cur_stream(),
-                       X.data_ptr<float>(), C.data_ptr<float>(), x_sq.data_ptr<float>(),
-                       c_sq.data_ptr<float>(), (int)n, d, k,
-                       labels.data_ptr<int32_t>(), min_dists.data_ptr<float>(),
-                       inertia.data_ptr<double>());
+  if (n > 0) {
+    // A/B-measured on MI355X (1M x 3000, k=1000): register-staged BK=64
+    // write-after-barrier 71.1 ms; glds 2-buffer 73.8 ms; glds 3-buffer
+    // 1-block/CU span 111 ms (single wave/SIMD starves the MFMA pipe).
+    // Register staging is the default; glds variants stay selectable.
+    static const char* v = getenv("SRML_KMEANS_VARIANT");
+    if (d % 4 == 0 && v && v[0] == '3')
+      hipLaunchKernelGGL(kmeans_assign_glds3_kernel, dim3(grid), dim3(256), 0,
+                         cur_stream(), X.data_ptr<float>(), C.data_ptr<float>(),
+                         x_sq.data_ptr<float>(), c_sq.data_ptr<float>(), (int)n, d, k,
+                         labels.data_ptr<int32_t>(), min_dists.data_ptr<float>(),
+                         inertia.data_ptr<double>());
+    else if (d % 4 == 0 && v && v[0] == '2')
+      hipLaunchKernelGGL(kmeans_assign_glds_kernel, dim3(grid), dim3(256), 0,
+                         cur_stream(), X.data_ptr<float>(), C.data_ptr<float>(),
+                         x_sq.data_ptr<float>(), c_sq.data_ptr<float>(), (int)n, d, k,
+                         labels.data_ptr<int32_t>(), min_dists.data_ptr<float>(),
+                         inertia.data_ptr<double>());
+    else
+      hipLaunchKernelGGL(kmeans_assign_kernel, dim3(grid), dim3(256), 0, cur_stream(),
+                         X.data_ptr<float>(), C.data_ptr<float>(), x_sq.data_ptr<float>(),
+                         c_sq.data_ptr<float>(), (int)n, d, k,
+                         labels.data_ptr<int32_t>(), min_dists.data_ptr<float>(),
+                         inertia.data_ptr<double>());
+  }
   return {labels, min_dists, inertia};
 }
 
