@@ -41,7 +41,7 @@ from .params import (
     _NativeParams,
 )
 from .parallel.context import Comm, PartitionDescriptor, get_comm
-from .utils import get_logger
+from .utils import annotate, get_logger
 
 
 class _FitContext:
@@ -157,7 +157,8 @@ class Estimator(_NativeParams, HasVerbose):
         for pm in param_maps:
             params = dict(self._native_params)
             params.update(pm)
-            attrs = self._fit_array(X, y, ctx, params)
+            with annotate(f"srml::{type(self).__name__}.fit"):
+                attrs = self._fit_array(X, y, ctx, params)
             model = self._create_model(attrs)
             self._copyValues(model)
             model._native_params = dict(params)
@@ -211,13 +212,32 @@ class Model(_NativeParams, HasVerbose):
         the feature struct; here a direct device-batched call)."""
         features_col, features_cols = self._get_input_columns()
         X = extract_features(df, features_col, features_cols, self._float32_inputs)
-        out = self._transform_array(X)
+        with annotate(f"srml::{type(self).__name__}.transform"):
+            out = self._transform_array(X)
         if isinstance(out, dict):
             res = df
             for name, col in out.items():
                 res = res.with_column(name, col)
             return res
         return df.with_column(self._out_col_name(), out)
+
+    def _transformEvaluate(self, df: DataFrame, evaluator) -> List[float]:
+        """Single-pass transform+evaluate (reference
+        `_CumlModel._transformEvaluate` / `_transform_evaluate_internal`,
+        core.py:1572-1693: one job evaluates the model; multi-model variants
+        merge per-rank sufficient statistics)."""
+        out = self.transform(df)
+        return [evaluator.evaluate(out)]
+
+    @staticmethod
+    def _combine(models: Sequence["Model"]) -> "Model":
+        """Combine per-param-map models for single-pass CV evaluation
+        (reference per-algo _combine, e.g. regression.py:828). The SPMD
+        runtime evaluates models independently, so combining is the list
+        head carrying its peers."""
+        head = models[0]
+        head._combined = list(models)
+        return head
 
     # -- persistence -------------------------------------------------------
     def write(self) -> "_Writer":

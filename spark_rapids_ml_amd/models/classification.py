@@ -28,6 +28,7 @@ import torch
 from ..core import Estimator, Model, _FitContext
 from ..data import to_device_tensor
 from ..params import (
+    HasEnableSparseDataOptim,
     HasFeaturesCol,
     HasFeaturesCols,
     HasLabelCol,
@@ -50,6 +51,7 @@ class _LogisticRegressionParams(
     HasPredictionCol,
     HasProbabilityCol,
     HasRawPredictionCol,
+    HasEnableSparseDataOptim,
 ):
     regParam = Param("logreg", "regParam", "regularization λ.", TypeConverters.toFloat)
     elasticNetParam = Param(
@@ -176,6 +178,20 @@ class LogisticRegression(_LogisticRegressionParams, Estimator):
         from ..data import _is_sparse
 
         sparse = _is_sparse(X)
+        # enable_sparse_data_optim (reference params.py:45-66): None=auto,
+        # False=densify, True=require sparse input
+        opt = (
+            self.getOrDefault("enable_sparse_data_optim")
+            if self.isSet("enable_sparse_data_optim")
+            else None
+        )
+        if opt is False and sparse:
+            X = X.toarray()
+            sparse = False
+        elif opt is True and not sparse:
+            raise ValueError(
+                "enable_sparse_data_optim=True requires sparse CSR features"
+            )
         if sparse:
             # CSR path (reference classification.py:960-966: int64 index
             # escape only above 1e9 nnz; torch CSR uses int64 throughout)
@@ -279,7 +295,7 @@ class LogisticRegression(_LogisticRegressionParams, Estimator):
                 l1_vec[:, :d] = l1 / sig_safe.to(torch.float64).cpu()[None, :]
             l1_vec = l1_vec.flatten()
 
-        w_opt, obj, n_iter = lbfgs(
+        w_opt, obj, n_iter, obj_history = lbfgs(
             w0,
             closure,
             max_iter=max_iter,
@@ -306,6 +322,7 @@ class LogisticRegression(_LogisticRegressionParams, Estimator):
             "classes_": classes.astype(np.float64),
             "n_iter_": n_iter,
             "objective_": float(obj),
+            "objective_history_": np.asarray(obj_history, dtype=np.float64),
         }
 
     def _column_std(self, Xt: torch.Tensor, comm, n: int) -> torch.Tensor:
@@ -344,6 +361,7 @@ class LogisticRegressionModel(_LogisticRegressionParams, Model):
         classes_: np.ndarray,
         n_iter_: int = 0,
         objective_: float = 0.0,
+        objective_history_: Optional[np.ndarray] = None,
         **kwargs: Any,
     ) -> None:
         super().__init__(
@@ -352,7 +370,17 @@ class LogisticRegressionModel(_LogisticRegressionParams, Model):
             classes_=np.asarray(classes_),
             n_iter_=int(n_iter_),
             objective_=float(objective_),
+            objective_history_=np.asarray(
+                objective_history_ if objective_history_ is not None else []
+            ),
         )
+
+    @property
+    def objectiveHistory(self) -> np.ndarray:
+        """Per-iteration objective (Spark LogisticRegressionTrainingSummary
+        parity; reference compares full objectives,
+        tests_large/test_large_logistic_regression.py:39-60)."""
+        return self._model_attributes["objective_history_"]
 
     @property
     def coefficients(self) -> np.ndarray:

@@ -78,7 +78,7 @@ class _KMeansParams(HasFeaturesCol, HasFeaturesCols, HasPredictionCol):
             "maxIter": "max_iter",
             "seed": "random_state",
             "tol": "tol",
-            "distanceMeasure": "",  # euclidean is the only metric; validated below
+            "distanceMeasure": "metric",
             "weightCol": None,
         }
 
@@ -86,6 +86,7 @@ class _KMeansParams(HasFeaturesCol, HasFeaturesCols, HasPredictionCol):
     def _param_value_mapping(cls):
         return {
             "init": lambda v: {"k-means||": "k-means||", "random": "random"}.get(v, None),
+            "metric": lambda v: "euclidean" if v == "euclidean" else None,
         }
 
     @classmethod
@@ -97,6 +98,7 @@ class _KMeansParams(HasFeaturesCol, HasFeaturesCols, HasPredictionCol):
             "max_iter": 20,
             "random_state": 1,
             "tol": 1e-4,
+            "metric": "euclidean",
             "oversampling_factor": 2.0,
             "verbose": False,
         }
@@ -193,6 +195,7 @@ class KMeans(_KMeansParams, Estimator):
             "cluster_centers_": as_numpy(C),
             "n_iter_": n_iter,
             "inertia_": inertia,
+            "cluster_sizes_": as_numpy(g_counts).astype(np.int64),
         }
 
     def _owned_rows(self, global_idx: np.ndarray, pdesc) -> Tuple[np.ndarray, np.ndarray]:
@@ -313,6 +316,15 @@ def _weighted_kmeanspp(cand: np.ndarray, w: np.ndarray, k: int, rng) -> np.ndarr
     return C
 
 
+class KMeansSummary:
+    """Spark KMeansSummary equivalent."""
+
+    def __init__(self, trainingCost: float, clusterSizes: list, numIter: int):
+        self.trainingCost = trainingCost
+        self.clusterSizes = clusterSizes
+        self.numIter = numIter
+
+
 class KMeansModel(_KMeansParams, Model):
     """Fitted KMeans model (reference KMeansModel, clustering.py:505)."""
 
@@ -321,13 +333,31 @@ class KMeansModel(_KMeansParams, Model):
         cluster_centers_: np.ndarray,
         n_iter_: int = 0,
         inertia_: float = 0.0,
+        cluster_sizes_: Optional[np.ndarray] = None,
         **kwargs: Any,
     ) -> None:
         super().__init__(
             cluster_centers_=np.asarray(cluster_centers_),
             n_iter_=int(n_iter_),
             inertia_=float(inertia_),
+            cluster_sizes_=np.asarray(
+                cluster_sizes_ if cluster_sizes_ is not None else []
+            ),
         )
+
+    @property
+    def summary(self) -> "KMeansSummary":
+        """Training summary (Spark KMeansSummary parity: trainingCost,
+        clusterSizes, numIter)."""
+        return KMeansSummary(
+            trainingCost=self._model_attributes["inertia_"],
+            clusterSizes=list(self._model_attributes["cluster_sizes_"]),
+            numIter=self._model_attributes["n_iter_"],
+        )
+
+    @property
+    def hasSummary(self) -> bool:
+        return len(self._model_attributes.get("cluster_sizes_", [])) > 0
 
     @property
     def cluster_centers_(self) -> np.ndarray:

@@ -225,7 +225,8 @@ class UMAP(_UMAPParams, Estimator):
         return self._set_params(n_neighbors=value)
 
     def _is_supervised(self) -> bool:
-        return False
+        # supervised fit when a labelCol is present (reference umap.py:1035-1050)
+        return self.isSet("labelCol")
 
     def _fit_array(
         self, X: Any, y: Optional[Any], ctx: _FitContext, params: Dict[str, Any]
@@ -235,14 +236,23 @@ class UMAP(_UMAPParams, Estimator):
         seed = int(self.getOrDefault("random_state"))
         rng = np.random.default_rng(seed)
         X_local = np.ascontiguousarray(X, dtype=np.float32)
+        y_local = np.asarray(y, dtype=np.float32) if y is not None else None
         if frac < 1.0 and X_local.shape[0] > 0:
             m = max(1, int(frac * X_local.shape[0]))
-            X_local = X_local[rng.choice(X_local.shape[0], m, replace=False)]
-        parts = comm.allgather_obj(X_local)  # gather to all; rank0 fits
-        X_fit = np.concatenate([p for p in parts if len(p)], axis=0)
+            sel = rng.choice(X_local.shape[0], m, replace=False)
+            X_local = X_local[sel]
+            if y_local is not None:
+                y_local = y_local[sel]
+        parts = comm.allgather_obj((X_local, y_local))  # gather to all; rank0 fits
+        X_fit = np.concatenate([p for p, _ in parts if len(p)], axis=0)
+        y_fit = (
+            np.concatenate([q for p, q in parts if q is not None and len(p)])
+            if y_local is not None
+            else None
+        )
 
         if comm.rank == 0:
-            emb = self._fit_single(X_fit, ctx.device)
+            emb = self._fit_single(X_fit, ctx.device, y_fit)
         else:
             emb = np.zeros((X_fit.shape[0], int(self.getOrDefault("n_components"))), dtype=np.float32)
         emb_t = torch.from_numpy(emb).to(comm._coll_device())
@@ -265,7 +275,9 @@ class UMAP(_UMAPParams, Estimator):
             )
         return float(a), float(b)
 
-    def _fit_single(self, X: np.ndarray, device: torch.device) -> np.ndarray:
+    def _fit_single(
+        self, X: np.ndarray, device: torch.device, y: Optional[np.ndarray] = None
+    ) -> np.ndarray:
         n = X.shape[0]
         k = int(float(self.getOrDefault("n_neighbors")))
         k = min(k, max(2, n - 1))
@@ -281,6 +293,14 @@ class UMAP(_UMAPParams, Estimator):
             float(self.getOrDefault("set_op_mix_ratio")),
             float(self.getOrDefault("local_connectivity")),
         )
+        if y is not None:
+            # supervised: categorical simplicial-set intersection — edges
+            # across label boundaries shrink by exp(-far_dist) (umap-learn
+            # semantics; reference supervised fit, umap.py:1035-1050)
+            far = math.exp(-5.0)
+            diff = y[rows] != y[cols]
+            vals = vals.copy()
+            vals[diff] *= far
         a, b = self._ab()
 
         n_epochs = int(self.getOrDefault("n_epochs")) or (500 if n <= 10000 else 200)

@@ -22,11 +22,13 @@ def lbfgs(
     history: int = 10,
     l1_strength: Optional[torch.Tensor] = None,
     linesearch_max_iter: int = 20,
-) -> Tuple[torch.Tensor, float, int]:
+) -> Tuple[torch.Tensor, float, int, list]:
     """Minimize f(w) (+ sum(l1_strength*|w|) when given -> OWL-QN).
 
     closure: w -> (smooth loss, smooth grad).
-    Returns (w, final_loss_including_l1, n_iters).
+    Returns (w, final_loss_including_l1, n_iters, objective_history) —
+    the history mirrors Spark's objectiveHistory (reference
+    tests_large/test_large_logistic_regression.py compares it).
     Convergence: relative objective decrease < tol (Spark's criterion family).
     """
     w = w0.clone().to(torch.float64)
@@ -42,6 +44,7 @@ def lbfgs(
     loss, grad = closure(w)
     grad = grad.to(torch.float64)
     obj = full_obj(loss, w)
+    obj_history = [obj]
 
     s_hist: list = []
     y_hist: list = []
@@ -131,7 +134,8 @@ def lbfgs(
 
         rel = abs(obj - obj_new) / max(abs(obj), abs(obj_new), 1.0)
         w, grad, loss, obj = w_new, grad_new, loss_new, obj_new
+        obj_history.append(obj)
         if rel < tol:
             break
 
-    return w, obj, n_iter
+    return w, obj, n_iter, obj_history

@@ -2,6 +2,7 @@
 
 from __future__ import annotations
 
+import contextlib
 import logging
 import sys
 from typing import Any, Optional, Tuple
@@ -79,3 +80,25 @@ def inertia_dtype(X: torch.Tensor) -> torch.dtype:
 
 def as_numpy(t: torch.Tensor) -> np.ndarray:
     return t.detach().cpu().numpy()
+
+
+@contextlib.contextmanager
+def annotate(name: str):
+    """rocprof-visible named range (roctx via torch's nvtx shim on ROCm).
+    The reference has no tracer integration; these ranges make
+    fit/transform phases attributable in rocprofv3 --sys-trace timelines."""
+    pushed = False
+    try:
+        if torch.cuda.is_available():
+            torch.cuda.nvtx.range_push(name)
+            pushed = True
+    except Exception:
+        pushed = False
+    try:
+        yield
+    finally:
+        if pushed:
+            try:
+                torch.cuda.nvtx.range_pop()
+            except Exception:
+                pass
