@@ -153,12 +153,21 @@ class Estimator(_NativeParams, HasVerbose):
         ctx = _FitContext(comm, pdesc, comm.device)
         logger = get_logger(self.__class__)
 
+        verbose = bool(self.getOrDefault("verbose")) if self.hasParam("verbose") else False
         models: List[Model] = []
         for pm in param_maps:
             params = dict(self._native_params)
             params.update(pm)
+            import time as _time
+
+            t0 = _time.perf_counter()
             with annotate(f"srml::{type(self).__name__}.fit"):
                 attrs = self._fit_array(X, y, ctx, params)
+            if verbose:
+                logger.info(
+                    f"{type(self).__name__} fit: rows={pdesc.m} cols={pdesc.n} "
+                    f"ranks={comm.world_size} took {_time.perf_counter() - t0:.3f}s"
+                )
             model = self._create_model(attrs)
             self._copyValues(model)
             model._native_params = dict(params)
