@@ -340,18 +340,15 @@ def _grow_tree(
                     H = hist.view(B, F, n_bins, n_classes)
                     gain, sbin, lval, rval, lcnt = _best_split_class(H, min_leaf)
                 else:
-                    cnt = torch.zeros(B * F * n_bins, dtype=torch.float32, device=dev)
-                    sm = torch.zeros_like(cnt)
-                    s2 = torch.zeros_like(cnt)
+                    # one fused (count, sum, sum^2) scatter instead of three
+                    hist3 = torch.zeros(B * F * n_bins, 3, dtype=torch.float32, device=dev)
                     flat = base.flatten()
-                    ones = torch.ones(flat.numel(), device=dev)
-                    cnt.index_add_(0, flat, ones)
-                    sm.index_add_(0, flat, yb[:, None].expand(-1, F).flatten())
-                    s2.index_add_(0, flat, (yb * yb)[:, None].expand(-1, F).flatten())
-                    H = torch.stack(
-                        [cnt.view(B, F, n_bins), sm.view(B, F, n_bins), s2.view(B, F, n_bins)],
-                        dim=-1,
-                    )
+                    src = torch.empty(bins.shape[0], 3, dtype=torch.float32, device=dev)
+                    src[:, 0] = 1.0
+                    src[:, 1] = yb
+                    src[:, 2] = yb * yb
+                    hist3.index_add_(0, flat, src.repeat_interleave(F, dim=0) if F > 1 else src)
+                    H = hist3.view(B, F, n_bins, 3)
                     gain, sbin, lval, rval, lcnt = _best_split_reg(H, min_leaf)
                 # best feature within chunk
                 g, fidx = _max_lastdim(gain)

@@ -48,30 +48,33 @@ def find_ab_params(spread: float = 1.0, min_dist: float = 0.1) -> Tuple[float, f
 
 def _smooth_knn(dists: np.ndarray, local_connectivity: float = 1.0) -> Tuple[np.ndarray, np.ndarray]:
     """Per-row (sigma, rho) by binary search: sum_j exp(-(d_ij-rho)/sigma) =
-    log2(k) (umap-learn smooth_knn_dist semantics)."""
+    log2(k) (umap-learn smooth_knn_dist semantics). Fully vectorized over
+    rows: 64 synchronized bisection steps on the whole matrix (the per-row
+    Python loop cost ~1 ms/row at 100k rows)."""
     n, k = dists.shape
     target = math.log2(k)
-    rho = np.zeros(n)
-    sigma = np.zeros(n)
-    for i in range(n):
-        row = dists[i]
-        nz = row[row > 0]
-        if len(nz) >= local_connectivity:
-            rho[i] = nz[int(local_connectivity) - 1] if int(local_connectivity) >= 1 else nz[0]
-        elif len(nz) > 0:
-            rho[i] = nz[-1]
-        lo, hi, mid = 0.0, np.inf, 1.0
-        for _ in range(64):
-            val = np.exp(-np.maximum(row - rho[i], 0.0) / mid).sum()
-            if abs(val - target) < 1e-5:
-                break
-            if val > target:
-                hi = mid
-                mid = (lo + hi) / 2.0
-            else:
-                lo = mid
-                mid = mid * 2 if hi == np.inf else (lo + hi) / 2.0
-        sigma[i] = max(mid, 1e-3 * (row.mean() if row.mean() > 0 else 1.0))
+    d64 = dists.astype(np.float64)
+    pos = d64 > 0
+    npos = pos.sum(axis=1)
+    lc = max(1, int(local_connectivity))
+    # rho: lc-th smallest positive distance (rows are sorted by construction)
+    first_pos = np.where(npos > 0, np.argmax(pos, axis=1), 0)
+    rho_idx = np.minimum(first_pos + lc - 1, np.maximum(first_pos, first_pos + npos - 1))
+    rho = np.take_along_axis(d64, rho_idx[:, None], axis=1)[:, 0]
+    rho[npos == 0] = 0.0
+
+    gap = np.maximum(d64 - rho[:, None], 0.0)
+    lo = np.zeros(n)
+    hi = np.full(n, np.inf)
+    mid = np.ones(n)
+    for _ in range(64):
+        val = np.exp(-gap / mid[:, None]).sum(axis=1)
+        high = val > target  # sigma too large -> shrink
+        hi = np.where(high, mid, hi)
+        lo = np.where(high, lo, mid)
+        mid = np.where(high, (lo + hi) / 2.0, np.where(np.isinf(hi), mid * 2, (lo + hi) / 2.0))
+    mean_d = d64.mean(axis=1)
+    sigma = np.maximum(mid, 1e-3 * np.where(mean_d > 0, mean_d, 1.0))
     return sigma, rho
 
 
@@ -326,7 +329,7 @@ class UMAP(_UMAPParams, Estimator):
     def _init_embedding(self, X, rows, cols, vals, dim, seed) -> np.ndarray:
         n = X.shape[0]
         init = self.getOrDefault("init")
-        if init == "spectral" and n > dim + 1:
+        if init == "spectral" and dim + 1 < n <= 50000:
             try:
                 import scipy.sparse as sp
                 from scipy.sparse.linalg import eigsh
