@@ -211,38 +211,66 @@ def _bin_data(X: torch.Tensor, edges: torch.Tensor) -> torch.Tensor:
 
 
 class _Tree:
-    """Flat-array binary tree. Arrays grow as nodes are added."""
+    """Flat-array binary tree with BULK growth (one numpy append per node
+    batch — the per-node Python loop was the depth-13 bottleneck)."""
 
     __slots__ = ("feature", "threshold", "left", "right", "is_leaf", "value", "n_nodes")
 
     def __init__(self, value_width: int):
-        self.feature: List[int] = []
-        self.threshold: List[float] = []
-        self.left: List[int] = []
-        self.right: List[int] = []
-        self.is_leaf: List[bool] = []
-        self.value: List[np.ndarray] = []
+        self.feature = [np.empty(0, np.int32)]
+        self.threshold = [np.empty(0, np.float32)]
+        self.left = [np.empty(0, np.int32)]
+        self.right = [np.empty(0, np.int32)]
+        self.is_leaf = [np.empty(0, bool)]
+        self.value = [np.empty((0, value_width), np.float32)]
         self.n_nodes = 0
 
-    def add_node(self, value: np.ndarray) -> int:
-        i = self.n_nodes
-        self.n_nodes += 1
-        self.feature.append(-1)
-        self.threshold.append(0.0)
-        self.left.append(-1)
-        self.right.append(-1)
-        self.is_leaf.append(True)
-        self.value.append(value)
-        return i
+    def add_leaves(self, values: np.ndarray) -> int:
+        """Append len(values) leaf nodes; returns the first new node id."""
+        m = values.shape[0]
+        first = self.n_nodes
+        self.n_nodes += m
+        self.feature.append(np.full(m, -1, np.int32))
+        self.threshold.append(np.zeros(m, np.float32))
+        self.left.append(np.full(m, -1, np.int32))
+        self.right.append(np.full(m, -1, np.int32))
+        self.is_leaf.append(np.ones(m, bool))
+        self.value.append(values.astype(np.float32))
+        return first
+
+    def _consolidate(self) -> None:
+        if len(self.feature) > 1:
+            self.feature = [np.concatenate(self.feature)]
+            self.threshold = [np.concatenate(self.threshold)]
+            self.left = [np.concatenate(self.left)]
+            self.right = [np.concatenate(self.right)]
+            self.is_leaf = [np.concatenate(self.is_leaf)]
+            self.value = [np.concatenate(self.value, axis=0)]
+
+    def set_splits(
+        self,
+        node_ids: np.ndarray,
+        feature: np.ndarray,
+        threshold: np.ndarray,
+        left: np.ndarray,
+        right: np.ndarray,
+    ) -> None:
+        self._consolidate()
+        self.feature[0][node_ids] = feature.astype(np.int32)
+        self.threshold[0][node_ids] = threshold.astype(np.float32)
+        self.left[0][node_ids] = left.astype(np.int32)
+        self.right[0][node_ids] = right.astype(np.int32)
+        self.is_leaf[0][node_ids] = False
 
     def to_arrays(self) -> Dict[str, np.ndarray]:
+        self._consolidate()
         return {
-            "feature": np.asarray(self.feature, dtype=np.int32),
-            "threshold": np.asarray(self.threshold, dtype=np.float32),
-            "left": np.asarray(self.left, dtype=np.int32),
-            "right": np.asarray(self.right, dtype=np.int32),
-            "is_leaf": np.asarray(self.is_leaf, dtype=bool),
-            "value": np.stack(self.value).astype(np.float32),
+            "feature": self.feature[0],
+            "threshold": self.threshold[0],
+            "left": self.left[0],
+            "right": self.right[0],
+            "is_leaf": self.is_leaf[0],
+            "value": self.value[0],
         }
 
 
@@ -274,7 +302,8 @@ def _grow_tree(
         yf = y.to(torch.float32)
         root_val = np.array([float(yf.mean().item()) if n else 0.0, float(n)], dtype=np.float32)
 
-    root = tree.add_node(root_val)
+    root = tree.add_leaves(root_val[None, :])
+    edges_np = as_numpy(edges)
     node_of_row = torch.zeros(n, dtype=torch.int64, device=dev)
     frontier = [root]  # node ids whose rows are still splittable
 
@@ -319,17 +348,19 @@ def _grow_tree(
                 scores = torch.rand((B, d), generator=gen, device=dev)
                 feat_sel = scores.argsort(dim=1)[:, :max_features]  # [B, mf]
                 mf = max_features
+                Xb_rows = None  # sampled path reads only selected bytes
             else:
                 feat_sel = None
                 mf = d
-
-            Xb_rows = Xb[rows]
+                Xb_rows = Xb[rows]
             for f0 in range(0, mf, feat_chunk):
                 f1 = min(mf, f0 + feat_chunk)
                 F = f1 - f0
                 if feat_sel is not None:
                     sel = feat_sel[loc][:, f0:f1]  # [m, F] per-row node's features
-                    bins = Xb_rows.gather(1, sel).to(torch.int64)
+                    # direct 2D gather: touches only m*F bytes of Xb instead
+                    # of materializing the [m, d] row block
+                    bins = Xb[rows[:, None], sel].to(torch.int64)
                 else:
                     bins = Xb_rows[:, f0:f1].to(torch.int64)
                 base = (loc[:, None] * F + torch.arange(F, device=dev)[None, :]) * n_bins + bins
@@ -365,50 +396,53 @@ def _grow_tree(
                 best_lval = torch.where(upd[:, None], lval[ar, fidx], best_lval)
                 best_rval = torch.where(upd[:, None], rval[ar, fidx], best_rval)
 
-            # materialize splits
+            # materialize splits: fully vectorized (one bulk leaf append +
+            # one set_splits per batch; the per-node loop was the depth-13
+            # wall-clock bottleneck)
             bg = as_numpy(best_gain)
             bf = as_numpy(best_feat)
             bb = as_numpy(best_bin)
             lv = as_numpy(best_lval)
             rv = as_numpy(best_rval)
-            edges_np = as_numpy(edges)
-            split_nodes = []
-            for i, nid in enumerate(batch):
-                if bg[i] <= min_gain or bf[i] < 0:
-                    continue
-                f, b = int(bf[i]), int(bb[i])
-                thr = float(edges_np[f, b])
-                lchild = tree.add_node(lv[i])
-                rchild = tree.add_node(rv[i])
-                tree.feature[nid] = f
-                tree.threshold[nid] = thr
-                tree.left[nid] = lchild
-                tree.right[nid] = rchild
-                tree.is_leaf[nid] = False
-                split_nodes.append((nid, f, b, lchild, rchild))
-                if depth + 1 < max_depth:
-                    lcount = lv[i][-1] if task == "regression" else lv[i].sum()
-                    rcount = rv[i][-1] if task == "regression" else rv[i].sum()
-                    if lcount >= 2 * min_leaf:
-                        new_frontier.append(lchild)
-                    if rcount >= 2 * min_leaf:
-                        new_frontier.append(rchild)
-            if split_nodes:
-                nid_t = torch.tensor([s[0] for s in split_nodes], device=dev)
-                f_t = torch.tensor([s[1] for s in split_nodes], device=dev)
-                b_t = torch.tensor([s[2] for s in split_nodes], device=dev)
-                l_t = torch.tensor([s[3] for s in split_nodes], device=dev)
-                r_t = torch.tensor([s[4] for s in split_nodes], device=dev)
-                lut2 = torch.full((tree.n_nodes,), -1, dtype=torch.int64, device=dev)
-                lut2[nid_t] = torch.arange(len(split_nodes), dtype=torch.int64, device=dev)
-                sl = lut2[node_of_row]
-                moved = sl >= 0
-                mrows = torch.nonzero(moved).flatten()
-                srel = sl[mrows]
-                go_left = Xb[mrows].gather(1, f_t[srel].view(-1, 1)).flatten().to(
-                    torch.int64
-                ) <= b_t[srel]
-                node_of_row[mrows] = torch.where(go_left, l_t[srel], r_t[srel])
+            batch_np = np.asarray(batch, dtype=np.int64)
+            valid = (bg > min_gain) & (bf >= 0)
+            idxs = np.nonzero(valid)[0]
+            if len(idxs) == 0:
+                continue
+            ns = len(idxs)
+            children = np.empty((2 * ns, lv.shape[1]), dtype=np.float32)
+            children[0::2] = lv[idxs]
+            children[1::2] = rv[idxs]
+            first = tree.add_leaves(children)
+            l_ids = first + 2 * np.arange(ns)
+            r_ids = l_ids + 1
+            fsel = bf[idxs].astype(np.int64)
+            bsel = bb[idxs].astype(np.int64)
+            tree.set_splits(
+                batch_np[idxs], fsel, edges_np[fsel, bsel], l_ids, r_ids
+            )
+            if depth + 1 < max_depth:
+                if task == "regression":
+                    lcount = lv[idxs][:, -1]
+                    rcount = rv[idxs][:, -1]
+                else:
+                    lcount = lv[idxs].sum(axis=1)
+                    rcount = rv[idxs].sum(axis=1)
+                new_frontier.extend(l_ids[lcount >= 2 * min_leaf].tolist())
+                new_frontier.extend(r_ids[rcount >= 2 * min_leaf].tolist())
+            # reroute rows of split nodes (touch only the split feature byte)
+            nid_t = torch.from_numpy(batch_np[idxs]).to(dev)
+            f_t = torch.from_numpy(fsel).to(dev)
+            b_t = torch.from_numpy(bsel).to(dev)
+            l_t = torch.from_numpy(l_ids).to(dev)
+            r_t = torch.from_numpy(r_ids).to(dev)
+            lut2 = torch.full((tree.n_nodes,), -1, dtype=torch.int64, device=dev)
+            lut2[nid_t] = torch.arange(ns, dtype=torch.int64, device=dev)
+            sl = lut2[node_of_row]
+            mrows = torch.nonzero(sl >= 0).flatten()
+            srel = sl[mrows]
+            go_left = Xb[mrows, f_t[srel]].to(torch.int64) <= b_t[srel]
+            node_of_row[mrows] = torch.where(go_left, l_t[srel], r_t[srel])
         frontier = new_frontier
 
     return tree.to_arrays()
