@@ -140,7 +140,7 @@ def init_comm(backend: Optional[str] = None, timeout_s: int = 300) -> Comm:
 
     if world > 1:
         if backend is None:
-            backend = "nccl" if use_cuda else "gloo"
+            backend = os.environ.get("SRML_BACKEND") or ("nccl" if use_cuda else "gloo")
         if not dist.is_initialized():
             dist.init_process_group(
                 backend=backend,

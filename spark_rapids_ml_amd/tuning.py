@@ -136,7 +136,9 @@ class CrossValidator(Params):
 
 class CrossValidatorModel(Params):
     """Holds the best model + per-param-map metrics (reference
-    tuning.py:141-157)."""
+    tuning.py:141-157). Persistence stores the best model under
+    bestModel/ plus metrics JSON (reference load() resolves the concrete
+    model class the same way, tuning.py:159-186)."""
 
     def __init__(
         self,
@@ -151,3 +153,37 @@ class CrossValidatorModel(Params):
 
     def transform(self, df: DataFrame) -> DataFrame:
         return self.bestModel.transform(df)
+
+    def save(self, path: str) -> None:
+        import json
+        import os
+
+        comm = get_comm()
+        self.bestModel.write().overwrite().save(os.path.join(path, "bestModel"))
+        if comm.rank == 0:
+            meta = {
+                "class": f"{type(self.bestModel).__module__}."
+                f"{type(self.bestModel).__qualname__}",
+                "avgMetrics": self.avgMetrics,
+                "stdMetrics": self.stdMetrics,
+            }
+            with open(os.path.join(path, "cv_metadata.json"), "w") as f:
+                json.dump(meta, f)
+        comm.barrier()
+
+    @classmethod
+    def load(cls, path: str) -> "CrossValidatorModel":
+        import importlib
+        import json
+        import os
+
+        with open(os.path.join(path, "cv_metadata.json")) as f:
+            meta = json.load(f)
+        mod_name, cls_name = meta["class"].rsplit(".", 1)
+        model_cls = getattr(importlib.import_module(mod_name), cls_name)
+        best = model_cls.load(os.path.join(path, "bestModel"))
+        return cls(
+            bestModel=best,
+            avgMetrics=meta["avgMetrics"],
+            stdMetrics=meta["stdMetrics"],
+        )

@@ -76,3 +76,21 @@ def test_evaluators_basic():
     assert auc > 0.9
     acc = MulticlassClassificationEvaluator(metricName="accuracy").evaluate(df)
     assert acc > 0.9
+
+
+def test_cv_model_persistence(tmp_path):
+    X, y = make_regression(n_samples=200, n_features=6, noise=1.0, random_state=0)
+    df = DataFrame.from_numpy(X.astype(np.float64), y.astype(np.float64))
+    lr = LinearRegression()
+    cv = CrossValidator(
+        estimator=lr,
+        estimatorParamMaps=ParamGridBuilder().addGrid(lr.getParam("regParam"), [0.0, 1.0]).build(),
+        evaluator=RegressionEvaluator(metricName="rmse"),
+        numFolds=2,
+    )
+    model = cv.fit(df)
+    path = str(tmp_path / "cv")
+    model.save(path)
+    loaded = CrossValidatorModel.load(path)
+    assert loaded.avgMetrics == model.avgMetrics
+    assert np.allclose(loaded.bestModel.coefficients, model.bestModel.coefficients)
