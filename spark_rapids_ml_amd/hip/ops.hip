@@ -568,6 +568,147 @@ __global__ __launch_bounds__(256) void kmeans_assign_glds3_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// kmeans_assign_sb3: BM=64 x BN=256, BK=32, SINGLE-buffer glds staging.
+// 41 KB LDS -> 3 blocks/CU (3 waves/SIMD cover each other's staging) and
+// X is re-streamed only k/256 times. Requires d%4==0.
+// ---------------------------------------------------------------------------
+
+constexpr int KS_BM = 64;
+constexpr int KS_BN = 256;
+constexpr int KS_BK = 32;
+constexpr int KS_X0 = 0;                       // 64*32 floats
+constexpr int KS_C0 = KS_BM * KS_BK;           // 256*32 floats
+constexpr int KS_BEST = KS_C0 + KS_BN * KS_BK; // u64[64] = 128 floats
+constexpr int KS_SCR = KS_BEST + 128;
+constexpr int KS_TOTAL = KS_SCR + 16;
+
+__global__ __launch_bounds__(256) void kmeans_assign_sb3_kernel(
+    const float* __restrict__ X, const float* __restrict__ C,
+    const float* __restrict__ x_sq, const float* __restrict__ c_sq,
+    int n, int d, int k,
+    int32_t* __restrict__ labels, float* __restrict__ min_dists,
+    double* __restrict__ inertia) {
+  __shared__ __attribute__((aligned(16))) float smem[KS_TOTAL];
+
+  const int i0 = blockIdx.x * KS_BM;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;   // wave = column quarter (wc)
+
+  unsigned long long* best = reinterpret_cast<unsigned long long*>(&smem[KS_BEST]);
+  for (int i = tid; i < KS_BM; i += blockDim.x) best[i] = ~0ULL;
+
+  const bool full_rows = (i0 + KS_BM <= n);
+  const int nsteps = (d + KS_BK - 1) / KS_BK;
+
+  // glds: instruction q covers rows [q*8, q*8+8) of a [rows][32]-float image
+  auto issue_glds = [&](const float* __restrict__ src, int base_row, int d0,
+                        int lds_off, int n_instr_per_wave) {
+    for (int qq = 0; qq < n_instr_per_wave; ++qq) {
+      const int q = wave * n_instr_per_wave + qq;
+      const int i = q * 8 + (lane >> 3);
+      const int g = (lane & 7) ^ (i & 7);
+      const float* gp = src + (int64_t)(base_row + i) * d + d0 + (g << 2);
+      auto lp = (__attribute__((address_space(3))) uint32_t*)(&smem[lds_off + q * 256]);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)gp, lp, 16, 0, 0);
+    }
+  };
+  auto issue_scalar = [&](const float* __restrict__ src, int base_row, int lim,
+                          int d0, int lds_off, int rows) {
+    for (int e = tid; e < rows * KS_BK; e += 256) {
+      int i = e >> 5, kd = e & 31;
+      int gr = base_row + i, gd = d0 + kd;
+      smem[lds_off + kg_swz(i, kd)] =
+          (gr < lim && gd < d) ? src[(int64_t)gr * d + gd] : 0.0f;
+    }
+  };
+
+  for (int j0 = 0; j0 < k; j0 += KS_BN) {
+    const bool pair_full_cols = full_rows && (j0 + KS_BN <= k);
+    f32x16 acc[2][2];
+#pragma unroll
+    for (int m = 0; m < 2; ++m)
+#pragma unroll
+      for (int nn = 0; nn < 2; ++nn)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) acc[m][nn][r] = 0.0f;
+
+    for (int step = 0; step < nsteps; ++step) {
+      const int d0 = step * KS_BK;
+      __syncthreads();  // previous MFMA reads done (and best[] init)
+      if (pair_full_cols && d0 + KS_BK <= d) {
+        issue_glds(X, i0, d0, KS_X0, 2);
+        issue_glds(C, j0, d0, KS_C0, 8);
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      } else {
+        issue_scalar(X, i0, n, d0, KS_X0, KS_BM);
+        issue_scalar(C, j0, k, d0, KS_C0, KS_BN);
+      }
+      __syncthreads();
+
+      const int ci = wave * 64 + (lane & 31);
+#pragma unroll 8
+      for (int kk = 0; kk < KS_BK / 2; ++kk) {
+        const int kd = 2 * kk + (lane >> 5);
+        float a0 = smem[KS_X0 + kg_swz(lane & 31, kd)];
+        float a1 = smem[KS_X0 + kg_swz((lane & 31) + 32, kd)];
+        float b0 = smem[KS_C0 + kg_swz(ci, kd)];
+        float b1 = smem[KS_C0 + kg_swz(ci + 32, kd)];
+        acc[0][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc[0][0], 0, 0, 0);
+        acc[0][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc[0][1], 0, 0, 0);
+        acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc[1][0], 0, 0, 0);
+        acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc[1][1], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+
+#pragma unroll
+    for (int m = 0; m < 2; ++m) {
+#pragma unroll
+      for (int nn = 0; nn < 2; ++nn) {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          int row = m * 32 + (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
+          int col = wave * 64 + nn * 32 + (lane & 31);
+          int gi = i0 + row, gj = j0 + col;
+          if (gi < n && gj < k) {
+            float dist = x_sq[gi] + c_sq[gj] - 2.0f * acc[m][nn][r];
+            dist = dist < 0.0f ? 0.0f : dist;
+            unsigned long long packed =
+                ((unsigned long long)__float_as_uint(dist) << 32) |
+                (unsigned int)gj;
+            atomicMin(&best[row], packed);
+          }
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  double* block_inertia = reinterpret_cast<double*>(&smem[KS_SCR]);
+  double partial = 0.0;
+  for (int i = tid; i < KS_BM; i += blockDim.x) {
+    int gi = i0 + i;
+    if (gi < n) {
+      unsigned long long p = best[i];
+      float dist = __uint_as_float((unsigned int)(p >> 32));
+      labels[gi] = (int32_t)(p & 0xffffffffu);
+      min_dists[gi] = dist;
+      partial += (double)dist;
+    }
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1)
+    partial += __shfl_down(partial, off, 64);
+  if (lane == 0) block_inertia[wave] = partial;
+  __syncthreads();
+  if (tid == 0)
+    atomicAdd(inertia, block_inertia[0] + block_inertia[1] + block_inertia[2] +
+                           block_inertia[3]);
+}
+
+// ---------------------------------------------------------------------------
 // label_accumulate: sums[label[i]] += X[i], counts[label[i]] += 1 via a
 // sort-based segmented reduction: rows pre-sorted by label (torch.sort in
 // the wrapper), each (label, split) block register-accumulates its column
@@ -847,7 +988,14 @@ std::vector<torch::Tensor> kmeans_assign(torch::Tensor X, torch::Tensor C,
     // 1-block/CU span 111 ms (single wave/SIMD starves the MFMA pipe).
     // Register staging is the default; glds variants stay selectable.
     static const char* v = getenv("SRML_KMEANS_VARIANT");
-    if (d % 4 == 0 && v && v[0] == '3')
+    if (d % 4 == 0 && v && v[0] == 's') {
+      const int grid_s = (int)((n + KS_BM - 1) / KS_BM);
+      hipLaunchKernelGGL(kmeans_assign_sb3_kernel, dim3(grid_s), dim3(256), 0,
+                         cur_stream(), X.data_ptr<float>(), C.data_ptr<float>(),
+                         x_sq.data_ptr<float>(), c_sq.data_ptr<float>(), (int)n, d, k,
+                         labels.data_ptr<int32_t>(), min_dists.data_ptr<float>(),
+                         inertia.data_ptr<double>());
+    } else if (d % 4 == 0 && v && v[0] == '3')
       hipLaunchKernelGGL(kmeans_assign_glds3_kernel, dim3(grid), dim3(256), 0,
                          cur_stream(), X.data_ptr<float>(), C.data_ptr<float>(),
                          x_sq.data_ptr<float>(), c_sq.data_ptr<float>(), (int)n, d, k,

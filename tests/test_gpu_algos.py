@@ -80,3 +80,61 @@ def test_knn_gpu_exact():
     sk_d, sk_i = sk.kneighbors(Q)
     # distances must match; index ties allowed
     assert np.allclose(np.asarray(knn_df["distances"]), sk_d, atol=1e-3)
+
+
+def test_umap_gpu_trustworthiness():
+    from sklearn.manifold import trustworthiness
+
+    from spark_rapids_ml_amd import UMAP
+
+    X, _ = make_blobs(n_samples=2000, n_features=32, centers=6, cluster_std=0.5, random_state=0)
+    X = X.astype(np.float32)
+    model = UMAP(n_neighbors=15, n_epochs=150, random_state=42).fit(DataFrame.from_numpy(X))
+    t = trustworthiness(X, model.embedding, n_neighbors=15)
+    assert t > 0.8
+
+
+def test_dbscan_gpu_matches_sklearn():
+    from sklearn.cluster import DBSCAN as SkDBSCAN
+    from sklearn.metrics import adjusted_rand_score
+
+    from spark_rapids_ml_amd import DBSCAN
+
+    X, _ = make_blobs(n_samples=3000, n_features=8, centers=5, cluster_std=0.4, random_state=0)
+    X = X.astype(np.float32)
+    model = DBSCAN(eps=1.5, min_samples=5).fit(DataFrame.from_numpy(X))
+    out = model.transform(DataFrame.from_numpy(X))
+    sk = SkDBSCAN(eps=1.5, min_samples=5).fit(X)
+    assert adjusted_rand_score(np.asarray(out["prediction"]), sk.labels_) == 1.0
+
+
+def test_ann_gpu_recall():
+    from sklearn.neighbors import NearestNeighbors as SkNN
+
+    from spark_rapids_ml_amd import ApproximateNearestNeighbors
+
+    rng = np.random.default_rng(0)
+    X = rng.normal(size=(20000, 64)).astype(np.float32)
+    model = ApproximateNearestNeighbors(
+        k=10, algorithm="ivfflat", algoParams={"nlist": 64, "nprobe": 16}
+    ).fit(DataFrame.from_numpy(X))
+    _, _, knn_df = model.kneighbors(DataFrame.from_numpy(X[:200]))
+    idx = np.asarray(knn_df["indices"])
+    sk = SkNN(n_neighbors=10).fit(X)
+    _, sk_idx = sk.kneighbors(X[:200])
+    hits = sum(len(set(a.tolist()) & set(b.tolist())) for a, b in zip(idx, sk_idx))
+    assert hits / sk_idx.size > 0.75
+
+
+def test_sparse_logreg_gpu():
+    import scipy.sparse as sp
+
+    rng = np.random.default_rng(0)
+    X = rng.normal(size=(20000, 128))
+    X[rng.random(X.shape) < 0.9] = 0.0
+    w = rng.normal(size=128)
+    y = (X @ w > 0).astype(np.float64)
+    Xs = sp.csr_matrix(X.astype(np.float32))
+    model = LogisticRegression(regParam=1e-4, maxIter=100).fit(DataFrame.from_numpy(Xs, y))
+    out = model.transform(DataFrame.from_numpy(Xs))
+    assert (np.asarray(out["prediction"]) == y).mean() > 0.95
