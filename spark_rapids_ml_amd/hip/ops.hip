@@ -915,6 +915,165 @@ __global__ void gram_mirror_kernel(float* __restrict__ out, int d) {
 }
 
 // ---------------------------------------------------------------------------
+// knn_select: fused brute-force k-NN — MFMA distance tiles + in-LDS
+// per-query top-k (reference NearestNeighborsMG's tiled ||x-y||^2 + top-k
+// merge, SURVEY.md §2.3b knn.py:763-774). No [nq, ni] distance matrix is
+// materialized: each 64-query block streams item tiles, keeps an unsorted
+// k-candidate pool per query in LDS (one merge lane per query, current-max
+// cached in registers), and writes its item-range partial top-k; partials
+// across item splits merge on the host side (torch.topk over [q, S*k]).
+// k <= 64.
+// ---------------------------------------------------------------------------
+
+constexpr int KN_QB = 64;    // queries per block
+constexpr int KN_IB = 128;   // items per tile
+constexpr int KN_BK = 32;    // d chunk
+constexpr int KN_KMAX = 64;
+
+__global__ __launch_bounds__(256) void knn_select_kernel(
+    const float* __restrict__ Q,     // [nq, d]
+    const float* __restrict__ I,     // [ni, d]
+    const float* __restrict__ q_sq,  // [nq]
+    const float* __restrict__ i_sq,  // [ni]
+    int nq, int ni, int d, int k, int isplit,
+    float* __restrict__ out_dist,    // [nq, isplit, k]
+    int32_t* __restrict__ out_idx) { // [nq, isplit, k]
+  // floats: lds_q [32][65], lds_i [32][129], dist [64][129], pool u64[64][k]
+  __shared__ __attribute__((aligned(16))) float smem[32 * 65 + 32 * 129 + 64 * 129 +
+                                                     2 * KN_QB * KN_KMAX];
+  constexpr int SQ = 0;
+  constexpr int SI = 32 * 65;
+  constexpr int SD = SI + 32 * 129;
+  constexpr int SP = SD + 64 * 129;
+
+  const int q0 = (blockIdx.x / isplit) * KN_QB;
+  const int split = blockIdx.x % isplit;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = wave >> 1;  // 0..1 query half (32 rows)
+  const int wc = wave & 1;   // 0..1 item half (64 cols)
+
+  unsigned long long* pool = reinterpret_cast<unsigned long long*>(&smem[SP]);
+  for (int e = tid; e < KN_QB * k; e += 256) pool[e] = ~0ULL;
+
+  const int per = (ni + isplit - 1) / isplit;
+  const int it_begin = split * per;
+  const int it_end = min(ni, it_begin + per);
+
+  // merge-lane registers: thread t (<64) owns query q0+t
+  float cur_max = __uint_as_float(0x7f7fffffu);  // FLT_MAX
+  int cur_pos = 0;
+  int filled = 0;
+
+  for (int ib = it_begin; ib < it_end; ib += KN_IB) {
+    f32x16 acc[2];
+#pragma unroll
+    for (int nn = 0; nn < 2; ++nn)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) acc[nn][r] = 0.0f;
+
+    for (int d0 = 0; d0 < d; d0 += KN_BK) {
+      // stage Q[q0:q0+64, d0:+32] -> lds_q[kd][row], I tile likewise
+      for (int e = tid; e < KN_QB * KN_BK; e += 256) {
+        int row = e >> 5, kd = e & 31;
+        int gq = q0 + row, gd = d0 + kd;
+        smem[SQ + kd * 65 + row] =
+            (gq < nq && gd < d) ? Q[(int64_t)gq * d + gd] : 0.0f;
+      }
+      for (int e = tid; e < KN_IB * KN_BK; e += 256) {
+        int row = e >> 5, kd = e & 31;
+        int gi = ib + row, gd = d0 + kd;
+        smem[SI + kd * 129 + row] =
+            (gi < it_end && gd < d) ? I[(int64_t)gi * d + gd] : 0.0f;
+      }
+      __syncthreads();
+#pragma unroll 8
+      for (int kk = 0; kk < KN_BK / 2; ++kk) {
+        const int kd = 2 * kk + (lane >> 5);
+        float a = smem[SQ + kd * 65 + wr * 32 + (lane & 31)];
+        float b0 = smem[SI + kd * 129 + wc * 64 + (lane & 31)];
+        float b1 = smem[SI + kd * 129 + wc * 64 + 32 + (lane & 31)];
+        acc[0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a, b0, acc[0], 0, 0, 0);
+        acc[1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a, b1, acc[1], 0, 0, 0);
+      }
+      __syncthreads();
+    }
+
+    // distances into SD[row][col]
+#pragma unroll
+    for (int nn = 0; nn < 2; ++nn) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        int row = wr * 32 + (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
+        int col = wc * 64 + nn * 32 + (lane & 31);
+        int gq = q0 + row, gi = ib + col;
+        float dist = 3.4e38f;
+        if (gq < nq && gi < it_end) {
+          dist = q_sq[gq] + i_sq[gi] - 2.0f * acc[nn][r];
+          dist = dist < 0.0f ? 0.0f : dist;
+        }
+        smem[SD + row * 129 + col] = dist;
+      }
+    }
+    __syncthreads();
+
+    // merge: lane t of waves 0-? — threads 0..63 each own one query row
+    if (tid < KN_QB) {
+      const int q = tid;
+      unsigned long long* mypool = pool + q * k;
+      const int lim = min(KN_IB, it_end - ib);
+      for (int j = 0; j < lim; ++j) {
+        float dd = smem[SD + q * 129 + j];
+        if (filled < k) {
+          mypool[filled] =
+              ((unsigned long long)__float_as_uint(dd) << 32) | (unsigned)(ib + j);
+          ++filled;
+          if (filled == k) {  // establish current max
+            cur_max = -1.0f;
+            for (int t2 = 0; t2 < k; ++t2) {
+              float v = __uint_as_float((unsigned)(mypool[t2] >> 32));
+              if (v > cur_max) { cur_max = v; cur_pos = t2; }
+            }
+          }
+        } else if (dd < cur_max) {
+          mypool[cur_pos] =
+              ((unsigned long long)__float_as_uint(dd) << 32) | (unsigned)(ib + j);
+          cur_max = -1.0f;
+          for (int t2 = 0; t2 < k; ++t2) {  // rescan for new max
+            float v = __uint_as_float((unsigned)(mypool[t2] >> 32));
+            if (v > cur_max) { cur_max = v; cur_pos = t2; }
+          }
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  // write partials (sorted ascending by simple selection — k<=64)
+  if (tid < KN_QB) {
+    const int q = tid;
+    if (q0 + q < nq) {
+      unsigned long long* mypool = pool + q * k;
+      for (int a = 0; a < k; ++a) {  // selection sort in LDS
+        int best_t = a;
+        unsigned long long bv = mypool[a];
+        for (int b = a + 1; b < k; ++b)
+          if (mypool[b] < bv) { bv = mypool[b]; best_t = b; }
+        unsigned long long tmp = mypool[a];
+        mypool[a] = bv;
+        mypool[best_t] = tmp;
+        bool invalid = (bv & 0xffffffffu) == 0xffffffffu;
+        float dv = invalid ? 3.4e38f : __uint_as_float((unsigned)(bv >> 32));
+        int64_t off = ((int64_t)(q0 + q) * isplit + split) * k + a;
+        out_dist[off] = dv;
+        out_idx[off] = invalid ? -1 : (int32_t)(bv & 0xffffffffu);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // softmax_residual_loss: per-row softmax (C>1) or sigmoid (C==1) residual
 // and summed log-loss. resid = softmax(scores) - onehot(y) (or p - y).
 // One wave per row chunk; memory-bound, fused to one pass.
@@ -1056,6 +1215,36 @@ torch::Tensor gram_f32(torch::Tensor A) {
   return out;
 }
 
+std::vector<torch::Tensor> knn_select(torch::Tensor Q, torch::Tensor I, int64_t k) {
+  TORCH_CHECK(Q.is_cuda() && I.is_cuda());
+  TORCH_CHECK(Q.dtype() == torch::kFloat32 && I.dtype() == torch::kFloat32);
+  TORCH_CHECK(Q.is_contiguous() && I.is_contiguous());
+  TORCH_CHECK(k >= 1 && k <= KN_KMAX, "knn_select supports k in [1, 64]");
+  const int nq = (int)Q.size(0);
+  const int ni = (int)I.size(0);
+  const int d = (int)Q.size(1);
+  auto q_sq = (Q * Q).sum(1);
+  auto i_sq = (I * I).sum(1);
+  const int qblocks = (nq + KN_QB - 1) / KN_QB;
+  int isplit = std::max(1, std::min((ni + KN_IB - 1) / KN_IB, 512 / std::max(1, qblocks)));
+  auto dist = torch::empty({nq, (int64_t)isplit, k}, Q.options());
+  auto idx = torch::empty({nq, (int64_t)isplit, k}, Q.options().dtype(torch::kInt32));
+  if (nq > 0 && ni > 0)
+    hipLaunchKernelGGL(knn_select_kernel, dim3((unsigned)(qblocks * isplit)), dim3(256),
+                       0, cur_stream(), Q.data_ptr<float>(), I.data_ptr<float>(),
+                       q_sq.data_ptr<float>(), i_sq.data_ptr<float>(), nq, ni, d,
+                       (int)k, isplit, dist.data_ptr<float>(), idx.data_ptr<int32_t>());
+  // merge partial top-k across item splits
+  auto dflat = dist.view({nq, (int64_t)isplit * k});
+  auto iflat = idx.view({nq, (int64_t)isplit * k});
+  int64_t kk = std::min<int64_t>(k, (int64_t)isplit * k);
+  auto top = dflat.topk(kk, /*dim=*/1, /*largest=*/false);
+  auto vals = std::get<0>(top);
+  auto order = std::get<1>(top);
+  auto ids = iflat.gather(1, order);
+  return {vals, ids.to(torch::kInt64)};
+}
+
 std::vector<torch::Tensor> softmax_residual_loss(torch::Tensor scores, torch::Tensor y) {
   TORCH_CHECK(scores.is_cuda() && y.is_cuda());
   TORCH_CHECK(scores.dtype() == torch::kFloat32);
@@ -1078,5 +1267,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("label_accumulate", &label_accumulate, "per-center sum/count scatter");
   m.def("gram_f32", &gram_f32, "A^T A via MFMA f32");
   m.def("softmax_residual_loss", &softmax_residual_loss, "fused softmax residual + loss");
+  m.def("knn_select", &knn_select, "fused MFMA distance + in-LDS top-k");
   m.attr("_is_hip") = true;
 }

@@ -34,7 +34,7 @@ from ..params import (
     TypeConverters,
     DictTypeConverters,
 )
-from ..ops.torch_ref import knn_topk
+from ..ops.knn import knn_topk
 from ..utils import as_numpy, get_logger
 
 

@@ -26,7 +26,7 @@ from ..params import (
     Param,
     TypeConverters,
 )
-from ..ops.torch_ref import knn_topk
+from ..ops.knn import knn_topk
 from ..utils import as_numpy, get_logger
 
 

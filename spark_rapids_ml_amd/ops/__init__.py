@@ -15,6 +15,7 @@ from . import torch_ref
 from .kmeans import kmeans_assign_reduce, kmeans_predict
 from .linalg import gram, xty_gram, eigh_sym, sign_flip
 from .glm import logistic_grad_loss, linear_grad_loss
+from .knn import knn_topk
 
 __all__ = [
     "has_hip_ops",
@@ -29,5 +30,6 @@ __all__ = [
     "eigh_sym",
     "sign_flip",
     "logistic_grad_loss",
+    "knn_topk",
     "linear_grad_loss",
 ]

@@ -94,3 +94,17 @@ def test_hip_required_on_gpu():
     assert has_hip_ops()
     X = torch.zeros(4, 4).cuda()
     assert use_hip(X)
+
+
+@pytest.mark.parametrize("nq,ni,d,k", [(500, 2000, 64, 8), (64, 128, 32, 5), (1000, 10000, 768, 64), (130, 999, 33, 16)])
+def test_knn_select_matches_ref(ext, nq, ni, d, k):
+    Q = _rand(nq, d, seed=3)
+    I = _rand(ni, d, seed=4)
+    d2, idx = ext.knn_select(Q, I, k)
+    ref_d, ref_i = torch_ref.knn_topk(Q, I, k)
+    dists = torch.sqrt(torch.clamp(d2, min=0))
+    assert torch.allclose(dists, ref_d, rtol=1e-3, atol=1e-3), (dists - ref_d).abs().max()
+    # indices equal up to distance ties
+    mism = idx != ref_i
+    if bool(mism.any()):
+        assert torch.allclose(dists[mism], ref_d[mism], rtol=1e-3, atol=1e-3)
