@@ -973,19 +973,35 @@ __global__ __launch_bounds__(256) void knn_select_kernel(
 #pragma unroll
       for (int r = 0; r < 16; ++r) acc[nn][r] = 0.0f;
 
+    const bool tile_full = (q0 + KN_QB <= nq) && (ib + KN_IB <= it_end);
     for (int d0 = 0; d0 < d; d0 += KN_BK) {
-      // stage Q[q0:q0+64, d0:+32] -> lds_q[kd][row], I tile likewise
-      for (int e = tid; e < KN_QB * KN_BK; e += 256) {
-        int row = e >> 5, kd = e & 31;
-        int gq = q0 + row, gd = d0 + kd;
-        smem[SQ + kd * 65 + row] =
-            (gq < nq && gd < d) ? Q[(int64_t)gq * d + gd] : 0.0f;
-      }
-      for (int e = tid; e < KN_IB * KN_BK; e += 256) {
-        int row = e >> 5, kd = e & 31;
-        int gi = ib + row, gd = d0 + kd;
-        smem[SI + kd * 129 + row] =
-            (gi < it_end && gd < d) ? I[(int64_t)gi * d + gd] : 0.0f;
+      // stage Q[q0:q0+64, d0:+32] -> lds_q[kd][row], I tile likewise.
+      // Guards hoisted to tile level (per-element selects de-pipeline
+      // hipcc, guide §5.4 trap (c)).
+      if (tile_full && d0 + KN_BK <= d) {
+#pragma unroll 4
+        for (int e = tid; e < KN_QB * KN_BK; e += 256) {
+          int row = e >> 5, kd = e & 31;
+          smem[SQ + kd * 65 + row] = Q[(int64_t)(q0 + row) * d + d0 + kd];
+        }
+#pragma unroll 4
+        for (int e = tid; e < KN_IB * KN_BK; e += 256) {
+          int row = e >> 5, kd = e & 31;
+          smem[SI + kd * 129 + row] = I[(int64_t)(ib + row) * d + d0 + kd];
+        }
+      } else {
+        for (int e = tid; e < KN_QB * KN_BK; e += 256) {
+          int row = e >> 5, kd = e & 31;
+          int gq = q0 + row, gd = d0 + kd;
+          smem[SQ + kd * 65 + row] =
+              (gq < nq && gd < d) ? Q[(int64_t)gq * d + gd] : 0.0f;
+        }
+        for (int e = tid; e < KN_IB * KN_BK; e += 256) {
+          int row = e >> 5, kd = e & 31;
+          int gi = ib + row, gd = d0 + kd;
+          smem[SI + kd * 129 + row] =
+              (gi < it_end && gd < d) ? I[(int64_t)gi * d + gd] : 0.0f;
+        }
       }
       __syncthreads();
 #pragma unroll 8
