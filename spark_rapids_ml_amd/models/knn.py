@@ -290,9 +290,30 @@ class ApproximateNearestNeighborsModel(_ANNModelParams, _NNModelBase):
             search = self._make_ivfflat_search(algo_params)
         elif algo in ("ivfpq", "ivf_pq"):
             search = self._make_ivfpq_search(algo_params)
+        elif algo == "cagra":
+            search = self._make_cagra_search(algo_params)
         else:
             raise ValueError(f"unsupported ANN algorithm {algo!r}")
         return self._kneighbors_impl(query_df, search)
+
+    # -- CAGRA-equivalent graph ANN ---------------------------------------
+    def _make_cagra_search(self, algo_params: Dict[str, Any]):
+        """Graph ANN (reference cuVS cagra, knn.py:1521-1524): nn-descent
+        graph build + fixed-hop beam search with itopk_size candidates."""
+
+        def search(Qt: torch.Tensor, X: np.ndarray, k: int):
+            from ..parallel.context import get_comm
+
+            device = get_comm().device
+            Xt = to_device_tensor(np.ascontiguousarray(X, dtype=np.float32), device)
+            degree = int(algo_params.get("graph_degree", 32))
+            build_iters = int(algo_params.get("nn_descent_niter", 3))
+            itopk = int(algo_params.get("itopk_size", max(64, 2 * k)))
+            hops = int(algo_params.get("max_iterations", 8))
+            G = _nn_descent(Xt, degree, build_iters)
+            return _graph_beam_search(Qt, Xt, G, k, itopk, hops)
+
+        return search
 
     # -- IVF-Flat ----------------------------------------------------------
     def _build_coarse(self, Xt: torch.Tensor, nlist: int, iters: int = 10):
@@ -415,3 +436,87 @@ def _ivf_search(
         best_d[qsel] = mvals
         best_i[qsel] = cat_i[rows, morder]
     return torch.sqrt(torch.clamp(best_d, min=0.0)), best_i
+
+
+def _dedup_topk(ids: torch.Tensor, dists: torch.Tensor, k: int) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Row-wise top-k by distance with duplicate-id suppression (duplicates
+    otherwise crowd out real candidates in graph build/search)."""
+    sid, order = torch.sort(ids, dim=1)
+    sd = dists.gather(1, order)
+    dup = torch.zeros_like(sid, dtype=torch.bool)
+    dup[:, 1:] = sid[:, 1:] == sid[:, :-1]
+    sd = torch.where(dup, torch.full_like(sd, 3.4e38), sd)
+    vals, o2 = torch.topk(sd, min(k, sd.shape[1]), dim=1, largest=False)
+    return sid.gather(1, o2), vals
+
+
+def _batched_dists(X: torch.Tensor, rows: torch.Tensor, cand: torch.Tensor) -> torch.Tensor:
+    """||X[rows][:,None] - X[cand]||^2 for cand [m, c] (chunked over m)."""
+    out = torch.empty(cand.shape, dtype=torch.float32, device=X.device)
+    chunk = max(1, (1 << 26) // max(1, cand.shape[1] * X.shape[1]))
+    for s in range(0, cand.shape[0], chunk):
+        e = min(cand.shape[0], s + chunk)
+        a = X[rows[s:e]][:, None, :]
+        b = X[cand[s:e]]
+        out[s:e] = ((a - b) ** 2).sum(dim=2)
+    return out
+
+
+def _nn_descent(X: torch.Tensor, degree: int, n_iter: int) -> torch.Tensor:
+    """Approximate kNN graph by nn-descent (reference cagra's graph build):
+    start from random neighbors, iteratively propose neighbors-of-neighbors,
+    keep the best `degree` per node. Returns [n, degree] int64."""
+    n = X.shape[0]
+    device = X.device
+    g = torch.Generator(device="cpu")
+    g.manual_seed(7)
+    G = torch.randint(0, n, (n, degree), generator=g).to(device)
+    rows = torch.arange(n, device=device)
+    Gd = _batched_dists(X, rows, G)
+    for _ in range(max(1, n_iter)):
+        # candidates: neighbors-of-neighbors (subsampled to bound degree^2)
+        sub = G[:, : min(degree, 8)]
+        cand = G[sub].reshape(n, -1)  # [n, sub*degree]
+        cd = _batched_dists(X, rows, cand)
+        cat_i = torch.cat([G, cand], dim=1)
+        cat_d = torch.cat([Gd, cd], dim=1)
+        # avoid self loops, then duplicate-free top-degree
+        cat_d = torch.where(cat_i == rows[:, None], torch.full_like(cat_d, 3.4e38), cat_d)
+        G, Gd = _dedup_topk(cat_i, cat_d, degree)
+    return G
+
+
+def _graph_beam_search(
+    Qt: torch.Tensor, Xt: torch.Tensor, G: torch.Tensor, k: int, itopk: int, hops: int
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Fixed-hop beam search over the kNN graph, vectorized over queries."""
+    nq = Qt.shape[0]
+    n = Xt.shape[0]
+    device = Qt.device
+    g = torch.Generator(device="cpu")
+    g.manual_seed(11)
+    beam = torch.randint(0, n, (nq, itopk), generator=g).to(device)
+    rowsq = torch.arange(nq, device=device)
+
+    def qdists(cand: torch.Tensor) -> torch.Tensor:
+        out = torch.empty(cand.shape, dtype=torch.float32, device=device)
+        chunk = max(1, (1 << 26) // max(1, cand.shape[1] * Qt.shape[1]))
+        for s in range(0, nq, chunk):
+            e = min(nq, s + chunk)
+            a = Qt[s:e][:, None, :]
+            b = Xt[cand[s:e]]
+            out[s:e] = ((a - b) ** 2).sum(dim=2)
+        return out
+
+    bd = qdists(beam)
+    expand = max(4, itopk // 8)  # nodes expanded per hop
+    for _ in range(hops):
+        top = beam.gather(1, bd.topk(min(expand, itopk), dim=1, largest=False).indices)
+        cand = G[top].reshape(nq, -1)
+        cd = qdists(cand)
+        cat_i = torch.cat([beam, cand], dim=1)
+        cat_d = torch.cat([bd, cd], dim=1)
+        beam, bd = _dedup_topk(cat_i, cat_d, itopk)
+
+    ids2, vals2 = _dedup_topk(beam, bd, min(k, bd.shape[1]))
+    return torch.sqrt(torch.clamp(vals2, min=0.0)), ids2

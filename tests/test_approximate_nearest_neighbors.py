@@ -59,3 +59,20 @@ def test_ann_unsupported_algo_raises():
     )
     with pytest.raises(ValueError):
         model.kneighbors(DataFrame.from_numpy(X[:5]))
+
+
+def test_cagra_recall():
+    X = _data(n=3000, d=16)
+    Q = X[:100]
+    model = ApproximateNearestNeighbors(
+        k=10,
+        algorithm="cagra",
+        algoParams={"graph_degree": 32, "itopk_size": 128, "max_iterations": 10},
+    ).fit(DataFrame.from_numpy(X))
+    _, _, knn_df = model.kneighbors(DataFrame.from_numpy(Q))
+    idx = np.asarray(knn_df["indices"])
+    from sklearn.neighbors import NearestNeighbors as SkNN
+
+    sk = SkNN(n_neighbors=10).fit(X)
+    _, sk_idx = sk.kneighbors(Q)
+    assert _recall(idx, sk_idx) > 0.6
