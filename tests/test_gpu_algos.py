@@ -116,14 +116,14 @@ def test_ann_gpu_recall():
     rng = np.random.default_rng(0)
     X = rng.normal(size=(20000, 64)).astype(np.float32)
     model = ApproximateNearestNeighbors(
-        k=10, algorithm="ivfflat", algoParams={"nlist": 64, "nprobe": 16}
+        k=10, algorithm="ivfflat", algoParams={"nlist": 64, "nprobe": 32}
     ).fit(DataFrame.from_numpy(X))
     _, _, knn_df = model.kneighbors(DataFrame.from_numpy(X[:200]))
     idx = np.asarray(knn_df["indices"])
     sk = SkNN(n_neighbors=10).fit(X)
     _, sk_idx = sk.kneighbors(X[:200])
     hits = sum(len(set(a.tolist()) & set(b.tolist())) for a, b in zip(idx, sk_idx))
-    assert hits / sk_idx.size > 0.75
+    assert hits / sk_idx.size > 0.8  # nprobe=nlist/2 on unclustered gaussians
 
 
 def test_sparse_logreg_gpu():
