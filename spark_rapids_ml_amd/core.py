@@ -24,20 +24,17 @@ import json
 import os
 import pickle
 from abc import abstractmethod
-from typing import Any, Callable, Dict, Iterator, List, Optional, Sequence, Tuple, Union
+from typing import Any, Dict, Iterator, List, Optional, Sequence, Tuple, Union
 
 import numpy as np
 import torch
 
 from .data import DataFrame, extract_features, _is_sparse
 from .params import (
-    HasFeaturesCol,
-    HasFeaturesCols,
     HasLabelCol,
     HasPredictionCol,
     HasVerbose,
     Param,
-    Params,
     _NativeParams,
 )
 from .parallel.context import Comm, PartitionDescriptor, get_comm

@@ -20,7 +20,7 @@ classification.py:1018-1028, 1127-1147).
 
 from __future__ import annotations
 
-from typing import Any, Dict, List, Optional, Tuple
+from typing import Any, Dict, Optional, Tuple
 
 import numpy as np
 import torch

@@ -13,7 +13,7 @@ feature.py:438-449; here the projection is computed Spark's way directly).
 
 from __future__ import annotations
 
-from typing import Any, Dict, List, Optional
+from typing import Any, Dict, Optional
 
 import numpy as np
 import torch

@@ -19,7 +19,7 @@ Spark groupBy aggregation, knn.py:1282-1322).
 from __future__ import annotations
 
 import math
-from typing import Any, Dict, List, Optional, Tuple
+from typing import Any, Dict, Optional, Tuple
 
 import numpy as np
 import torch

@@ -26,7 +26,7 @@ w̃_j/σ_xj (raw-coefficient penalization, Spark's behavior).
 
 from __future__ import annotations
 
-from typing import Any, Dict, List, Optional, Tuple
+from typing import Any, Dict, Optional, Tuple
 
 import numpy as np
 import torch
@@ -43,11 +43,7 @@ from ..params import (
 )
 from ..ops import gram
 from ..utils import as_numpy
-from .tree import (
-    _RandomForestEstimator,
-    _RandomForestModel,
-    _RandomForestParams,
-)
+from .tree import _RandomForestEstimator, _RandomForestModel
 
 
 class _LinearRegressionParams(HasFeaturesCol, HasFeaturesCols, HasLabelCol, HasPredictionCol):

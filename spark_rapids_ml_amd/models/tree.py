@@ -32,8 +32,6 @@ from ..params import (
     HasFeaturesCols,
     HasLabelCol,
     HasPredictionCol,
-    HasProbabilityCol,
-    HasRawPredictionCol,
     Param,
     TypeConverters,
 )

@@ -17,7 +17,7 @@ import numpy as np
 import torch
 
 from ..core import Estimator, Model, _FitContext
-from ..data import DataFrame, extract_features, to_device_tensor
+from ..data import to_device_tensor
 from ..params import (
     HasFeaturesCol,
     HasFeaturesCols,

@@ -9,7 +9,7 @@ eigenvector sign convention is reproduced by `sign_flip`.
 
 from __future__ import annotations
 
-from typing import Optional, Tuple
+from typing import Tuple
 
 import torch
 

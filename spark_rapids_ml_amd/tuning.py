@@ -12,9 +12,9 @@ is preserved, which is where the reference's speedup comes from.
 from __future__ import annotations
 
 import numpy as np
-from typing import Any, Dict, List, Optional, Sequence, Tuple
+from typing import Any, Dict, List, Optional, Tuple
 
-from .core import Estimator, Model, _Reader, _Writer
+from .core import Estimator, Model
 from .data import DataFrame
 from .evaluation import Evaluator
 from .params import Param, Params, TypeConverters
