@@ -369,15 +369,16 @@ def _grow_tree(
                     H = hist.view(B, F, n_bins, n_classes)
                     gain, sbin, lval, rval, lcnt = _best_split_class(H, min_leaf)
                 else:
-                    # one fused (count, sum, sum^2) scatter instead of three
-                    hist3 = torch.zeros(B * F * n_bins, 3, dtype=torch.float32, device=dev)
+                    # (count, sum) only: the sum-of-squares terms cancel in
+                    # the variance-reduction gain (ls²/lc + rs²/rc - ts²/tc),
+                    # so the scatter is 2 channels, not 3
+                    hist2 = torch.zeros(B * F * n_bins, 2, dtype=torch.float32, device=dev)
                     flat = base.flatten()
-                    src = torch.empty(bins.shape[0], 3, dtype=torch.float32, device=dev)
+                    src = torch.empty(bins.shape[0], 2, dtype=torch.float32, device=dev)
                     src[:, 0] = 1.0
                     src[:, 1] = yb
-                    src[:, 2] = yb * yb
-                    hist3.index_add_(0, flat, src.repeat_interleave(F, dim=0) if F > 1 else src)
-                    H = hist3.view(B, F, n_bins, 3)
+                    hist2.index_add_(0, flat, src.repeat_interleave(F, dim=0) if F > 1 else src)
+                    H = hist2.view(B, F, n_bins, 2)
                     gain, sbin, lval, rval, lcnt = _best_split_reg(H, min_leaf)
                 # best feature within chunk
                 g, fidx = _max_lastdim(gain)
@@ -478,17 +479,20 @@ def _best_split_class(H: torch.Tensor, min_leaf: int):
 
 
 def _best_split_reg(H: torch.Tensor, min_leaf: int):
-    """H: [B,F,nb,3] = (count,sum,sum2). Variance-reduction gain."""
+    """H: [B,F,nb,2] = (count,sum). Variance-reduction gain — the
+    sum-of-squares terms cancel across parent/children, leaving
+    (ls²/lc + rs²/rc - ts²/tc)/tc, so the histogram needs no y² channel."""
     cs = H.cumsum(dim=2)
     total = cs[:, :, -1:, :]
     left = cs[:, :, :-1, :]
     right = total - left
-    lc, ls, ls2 = left[..., 0], left[..., 1], left[..., 2]
-    rc, rs, rs2 = right[..., 0], right[..., 1], right[..., 2]
-    tc, ts, ts2 = total[..., 0], total[..., 1], total[..., 2]
-    var = lambda c, s, s2: torch.clamp(s2 / torch.clamp(c, min=1e-12) - (s / torch.clamp(c, min=1e-12)) ** 2, min=0.0)
+    lc, ls = left[..., 0], left[..., 1]
+    rc, rs = right[..., 0], right[..., 1]
+    tc, ts = total[..., 0], total[..., 1]
+    lcc = torch.clamp(lc, min=1e-12)
+    rcc = torch.clamp(rc, min=1e-12)
     tcc = torch.clamp(tc, min=1e-12)
-    gain = var(tc, ts, ts2) - (lc / tcc) * var(lc, ls, ls2) - (rc / tcc) * var(rc, rs, rs2)
+    gain = (ls * ls / lcc + rs * rs / rcc - ts * ts / tcc) / tcc
     valid = (lc >= min_leaf) & (rc >= min_leaf)
     gain = torch.where(valid, gain, torch.full_like(gain, -1.0))
     best_gain, best_bin = _max_lastdim(gain)
