@@ -1090,6 +1090,71 @@ __global__ __launch_bounds__(256) void knn_select_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// rf_histogram: per-(node, feature, bin[, class]) histograms for random
+// forest split finding (reference: cuML RF histogram build, SURVEY.md §2.3b
+// tree.py:384-389). Rows pre-sorted by node (perm + seg_off, as in
+// segment_sum); each (node, row-split) block accumulates its feature-chunk
+// histogram in LDS (fast LDS atomics) and adds it to global once.
+// Classification: C channels of class counts. Regression (n_classes==0):
+// 2 channels (count, sum) — sum-of-squares cancels in the gain.
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(256) void rf_histogram_kernel(
+    const uint8_t* __restrict__ Xb,   // [n, d] binned
+    const int64_t* __restrict__ perm, // [m] rows sorted by node
+    const int64_t* __restrict__ seg_off,  // [B+1]
+    const int32_t* __restrict__ feat_sel, // [B, mf] or nullptr (identity)
+    const int32_t* __restrict__ y_cls,    // [n] class ids (classification)
+    const float* __restrict__ y_reg,      // [n] targets (regression)
+    int d, int mf, int f0, int FC, int n_bins, int C, int split,
+    float* __restrict__ out) {            // [B, FC, n_bins, C]
+  extern __shared__ __attribute__((aligned(16))) float lhist[];  // [FC][nb][C]
+  const int b = blockIdx.x / split;
+  const int slice = blockIdx.x % split;
+  const int64_t s0 = seg_off[b], e0 = seg_off[b + 1];
+  const int64_t len = e0 - s0;
+  if (len == 0) return;
+  const int64_t chunk = (len + split - 1) / split;
+  const int64_t rs = s0 + slice * chunk;
+  const int64_t re = min(e0, rs + chunk);
+  if (rs >= re) return;
+
+  const int tid = threadIdx.x;
+  const int nfc = FC * n_bins * C;
+  for (int e = tid; e < nfc; e += 256) lhist[e] = 0.0f;
+  // feature ids for this chunk into registers-ish LDS tail? read direct.
+  __syncthreads();
+
+  const bool classif = (y_cls != nullptr);
+  for (int64_t r = rs + tid; r < re; r += 256) {
+    const int64_t row = perm[r];
+    const uint8_t* xrow = Xb + row * (int64_t)d;
+    const int yc = classif ? y_cls[row] : 0;
+    const float yv = classif ? 1.0f : y_reg[row];
+    for (int q = 0; q < FC; ++q) {
+      const int f = feat_sel ? feat_sel[(int64_t)b * mf + f0 + q] : (f0 + q);
+      const int bin = xrow[f];
+      if (classif) {
+        atomicAdd(&lhist[(q * n_bins + bin) * C + yc], 1.0f);
+      } else {
+        float* cell = &lhist[(q * n_bins + bin) * 2];
+        atomicAdd(cell, 1.0f);
+        atomicAdd(cell + 1, yv);
+      }
+    }
+  }
+  __syncthreads();
+
+  float* dst = out + (int64_t)b * FC * n_bins * C;
+  if (split > 1) {
+    for (int e = tid; e < nfc; e += 256)
+      if (lhist[e] != 0.0f) atomicAdd(&dst[e], lhist[e]);
+  } else {
+    for (int e = tid; e < nfc; e += 256) dst[e] = lhist[e];
+  }
+}
+
+// ---------------------------------------------------------------------------
 // softmax_residual_loss: per-row softmax (C>1) or sigmoid (C==1) residual
 // and summed log-loss. resid = softmax(scores) - onehot(y) (or p - y).
 // One wave per row chunk; memory-bound, fused to one pass.
@@ -1278,11 +1343,37 @@ std::vector<torch::Tensor> softmax_residual_loss(torch::Tensor scores, torch::Te
   return {resid, loss.squeeze(0)};
 }
 
+torch::Tensor rf_histogram(torch::Tensor Xb, torch::Tensor perm, torch::Tensor seg_off,
+                           torch::Tensor feat_sel, torch::Tensor y, int64_t f0,
+                           int64_t FC, int64_t n_bins, int64_t n_classes) {
+  TORCH_CHECK(Xb.is_cuda() && Xb.dtype() == torch::kUInt8 && Xb.is_contiguous());
+  TORCH_CHECK(perm.dtype() == torch::kInt64 && seg_off.dtype() == torch::kInt64);
+  const int d = (int)Xb.size(1);
+  const int B = (int)seg_off.size(0) - 1;
+  const bool classif = n_classes > 0;
+  const int C = classif ? (int)n_classes : 2;
+  const int mf = feat_sel.numel() > 0 ? (int)feat_sel.size(1) : 0;
+  auto out = torch::zeros({(int64_t)B, FC, n_bins, (int64_t)C}, Xb.options().dtype(torch::kFloat32));
+  const size_t lds = (size_t)FC * n_bins * C * 4;
+  TORCH_CHECK(lds <= 160 * 1024, "feature chunk too large for LDS");
+  int split = std::max(1, (int)(1024 / std::max(1, B)));
+  hipLaunchKernelGGL(rf_histogram_kernel, dim3((unsigned)(B * split)), dim3(256), lds,
+                     cur_stream(), Xb.data_ptr<uint8_t>(), perm.data_ptr<int64_t>(),
+                     seg_off.data_ptr<int64_t>(),
+                     mf > 0 ? feat_sel.data_ptr<int32_t>() : nullptr,
+                     classif ? y.data_ptr<int32_t>() : nullptr,
+                     classif ? nullptr : y.data_ptr<float>(),
+                     d, mf, (int)f0, (int)FC, (int)n_bins, C, split,
+                     out.data_ptr<float>());
+  return out;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("kmeans_assign", &kmeans_assign, "fused MFMA distance + argmin");
   m.def("label_accumulate", &label_accumulate, "per-center sum/count scatter");
   m.def("gram_f32", &gram_f32, "A^T A via MFMA f32");
   m.def("softmax_residual_loss", &softmax_residual_loss, "fused softmax residual + loss");
   m.def("knn_select", &knn_select, "fused MFMA distance + in-LDS top-k");
+  m.def("rf_histogram", &rf_histogram, "LDS-privatized RF split histograms");
   m.attr("_is_hip") = true;
 }

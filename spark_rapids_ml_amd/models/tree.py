@@ -303,6 +303,18 @@ def _grow_tree(
     root = tree.add_leaves(root_val[None, :])
     edges_np = as_numpy(edges)
     node_of_row = torch.zeros(n, dtype=torch.int64, device=dev)
+
+    # HIP histogram path (LDS-privatized, rows segment-sorted by node)
+    from ..ops.dispatch import has_hip_ops, use_hip
+
+    hip_hist = Xb.is_cuda and use_hip(Xb) and has_hip_ops()
+    if hip_hist:
+        from ..ops.dispatch import hip_ops
+
+        ext = hip_ops()
+        y32 = y_idx.to(torch.int32) if task == "classification" else yf
+        C_ch = n_classes if task == "classification" else 2
+        fc_kernel = max(1, min(512, (150 * 1024 // 4) // max(1, n_bins * C_ch)))
     frontier = [root]  # node ids whose rows are still splittable
 
     for depth in range(max_depth):
@@ -350,28 +362,56 @@ def _grow_tree(
             else:
                 feat_sel = None
                 mf = d
-                Xb_rows = Xb[rows]
-            for f0 in range(0, mf, feat_chunk):
-                f1 = min(mf, f0 + feat_chunk)
+                Xb_rows = None
+            if hip_hist:
+                sl, perm_rel = loc.sort()
+                perm = rows[perm_rel].contiguous()
+                seg_off = torch.searchsorted(
+                    sl, torch.arange(B + 1, dtype=torch.int64, device=dev)
+                ).contiguous()
+                fsel32 = (
+                    feat_sel.to(torch.int32).contiguous()
+                    if feat_sel is not None
+                    else torch.empty((0, 0), dtype=torch.int32, device=dev)
+                )
+                fc_step = fc_kernel
+            else:
+                if feat_sel is None:
+                    Xb_rows = Xb[rows]
+                fc_step = feat_chunk
+            for f0 in range(0, mf, fc_step):
+                f1 = min(mf, f0 + fc_step)
                 F = f1 - f0
-                if feat_sel is not None:
-                    sel = feat_sel[loc][:, f0:f1]  # [m, F] per-row node's features
-                    # direct 2D gather: touches only m*F bytes of Xb instead
-                    # of materializing the [m, d] row block
-                    bins = Xb[rows[:, None], sel].to(torch.int64)
-                else:
-                    bins = Xb_rows[:, f0:f1].to(torch.int64)
-                base = (loc[:, None] * F + torch.arange(F, device=dev)[None, :]) * n_bins + bins
-                if task == "classification":
+                if hip_hist:
+                    H = ext.rf_histogram(
+                        Xb, perm, seg_off, fsel32, y32, f0, F, n_bins,
+                        n_classes if task == "classification" else 0,
+                    )
+                    if task == "classification":
+                        gain, sbin, lval, rval, lcnt = _best_split_class(H, min_leaf)
+                    else:
+                        gain, sbin, lval, rval, lcnt = _best_split_reg(H, min_leaf)
+                elif task == "classification":
+                    if feat_sel is not None:
+                        sel = feat_sel[loc][:, f0:f1]
+                        bins = Xb[rows[:, None], sel].to(torch.int64)
+                    else:
+                        bins = Xb_rows[:, f0:f1].to(torch.int64)
+                    base = (loc[:, None] * F + torch.arange(F, device=dev)[None, :]) * n_bins + bins
                     hist = torch.zeros(B * F * n_bins * n_classes, dtype=torch.float32, device=dev)
                     idx = base * n_classes + yb[:, None]
                     hist.index_add_(0, idx.flatten(), torch.ones(idx.numel(), device=dev))
                     H = hist.view(B, F, n_bins, n_classes)
                     gain, sbin, lval, rval, lcnt = _best_split_class(H, min_leaf)
                 else:
+                    if feat_sel is not None:
+                        sel = feat_sel[loc][:, f0:f1]
+                        bins = Xb[rows[:, None], sel].to(torch.int64)
+                    else:
+                        bins = Xb_rows[:, f0:f1].to(torch.int64)
+                    base = (loc[:, None] * F + torch.arange(F, device=dev)[None, :]) * n_bins + bins
                     # (count, sum) only: the sum-of-squares terms cancel in
-                    # the variance-reduction gain (ls²/lc + rs²/rc - ts²/tc),
-                    # so the scatter is 2 channels, not 3
+                    # the variance-reduction gain (ls²/lc + rs²/rc - ts²/tc)
                     hist2 = torch.zeros(B * F * n_bins, 2, dtype=torch.float32, device=dev)
                     flat = base.flatten()
                     src = torch.empty(bins.shape[0], 2, dtype=torch.float32, device=dev)

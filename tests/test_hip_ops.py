@@ -108,3 +108,43 @@ def test_knn_select_matches_ref(ext, nq, ni, d, k):
     mism = idx != ref_i
     if bool(mism.any()):
         assert torch.allclose(dists[mism], ref_d[mism], rtol=1e-3, atol=1e-3)
+
+
+@pytest.mark.parametrize("task,nc", [("classification", 3), ("regression", 0)])
+def test_rf_histogram_matches_ref(ext, task, nc):
+    g = torch.Generator().manual_seed(0)
+    n, d, nb, B, mf = 5000, 40, 16, 7, 6
+    Xb = torch.randint(0, nb, (n, d), generator=g, dtype=torch.uint8).cuda()
+    loc = torch.randint(0, B, (n,), generator=g).cuda()
+    feat_sel = torch.stack([torch.randperm(d, generator=g)[:mf] for _ in range(B)]).cuda()
+    sl, perm_rel = loc.sort()
+    perm = perm_rel.to(torch.int64).contiguous()
+    seg_off = torch.searchsorted(sl, torch.arange(B + 1).cuda()).contiguous()
+    if task == "classification":
+        y = torch.randint(0, nc, (n,), generator=g, dtype=torch.int32).cuda()
+        H = ext.rf_histogram(Xb, perm, seg_off, feat_sel.to(torch.int32), y, 0, mf, nb, nc)
+        ref = torch.zeros(B, mf, nb, nc).cuda()
+        for b in range(B):
+            m = loc == b
+            sel = feat_sel[b]
+            for q in range(mf):
+                bins = Xb[m][:, sel[q]].long()
+                for c in range(nc):
+                    ref[b, q].index_add_(
+                        0, bins[y[m] == c], torch.ones(int((y[m] == c).sum())).cuda()
+                    ) if False else None
+                ref[b, q] = torch.zeros(nb, nc).cuda().index_put_(
+                    (bins, y[m].long()), torch.ones(len(bins)).cuda(), accumulate=True
+                )
+    else:
+        y = torch.randn(n, generator=g).cuda()
+        H = ext.rf_histogram(Xb, perm, seg_off, feat_sel.to(torch.int32), y, 0, mf, nb, 0)
+        ref = torch.zeros(B, mf, nb, 2).cuda()
+        for b in range(B):
+            m = loc == b
+            sel = feat_sel[b]
+            for q in range(mf):
+                bins = Xb[m][:, sel[q]].long()
+                ref[b, q, :, 0].index_add_(0, bins, torch.ones(len(bins)).cuda())
+                ref[b, q, :, 1].index_add_(0, bins, y[m])
+    assert torch.allclose(H, ref, rtol=1e-4, atol=1e-2), (H - ref).abs().max()
