@@ -374,6 +374,28 @@ class KMeansModel(_KMeansParams, Model):
     def trainingCost(self) -> float:
         return self._model_attributes["inertia_"]
 
+    def computeCost(self, df) -> float:
+        """Sum of squared distances to the nearest center over the
+        dataframe (Spark's deprecated-but-present KMeansModel.computeCost;
+        globally reduced over ranks)."""
+        from ..data import extract_features
+        from ..parallel.context import get_comm
+
+        comm = get_comm()
+        features_col, features_cols = self._get_input_columns()
+        X = extract_features(df, features_col, features_cols, self._float32_inputs)
+        Xt = to_device_tensor(np.ascontiguousarray(X, dtype=np.float32), comm.device)
+        C = torch.from_numpy(
+            np.ascontiguousarray(self.cluster_centers_, dtype=np.float32)
+        ).to(comm.device)
+        from ..ops.torch_ref import pairwise_sq_dists
+
+        local = 0.0
+        for s0 in range(0, Xt.shape[0], 1 << 16):
+            d2 = pairwise_sq_dists(Xt[s0 : s0 + (1 << 16)], C)
+            local += float(d2.min(dim=1).values.sum().item())
+        return comm.allreduce_scalar(local)
+
     def predict(self, vector: np.ndarray) -> int:
         d = ((self.cluster_centers_ - np.asarray(vector)[None, :]) ** 2).sum(axis=1)
         return int(d.argmin())

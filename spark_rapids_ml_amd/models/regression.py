@@ -369,7 +369,19 @@ class LinearRegressionModel(_LinearRegressionParams, Model):
 
     @property
     def hasSummary(self) -> bool:
-        return False
+        return self._model_attributes.get("n_rows_", 0) > 0
+
+    @property
+    def summary(self) -> "LinearRegressionTrainingSummary":
+        """Training summary (Spark LinearRegressionTrainingSummary parity:
+        rootMeanSquaredError/meanSquaredError from the training SSE)."""
+        n = max(1, self._model_attributes["n_rows_"])
+        sse = self._model_attributes["sse_"]
+        return LinearRegressionTrainingSummary(
+            meanSquaredError=sse / n,
+            rootMeanSquaredError=float(np.sqrt(sse / n)),
+            numInstances=n,
+        )
 
     def setFeaturesCol(self, value) -> "LinearRegressionModel":
         if isinstance(value, (list, tuple)):
@@ -417,3 +429,13 @@ class RandomForestRegressionModel(_RandomForestModel):
     """Fitted RF regression model (reference regression.py:1055)."""
 
     _task = "regression"
+
+
+class LinearRegressionTrainingSummary:
+    """Spark LinearRegressionTrainingSummary equivalent (training-set
+    metrics from the fit's sufficient statistics)."""
+
+    def __init__(self, meanSquaredError: float, rootMeanSquaredError: float, numInstances: int):
+        self.meanSquaredError = meanSquaredError
+        self.rootMeanSquaredError = rootMeanSquaredError
+        self.numInstances = numInstances

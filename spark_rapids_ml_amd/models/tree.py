@@ -212,7 +212,7 @@ class _Tree:
     """Flat-array binary tree with BULK growth (one numpy append per node
     batch — the per-node Python loop was the depth-13 bottleneck)."""
 
-    __slots__ = ("feature", "threshold", "left", "right", "is_leaf", "value", "n_nodes")
+    __slots__ = ("feature", "threshold", "left", "right", "is_leaf", "gain", "value", "n_nodes")
 
     def __init__(self, value_width: int):
         self.feature = [np.empty(0, np.int32)]
@@ -220,6 +220,7 @@ class _Tree:
         self.left = [np.empty(0, np.int32)]
         self.right = [np.empty(0, np.int32)]
         self.is_leaf = [np.empty(0, bool)]
+        self.gain = [np.empty(0, np.float32)]
         self.value = [np.empty((0, value_width), np.float32)]
         self.n_nodes = 0
 
@@ -233,6 +234,7 @@ class _Tree:
         self.left.append(np.full(m, -1, np.int32))
         self.right.append(np.full(m, -1, np.int32))
         self.is_leaf.append(np.ones(m, bool))
+        self.gain.append(np.zeros(m, np.float32))
         self.value.append(values.astype(np.float32))
         return first
 
@@ -243,6 +245,7 @@ class _Tree:
             self.left = [np.concatenate(self.left)]
             self.right = [np.concatenate(self.right)]
             self.is_leaf = [np.concatenate(self.is_leaf)]
+            self.gain = [np.concatenate(self.gain)]
             self.value = [np.concatenate(self.value, axis=0)]
 
     def set_splits(
@@ -252,6 +255,7 @@ class _Tree:
         threshold: np.ndarray,
         left: np.ndarray,
         right: np.ndarray,
+        gain: Optional[np.ndarray] = None,
     ) -> None:
         self._consolidate()
         self.feature[0][node_ids] = feature.astype(np.int32)
@@ -259,6 +263,8 @@ class _Tree:
         self.left[0][node_ids] = left.astype(np.int32)
         self.right[0][node_ids] = right.astype(np.int32)
         self.is_leaf[0][node_ids] = False
+        if gain is not None:
+            self.gain[0][node_ids] = gain.astype(np.float32)
 
     def to_arrays(self) -> Dict[str, np.ndarray]:
         self._consolidate()
@@ -268,6 +274,7 @@ class _Tree:
             "left": self.left[0],
             "right": self.right[0],
             "is_leaf": self.is_leaf[0],
+            "gain": self.gain[0],
             "value": self.value[0],
         }
 
@@ -458,7 +465,8 @@ def _grow_tree(
             fsel = bf[idxs].astype(np.int64)
             bsel = bb[idxs].astype(np.int64)
             tree.set_splits(
-                batch_np[idxs], fsel, edges_np[fsel, bsel], l_ids, r_ids
+                batch_np[idxs], fsel, edges_np[fsel, bsel], l_ids, r_ids,
+                gain=bg[idxs],
             )
             if depth + 1 < max_depth:
                 if task == "regression":
@@ -700,6 +708,58 @@ class _RandomForestModel(_RandomForestParams, Model):
     @property
     def numClasses(self) -> int:
         return self._model_attributes["n_classes_"]
+
+    @property
+    def treeWeights(self) -> List[float]:
+        return [1.0] * self.numTrees
+
+    @property
+    def totalNumNodes(self) -> int:
+        return int(sum(t["feature"].shape[0] for t in self.trees))
+
+    @property
+    def featureImportances(self) -> np.ndarray:
+        """Impurity-decrease feature importances (Spark semantics: per-tree
+        gain x node-count sums, normalized per tree then averaged)."""
+        total = np.zeros(self.numFeatures)
+        for t in self.trees:
+            imp = np.zeros(self.numFeatures)
+            internal = ~t["is_leaf"]
+            if self._task == "classification":
+                counts = t["value"].sum(axis=1)
+            else:
+                counts = t["value"][:, 1]
+            f = t["feature"][internal]
+            w = t.get("gain", np.zeros(len(t["feature"])))[internal] * counts[internal]
+            np.add.at(imp, f, w)
+            tot = imp.sum()
+            if tot > 0:
+                total += imp / tot
+        tot = total.sum()
+        return total / tot if tot > 0 else total
+
+    def toDebugString(self) -> str:
+        """Human-readable forest dump (Spark toDebugString parity)."""
+        lines = [f"{type(self).__name__} with {self.numTrees} trees"]
+        for ti, t in enumerate(self.trees):
+            lines.append(f"  Tree {ti} ({t['feature'].shape[0]} nodes):")
+
+            def rec(node: int, depth: int) -> None:
+                pad = "    " * (depth + 1)
+                if t["is_leaf"][node]:
+                    lines.append(f"{pad}Predict: {t['value'][node].tolist()}")
+                else:
+                    lines.append(
+                        f"{pad}If (feature {t['feature'][node]} <= "
+                        f"{t['threshold'][node]:.6g})"
+                    )
+                    rec(int(t["left"][node]), depth + 1)
+                    lines.append(f"{pad}Else")
+                    rec(int(t["right"][node]), depth + 1)
+
+            if t["feature"].shape[0] <= 2047:  # bound output size
+                rec(0, 1)
+        return "\n".join(lines)
 
     def dump_as_json(self) -> str:
         """Forest dump for Spark-tree translation parity (reference

@@ -143,3 +143,20 @@ def test_rf_distributed_trees_split_across_ranks():
     for n_trees, acc in results:
         assert n_trees == 10  # merged forest has all trees
         assert acc > 0.85
+
+
+def test_rf_feature_importances_and_debug_string():
+    X, y = _cls_data(n=400, d=8)
+    # feature 0 carries all signal
+    X[:, 0] = y * 4 + np.random.default_rng(0).normal(0, 0.1, len(y))
+    model = RandomForestClassifier(
+        numTrees=10, maxDepth=4, seed=0, featureSubsetStrategy="all"
+    ).fit(DataFrame.from_numpy(X, y))
+    imp = model.featureImportances
+    assert imp.shape == (8,)
+    assert abs(imp.sum() - 1.0) < 1e-6
+    assert imp.argmax() == 0
+    s = model.toDebugString()
+    assert "Tree 0" in s and "feature" in s
+    assert model.treeWeights == [1.0] * 10
+    assert model.totalNumNodes > 10
