@@ -527,10 +527,9 @@ class DBSCANModel(_DBSCANParams, Model):
         device = comm.device
 
         # replicate dataset (reference broadcasts the df in <=8GB chunks,
-        # clustering.py:1152-1159; allgather over xGMI here)
-        parts = comm.allgather_obj(np.ascontiguousarray(X_local, dtype=np.float32))
-        X_full = np.concatenate(parts, axis=0)
-        Xf = to_device_tensor(X_full, device)
+        # clustering.py:1152-1159): ONE tensor all-gather over xGMI
+        X_dev = to_device_tensor(np.ascontiguousarray(X_local, dtype=np.float32), device)
+        Xf = torch.cat(comm.allgather_rows(X_dev), dim=0)
         n = Xf.shape[0]
         off = pdesc.row_offset()
         n_local = X_local.shape[0]
@@ -551,12 +550,10 @@ class DBSCANModel(_DBSCANParams, Model):
             )
             core_local[s:e] = (d2 <= eps2).sum(dim=1) >= min_samples
 
-        # global core mask
-        core_full = torch.zeros(n, dtype=torch.bool, device=device)
-        core_full[off : off + n_local] = core_local
-        comm.allreduce(comm.to_coll(core_full.to(torch.int32)), "max")
-        core_parts = comm.allgather_obj(as_numpy(core_local))
-        core_full = torch.from_numpy(np.concatenate(core_parts)).to(device)
+        # global core mask (tensor all-gather of per-rank slices)
+        core_full = torch.cat(
+            comm.allgather_rows(core_local.to(torch.uint8)), dim=0
+        ).to(torch.bool)
 
         # min-label propagation over the core-core graph with pointer jumping
         labels = torch.arange(n, dtype=torch.int64, device=device)

@@ -96,10 +96,29 @@ class _NNModelBase(_NNParams, Model):
         q_offset: int,
         k: int,
     ) -> Tuple[np.ndarray, np.ndarray]:
-        """All-gather per-rank partial top-k and merge for the local queries."""
-        parts = comm.allgather_obj((my_dists, my_ids))
-        d_cat = np.concatenate([p[0] for p in parts], axis=1)  # [nq_total, P*k']
-        i_cat = np.concatenate([p[1] for p in parts], axis=1)
+        """All-gather per-rank partial top-k (one fused tensor collective —
+        the reference moves these with UCX p2p, knn.py:763-774) and merge
+        for the local queries. Item ids ride as int32 bit-cast into the
+        float buffer (one collective instead of a pickled-object gather)."""
+        import torch as _torch
+
+        kp = my_dists.shape[1]
+        buf = np.concatenate(
+            [
+                np.ascontiguousarray(my_dists, dtype=np.float32),
+                my_ids.astype(np.int32).view(np.float32),
+            ],
+            axis=1,
+        )
+        parts = comm.allgather_rows(_torch.from_numpy(buf))
+        d_cat = np.concatenate([as_numpy(p)[:, :kp] for p in parts], axis=1)
+        i_cat = np.concatenate(
+            [
+                np.ascontiguousarray(as_numpy(p)[:, kp:]).view(np.int32).astype(np.int64)
+                for p in parts
+            ],
+            axis=1,
+        )
         dl = d_cat[q_offset : q_offset + nq_local]
         il = i_cat[q_offset : q_offset + nq_local]
         order = np.argsort(dl, axis=1, kind="stable")[:, :k]

@@ -90,6 +90,24 @@ class Comm:
         dist.all_gather_object(out, obj)
         return out
 
+    def allgather_rows(self, t: torch.Tensor) -> List[torch.Tensor]:
+        """Variable-row-count tensor all-gather: pads local rows to the max,
+        runs ONE tensor all-gather (RCCL over xGMI on GPU — large payloads
+        like DBSCAN's replicated dataset or kNN's query/partial buffers must
+        not ride the pickled-object path), and trims per-rank results."""
+        if not self.initialized:
+            return [t]
+        n_local = t.shape[0]
+        counts = self.allgather_obj(int(n_local))
+        n_max = max(counts)
+        shape = (n_max,) + tuple(t.shape[1:])
+        padded = torch.zeros(shape, dtype=t.dtype, device=self._coll_device())
+        if n_local > 0:
+            padded[:n_local] = self.to_coll(t)
+        out = [torch.empty_like(padded) for _ in range(self.world_size)]
+        dist.all_gather(out, padded.contiguous())
+        return [o[: counts[r]].to(t.device) for r, o in enumerate(out)]
+
     def broadcast(self, t: torch.Tensor, src: int = 0) -> torch.Tensor:
         if self.initialized:
             dist.broadcast(t, src=src)
