@@ -136,12 +136,15 @@ class _NNModelBase(_NNParams, Model):
         Q, qids, query_df = self._query_xy(query_df)
 
         # replicate queries (reference broadcasts queries, knn.py:1250-1280)
-        q_parts = comm.allgather_obj((np.ascontiguousarray(Q, dtype=np.float32), qids))
-        Q_full = np.concatenate([p[0] for p in q_parts], axis=0)
-        q_offset = sum(p[0].shape[0] for p in q_parts[: comm.rank])
+        # — tensor all-gather over xGMI, not pickled objects
+        import torch as _torch
 
         device = comm.device
-        Qt = to_device_tensor(Q_full, device)
+        q_parts = comm.allgather_rows(
+            to_device_tensor(np.ascontiguousarray(Q, dtype=np.float32), device)
+        )
+        Qt = _torch.cat(q_parts, dim=0)
+        q_offset = sum(p.shape[0] for p in q_parts[: comm.rank])
         if X.shape[0] > 0:
             dists, idx = search_fn(Qt, X, k)
             ids_local = item_ids[as_numpy(idx)]

@@ -246,13 +246,17 @@ class UMAP(_UMAPParams, Estimator):
             X_local = X_local[sel]
             if y_local is not None:
                 y_local = y_local[sel]
-        parts = comm.allgather_obj((X_local, y_local))  # gather to all; rank0 fits
-        X_fit = np.concatenate([p for p, _ in parts if len(p)], axis=0)
-        y_fit = (
-            np.concatenate([q for p, q in parts if q is not None and len(p)])
-            if y_local is not None
-            else None
+        # gather to all ranks via tensor collectives (rank 0 fits)
+        X_parts = comm.allgather_rows(
+            torch.from_numpy(X_local).to(comm._coll_device())
         )
+        X_fit = np.concatenate([p.cpu().numpy() for p in X_parts], axis=0)
+        y_fit = None
+        if y_local is not None:
+            y_parts = comm.allgather_rows(
+                torch.from_numpy(y_local).to(comm._coll_device())
+            )
+            y_fit = np.concatenate([p.cpu().numpy() for p in y_parts])
 
         if comm.rank == 0:
             emb = self._fit_single(X_fit, ctx.device, y_fit)
