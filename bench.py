@@ -74,8 +74,7 @@ def main() -> None:
         buf[:, :cols] = sums
         buf[:, cols] = counts
         buf[0, cols + 1] = inertia
-        comm.allreduce(comm.to_coll(buf))
-        buf = buf.to(dev)
+        buf = comm.allreduce_t(buf)
         g_counts = buf[:, cols]
         nonempty = g_counts > 0
         C_new = C.clone()

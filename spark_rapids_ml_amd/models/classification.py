@@ -274,8 +274,7 @@ class LogisticRegression(_LogisticRegressionParams, Estimator):
             buf = torch.zeros(C_out * ncol + 1, dtype=torch.float64, device=grad.device)
             buf[:-1] = grad.to(torch.float64).flatten()
             buf[-1] = loss.to(torch.float64)
-            comm.allreduce(comm.to_coll(buf))
-            buf = buf.cpu()
+            buf = comm.allreduce_t(buf).cpu()
             g = buf[:-1] / n
             total_loss = float(buf[-1]) / n
             if l2 > 0:
@@ -339,8 +338,7 @@ class LogisticRegression(_LogisticRegressionParams, Estimator):
                     xb = Xt[s0 : s0 + (1 << 18)]
                     buf[0] += xb.sum(dim=0, dtype=torch.float64)
                     buf[1] += (xb * xb).sum(dim=0, dtype=torch.float64)
-        comm.allreduce(comm.to_coll(buf))
-        buf = buf.to(Xt.device)
+        buf = comm.allreduce_t(buf)
         mean = buf[0] / n
         var = torch.clamp((buf[1] - n * mean * mean) / max(1, n - 1), min=0.0)
         return torch.sqrt(var)

@@ -176,8 +176,7 @@ class KMeans(_KMeansParams, Estimator):
             buf[:, : pdesc.n] = sums
             buf[:, pdesc.n] = counts
             buf[0, pdesc.n + 1] = local_inertia
-            comm.allreduce(comm.to_coll(buf))
-            buf = buf.to(Xt.device)
+            buf = comm.allreduce_t(buf)
             g_sums = buf[:, : pdesc.n]
             g_counts = buf[:, pdesc.n]
             inertia = float(buf[0, pdesc.n + 1].item())
@@ -215,8 +214,7 @@ class KMeans(_KMeansParams, Estimator):
             C[torch.from_numpy(pos).to(Xt.device)] = Xt[
                 torch.from_numpy(local).to(Xt.device)
             ]
-        comm.allreduce(comm.to_coll(C))
-        return C.to(Xt.device)
+        return comm.allreduce_t(C)
 
     def _init_scalable_kmeanspp(
         self, Xt, x_sq, k, seed, comm, pdesc, rounds: int, oversample: float
@@ -230,8 +228,7 @@ class KMeans(_KMeansParams, Estimator):
         cand = torch.zeros((1, pdesc.n), dtype=Xt.dtype, device=Xt.device)
         if len(pos):
             cand[0] = Xt[int(local[0])]
-        comm.allreduce(comm.to_coll(cand))
-        cand = cand.to(Xt.device)
+        cand = comm.allreduce_t(cand)
 
         l = max(1.0, oversample * k)
         for _ in range(max(1, rounds)):
@@ -257,7 +254,7 @@ class KMeans(_KMeansParams, Estimator):
 
         # weight candidates by how many points they own
         labels, _, counts, _ = kmeans_assign_reduce(Xt, cand, x_sq)
-        counts = comm.allreduce(comm.to_coll(counts.clone())).to(Xt.device)
+        counts = comm.allreduce_t(counts.clone())
         w = as_numpy(counts).astype(np.float64)
         cand_np = as_numpy(cand).astype(np.float64)
         centers = _weighted_kmeanspp(cand_np, w, k, np.random.default_rng(seed + 7))
@@ -582,8 +579,7 @@ class DBSCANModel(_DBSCANParams, Model):
                 )
             new_full = torch.full((n,), torch.iinfo(torch.int64).max, dtype=torch.int64, device=device)
             new_full[off : off + n_local] = new_local
-            comm.allreduce(comm.to_coll(new_full), "min")
-            new_full = new_full.to(device)
+            new_full = comm.allreduce_t(new_full, "min")
             new_full[~core_full] = torch.iinfo(torch.int64).max
             # pointer jumping: label <- label[label] for core points
             core_ids = torch.nonzero(core_full).flatten()

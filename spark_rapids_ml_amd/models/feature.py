@@ -97,8 +97,7 @@ class PCA(_PCAParams, Estimator):
             g = gram(Xt)  # f32 MFMA on GPU
             buf[:d] = g.to(torch.float64)
             buf[d] = Xt.sum(dim=0).to(torch.float64)
-        comm.allreduce(comm.to_coll(buf))
-        buf = buf.to(Xt.device)
+        buf = comm.allreduce_t(buf)
         mean = buf[d] / pdesc.m
         cov = cov_from_gram(buf[:d], mean, pdesc.m, ddof=1)
 

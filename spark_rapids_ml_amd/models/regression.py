@@ -166,8 +166,7 @@ class LinearRegression(_LinearRegressionParams, Estimator):
             buf[d + 1] = Xt.sum(dim=0).to(torch.float64)
             buf[d + 2, 0] = yt.sum().to(torch.float64)
             buf[d + 2, 1] = (yt * yt).sum().to(torch.float64)
-        comm.allreduce(comm.to_coll(buf))
-        buf = buf.to(ctx.device)
+        buf = comm.allreduce_t(buf)
         m = {
             "G": buf[:d],
             "Xty": buf[d],

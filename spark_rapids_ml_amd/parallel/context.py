@@ -66,6 +66,17 @@ class Comm:
             dist.all_reduce(t, op=ops[op])
         return t
 
+    def allreduce_t(self, t: torch.Tensor, op: str = "sum") -> torch.Tensor:
+        """All-reduce returning the reduced tensor ON t's device. Needed
+        because `allreduce(to_coll(t))` reduces a COPY when the collective
+        device differs from t's (gloo collectives with GPU tensors) and the
+        original would silently stay unreduced."""
+        ct = self.to_coll(t)
+        self.allreduce(ct, op)
+        if ct is t:
+            return t
+        return ct.to(t.device)
+
     def allreduce_scalar(self, value: float, op: str = "sum") -> float:
         if not self.initialized:
             return float(value)

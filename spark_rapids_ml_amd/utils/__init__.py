@@ -56,8 +56,7 @@ def standardize_dataset(
             xb = X[s0 : s0 + (1 << 18)]
             acc[0] += xb.sum(dim=0, dtype=torch.float64)
             acc[1] += (xb * xb).sum(dim=0, dtype=torch.float64)
-    comm.allreduce(comm.to_coll(acc))
-    acc = acc.to(X.device)
+    acc = comm.allreduce_t(acc)
     mean = acc[0] / n_total
     if n_total > 1:
         var = (acc[1] - n_total * mean * mean) / (n_total - 1)
