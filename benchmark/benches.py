@@ -101,10 +101,10 @@ class BenchLogisticRegression(BenchmarkBase):
 
     def make_data(self, args):
         if args.sparse_density:
-            X, y = gen_data.gen_sparse_regression(
-                args.num_rows, args.num_cols, density=args.sparse_density, seed=args.seed
+            nnz_row = max(1, int(args.sparse_density * args.num_cols))
+            X, y = gen_data.gen_sparse_classification_fast(
+                args.num_rows, args.num_cols, nnz_per_row=nnz_row, seed=args.seed
             )
-            y = (y > np.median(y)).astype(np.float64)
             return DataFrame.from_numpy(X, y)
         X, y = gen_data.gen_classification(
             args.num_rows, args.num_cols, n_classes=args.n_classes, seed=args.seed

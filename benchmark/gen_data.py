@@ -117,12 +117,37 @@ def gen_sparse_regression(
     return X.astype(dtype), y.astype(np.float64)
 
 
+def gen_sparse_classification_fast(
+    num_rows: int, num_cols: int, nnz_per_row: int = 20, seed: int = 0, dtype: str = "float32",
+):
+    """Direct CSR construction (scipy.sparse.random is too slow past ~1e8
+    nnz): fixed nnz per row, sorted column draws, synthetic linear labels."""
+    import scipy.sparse as sp
+
+    comm = get_comm()
+    rng = np.random.default_rng(seed)
+    w = rng.normal(size=num_cols)
+    rng_local = np.random.default_rng(seed + 104729 * (comm.rank + 1))
+    n_local = _shard(num_rows, comm.rank, comm.world_size)
+    cols = np.sort(
+        rng_local.integers(0, num_cols, size=(n_local, nnz_per_row), dtype=np.int32),
+        axis=1,
+    )
+    data = rng_local.normal(size=(n_local, nnz_per_row)).astype(dtype)
+    indptr = np.arange(0, (n_local + 1) * nnz_per_row, nnz_per_row, dtype=np.int64)
+    X = sp.csr_matrix((data.ravel(), cols.ravel(), indptr), shape=(n_local, num_cols))
+    margin = (data * w[cols]).sum(axis=1)
+    y = (margin + 0.3 * rng_local.normal(size=n_local) > 0).astype(np.float64)
+    return X, y
+
+
 GENERATORS = {
     "blobs": gen_blobs,
     "low_rank_matrix": gen_low_rank_matrix,
     "regression": gen_regression,
     "classification": gen_classification,
     "sparse_regression": gen_sparse_regression,
+    "sparse_classification_fast": gen_sparse_classification_fast,
 }
 
 
