@@ -52,10 +52,10 @@ def standardize_dataset(
     d = X.shape[1]
     acc = torch.zeros((2, d), dtype=torch.float64, device=X.device)
     if X.shape[0] > 0:
-        for s0 in range(0, X.shape[0], 1 << 18):  # no full f64 copy
+        for s0 in range(0, X.shape[0], 1 << 18):  # f32 chunk sums, f64 acc
             xb = X[s0 : s0 + (1 << 18)]
-            acc[0] += xb.sum(dim=0, dtype=torch.float64)
-            acc[1] += (xb * xb).sum(dim=0, dtype=torch.float64)
+            acc[0] += xb.sum(dim=0).to(torch.float64)
+            acc[1] += (xb * xb).sum(dim=0).to(torch.float64)
     acc = comm.allreduce_t(acc)
     mean = acc[0] / n_total
     if n_total > 1:
