@@ -243,22 +243,19 @@ class KMeans(_KMeansParams, Estimator):
                 generator=_dev_gen(probs.device, seed + 13 * pdesc.rank + 1),
             )
             picked = Xt[draws < probs]
-            gathered = comm.allgather_obj(as_numpy(picked))
-            new_c = np.concatenate([g for g in gathered if len(g)], axis=0) if any(
-                len(g) for g in gathered
-            ) else np.zeros((0, pdesc.n), dtype=np.float32)
-            if len(new_c):
-                cand = torch.cat(
-                    [cand, torch.from_numpy(new_c).to(Xt.device, Xt.dtype)], dim=0
-                )
+            parts = comm.allgather_rows(picked)
+            new_c = [p for p in parts if p.shape[0] > 0]
+            if new_c:
+                cand = torch.cat([cand] + new_c, dim=0)
 
-        # weight candidates by how many points they own
+        # weight candidates by how many points they own, then deterministic
+        # weighted k-means++ on the (small) pool — identical on every rank
         labels, _, counts, _ = kmeans_assign_reduce(Xt, cand, x_sq)
         counts = comm.allreduce_t(counts.clone())
-        w = as_numpy(counts).astype(np.float64)
-        cand_np = as_numpy(cand).astype(np.float64)
-        centers = _weighted_kmeanspp(cand_np, w, k, np.random.default_rng(seed + 7))
-        return torch.from_numpy(centers).to(Xt.device, Xt.dtype)
+        centers = _weighted_kmeanspp_t(
+            cand, counts.to(torch.float32), k, seed + 7
+        )
+        return centers.to(Xt.dtype)
 
     def _create_model(self, attrs: Dict[str, Any]) -> "KMeansModel":
         return KMeansModel(**attrs)
@@ -281,35 +278,49 @@ def _min_sq_dist(Xt: torch.Tensor, C: torch.Tensor, x_sq: torch.Tensor) -> torch
     return out
 
 
-def _weighted_kmeanspp(cand: np.ndarray, w: np.ndarray, k: int, rng) -> np.ndarray:
-    """Deterministic weighted k-means++ over the (small) candidate pool,
-    followed by a few weighted Lloyd refinement steps."""
-    n = cand.shape[0]
+def _weighted_kmeanspp_t(
+    cand: torch.Tensor, w: torch.Tensor, k: int, seed: int
+) -> torch.Tensor:
+    """Deterministic weighted k-means++ over the candidate pool, followed by
+    a few weighted Lloyd refinement steps — all on-device with matmul
+    distances (the pool is ~oversampling*k*rounds candidates; a naive
+    broadcasted distance tensor would be GBs at k=1000, d=3000)."""
+    n, d = cand.shape
+    dev = cand.device
     if n <= k:
-        # pad by repeating candidates (degenerate but non-fatal)
         reps = int(np.ceil(k / max(1, n)))
-        pool = np.tile(cand, (reps, 1))[:k]
-        return pool
-    wsum = w.sum()
-    probs = w / wsum if wsum > 0 else np.full(n, 1.0 / n)
-    centers = [cand[rng.choice(n, p=probs)]]
-    d2 = ((cand - centers[0]) ** 2).sum(axis=1)
+        return cand.repeat(reps, 1)[:k].clone()
+    rng = np.random.default_rng(seed)
+    wn = as_numpy(w).astype(np.float64)
+    wsum = wn.sum()
+    probs = wn / wsum if wsum > 0 else np.full(n, 1.0 / n)
+    c_sq = (cand * cand).sum(dim=1)
+
+    def dist_to(center: torch.Tensor) -> torch.Tensor:
+        return torch.clamp(
+            c_sq + (center * center).sum() - 2.0 * (cand @ center), min=0.0
+        )
+
+    first = int(rng.choice(n, p=probs))
+    chosen = [first]
+    d2 = dist_to(cand[first])
     for _ in range(1, k):
-        scores = w * d2
+        scores = as_numpy(w.to(torch.float64) * d2.to(torch.float64))
         tot = scores.sum()
-        if tot <= 0:
-            centers.append(cand[rng.integers(0, n)])
-            continue
-        centers.append(cand[rng.choice(n, p=scores / tot)])
-        d2 = np.minimum(d2, ((cand - centers[-1]) ** 2).sum(axis=1))
-    C = np.stack(centers)
-    for _ in range(5):  # weighted Lloyd refinement on the pool
-        d = ((cand[:, None, :] - C[None, :, :]) ** 2).sum(axis=2)
-        lb = d.argmin(axis=1)
-        for j in range(k):
-            m = lb == j
-            if w[m].sum() > 0:
-                C[j] = (cand[m] * w[m, None]).sum(axis=0) / w[m].sum()
+        nxt = int(rng.choice(n, p=scores / tot)) if tot > 0 else int(rng.integers(0, n))
+        chosen.append(nxt)
+        d2 = torch.minimum(d2, dist_to(cand[nxt]))
+    C = cand[torch.tensor(chosen, dtype=torch.int64, device=dev)].clone()
+    # weighted Lloyd refinement on the pool (matmul distances)
+    for _ in range(5):
+        dmat = c_sq[:, None] + (C * C).sum(dim=1)[None, :] - 2.0 * (cand @ C.T)
+        lb = dmat.argmin(dim=1)
+        sums = torch.zeros_like(C)
+        wsum_t = torch.zeros(k, dtype=w.dtype, device=dev)
+        sums.index_add_(0, lb, cand * w[:, None])
+        wsum_t.index_add_(0, lb, w)
+        ne = wsum_t > 0
+        C[ne] = sums[ne] / wsum_t[ne, None]
     return C
 
 
