@@ -133,8 +133,42 @@ class DataFrame:
         import pyarrow as pa
 
         table = pa.concat_tables(tables)
-        pdf = table.to_pandas()
-        return cls.from_pandas(pdf, vector_cols=vector_cols)
+        return cls._from_arrow(table, vector_cols)
+
+    @classmethod
+    def _from_arrow(cls, table: Any, vector_cols: Sequence[str] = ()) -> "DataFrame":
+        """Arrow table -> columns without the pandas hop: list columns become
+        2-D arrays straight from the flattened value buffers (per-row Python
+        lists cost ~1 µs/row — unusable at 1M×3000)."""
+        import pyarrow as pa
+
+        cols: Dict[str, Column] = {}
+        n = table.num_rows
+        for name in table.column_names:
+            arr = table.column(name).combine_chunks()
+            if isinstance(arr, pa.ChunkedArray):
+                arr = arr.chunk(0) if arr.num_chunks else pa.array([], type=arr.type)
+            t = arr.type
+            if pa.types.is_fixed_size_list(t):
+                width = t.list_size
+                vals = arr.values.to_numpy(zero_copy_only=False)
+                cols[name] = np.ascontiguousarray(vals.reshape(n, width))
+            elif pa.types.is_list(t) or pa.types.is_large_list(t):
+                offsets = arr.offsets.to_numpy(zero_copy_only=False)
+                widths = np.diff(offsets)
+                vals = arr.values.to_numpy(zero_copy_only=False)
+                if n > 0 and (widths == widths[0]).all():
+                    # uniform row width: one reshape off the value buffer
+                    cols[name] = np.ascontiguousarray(
+                        vals[offsets[0] : offsets[-1]].reshape(n, int(widths[0]))
+                    )
+                else:
+                    cols[name] = np.stack(
+                        [vals[offsets[i] : offsets[i + 1]] for i in range(n)]
+                    )
+            else:
+                cols[name] = arr.to_numpy(zero_copy_only=False)
+        return cls(cols)
 
     def write_parquet(self, path: str) -> None:
         """Each rank writes its shard as part-<rank>.parquet."""

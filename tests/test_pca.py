@@ -96,3 +96,11 @@ def test_pca_distributed_matches_single():
     single = PCA(k=3).fit(DataFrame.from_numpy(X))
     assert np.allclose(comp0, np.asarray(single.components_), atol=1e-6)
     assert np.allclose(mean0, X.mean(axis=0), atol=1e-8)
+
+
+def test_pca_input_col_api():
+    X = _data(n=100)
+    df = DataFrame({"feat_vec": X})
+    model = PCA(k=2).setInputCol("feat_vec").setOutputCol("proj").fit(df)
+    out = model.transform(df)
+    assert np.asarray(out["proj"]).shape == (100, 2)
