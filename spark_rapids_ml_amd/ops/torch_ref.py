@@ -86,21 +86,44 @@ def knn_topk(
     I: torch.Tensor,
     k: int,
     chunk: int = 8192,
+    item_chunk_elems: int = 1 << 31,
 ) -> Tuple[torch.Tensor, torch.Tensor]:
     """Brute-force kNN of queries Q against items I: returns (dists [q,k],
-    idx int64 [q,k]) with squared-euclid computed f32 then sqrt'ed."""
-    nq = Q.shape[0]
-    k = min(k, I.shape[0])
+    idx int64 [q,k]) with squared-euclid computed f32 then sqrt'ed.
+    Items are chunked too (running top-k merge) so the distance block stays
+    bounded — 10M items × an 8k query chunk would otherwise materialize
+    hundreds of GB."""
+    nq, ni = Q.shape[0], I.shape[0]
+    k = min(k, ni)
     dists = torch.empty((nq, k), dtype=torch.float32, device=Q.device)
     idx = torch.empty((nq, k), dtype=torch.int64, device=Q.device)
-    i_sq = (I**2).sum(dim=1)
+    ichunk = max(k, min(ni, item_chunk_elems // max(1, chunk)))
     for s in range(0, nq, chunk):
         e = min(nq, s + chunk)
         qb = Q[s:e]
-        d2 = (qb**2).sum(dim=1)[:, None] + i_sq[None, :] - 2.0 * (qb @ I.T)
-        vals, ids = torch.topk(d2, k, dim=1, largest=False)
-        dists[s:e] = torch.sqrt(torch.clamp(vals, min=0.0))
-        idx[s:e] = ids
+        q_sq = (qb**2).sum(dim=1)[:, None]
+        best_d: torch.Tensor = None  # type: ignore[assignment]
+        best_i: torch.Tensor = None  # type: ignore[assignment]
+        for s2 in range(0, ni, ichunk):
+            e2 = min(ni, s2 + ichunk)
+            ib = I[s2:e2]
+            d2 = q_sq + (ib**2).sum(dim=1)[None, :] - 2.0 * (qb @ ib.T)
+            kk = min(k, e2 - s2)
+            vals, ids = torch.topk(d2, kk, dim=1, largest=False)
+            ids = ids + s2
+            if best_d is None:
+                best_d, best_i = vals, ids
+            else:
+                cat_d = torch.cat([best_d, vals], dim=1)
+                cat_i = torch.cat([best_i, ids], dim=1)
+                best_d, order = torch.topk(cat_d, min(k, cat_d.shape[1]), dim=1, largest=False)
+                best_i = cat_i.gather(1, order)
+        if best_d.shape[1] < k:  # ni < k edge
+            pad = k - best_d.shape[1]
+            best_d = torch.nn.functional.pad(best_d, (0, pad), value=float("inf"))
+            best_i = torch.nn.functional.pad(best_i, (0, pad), value=-1)
+        dists[s:e] = torch.sqrt(torch.clamp(best_d, min=0.0))
+        idx[s:e] = best_i
     return dists, idx
 
 
