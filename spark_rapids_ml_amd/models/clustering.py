@@ -543,9 +543,15 @@ class DBSCANModel(_DBSCANParams, Model):
         n_local = X_local.shape[0]
         eps2 = eps * eps
 
-        # batched adjacency: rows [off, off+n_local) vs all
+        # batched adjacency: rows [off, off+n_local) vs all. The masked
+        # label tensors are [chunk, n] int64, so the chunk is bounded by
+        # bytes (reference max_mbytes_per_batch, clustering.py:673-682),
+        # defaulting to ~4 GB per intermediate.
         mb = self.getOrDefault("max_mbytes_per_batch")
-        chunk = max(1, int((mb * 1e6 / (4 * max(1, n))) if mb else 4096))
+        if mb:
+            chunk = max(1, int(mb * 1e6 / (8 * max(1, n))))
+        else:
+            chunk = max(64, min(4096, (4 << 30) // (8 * max(1, n))))
         x_sq = (Xf * Xf).sum(dim=1)
 
         core_local = torch.zeros(n_local, dtype=torch.bool, device=device)
