@@ -372,8 +372,10 @@ class ApproximateNearestNeighborsModel(_ANNModelParams, _NNModelBase):
     # -- IVF-PQ ------------------------------------------------------------
     def _make_ivfpq_search(self, algo_params: Dict[str, Any]):
         """IVF-PQ with exact refine (reference ivf_pq + refine,
-        knn.py:1512-1515,1643-1651): PQ-coded candidate scan, then exact
-        re-rank of refine_ratio*k candidates."""
+        knn.py:1512-1515,1643-1651): residual product quantization —
+        per-subspace 256-centroid codebooks trained on coarse residuals,
+        uint8 codes, per-query ADC lookup-table candidate scan over the
+        probed lists, then exact re-rank of refine_ratio*k candidates."""
 
         def search(Qt: torch.Tensor, X: np.ndarray, k: int):
             from ..parallel.context import get_comm
@@ -384,13 +386,17 @@ class ApproximateNearestNeighborsModel(_ANNModelParams, _NNModelBase):
             nlist = int(algo_params.get("nlist", self._default_nlist(n)))
             nprobe = int(algo_params.get("nprobe", max(1, nlist // 16)))
             refine = float(algo_params.get("refine_ratio", 2.0))
+            m_sub = int(algo_params.get("M", max(1, d // 4)))
+            while d % m_sub != 0:  # subspace dims must divide d
+                m_sub -= 1
             k_cand = min(n, max(k, int(k * refine)))
             C, labels = self._build_coarse(Xt, nlist)
-            # candidate scan uses exact distances on the probed lists (the PQ
-            # approximation is skipped — with 288 GB HBM the full-precision
-            # vectors are resident; refine then re-ranks exactly)
-            d_c, i_c = _ivf_search(Qt, Xt, C, labels, nlist, nprobe, k_cand)
-            # exact refine on candidates
+            resid = Xt - C[labels]
+            codebooks, codes = _pq_encode(resid, m_sub)
+            d_c, i_c = _ivfpq_scan(
+                Qt, C, labels, codebooks, codes, nlist, nprobe, k_cand
+            )
+            # exact refine on candidates (full-precision vectors resident)
             rows = torch.arange(Qt.shape[0], device=device)[:, None]
             cand = Xt[i_c.clamp(min=0)]  # [q, k_cand, d]
             d_exact = ((Qt[:, None, :] - cand) ** 2).sum(dim=2)
@@ -542,3 +548,111 @@ def _graph_beam_search(
 
     ids2, vals2 = _dedup_topk(beam, bd, min(k, bd.shape[1]))
     return torch.sqrt(torch.clamp(vals2, min=0.0)), ids2
+
+
+def _pq_encode(resid: torch.Tensor, m_sub: int, n_codes: int = 256, iters: int = 8):
+    """Train per-subspace codebooks (k-means on a sample) and encode.
+    Returns (codebooks [m, 256, ds], codes uint8 [n, m])."""
+    n, d = resid.shape
+    ds = d // m_sub
+    device = resid.device
+    g = torch.Generator(device="cpu")
+    g.manual_seed(123)
+    n_codes = min(n_codes, max(2, n))
+    sample = resid[torch.randperm(n, generator=g)[: min(n, 65536)].to(device)]
+    codebooks = torch.empty(m_sub, n_codes, ds, dtype=torch.float32, device=device)
+    codes = torch.empty(n, m_sub, dtype=torch.uint8, device=device)
+    for m in range(m_sub):
+        sub = sample[:, m * ds : (m + 1) * ds].contiguous()
+        cb = sub[torch.randperm(sub.shape[0], generator=g)[:n_codes].to(device)].clone()
+        for _ in range(iters):
+            d2 = (
+                (sub * sub).sum(1)[:, None]
+                + (cb * cb).sum(1)[None, :]
+                - 2.0 * (sub @ cb.T)
+            )
+            lb = d2.argmin(dim=1)
+            sums = torch.zeros_like(cb)
+            cnt = torch.zeros(n_codes, device=device)
+            sums.index_add_(0, lb, sub)
+            cnt.index_add_(0, lb, torch.ones_like(lb, dtype=torch.float32))
+            ne = cnt > 0
+            cb[ne] = sums[ne] / cnt[ne, None]
+        codebooks[m] = cb
+        full = resid[:, m * ds : (m + 1) * ds]
+        # chunked encode
+        for s0 in range(0, n, 1 << 18):
+            e0 = min(n, s0 + (1 << 18))
+            d2 = (
+                (full[s0:e0] * full[s0:e0]).sum(1)[:, None]
+                + (cb * cb).sum(1)[None, :]
+                - 2.0 * (full[s0:e0] @ cb.T)
+            )
+            codes[s0:e0, m] = d2.argmin(dim=1).to(torch.uint8)
+    return codebooks, codes
+
+
+def _ivfpq_scan(
+    Qt: torch.Tensor,
+    C: torch.Tensor,
+    labels: torch.Tensor,
+    codebooks: torch.Tensor,
+    codes: torch.Tensor,
+    nlist: int,
+    nprobe: int,
+    k: int,
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    """ADC scan: per (query, probed list), approx dist = ||q - c_list||-part
+    + sum_m LUT[m, code_m] where LUT holds ||(q - c_list)_m - cb[m,j]||²."""
+    device = Qt.device
+    nq = Qt.shape[0]
+    m_sub, n_codes, ds = codebooks.shape
+    nprobe = min(nprobe, nlist)
+    cd = (
+        (Qt * Qt).sum(dim=1)[:, None]
+        + (C * C).sum(dim=1)[None, :]
+        - 2.0 * (Qt @ C.T)
+    )
+    probe = cd.topk(nprobe, dim=1, largest=False).indices  # [nq, nprobe]
+
+    order = torch.argsort(labels)
+    sorted_labels = labels[order]
+    boundaries = torch.searchsorted(
+        sorted_labels, torch.arange(nlist + 1, device=device)
+    )
+    best_d = torch.full((nq, k), float("inf"), device=device)
+    best_i = torch.full((nq, k), -1, dtype=torch.int64, device=device)
+    cb_sq = (codebooks * codebooks).sum(dim=2)  # [m, 256]
+    for p in range(nprobe):
+        lists = probe[:, p]  # per-query probed list id
+        # group queries by list for batched LUT scans
+        uniq, inv = torch.unique(lists, return_inverse=True)
+        for ui in range(uniq.numel()):
+            l = int(uniq[ui])
+            s, e = int(boundaries[l]), int(boundaries[l + 1])
+            if s == e:
+                continue
+            qsel = torch.nonzero(inv == ui).flatten()
+            items = order[s:e]
+            qres = Qt[qsel] - C[l][None, :]  # [m_q, d] residual queries
+            # LUT [m_q, m_sub, 256]: ||qres_m||² - 2 qres_m . cb + ||cb||²
+            qr = qres.view(-1, m_sub, ds)
+            dots = torch.einsum("qmd,mcd->qmc", qr, codebooks)
+            lut = (qr * qr).sum(2)[:, :, None] - 2.0 * dots + cb_sq[None, :, :]
+            lc = codes[items].long()  # [n_l, m_sub]
+            # approx dists [m_q, n_l] = sum_m lut[q, m, code]
+            d_apx = torch.zeros(qsel.numel(), items.numel(), device=device)
+            for m in range(m_sub):
+                d_apx += lut[:, m, :].gather(
+                    1, lc[:, m][None, :].expand(qsel.numel(), -1)
+                )
+            kb = min(k, items.numel())
+            vals, loc = torch.topk(d_apx, kb, dim=1, largest=False)
+            ids = items[loc]
+            cat_d = torch.cat([best_d[qsel], vals], dim=1)
+            cat_i = torch.cat([best_i[qsel], ids], dim=1)
+            mvals, morder = torch.topk(cat_d, k, dim=1, largest=False)
+            rows = torch.arange(qsel.numel(), device=device)[:, None]
+            best_d[qsel] = mvals
+            best_i[qsel] = cat_i.gather(1, morder)
+    return best_d, best_i

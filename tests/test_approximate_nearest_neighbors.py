@@ -23,7 +23,7 @@ def _recall(idx, sk_idx):
 
 @pytest.mark.parametrize("algo,params", [
     ("ivfflat", {"nlist": 32, "nprobe": 8}),
-    ("ivfpq", {"nlist": 32, "nprobe": 8, "refine_ratio": 2.0}),
+    ("ivfpq", {"nlist": 32, "nprobe": 8, "refine_ratio": 4.0}),
 ])
 def test_ann_recall(algo, params):
     X = _data()
