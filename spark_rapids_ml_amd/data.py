@@ -239,6 +239,29 @@ class DataFrame:
         cols[name] = col
         return DataFrame(cols)
 
+    def randomSplit(self, weights: Sequence[float], seed: int = 0) -> List["DataFrame"]:
+        """Split local rows by weight fractions (pyspark DataFrame.randomSplit
+        semantics; each rank splits its shard with a rank-salted seed)."""
+        from .parallel.context import get_comm
+
+        rng = np.random.default_rng(seed + 1000003 * get_comm().rank)
+        w = np.asarray(weights, dtype=np.float64)
+        w = w / w.sum()
+        edges = np.cumsum(w)
+        draws = rng.random(self._n)
+        out = []
+        lo = 0.0
+        for hi in edges:
+            out.append(self.take_local(np.nonzero((draws >= lo) & (draws < hi))[0]))
+            lo = hi
+        return out
+
+    def sample(self, fraction: float, seed: int = 0) -> "DataFrame":
+        from .parallel.context import get_comm
+
+        rng = np.random.default_rng(seed + 1000003 * get_comm().rank)
+        return self.take_local(np.nonzero(rng.random(self._n) < fraction)[0])
+
     def take_local(self, idx: np.ndarray) -> "DataFrame":
         return DataFrame({n: c[idx] for n, c in self._cols.items()})
 

@@ -94,3 +94,32 @@ def test_cv_model_persistence(tmp_path):
     loaded = CrossValidatorModel.load(path)
     assert loaded.avgMetrics == model.avgMetrics
     assert np.allclose(loaded.bestModel.coefficients, model.bestModel.coefficients)
+
+
+def _dist_cv(seed: int):
+    from spark_rapids_ml_amd.parallel.context import get_comm
+
+    comm = get_comm()
+    X, y = make_regression(n_samples=300, n_features=6, noise=1.0, random_state=seed)
+    sl = slice(comm.rank, None, comm.world_size)
+    df = DataFrame.from_numpy(X[sl].astype(np.float64), y[sl].astype(np.float64))
+    lr = LinearRegression()
+    cv = CrossValidator(
+        estimator=lr,
+        estimatorParamMaps=ParamGridBuilder().addGrid(lr.getParam("regParam"), [0.0, 50.0]).build(),
+        evaluator=RegressionEvaluator(metricName="rmse"),
+        numFolds=2,
+    )
+    model = cv.fit(df)
+    return model.avgMetrics, float(model.bestModel.getOrDefault("regParam"))
+
+
+def test_cv_distributed():
+    from .dist_utils import run_distributed
+
+    results = run_distributed(_dist_cv, world_size=2, args=(0,))
+    for metrics, best in results:
+        assert len(metrics) == 2
+        assert best == 0.0
+    # ranks agree on metrics (collectives aligned)
+    assert np.allclose(results[0][0], results[1][0])
