@@ -483,7 +483,11 @@ class _NativeParams(_NativeClass, Params):
         super().__init__(**kwargs)
         self._native_params: Dict[str, Any] = dict(self._get_native_params_default())
         self._num_workers: Optional[int] = None
-        self._fallback_enabled = False
+        # reference spark.rapids.ml.cpu.fallback.enabled (params.py:276-285):
+        # unsupported params log-and-ignore instead of raising
+        from .config import get_conf
+
+        self._fallback_enabled = bool(get_conf("cpu_fallback_enabled"))
 
     @property
     def native_params(self) -> Dict[str, Any]:
