@@ -224,6 +224,10 @@ class UMAP(_UMAPParams, Estimator):
     def setOutputCol(self, value: str) -> "UMAP":
         return self._set_params(outputCol=value)
 
+    def setLabelCol(self, value: str) -> "UMAP":
+        """Supervised fit label column (reference umap.py:1035-1050)."""
+        return self._set_params(labelCol=value)
+
     def setK(self, value: int) -> "UMAP":
         return self._set_params(n_neighbors=value)
 

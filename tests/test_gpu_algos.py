@@ -138,3 +138,39 @@ def test_sparse_logreg_gpu():
     model = LogisticRegression(regParam=1e-4, maxIter=100).fit(DataFrame.from_numpy(Xs, y))
     out = model.transform(DataFrame.from_numpy(Xs))
     assert (np.asarray(out["prediction"]) == y).mean() > 0.95
+
+
+def test_ann_pq_and_cagra_gpu():
+    from sklearn.neighbors import NearestNeighbors as SkNN
+
+    from spark_rapids_ml_amd import ApproximateNearestNeighbors
+
+    rng = np.random.default_rng(1)
+    X = rng.normal(size=(20000, 64)).astype(np.float32)
+    sk = SkNN(n_neighbors=10).fit(X)
+    _, sk_idx = sk.kneighbors(X[:200])
+    for algo, params, floor in [
+        ("ivfpq", {"nlist": 64, "nprobe": 16, "refine_ratio": 4.0}, 0.6),
+        ("cagra", {"graph_degree": 32, "itopk_size": 128, "max_iterations": 10}, 0.6),
+    ]:
+        model = ApproximateNearestNeighbors(k=10, algorithm=algo, algoParams=params).fit(
+            DataFrame.from_numpy(X)
+        )
+        _, _, knn_df = model.kneighbors(DataFrame.from_numpy(X[:200]))
+        idx = np.asarray(knn_df["indices"])
+        hits = sum(len(set(a.tolist()) & set(b.tolist())) for a, b in zip(idx, sk_idx))
+        assert hits / sk_idx.size > floor, f"{algo}: {hits / sk_idx.size}"
+
+
+def test_umap_supervised_gpu():
+    from spark_rapids_ml_amd import UMAP
+
+    X, y = make_blobs(n_samples=3000, n_features=16, centers=4, cluster_std=2.0, random_state=0)
+    df = DataFrame.from_numpy(X.astype(np.float32), y.astype(np.float64))
+    model = UMAP(n_neighbors=10, n_epochs=100, random_state=0).setLabelCol("label").fit(df)
+    emb = model.embedding
+    # supervised embedding should separate classes decently even with overlap
+    from sklearn.neighbors import KNeighborsClassifier
+
+    acc = KNeighborsClassifier(5).fit(emb, y).score(emb, y)
+    assert acc > 0.9
