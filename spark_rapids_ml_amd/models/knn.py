@@ -333,7 +333,9 @@ class ApproximateNearestNeighborsModel(_ANNModelParams, _NNModelBase):
             itopk = int(algo_params.get("itopk_size", max(64, 2 * k)))
             hops = int(algo_params.get("max_iterations", 8))
             G = _nn_descent(Xt, degree, build_iters)
-            return _graph_beam_search(Qt, Xt, G, k, itopk, hops)
+            # cagra's "optimize": augment the search graph with reverse edges
+            G_search = torch.cat([G, _reverse_edges(G, degree // 2)], dim=1)
+            return _graph_beam_search(Qt, Xt, G_search, k, itopk, hops)
 
         return search
 
@@ -490,9 +492,28 @@ def _batched_dists(X: torch.Tensor, rows: torch.Tensor, cand: torch.Tensor) -> t
     return out
 
 
+def _reverse_edges(G: torch.Tensor, cap: int) -> torch.Tensor:
+    """Per-node reverse neighbors (who points at me), capped; filler = own
+    index (harmless duplicate under dedup). [n, cap] int64."""
+    n, deg = G.shape
+    device = G.device
+    src = torch.arange(n, device=device).repeat_interleave(deg)
+    dst = G.reshape(-1)
+    order = torch.argsort(dst)
+    dst_s = dst[order]
+    src_s = src[order]
+    firsts = torch.searchsorted(dst_s, torch.arange(n, device=device))
+    pos = torch.arange(dst_s.numel(), device=device) - firsts[dst_s]
+    keep = pos < cap
+    rev = torch.arange(n, device=device)[:, None].expand(n, cap).contiguous()
+    rev[dst_s[keep], pos[keep]] = src_s[keep]
+    return rev
+
+
 def _nn_descent(X: torch.Tensor, degree: int, n_iter: int) -> torch.Tensor:
     """Approximate kNN graph by nn-descent (reference cagra's graph build):
-    start from random neighbors, iteratively propose neighbors-of-neighbors,
+    start from random neighbors, iteratively propose neighbors-of-neighbors
+    in BOTH directions (forward + reverse join, the core of nn-descent),
     keep the best `degree` per node. Returns [n, degree] int64."""
     n = X.shape[0]
     device = X.device
@@ -502,13 +523,13 @@ def _nn_descent(X: torch.Tensor, degree: int, n_iter: int) -> torch.Tensor:
     rows = torch.arange(n, device=device)
     Gd = _batched_dists(X, rows, G)
     for _ in range(max(1, n_iter)):
-        # candidates: neighbors-of-neighbors (subsampled to bound degree^2)
-        sub = G[:, : min(degree, 8)]
-        cand = G[sub].reshape(n, -1)  # [n, sub*degree]
+        sub_f = G[:, : min(degree, 8)]
+        rev = _reverse_edges(G, min(degree, 8))
+        join = torch.cat([sub_f, rev], dim=1)  # [n, <=16]
+        cand = torch.cat([G[join].reshape(n, -1), rev], dim=1)
         cd = _batched_dists(X, rows, cand)
         cat_i = torch.cat([G, cand], dim=1)
         cat_d = torch.cat([Gd, cd], dim=1)
-        # avoid self loops, then duplicate-free top-degree
         cat_d = torch.where(cat_i == rows[:, None], torch.full_like(cat_d, 3.4e38), cat_d)
         G, Gd = _dedup_topk(cat_i, cat_d, degree)
     return G
