@@ -295,7 +295,21 @@ class UMAP(_UMAPParams, Estimator):
         dim = int(self.getOrDefault("n_components"))
         seed = int(self.getOrDefault("random_state"))
         Xt = to_device_tensor(X, device)
-        d, i = knn_topk(Xt, Xt, k + 1)
+        if n <= 50000:
+            d, i = knn_topk(Xt, Xt, k + 1)
+        else:
+            # nn-descent graph (reference switches off brute force above 50k
+            # rows, umap.py:359-366); reuse the CAGRA build
+            from .knn import _nn_descent, _batched_dists
+
+            G = _nn_descent(Xt, max(k + 1, 32), 5)
+            rows = torch.arange(n, device=Xt.device)
+            Gd = _batched_dists(Xt, rows, G)
+            vals, order = torch.sort(Gd, dim=1)
+            i = torch.cat([rows[:, None], G.gather(1, order)[:, :k]], dim=1)
+            d = torch.cat(
+                [torch.zeros(n, 1, device=Xt.device), torch.sqrt(vals[:, :k])], dim=1
+            )
         knn_d, knn_i = as_numpy(d)[:, 1:], as_numpy(i)[:, 1:]  # drop self
 
         rows, cols, vals = _fuzzy_simplicial_set(
