@@ -394,6 +394,31 @@ def _grow_tree(
                         Xb, perm, seg_off, fsel32, y32, f0, F, n_bins,
                         n_classes if task == "classification" else 0,
                     )
+                    if H.shape[3] <= 16:
+                        # fused gain scan + per-node best (kernel)
+                        g, fidx, sbin_b, lval_b, rval_b = ext.rf_best_split(
+                            H, min_leaf, task == "classification"
+                        )
+                        fidx = fidx.to(torch.int64)
+                        upd = g > best_gain
+                        best_gain = torch.where(upd, g, best_gain)
+                        ar = torch.arange(B, device=dev)
+                        if feat_sel is not None:
+                            chosen = feat_sel[ar, (fidx + f0).clamp(min=0)]
+                        else:
+                            chosen = fidx + f0
+                        best_feat = torch.where(upd, chosen, best_feat)
+                        best_bin = torch.where(upd, sbin_b.to(torch.int64), best_bin)
+                        if task == "regression":
+                            # kernel emits raw (count, sum); the tree stores
+                            # (mean, count) leaf values
+                            lc = torch.clamp(lval_b[:, 0], min=1e-12)
+                            rc = torch.clamp(rval_b[:, 0], min=1e-12)
+                            lval_b = torch.stack([lval_b[:, 1] / lc, lval_b[:, 0]], dim=1)
+                            rval_b = torch.stack([rval_b[:, 1] / rc, rval_b[:, 0]], dim=1)
+                        best_lval = torch.where(upd[:, None], lval_b, best_lval)
+                        best_rval = torch.where(upd[:, None], rval_b, best_rval)
+                        continue
                     if task == "classification":
                         gain, sbin, lval, rval, lcnt = _best_split_class(H, min_leaf)
                     else:

@@ -174,3 +174,17 @@ def test_umap_supervised_gpu():
 
     acc = KNeighborsClassifier(5).fit(emb, y).score(emb, y)
     assert acc > 0.9
+
+
+def test_rfr_gpu_r2():
+    from spark_rapids_ml_amd import RandomForestRegressor
+
+    X, y = make_regression(n_samples=5000, n_features=20, n_informative=10, noise=5.0, random_state=0)
+    X = X.astype(np.float32)
+    model = RandomForestRegressor(numTrees=20, maxDepth=8, seed=0).fit(
+        DataFrame.from_numpy(X, y.astype(np.float64))
+    )
+    out = model.transform(DataFrame.from_numpy(X))
+    pred = np.asarray(out["prediction"])
+    r2 = 1 - ((pred - y) ** 2).sum() / ((y - y.mean()) ** 2).sum()
+    assert r2 > 0.8, r2
