@@ -64,6 +64,16 @@ class _NNModelBase(_NNParams, Model):
         super().__init__(**kwargs)
         self._item_df = item_df
 
+    def write(self):
+        # persistence explicitly unsupported, matching the reference
+        # (knn.py:384-408,484-508): the model is the raw item dataframe
+        raise NotImplementedError(
+            f"{type(self).__name__} does not support saving (reference parity)"
+        )
+
+    def save(self, path: str) -> None:
+        self.write()
+
     def _transform_array(self, X: Any):  # pragma: no cover
         raise NotImplementedError("use kneighbors()/approxSimilarityJoin()")
 

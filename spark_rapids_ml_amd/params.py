@@ -519,7 +519,6 @@ class _NativeParams(_NativeClass, Params):
     def _set_params(self: P, **kwargs: Any) -> P:
         """Set Spark params and/or native params, keeping both in sync
         (reference params.py:430-487)."""
-        mapping = self._param_mapping()
         for k, v in kwargs.items():
             if k == "num_workers":
                 self._num_workers = v

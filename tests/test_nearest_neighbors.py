@@ -73,3 +73,10 @@ def test_knn_distributed_matches_single():
     for r, (idx, dist) in enumerate(results):
         assert np.array_equal(idx, sk_idx[r::2])
         assert np.allclose(dist, sk_dist[r::2], atol=1e-4)
+
+
+def test_knn_model_persistence_unsupported():
+    X = _data(n=50)
+    model = NearestNeighbors(k=3).fit(DataFrame.from_numpy(X))
+    with pytest.raises(NotImplementedError):
+        model.save("/tmp/should_not_exist")
