@@ -59,6 +59,19 @@ class ModelRegistry:
         return {mid: type(m).__name__ for mid, m in self._models.items()}
 
 
+def _check_path(path: str) -> str:
+    """Optional confinement: with SRML_SERVER_DATA_ROOT set, data/output
+    paths must resolve under it (the service reads/writes as its own user)."""
+    import os
+
+    root = os.environ.get("SRML_SERVER_DATA_ROOT")
+    if root:
+        real = os.path.realpath(path)
+        if not real.startswith(os.path.realpath(root) + os.sep):
+            raise ValueError(f"path {path!r} outside SRML_SERVER_DATA_ROOT")
+    return path
+
+
 def _resolve_estimator(name: str, params: Dict[str, Any]) -> Estimator:
     if name not in _ESTIMATORS:
         raise ValueError(f"unsupported estimator {name!r}")
@@ -116,7 +129,7 @@ def create_app(registry: Optional[ModelRegistry] = None):
     def fit(req: FitRequest):
         try:
             est = _resolve_estimator(req.estimator, req.params)
-            df = DataFrame.read_parquet(req.data_path, vector_cols=["features"])
+            df = DataFrame.read_parquet(_check_path(req.data_path), vector_cols=["features"])
             model = est.fit(df)
             mid = reg.put(model)
             return {"model_id": mid, "attributes": _attrs_json(model)}
@@ -127,9 +140,9 @@ def create_app(registry: Optional[ModelRegistry] = None):
     def transform(req: TransformRequest):
         try:
             model = reg.get(req.model_id)
-            df = DataFrame.read_parquet(req.data_path, vector_cols=["features"])
+            df = DataFrame.read_parquet(_check_path(req.data_path), vector_cols=["features"])
             out = model.transform(df)
-            out.write_parquet(req.output_path)
+            out.write_parquet(_check_path(req.output_path))
             return {"output_path": req.output_path, "columns": out.columns}
         except (ValueError, KeyError, FileNotFoundError) as e:
             raise HTTPException(status_code=400, detail=str(e))
@@ -141,7 +154,7 @@ def create_app(registry: Optional[ModelRegistry] = None):
     @app.post("/models/{model_id}/save")
     def save(model_id: str, req: SaveRequest):
         try:
-            reg.get(model_id).write().overwrite().save(req.path)
+            reg.get(model_id).write().overwrite().save(_check_path(req.path))
             return {"path": req.path}
         except KeyError as e:
             raise HTTPException(status_code=404, detail=str(e))

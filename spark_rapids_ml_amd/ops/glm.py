@@ -44,7 +44,7 @@ def logistic_grad_loss(
     if fit_intercept:
         scores = scores + W[:, d][None, :]
 
-    if use_hip(scores):
+    if use_hip(scores) and y_idx.dtype == torch.int64:
         ext = hip_ops()
         resid, loss = ext.softmax_residual_loss(
             scores.contiguous().to(torch.float32), y_idx.contiguous()

@@ -23,7 +23,7 @@ def kmeans_assign_reduce(
     x_sq: Optional[torch.Tensor] = None,
 ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor, float]:
     """Returns (labels i32 [n], sums [k,d], counts [k], inertia)."""
-    if use_hip(X):
+    if use_hip(X) and X.dtype == torch.float32:
         ext = hip_ops()
         if x_sq is None:
             x_sq = _xsq(X)
@@ -36,7 +36,7 @@ def kmeans_assign_reduce(
 
 
 def kmeans_predict(X: torch.Tensor, C: torch.Tensor) -> torch.Tensor:
-    if use_hip(X):
+    if use_hip(X) and X.dtype == torch.float32:
         ext = hip_ops()
         labels, _, _ = ext.kmeans_assign(X.contiguous(), C.contiguous(), _xsq(X))
         return labels

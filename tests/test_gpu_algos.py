@@ -188,3 +188,13 @@ def test_rfr_gpu_r2():
     pred = np.asarray(out["prediction"])
     r2 = 1 - ((pred - y) ** 2).sum() / ((y - y.mean()) ** 2).sum()
     assert r2 > 0.8, r2
+
+
+def test_kmeans_gpu_float64_inputs():
+    # float32_inputs=False keeps f64 end to end: GPU falls back to the
+    # torch path (the f32 MFMA kernel must not receive f64)
+    X, _ = make_blobs(n_samples=2000, n_features=16, centers=4, random_state=0)
+    model = KMeans(k=4, maxIter=10, seed=1, float32_inputs=False).fit(
+        DataFrame.from_numpy(X.astype(np.float64))
+    )
+    assert model.cluster_centers_.dtype == np.float64
