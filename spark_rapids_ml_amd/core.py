@@ -52,7 +52,21 @@ class _FitContext:
         # shared across the param maps of one fit call so a multi-model sweep
         # reuses one data pass (reference single-pass fitMultiple,
         # core.py:1177-1228 / regression.py:657-674)
-        self.cache: Dict[str, Any] = {}
+        self.cache: Dict[Any, Any] = {}
+
+    def device_tensor(self, X: Any, dtype: Any = None) -> torch.Tensor:
+        """Ingest the local feature shard once per fit call (cached across
+        the param maps of a fitMultiple sweep)."""
+        from .data import to_device_tensor
+
+        key = ("device_tensor", id(X), str(dtype))
+        if key not in self.cache:
+            arr = np.ascontiguousarray(X)
+            t = to_device_tensor(arr, self.device)
+            if dtype is not None:
+                t = t.to(dtype)
+            self.cache[key] = t
+        return self.cache[key]
 
 
 class Estimator(_NativeParams, HasVerbose):

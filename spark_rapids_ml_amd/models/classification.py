@@ -205,7 +205,7 @@ class LogisticRegression(_LogisticRegressionParams, Estimator):
                 size=Xcsr.shape,
             ).to(ctx.device)
         else:
-            Xt = to_device_tensor(np.ascontiguousarray(X), ctx.device)
+            Xt = ctx.device_tensor(X)
 
         # classes: global sorted unique labels (reference allgathers classes_)
         local_classes = np.unique(np.asarray(y)) if len(np.asarray(y)) else np.array([])

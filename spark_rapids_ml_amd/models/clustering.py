@@ -154,8 +154,10 @@ class KMeans(_KMeansParams, Estimator):
         if pdesc.m < k:
             raise ValueError(f"n_clusters={k} > total rows {pdesc.m}")
 
-        Xt = to_device_tensor(np.ascontiguousarray(X), ctx.device)
-        x_sq = (Xt * Xt).sum(dim=1)
+        Xt = ctx.device_tensor(X)
+        if "x_sq" not in ctx.cache:
+            ctx.cache["x_sq"] = (Xt * Xt).sum(dim=1)
+        x_sq = ctx.cache["x_sq"]
 
         if params["init"] == "random":
             C = self._init_random(Xt, k, seed, comm, pdesc)

@@ -90,7 +90,7 @@ class PCA(_PCAParams, Estimator):
         if k > min(pdesc.m, d):
             raise ValueError(f"k={k} > min(n_rows={pdesc.m}, n_cols={d})")
 
-        Xt = to_device_tensor(np.ascontiguousarray(X), ctx.device)
+        Xt = ctx.device_tensor(X)
         # fused partials: [d+1, d] buffer = [Gram (d×d) ; colsum (1×d)]
         buf = torch.zeros((d + 1, d), dtype=torch.float64, device=Xt.device)
         if Xt.shape[0] > 0:

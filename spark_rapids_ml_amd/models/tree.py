@@ -631,7 +631,7 @@ class _RandomForestEstimator(_RandomForestParams, Estimator):
         bootstrap = bool(params["bootstrap"])
         max_samples = float(params["max_samples"])
 
-        Xt = to_device_tensor(np.ascontiguousarray(X, dtype=np.float32), ctx.device)
+        Xt = ctx.device_tensor(np.asarray(X, dtype=np.float32))
         n, d = Xt.shape
 
         if self._task == "classification":
