@@ -230,8 +230,9 @@ class LogisticRegression(_LogisticRegressionParams, Estimator):
                 "num_iters_hist_": [],
             }
 
-        cls_to_idx = {float(c): i for i, c in enumerate(classes)}
-        y_idx = np.array([cls_to_idx[float(v)] for v in np.asarray(y)], dtype=np.int64)
+        # vectorized label -> class-index mapping (a python dict loop costs
+        # ~0.3 µs/row — seconds at 50M rows)
+        y_idx = np.searchsorted(classes, np.asarray(y)).astype(np.int64)
         y_t = to_device_tensor(y_idx, ctx.device)
 
         multinomial = (family == "multinomial") or (family == "auto" and n_classes > 2)
