@@ -198,10 +198,13 @@ class LogisticRegression(_LogisticRegressionParams, Estimator):
             import scipy.sparse as sp
 
             Xcsr = X.tocsr()
+            # int32 indices below the 2^31 nnz escape hatch (reference
+            # classification.py:960-966,1054-1055), int64 above
+            idx_t = np.int32 if Xcsr.nnz < 2**31 else np.int64
             Xt = torch.sparse_csr_tensor(
-                torch.from_numpy(Xcsr.indptr.astype(np.int64)),
-                torch.from_numpy(Xcsr.indices.astype(np.int64)),
-                torch.from_numpy(Xcsr.data.astype(np.float32)),
+                torch.from_numpy(np.ascontiguousarray(Xcsr.indptr, dtype=idx_t)),
+                torch.from_numpy(np.ascontiguousarray(Xcsr.indices, dtype=idx_t)),
+                torch.from_numpy(np.ascontiguousarray(Xcsr.data, dtype=np.float32)),
                 size=Xcsr.shape,
             ).to(ctx.device)
         else:
