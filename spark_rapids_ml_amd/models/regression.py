@@ -297,38 +297,44 @@ def _cd_elasticnet(
 
     obj(w) = 1/2 wᵀQw - rᵀw + λ Σ_j s_j[(1-α)/2 w_j² + α|w_j|]
     update: w_j <- S(r_j - Σ_{i≠j} Q_ij w_i, λ α s_j) / (Q_jj + λ(1-α) s_j)
+
+    Runs on numpy: the replicated d×d Gram is tiny and the per-coordinate
+    loop would otherwise pay a device sync per coordinate.
     """
-    d = Q.shape[0]
-    w = torch.zeros(d, dtype=torch.float64, device=Q.device)
-    Q = Q.to(torch.float64)
-    r = r.to(torch.float64)
-    qdiag = torch.diagonal(Q)
-    denom = qdiag + lam * (1.0 - l1r) * pen_scale
-    denom = torch.clamp(denom, min=1e-30)
-    thresh = lam * l1r * pen_scale
-    Qw = torch.zeros_like(w)
+    import numpy as _np
+
+    dev = Q.device
+    Qn = Q.detach().cpu().to(torch.float64).numpy()
+    rn = r.detach().cpu().to(torch.float64).numpy()
+    pn = pen_scale.detach().cpu().to(torch.float64).numpy()
+    d = Qn.shape[0]
+    w = _np.zeros(d)
+    qdiag = _np.diagonal(Qn).copy()
+    denom = _np.maximum(qdiag + lam * (1.0 - l1r) * pn, 1e-30)
+    thresh = lam * l1r * pn
+    Qw = _np.zeros(d)
     for sweep in range(max_iter):
         w_max = 0.0
         d_max = 0.0
         for j in range(d):
-            wj_old = float(w[j])
-            rho = float(r[j] - Qw[j] + qdiag[j] * wj_old)
-            t = float(thresh[j])
+            wj_old = w[j]
+            rho = rn[j] - Qw[j] + qdiag[j] * wj_old
+            t = thresh[j]
             if rho > t:
-                wj = (rho - t) / float(denom[j])
+                wj = (rho - t) / denom[j]
             elif rho < -t:
-                wj = (rho + t) / float(denom[j])
+                wj = (rho + t) / denom[j]
             else:
                 wj = 0.0
             diff = wj - wj_old
             if diff != 0.0:
-                Qw += Q[:, j] * diff
+                Qw += Qn[:, j] * diff
                 w[j] = wj
             w_max = max(w_max, abs(wj))
             d_max = max(d_max, abs(diff))
         if w_max == 0.0 or d_max / max(w_max, 1e-30) < tol:
             break
-    return w
+    return torch.from_numpy(w).to(dev)
 
 
 class LinearRegressionModel(_LinearRegressionParams, Model):
