@@ -110,3 +110,25 @@ def test_kmeans_distributed_matches_single():
     assert np.allclose(
         _sorted_centers(c0), _sorted_centers(single.cluster_centers_), atol=0.05
     )
+
+
+def test_kmeans_rejects_noneuclidean():
+    with pytest.raises(ValueError):
+        KMeans(distanceMeasure="cosine")
+
+
+def test_kmeans_compute_cost():
+    X, _ = _make(n=300)
+    df = DataFrame.from_numpy(X)
+    model = KMeans(k=4, maxIter=20, seed=1, tol=1e-6).fit(df)
+    cost = model.computeCost(df)
+    assert np.isclose(cost, model.trainingCost, rtol=0.05)
+
+
+def test_kmeans_summary():
+    X, _ = _make(n=300)
+    model = KMeans(k=4, maxIter=10, seed=1).fit(DataFrame.from_numpy(X))
+    s = model.summary
+    assert s.numIter >= 1
+    assert sum(s.clusterSizes) == 300
+    assert s.trainingCost == model.trainingCost
