@@ -102,3 +102,36 @@ def test_server_fit_transform_roundtrip(client, tmp_path):
 def test_server_bad_estimator(client):
     r = client.post("/fit", json={"estimator": "Nope", "params": {}, "data_path": "/none"})
     assert r.status_code == 400
+
+
+def test_launch_single_rank(tmp_path):
+    script = tmp_path / "app.py"
+    script.write_text(
+        "import spark_rapids_ml\n"
+        "from spark_rapids_ml_amd.parallel.context import get_comm\n"
+        "print('LAUNCH_OK', get_comm().world_size)\n"
+    )
+    r = subprocess.run(
+        [sys.executable, "-m", "spark_rapids_ml_amd.launch", "--gpus", "0", str(script)],
+        capture_output=True, text=True, timeout=120,
+    )
+    assert r.returncode == 0, r.stderr
+    assert "LAUNCH_OK 1" in r.stdout
+
+
+def test_launch_two_ranks_gloo(tmp_path):
+    script = tmp_path / "app.py"
+    script.write_text(
+        "from spark_rapids_ml_amd.parallel.context import get_comm\n"
+        "c = get_comm()\n"
+        "print(f'LAUNCH_RANK {c.rank}/{c.world_size}')\n"
+    )
+    env = dict(os.environ, SRML_BACKEND="gloo")
+    r = subprocess.run(
+        [sys.executable, "-m", "spark_rapids_ml_amd.launch", "--gpus", "2",
+         "--master-port", "29719", str(script)],
+        env=env, capture_output=True, text=True, timeout=240,
+    )
+    assert r.returncode == 0, r.stderr
+    assert "LAUNCH_RANK 0/2" in r.stdout
+    assert "LAUNCH_RANK 1/2" in r.stdout

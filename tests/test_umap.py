@@ -60,3 +60,15 @@ def test_find_ab_params():
     # umap-learn reference values for spread=1, min_dist=0.1
     assert abs(a - 1.577) < 0.05
     assert abs(b - 0.895) < 0.05
+
+
+def test_umap_model_persistence(tmp_model_path):
+    from spark_rapids_ml_amd.models.umap import UMAPModel
+
+    X, _ = _data(n=300)
+    model = UMAP(n_neighbors=10, n_epochs=20, random_state=1).fit(DataFrame.from_numpy(X))
+    model.save(tmp_model_path)
+    loaded = UMAPModel.load(tmp_model_path)
+    assert np.allclose(loaded.embedding, model.embedding)
+    out = loaded.transform(DataFrame.from_numpy(X[:40]))
+    assert np.asarray(out["embedding"]).shape == (40, 2)
