@@ -160,8 +160,8 @@ class _NNModelBase(_NNParams, Model):
             ids_local = item_ids[as_numpy(idx)]
             d_np = as_numpy(dists)
         else:
-            d_np = np.full((Q_full.shape[0], k), np.inf, dtype=np.float32)
-            ids_local = np.full((Q_full.shape[0], k), -1, dtype=np.int64)
+            d_np = np.full((Qt.shape[0], k), np.inf, dtype=np.float32)
+            ids_local = np.full((Qt.shape[0], k), -1, dtype=np.int64)
         # pad to k columns for even merge
         if d_np.shape[1] < k:
             pad = k - d_np.shape[1]

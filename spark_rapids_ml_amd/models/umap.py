@@ -266,7 +266,9 @@ class UMAP(_UMAPParams, Estimator):
             emb = self._fit_single(X_fit, ctx.device, y_fit)
         else:
             emb = np.zeros((X_fit.shape[0], int(self.getOrDefault("n_components"))), dtype=np.float32)
-        emb_t = torch.from_numpy(emb).to(comm._coll_device())
+        # spectral init can leave emb F-ordered; the broadcast ships the raw
+        # buffer, so receivers would see a transposed layout — force C-order
+        emb_t = torch.from_numpy(np.ascontiguousarray(emb)).to(comm._coll_device())
         comm.broadcast(emb_t, src=0)
         emb = as_numpy(emb_t)
 

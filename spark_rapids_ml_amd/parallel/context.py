@@ -63,7 +63,14 @@ class Comm:
         over the xGMI mesh (SURVEY.md §5 comm backend)."""
         if self.initialized:
             ops = {"sum": dist.ReduceOp.SUM, "max": dist.ReduceOp.MAX, "min": dist.ReduceOp.MIN}
-            dist.all_reduce(t, op=ops[op])
+            if t.is_contiguous():
+                dist.all_reduce(t, op=ops[op])
+            else:
+                # collectives ship the raw buffer; a strided view would be
+                # reduced element-order-mismatched across ranks
+                ct = t.contiguous()
+                dist.all_reduce(ct, op=ops[op])
+                t.copy_(ct)
         return t
 
     def allreduce_t(self, t: torch.Tensor, op: str = "sum") -> torch.Tensor:
@@ -121,7 +128,12 @@ class Comm:
 
     def broadcast(self, t: torch.Tensor, src: int = 0) -> torch.Tensor:
         if self.initialized:
-            dist.broadcast(t, src=src)
+            if t.is_contiguous():
+                dist.broadcast(t, src=src)
+            else:
+                ct = t.contiguous()
+                dist.broadcast(ct, src=src)
+                t.copy_(ct)
         return t
 
     def broadcast_obj(self, obj: Any, src: int = 0) -> Any:

@@ -57,9 +57,59 @@ class NoOpTransformer(Transformer):
         return df
 
 
+def _save_stages(path: str, stages: List[Any]) -> None:
+    import json
+    import os
+
+    from .parallel.context import get_comm
+
+    comm = get_comm()
+    meta: List[Any] = []
+    for i, st in enumerate(stages):
+        cls_name = f"{type(st).__module__}.{type(st).__qualname__}"
+        if isinstance(st, (Estimator, Model)):
+            st.save(os.path.join(path, f"stage_{i}"))
+            meta.append({"class": cls_name, "kind": "dir"})
+        else:
+            params = {
+                p.name: st.getOrDefault(p.name)
+                for p in st.params
+                if st.isSet(p.name) or st.hasDefault(p.name)
+            }
+            meta.append({"class": cls_name, "kind": "params", "params": params})
+    if comm.rank == 0:
+        os.makedirs(path, exist_ok=True)
+        with open(os.path.join(path, "pipeline_metadata.json"), "w") as f:
+            json.dump(meta, f)
+    comm.barrier()
+
+
+def _load_stages(path: str) -> List[Any]:
+    import importlib
+    import json
+    import os
+
+    with open(os.path.join(path, "pipeline_metadata.json")) as f:
+        meta = json.load(f)
+    stages: List[Any] = []
+    for i, ent in enumerate(meta):
+        mod_name, cls_name = ent["class"].rsplit(".", 1)
+        st_cls = getattr(importlib.import_module(mod_name), cls_name)
+        if ent["kind"] == "dir":
+            stages.append(st_cls.load(os.path.join(path, f"stage_{i}")))
+        else:
+            st = st_cls()
+            for k, v in ent["params"].items():
+                st._set(**{k: v})
+            stages.append(st)
+    return stages
+
+
 class Pipeline(Params):
     """Sequential stages with the assembler bypass (reference Pipeline,
-    pipeline.py:59+)."""
+    pipeline.py:59+). save/load mirror pyspark Pipeline persistence:
+    estimator/model stages persist via their own writers, plain
+    transformers via a params JSON."""
 
     def __init__(self, stages: Optional[List[Any]] = None) -> None:
         super().__init__()
@@ -109,6 +159,13 @@ class Pipeline(Params):
                 raise TypeError(f"stage {st!r} is not an Estimator/Transformer")
         return PipelineModel(fitted)
 
+    def save(self, path: str) -> None:
+        _save_stages(path, self._stages)
+
+    @classmethod
+    def load(cls, path: str) -> "Pipeline":
+        return cls(_load_stages(path))
+
 
 class PipelineModel(Params):
     def __init__(self, stages: List[Any]) -> None:
@@ -120,3 +177,10 @@ class PipelineModel(Params):
         for st in self.stages:
             cur = st.transform(cur)
         return cur
+
+    def save(self, path: str) -> None:
+        _save_stages(path, self.stages)
+
+    @classmethod
+    def load(cls, path: str) -> "PipelineModel":
+        return cls(_load_stages(path))

@@ -76,3 +76,45 @@ def test_cagra_recall():
     sk = SkNN(n_neighbors=10).fit(X)
     _, sk_idx = sk.kneighbors(Q)
     assert _recall(idx, sk_idx) > 0.6
+
+
+def _dist_ann(algo):
+    from spark_rapids_ml_amd.parallel.context import get_comm
+
+    comm = get_comm()
+    rng = np.random.default_rng(0)
+    X = rng.normal(size=(600, 16)).astype(np.float32)
+    Q = rng.normal(size=(30, 16)).astype(np.float32)
+    ids = np.arange(600, dtype=np.int64)
+    df = DataFrame({
+        "features": X[comm.rank :: comm.world_size],
+        "id": ids[comm.rank :: comm.world_size],
+    })
+    qdf = DataFrame({"features": Q[comm.rank :: comm.world_size]})
+    model = ApproximateNearestNeighbors(
+        k=8, algorithm=algo, algoParams={"nlist": 8, "nprobe": 8}, idCol="id"
+    ).fit(df)
+    _, _, knn_df = model.kneighbors(qdf)
+    return np.asarray(knn_df["indices"]), np.asarray(knn_df["distances"])
+
+
+def test_ann_distributed_recall():
+    """Each rank searches its local index over the replicated queries and the
+    partial top-k merge assembles global neighbors (reference ANN is
+    comms-free per partition + Spark groupBy merge, knn.py:1282-1322)."""
+    from sklearn.neighbors import NearestNeighbors as SkNN
+
+    from tests.dist_utils import run_distributed
+
+    results = run_distributed(_dist_ann, world_size=2, args=("ivfflat",))
+    rng = np.random.default_rng(0)
+    X = rng.normal(size=(600, 16)).astype(np.float32)
+    Q = rng.normal(size=(30, 16)).astype(np.float32)
+    _, sk_idx = SkNN(n_neighbors=8).fit(X).kneighbors(Q)
+    hits = total = 0
+    for r, (idx, _d) in enumerate(results):
+        truth = sk_idx[r::2]
+        for row_pred, row_true in zip(idx, truth):
+            hits += len(set(row_pred) & set(row_true))
+            total += len(row_true)
+    assert hits / total > 0.85, f"recall {hits/total}"

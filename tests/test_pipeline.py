@@ -5,7 +5,7 @@ import numpy as np
 import pytest
 from sklearn.datasets import make_classification
 
-from spark_rapids_ml_amd import LogisticRegression
+from spark_rapids_ml_amd import KMeans, LogisticRegression
 from spark_rapids_ml_amd.data import DataFrame
 from spark_rapids_ml_amd.pipeline import NoOpTransformer, Pipeline, VectorAssembler
 
@@ -45,3 +45,26 @@ def test_pipeline_without_assembler():
     model = Pipeline(stages=[lr]).fit(df)
     out = model.transform(df)
     assert (np.asarray(out["prediction"]) == y).mean() > 0.65
+
+
+def test_pipeline_persistence(tmp_model_path):
+    import os
+
+    from spark_rapids_ml_amd.pipeline import PipelineModel
+
+    rng = np.random.default_rng(0)
+    df = DataFrame({
+        "a": rng.normal(size=200).astype(np.float32),
+        "b": rng.normal(size=200).astype(np.float32),
+    })
+    pipe = Pipeline([VectorAssembler(["a", "b"]), KMeans(k=3, maxIter=10, seed=1)])
+    pipe.save(os.path.join(tmp_model_path, "pipe"))
+    loaded_pipe = Pipeline.load(os.path.join(tmp_model_path, "pipe"))
+    assert loaded_pipe.getStages()[0].getInputCols() == ["a", "b"]
+    assert loaded_pipe.getStages()[1].getOrDefault("k") == 3
+
+    pm = pipe.fit(df)
+    pm.save(os.path.join(tmp_model_path, "pm"))
+    loaded = PipelineModel.load(os.path.join(tmp_model_path, "pm"))
+    out = loaded.transform(df)
+    assert np.array_equal(np.asarray(out["prediction"]), np.asarray(pm.transform(df)["prediction"]))

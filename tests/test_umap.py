@@ -72,3 +72,26 @@ def test_umap_model_persistence(tmp_model_path):
     assert np.allclose(loaded.embedding, model.embedding)
     out = loaded.transform(DataFrame.from_numpy(X[:40]))
     assert np.asarray(out["embedding"]).shape == (40, 2)
+
+
+def _dist_umap(_):
+    from spark_rapids_ml_amd.parallel.context import get_comm
+    from tests.test_umap import _data
+
+    comm = get_comm()
+    X, _y = _data(n=400)
+    shard = X[comm.rank :: comm.world_size]
+    model = UMAP(n_neighbors=10, n_epochs=30, random_state=1).fit(DataFrame.from_numpy(shard))
+    out = model.transform(DataFrame.from_numpy(shard[:20]))
+    return model.embedding, np.asarray(out["embedding"])
+
+
+def test_umap_distributed_model_replicated():
+    from tests.dist_utils import run_distributed
+
+    results = run_distributed(_dist_umap, world_size=2, args=(None,))
+    emb0, out0 = results[0]
+    emb1, out1 = results[1]
+    # fit on gathered data, broadcast: both ranks hold the identical model
+    assert np.allclose(emb0, emb1)
+    assert out0.shape == (20, 2) and out1.shape == (20, 2)

@@ -44,3 +44,36 @@ def _dist_empty_shard(algo: str):
 def test_empty_shard_rank(algo):
     results = run_distributed(_dist_empty_shard, world_size=2, args=(algo,))
     assert np.allclose(results[0], results[1], atol=1e-5)
+
+
+def _dist_knn_empty_items(_):
+    from spark_rapids_ml_amd import NearestNeighbors
+    from spark_rapids_ml_amd.parallel.context import get_comm
+
+    comm = get_comm()
+    rng = np.random.default_rng(0)
+    X = rng.normal(size=(100, 8)).astype(np.float32)
+    Q = rng.normal(size=(20, 8)).astype(np.float32)
+    ids = np.arange(100, dtype=np.int64)
+    # rank 1 holds NO items but does hold queries
+    if comm.rank == 0:
+        df = DataFrame({"features": X, "id": ids})
+    else:
+        df = DataFrame({"features": X[:0], "id": ids[:0]})
+    qdf = DataFrame({"features": Q[comm.rank :: comm.world_size]})
+    model = NearestNeighbors(k=4, idCol="id").fit(df)
+    _, _, knn_df = model.kneighbors(qdf)
+    return np.asarray(knn_df["indices"]), np.asarray(knn_df["distances"])
+
+
+def test_knn_empty_item_shard():
+    from sklearn.neighbors import NearestNeighbors as SkNN
+
+    results = run_distributed(_dist_knn_empty_items, world_size=2, args=(None,))
+    rng = np.random.default_rng(0)
+    X = rng.normal(size=(100, 8)).astype(np.float32)
+    Q = rng.normal(size=(20, 8)).astype(np.float32)
+    sk_dist, sk_idx = SkNN(n_neighbors=4).fit(X).kneighbors(Q)
+    for r, (idx, dist) in enumerate(results):
+        assert np.array_equal(idx, sk_idx[r::2])
+        assert np.allclose(dist, sk_dist[r::2], atol=1e-4)
