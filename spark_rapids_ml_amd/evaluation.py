@@ -94,9 +94,10 @@ class MulticlassClassificationEvaluator(Evaluator, HasLabelCol, HasPredictionCol
         lab = np.asarray(df[self.getOrDefault("labelCol")], dtype=np.float64)
         pred = np.asarray(df[self.getOrDefault("predictionCol")], dtype=np.float64)
         conf: Dict = {}
-        for l, p in zip(lab, pred):
-            key = (float(l), float(p))
-            conf[key] = conf.get(key, 0.0) + 1.0
+        if len(lab):
+            pairs = np.stack([lab, pred], axis=1)
+            uniq, counts = np.unique(pairs, axis=0, return_counts=True)
+            conf = {(float(a), float(b)): float(c) for (a, b), c in zip(uniq, counts)}
         log_loss = None
         if self.getOrDefault("metricName") == "logLoss":
             probs = np.asarray(df[self.getOrDefault("probabilityCol")], dtype=np.float64)
