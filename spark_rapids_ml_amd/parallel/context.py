@@ -205,8 +205,12 @@ def get_comm() -> Comm:
 
 
 def shutdown_comm(abort: bool = False) -> None:
-    """Tear down the process group (reference cuml_context.py:158-175:
-    destroy on clean exit, abort on exception to unblock surviving ranks)."""
+    """Tear down the process group (reference cuml_context.py:158-175 destroys
+    on clean exit and aborts on exception to unblock surviving ranks).
+    torch.distributed exposes one teardown — destroy_process_group — which on
+    an exception path has abort semantics for peers (their next collective
+    fails fast instead of blocking on a dead rank); the flag is accepted for
+    reference-API parity."""
     global _COMM
     if dist.is_available() and dist.is_initialized():
         dist.destroy_process_group()
