@@ -37,6 +37,7 @@ from ..params import (
     HasRawPredictionCol,
     Param,
     TypeConverters,
+    HasWeightCol,
 )
 from ..ops.glm import logistic_grad_loss
 from ..ops.lbfgs import lbfgs
@@ -52,6 +53,7 @@ class _LogisticRegressionParams(
     HasProbabilityCol,
     HasRawPredictionCol,
     HasEnableSparseDataOptim,
+    HasWeightCol,
 ):
     regParam = Param("logreg", "regParam", "regularization λ.", TypeConverters.toFloat)
     elasticNetParam = Param(
@@ -415,6 +417,36 @@ class LogisticRegressionModel(_LogisticRegressionParams, Model):
     @property
     def numFeatures(self) -> int:
         return self._model_attributes["coef_"].shape[1]
+
+    @property
+    def coef_(self) -> np.ndarray:
+        """cuml-style coefficient matrix (reference exposes the native model
+        attributes alongside the Spark API)."""
+        return self._model_attributes["coef_"]
+
+    @property
+    def intercept_(self) -> np.ndarray:
+        return self._model_attributes["intercept_"]
+
+    @property
+    def n_cols(self) -> int:
+        return self.numFeatures
+
+    def predict(self, value) -> float:
+        """Single-vector prediction (pyspark LogisticRegressionModel.predict)."""
+        out = self._transform_array(np.asarray(value, dtype=np.float64).reshape(1, -1))
+        return float(out[self.getOrDefault("predictionCol")][0])
+
+    def predictRaw(self, value) -> np.ndarray:
+        out = self._transform_array(np.asarray(value, dtype=np.float64).reshape(1, -1))
+        return np.asarray(out[self.getOrDefault("rawPredictionCol")][0])
+
+    def predictProbability(self, value) -> np.ndarray:
+        out = self._transform_array(np.asarray(value, dtype=np.float64).reshape(1, -1))
+        return np.asarray(out[self.getOrDefault("probabilityCol")][0])
+
+    def setRawPredictionCol(self, value: str) -> "LogisticRegressionModel":
+        return self._set_params(rawPredictionCol=value)
 
     def setFeaturesCol(self, value) -> "LogisticRegressionModel":
         if isinstance(value, (list, tuple)):

@@ -32,6 +32,7 @@ from ..params import (
     Param,
     Params,
     TypeConverters,
+    HasWeightCol,
 )
 from ..data import to_device_tensor
 from ..ops import kmeans_assign_reduce, kmeans_predict
@@ -39,7 +40,7 @@ from ..ops.torch_ref import pairwise_sq_dists
 from ..utils import as_numpy, get_logger
 
 
-class _KMeansParams(HasFeaturesCol, HasFeaturesCols, HasPredictionCol):
+class _KMeansParams(HasFeaturesCol, HasFeaturesCols, HasPredictionCol, HasWeightCol):
     """Spark KMeans params + native mapping (reference KMeansClass,
     clustering.py:86-125)."""
 

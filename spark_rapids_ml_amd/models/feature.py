@@ -167,8 +167,25 @@ class PCAModel(_PCAParams, Model):
         return self._model_attributes["explained_variance_ratio_"]
 
     @property
+    def explained_variance_(self) -> np.ndarray:
+        return self._model_attributes["explained_variance_"]
+
+    @property
+    def explained_variance_ratio_(self) -> np.ndarray:
+        return self._model_attributes["explained_variance_ratio_"]
+
+    @property
+    def singular_values_(self) -> np.ndarray:
+        return self._model_attributes["singular_values_"]
+
+    @property
     def mean_(self) -> np.ndarray:
         return self._model_attributes["mean_"]
+
+    @property
+    def mean(self) -> list:
+        """Column means as a list (reference PCAModel.mean, feature.py:291)."""
+        return self._model_attributes["mean_"].tolist()
 
     def setInputCol(self, value) -> "PCAModel":
         if isinstance(value, (list, tuple)):

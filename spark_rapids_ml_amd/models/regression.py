@@ -40,13 +40,14 @@ from ..params import (
     HasPredictionCol,
     Param,
     TypeConverters,
+    HasWeightCol,
 )
 from ..ops import gram
 from ..utils import as_numpy
 from .tree import _RandomForestEstimator, _RandomForestModel
 
 
-class _LinearRegressionParams(HasFeaturesCol, HasFeaturesCols, HasLabelCol, HasPredictionCol):
+class _LinearRegressionParams(HasFeaturesCol, HasFeaturesCols, HasLabelCol, HasPredictionCol, HasWeightCol):
     regParam = Param("linreg", "regParam", "regularization strength λ.", TypeConverters.toFloat)
     elasticNetParam = Param(
         "linreg", "elasticNetParam", "L1 ratio α in [0,1].", TypeConverters.toFloat

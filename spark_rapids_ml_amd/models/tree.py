@@ -34,11 +34,12 @@ from ..params import (
     HasPredictionCol,
     Param,
     TypeConverters,
+    HasWeightCol,
 )
 from ..utils import as_numpy, get_logger
 
 
-class _RandomForestParams(HasFeaturesCol, HasFeaturesCols, HasLabelCol, HasPredictionCol):
+class _RandomForestParams(HasFeaturesCol, HasFeaturesCols, HasLabelCol, HasPredictionCol, HasWeightCol):
     numTrees = Param("rf", "numTrees", "number of trees.", TypeConverters.toInt)
     maxDepth = Param("rf", "maxDepth", "max tree depth.", TypeConverters.toInt)
     maxBins = Param("rf", "maxBins", "max histogram bins (<=256).", TypeConverters.toInt)
@@ -56,6 +57,7 @@ class _RandomForestParams(HasFeaturesCol, HasFeaturesCols, HasLabelCol, HasPredi
     seed = Param("rf", "seed", "random seed.", TypeConverters.toInt)
     bootstrap = Param("rf", "bootstrap", "bootstrap rows.", TypeConverters.toBoolean)
     impurity = Param("rf", "impurity", "gini|entropy|variance.", TypeConverters.toString)
+    leafCol = Param("rf", "leafCol", "leaf index column (unsupported).", TypeConverters.toString)
 
     def __init__(self, **kwargs: Any) -> None:
         super().__init__(**kwargs)
@@ -85,6 +87,8 @@ class _RandomForestParams(HasFeaturesCol, HasFeaturesCols, HasLabelCol, HasPredi
             "seed": "random_state",
             "bootstrap": "bootstrap",
             "impurity": "split_criterion",
+            "weightCol": None,
+            "leafCol": None,
         }
 
     @classmethod
@@ -103,6 +107,9 @@ class _RandomForestParams(HasFeaturesCol, HasFeaturesCols, HasLabelCol, HasPredi
             "max_batch_size": 4096,
             "verbose": False,
         }
+
+    def setLeafCol(self, value: str) -> "_RandomForestParams":
+        raise ValueError("'leafCol' is not supported (reference tree.py:110 parity).")
 
     def getNumTrees(self) -> int:
         return self.getOrDefault("numTrees")
@@ -842,6 +849,25 @@ class _RandomForestModel(_RandomForestParams, Model):
             else:
                 acc += v[:, 0]
         return acc
+
+    def predict(self, value) -> float:
+        """Single-vector prediction (pyspark RandomForest*Model.predict)."""
+        out = self._transform_array(np.asarray(value, dtype=np.float32).reshape(1, -1))
+        if isinstance(out, dict):
+            return float(out[self.getOrDefault("predictionCol")][0])
+        return float(out[0])
+
+    def predictProbability(self, value) -> np.ndarray:
+        if self._task != "classification":
+            raise AttributeError("predictProbability is classification-only")
+        out = self._transform_array(np.asarray(value, dtype=np.float32).reshape(1, -1))
+        return np.asarray(out[self.getOrDefault("probabilityCol")][0])
+
+    def predictRaw(self, value) -> np.ndarray:
+        if self._task != "classification":
+            raise AttributeError("predictRaw is classification-only")
+        out = self._transform_array(np.asarray(value, dtype=np.float32).reshape(1, -1))
+        return np.asarray(out[self.getOrDefault("rawPredictionCol")][0])
 
     def _transform_array(self, X: Any):
         acc = self._predict_raw(X)

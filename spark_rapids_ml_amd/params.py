@@ -208,6 +208,24 @@ class Params:
             return self._paramMap[param]
         return self._defaultParamMap[param]
 
+    def __getattr__(self, name: str) -> Any:
+        """Auto-generate pyspark-style `getX()` getters for declared params
+        (pyspark writes these out by hand in shared.py; here any param
+        `seed` answers `getSeed()` etc.). Called only when normal attribute
+        lookup fails, so explicit getters/properties win."""
+        if name.startswith("get") and len(name) > 3:
+            pname = name[3].lower() + name[4:]
+            try:
+                bound = object.__getattribute__(self, "_bound_params")
+            except AttributeError:
+                bound = {}
+            target = pname if pname in bound else (name[3:] if name[3:] in bound else None)
+            if target is not None:
+                return lambda: self.getOrDefault(target)
+        raise AttributeError(
+            f"{type(self).__name__!r} object has no attribute {name!r}"
+        )
+
     def _set(self, **kwargs: Any) -> "Params":
         for k, v in kwargs.items():
             p = self.getParam(k)
@@ -364,6 +382,21 @@ class HasRawPredictionCol(Params):
 
     def getRawPredictionCol(self) -> str:
         return self.getOrDefault(self.rawPredictionCol)
+
+
+class HasWeightCol(Params):
+    """Instance weights are NOT supported by the native solvers, matching the
+    reference (weightCol is None-mapped everywhere; its setters raise,
+    e.g. reference clustering.py:357-359). Declared so scripts touching the
+    param get the reference's error, not an unknown-param error."""
+
+    weightCol = Param("shared", "weightCol", "instance weight column (unsupported).", TypeConverters.toString)
+
+    def __init__(self, **kwargs: Any) -> None:
+        super().__init__(**kwargs)
+
+    def setWeightCol(self, value: str) -> "HasWeightCol":
+        raise ValueError("'weightCol' is not supported (reference parity).")
 
 
 class HasOutputCol(Params):

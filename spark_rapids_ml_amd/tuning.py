@@ -87,6 +87,9 @@ class CrossValidator(Params):
     def setNumFolds(self, value: int) -> "CrossValidator":
         return self._set(numFolds=value)  # type: ignore[return-value]
 
+    def setCollectSubModels(self, value: bool) -> "CrossValidator":
+        return self._set(collectSubModels=value)  # type: ignore[return-value]
+
     def getEstimator(self) -> Estimator:
         return self._estimator
 
@@ -117,18 +120,23 @@ class CrossValidator(Params):
         epm = self._est_param_maps
         n_folds = self.getOrDefault("numFolds")
         metrics = np.zeros((len(epm), n_folds))
+        collect = bool(self.getOrDefault("collectSubModels"))
+        sub = [[None] * len(epm) for _ in range(n_folds)] if collect else None
         for fold, (train, test) in enumerate(self._kfold(df)):
             # single data pass over all param maps (reference tuning.py:123-130)
             for idx, model in self._estimator.fitMultiple(train, epm):
                 out = model.transform(test)
                 metrics[idx, fold] = self._evaluator.evaluate(out)
+                if collect:
+                    sub[fold][idx] = model
         avg = metrics.mean(axis=1)
         std = metrics.std(axis=1)
         best = int(np.argmax(avg) if self._evaluator.isLargerBetter() else np.argmin(avg))
         logger.info(f"CV best param map index {best}: avg metric {avg[best]:.6f}")
         best_model = self._estimator.fit(df, epm[best])
         cvm = CrossValidatorModel(
-            bestModel=best_model, avgMetrics=avg.tolist(), stdMetrics=std.tolist()
+            bestModel=best_model, avgMetrics=avg.tolist(), stdMetrics=std.tolist(),
+            subModels=sub,
         )
         self._copyValues(cvm)
         return cvm
@@ -145,9 +153,11 @@ class CrossValidatorModel(Params):
         bestModel: Optional[Model] = None,
         avgMetrics: Optional[List[float]] = None,
         stdMetrics: Optional[List[float]] = None,
+        subModels: Optional[List[List[Model]]] = None,
     ) -> None:
         super().__init__()
         self.bestModel = bestModel
+        self.subModels = subModels
         self.avgMetrics = avgMetrics or []
         self.stdMetrics = stdMetrics or []
 

@@ -71,3 +71,99 @@ def test_linreg_param_mapping():
     assert lr.native_params["alpha"] == 0.5
     assert lr.native_params["l1_ratio"] == 0.3
     assert lr.native_params["max_iter"] == 7
+
+
+def test_auto_generated_getters():
+    from spark_rapids_ml_amd import KMeans, LinearRegression
+
+    km = KMeans(k=7, maxIter=13, seed=5)
+    assert km.getK() == 7
+    assert km.getMaxIter() == 13
+    assert km.getSeed() == 5
+    lr = LinearRegression(regParam=0.25)
+    assert lr.getRegParam() == 0.25
+    with pytest.raises(AttributeError):
+        km.getNoSuchParam()
+
+
+def test_weightcol_unsupported_everywhere():
+    from spark_rapids_ml_amd import (
+        KMeans,
+        LinearRegression,
+        LogisticRegression,
+        RandomForestClassifier,
+    )
+
+    for cls in (KMeans, LinearRegression, LogisticRegression, RandomForestClassifier):
+        inst = cls()
+        assert inst.hasParam("weightCol")
+        with pytest.raises(ValueError):
+            inst.setWeightCol("w")
+    with pytest.raises(ValueError):
+        RandomForestClassifier().setLeafCol("leaf")
+
+
+def test_model_native_attribute_parity():
+    import numpy as np
+    from sklearn.datasets import make_classification
+
+    from spark_rapids_ml_amd import PCA, LogisticRegression
+    from spark_rapids_ml_amd.data import DataFrame
+
+    X, y = make_classification(n_samples=200, n_features=6, random_state=0)
+    X = X.astype(np.float32)
+    pca = PCA(k=3).fit(DataFrame.from_numpy(X))
+    assert pca.singular_values_.shape == (3,)
+    assert pca.explained_variance_ratio_.shape == (3,)
+    assert len(pca.mean) == 6
+    lr = LogisticRegression(maxIter=30).fit(DataFrame.from_numpy(X, y.astype(np.float64)))
+    assert lr.coef_.shape == (1, 6)
+    assert lr.n_cols == 6
+    v = X[0]
+    pred = lr.predict(v)
+    raw = lr.predictRaw(v)
+    prob = lr.predictProbability(v)
+    assert pred in (0.0, 1.0)
+    assert raw.shape == (2,) and prob.shape == (2,)
+    assert abs(prob.sum() - 1.0) < 1e-6
+
+
+def test_cv_collect_sub_models():
+    import numpy as np
+
+    from spark_rapids_ml_amd import LinearRegression
+    from spark_rapids_ml_amd.data import DataFrame
+    from spark_rapids_ml_amd.evaluation import RegressionEvaluator
+    from spark_rapids_ml_amd.tuning import CrossValidator, ParamGridBuilder
+
+    rng = np.random.default_rng(0)
+    X = rng.normal(size=(300, 5))
+    y = X @ rng.normal(size=5) + 0.01 * rng.normal(size=300)
+    df = DataFrame.from_numpy(X, y)
+    lr = LinearRegression()
+    grid = ParamGridBuilder().addGrid(lr.getParam("regParam"), [0.0, 0.1]).build()
+    cv = CrossValidator(lr, grid, RegressionEvaluator(), numFolds=2, collectSubModels=False)
+    cv.setCollectSubModels(True)
+    m = cv.fit(df)
+    assert m.subModels is not None
+    assert len(m.subModels) == 2 and len(m.subModels[0]) == 2
+    assert all(sm is not None for fold in m.subModels for sm in fold)
+
+
+def test_rf_single_vector_predict():
+    import numpy as np
+    from sklearn.datasets import make_classification
+
+    from spark_rapids_ml_amd import RandomForestClassifier
+    from spark_rapids_ml_amd.data import DataFrame
+
+    X, y = make_classification(n_samples=300, n_features=8, random_state=0)
+    X = X.astype(np.float32)
+    model = RandomForestClassifier(numTrees=5, maxDepth=4, seed=1).fit(
+        DataFrame.from_numpy(X, y.astype(np.float64))
+    )
+    v = X[0]
+    assert model.predict(v) in (0.0, 1.0)
+    p = model.predictProbability(v)
+    assert p.shape == (2,) and abs(p.sum() - 1.0) < 1e-6
+    assert model.predictRaw(v).shape == (2,)
