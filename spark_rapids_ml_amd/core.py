@@ -249,6 +249,15 @@ class Model(_NativeParams, HasVerbose):
         out = self.transform(df)
         return [evaluator.evaluate(out)]
 
+    def cpu(self):
+        """CPU-model conversion (reference `cpu()` builds the Spark JVM model
+        via py4j, e.g. feature.py:375-389). There is no JVM here; models run
+        on CPU natively when no GPU is present, so the base conversion is the
+        model itself. Subclasses with an exact scikit-learn counterpart
+        (KMeans, PCA, linear/logistic regression) return a fitted sklearn
+        object instead."""
+        return self
+
     @staticmethod
     def _combine(models: Sequence["Model"]) -> "Model":
         """Combine per-param-map models for single-pass CV evaluation

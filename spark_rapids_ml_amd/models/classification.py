@@ -448,6 +448,46 @@ class LogisticRegressionModel(_LogisticRegressionParams, Model):
     def setRawPredictionCol(self, value: str) -> "LogisticRegressionModel":
         return self._set_params(rawPredictionCol=value)
 
+    def cpu(self):
+        """Fitted sklearn.linear_model.LogisticRegression equivalent."""
+        from sklearn.linear_model import LogisticRegression as SkLogReg
+
+        sk = SkLogReg()
+        sk.coef_ = np.asarray(self._model_attributes["coef_"], dtype=np.float64)
+        sk.intercept_ = np.asarray(self._model_attributes["intercept_"], dtype=np.float64)
+        sk.classes_ = np.asarray(self.classes_)
+        sk.n_features_in_ = sk.coef_.shape[1]
+        return sk
+
+    @property
+    def hasSummary(self) -> bool:
+        """True when the fit recorded a training summary (Spark parity)."""
+        return len(self._model_attributes.get("objective_history_", [])) > 0
+
+    @property
+    def summary(self) -> "LogisticRegressionTrainingSummary":
+        if not self.hasSummary:
+            raise RuntimeError("No training summary available on this model")
+        hist = np.asarray(self._model_attributes["objective_history_"])
+        return LogisticRegressionTrainingSummary(
+            objectiveHistory=hist, totalIterations=len(hist), accuracy=None
+        )
+
+    def evaluate(self, df) -> "LogisticRegressionTrainingSummary":
+        """Classification metrics on a dataset (Spark model.evaluate parity)."""
+        from ..evaluation import MulticlassClassificationEvaluator
+
+        ev = MulticlassClassificationEvaluator(
+            labelCol=self.getOrDefault("labelCol"),
+            predictionCol=self.getOrDefault("predictionCol"),
+            metricName="accuracy",
+        )
+        acc = ev.evaluate(self.transform(df))
+        hist = np.asarray(self._model_attributes.get("objective_history_", []))
+        return LogisticRegressionTrainingSummary(
+            objectiveHistory=hist, totalIterations=len(hist), accuracy=acc
+        )
+
     def setFeaturesCol(self, value) -> "LogisticRegressionModel":
         if isinstance(value, (list, tuple)):
             return self._set_params(featuresCols=list(value))
@@ -551,3 +591,29 @@ class RandomForestClassificationModel(_RFClassifierParams, _RandomForestModel):
     """Fitted RF classification model (reference classification.py:534)."""
 
     _task = "classification"
+
+    @property
+    def hasSummary(self) -> bool:
+        """RF fits keep no training summary (matching pyspark models loaded
+        from storage); use evaluate(df) for dataset metrics."""
+        return False
+
+    def evaluate(self, df):
+        from ..evaluation import MulticlassClassificationEvaluator
+
+        ev = MulticlassClassificationEvaluator(
+            labelCol=self.getOrDefault("labelCol"),
+            predictionCol=self.getOrDefault("predictionCol"),
+            metricName="accuracy",
+        )
+        return ev.evaluate(self.transform(df))
+
+
+class LogisticRegressionTrainingSummary:
+    """Spark LogisticRegressionTrainingSummary equivalent (training objective
+    trace; accuracy filled when produced by model.evaluate)."""
+
+    def __init__(self, objectiveHistory, totalIterations: int, accuracy=None):
+        self.objectiveHistory = objectiveHistory
+        self.totalIterations = totalIterations
+        self.accuracy = accuracy

@@ -385,6 +385,18 @@ class KMeansModel(_KMeansParams, Model):
     def trainingCost(self) -> float:
         return self._model_attributes["inertia_"]
 
+    def cpu(self):
+        """Fitted sklearn.cluster.KMeans with this model's centers
+        (reference cpu() builds the Spark KMeansModel via py4j)."""
+        from sklearn.cluster import KMeans as SkKMeans
+
+        sk = SkKMeans(n_clusters=len(self.cluster_centers_), n_init=1)
+        sk.cluster_centers_ = np.asarray(self.cluster_centers_, dtype=np.float64)
+        sk._n_features_out = sk.cluster_centers_.shape[0]
+        sk._n_threads = 1
+        sk.n_features_in_ = sk.cluster_centers_.shape[1]
+        return sk
+
     def computeCost(self, df) -> float:
         """Sum of squared distances to the nearest center over the
         dataframe (Spark's deprecated-but-present KMeansModel.computeCost;
@@ -442,6 +454,12 @@ class _DBSCANParams(HasFeaturesCol, HasFeaturesCols, HasPredictionCol, HasIDCol)
         "dbscan", "min_samples", "core point neighbor threshold.", TypeConverters.toInt
     )
     metric = Param("dbscan", "metric", "distance metric.", TypeConverters.toString)
+    algorithm = Param(
+        "dbscan", "algorithm",
+        "brute|rbc (reference clustering.py:686-695; rbc is an accelerator "
+        "with identical results — this implementation's batched exact sweep "
+        "serves both).", TypeConverters.toString,
+    )
     max_mbytes_per_batch = Param(
         "dbscan",
         "max_mbytes_per_batch",
@@ -451,7 +469,10 @@ class _DBSCANParams(HasFeaturesCol, HasFeaturesCols, HasPredictionCol, HasIDCol)
 
     def __init__(self, **kwargs: Any) -> None:
         super().__init__(**kwargs)
-        self._setDefault(eps=0.5, min_samples=5, metric="euclidean", max_mbytes_per_batch=None)
+        self._setDefault(
+            eps=0.5, min_samples=5, metric="euclidean",
+            max_mbytes_per_batch=None, algorithm="brute",
+        )
 
     @classmethod
     def _param_mapping(cls) -> Dict[str, Optional[str]]:
@@ -460,6 +481,7 @@ class _DBSCANParams(HasFeaturesCol, HasFeaturesCols, HasPredictionCol, HasIDCol)
             "min_samples": "min_samples",
             "metric": "metric",
             "max_mbytes_per_batch": "max_mbytes_per_batch",
+            "algorithm": "algorithm",
         }
 
     @classmethod
@@ -469,6 +491,7 @@ class _DBSCANParams(HasFeaturesCol, HasFeaturesCols, HasPredictionCol, HasIDCol)
             "min_samples": 5,
             "metric": "euclidean",
             "max_mbytes_per_batch": None,
+            "algorithm": "brute",
             "verbose": False,
         }
 

@@ -187,6 +187,24 @@ class PCAModel(_PCAParams, Model):
         """Column means as a list (reference PCAModel.mean, feature.py:291)."""
         return self._model_attributes["mean_"].tolist()
 
+    def cpu(self):
+        """Fitted sklearn.decomposition.PCA equivalent (reference cpu()
+        builds the Spark PCAModel via py4j, feature.py:375-389)."""
+        from sklearn.decomposition import PCA as SkPCA
+
+        k = self.components_.shape[0]
+        sk = SkPCA(n_components=k)
+        sk.components_ = np.asarray(self.components_, dtype=np.float64)
+        sk.explained_variance_ = np.asarray(self.explained_variance_, dtype=np.float64)
+        sk.explained_variance_ratio_ = np.asarray(self.explained_variance_ratio_, dtype=np.float64)
+        sk.singular_values_ = np.asarray(self.singular_values_, dtype=np.float64)
+        sk.mean_ = np.asarray(self.mean_, dtype=np.float64)
+        sk.n_components_ = k
+        sk.n_features_in_ = sk.components_.shape[1]
+        sk.n_samples_ = int(self._model_attributes.get("n_rows_", 0))
+        sk.noise_variance_ = 0.0
+        return sk
+
     def setInputCol(self, value) -> "PCAModel":
         if isinstance(value, (list, tuple)):
             return self._set_params(featuresCols=list(value))

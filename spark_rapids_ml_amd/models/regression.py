@@ -389,6 +389,42 @@ class LinearRegressionModel(_LinearRegressionParams, Model):
             numInstances=n,
         )
 
+    def cpu(self):
+        """Fitted sklearn.linear_model.LinearRegression equivalent."""
+        from sklearn.linear_model import LinearRegression as SkLR
+
+        sk = SkLR()
+        sk.coef_ = np.asarray(self.coefficients, dtype=np.float64)
+        sk.intercept_ = float(self.intercept)
+        sk.n_features_in_ = len(sk.coef_)
+        return sk
+
+    @property
+    def scale(self) -> float:
+        """Spark LinearRegressionModel.scale — 1.0 except for huber loss,
+        which neither this build nor the reference supports."""
+        return 1.0
+
+    def evaluate(self, df) -> "LinearRegressionTrainingSummary":
+        """Regression metrics on a dataset (Spark model.evaluate parity):
+        transforms locally, merges sufficient statistics across ranks."""
+        from ..evaluation import RegressionEvaluator
+
+        ev = RegressionEvaluator(
+            labelCol=self.getOrDefault("labelCol"),
+            predictionCol=self.getOrDefault("predictionCol"),
+        )
+        from ..parallel.context import get_comm
+
+        out = self.transform(df)
+        mse = ev.setMetricName("mse").evaluate(out)
+        n = int(get_comm().allreduce_scalar(float(len(df))))
+        return LinearRegressionTrainingSummary(
+            meanSquaredError=mse,
+            rootMeanSquaredError=float(np.sqrt(mse)),
+            numInstances=n,
+        )
+
     def setFeaturesCol(self, value) -> "LinearRegressionModel":
         if isinstance(value, (list, tuple)):
             return self._set_params(featuresCols=list(value))

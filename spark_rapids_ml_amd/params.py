@@ -209,19 +209,27 @@ class Params:
         return self._defaultParamMap[param]
 
     def __getattr__(self, name: str) -> Any:
-        """Auto-generate pyspark-style `getX()` getters for declared params
-        (pyspark writes these out by hand in shared.py; here any param
-        `seed` answers `getSeed()` etc.). Called only when normal attribute
-        lookup fails, so explicit getters/properties win."""
-        if name.startswith("get") and len(name) > 3:
-            pname = name[3].lower() + name[4:]
+        """Auto-generate pyspark-style `getX()` getters and `setX(v)` setters
+        for declared params (pyspark writes these out by hand in shared.py;
+        here any param `seed` answers `getSeed()`/`setSeed(v)`, and a
+        snake_case param `min_samples` answers `getMinSamples()`). Called only
+        when normal attribute lookup fails, so explicit getters/setters/
+        properties — including ones that must raise — win."""
+        if (name.startswith("get") or name.startswith("set")) and len(name) > 3:
             try:
                 bound = object.__getattribute__(self, "_bound_params")
             except AttributeError:
                 bound = {}
-            target = pname if pname in bound else (name[3:] if name[3:] in bound else None)
+            camel = name[3].lower() + name[4:]
+            import re as _re
+
+            snake = _re.sub(r"(?<!^)([A-Z])", r"_\1", name[3:]).lower()
+            target = next((c for c in (camel, name[3:], snake) if c in bound), None)
             if target is not None:
-                return lambda: self.getOrDefault(target)
+                if name.startswith("get"):
+                    return lambda: self.getOrDefault(target)
+                setter = getattr(self, "_set_params", None) or self._set
+                return lambda value: setter(**{target: value})
         raise AttributeError(
             f"{type(self).__name__!r} object has no attribute {name!r}"
         )

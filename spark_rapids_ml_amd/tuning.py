@@ -99,6 +99,66 @@ class CrossValidator(Params):
     def getEstimatorParamMaps(self) -> List[Dict[Any, Any]]:
         return self._est_param_maps
 
+    def save(self, path: str) -> None:
+        """Persist estimator, evaluator, grid and CV params (the reference
+        relies on pyspark's CV persistence + a load-time classpath fix,
+        tuning.py:159-186)."""
+        import json
+        import os
+
+        comm = get_comm()
+        self._estimator.save(os.path.join(path, "estimator"))
+        if comm.rank == 0:
+            meta = {
+                "estimator_class": f"{type(self._estimator).__module__}."
+                f"{type(self._estimator).__qualname__}",
+                "evaluator_class": f"{type(self._evaluator).__module__}."
+                f"{type(self._evaluator).__qualname__}",
+                "evaluator_params": {
+                    p.name: self._evaluator.getOrDefault(p.name)
+                    for p in self._evaluator.params
+                    if self._evaluator.isDefined(p.name)
+                },
+                "param_maps": [
+                    {(p.name if isinstance(p, Param) else str(p)): v for p, v in pm.items()}
+                    for pm in self._est_param_maps
+                ],
+                "numFolds": self.getOrDefault("numFolds"),
+                "seed": self.getOrDefault("seed"),
+                "collectSubModels": self.getOrDefault("collectSubModels"),
+            }
+            with open(os.path.join(path, "cv_estimator_metadata.json"), "w") as f:
+                json.dump(meta, f)
+        comm.barrier()
+
+    @classmethod
+    def load(cls, path: str) -> "CrossValidator":
+        import importlib
+        import json
+        import os
+
+        with open(os.path.join(path, "cv_estimator_metadata.json")) as f:
+            meta = json.load(f)
+
+        def _resolve(classpath: str):
+            mod, name = classpath.rsplit(".", 1)
+            return getattr(importlib.import_module(mod), name)
+
+        est = _resolve(meta["estimator_class"]).load(os.path.join(path, "estimator"))
+        ev = _resolve(meta["evaluator_class"])(**meta["evaluator_params"])
+        maps = [
+            {est.getParam(name): v for name, v in pm.items()}
+            for pm in meta["param_maps"]
+        ]
+        return cls(
+            estimator=est,
+            estimatorParamMaps=maps,
+            evaluator=ev,
+            numFolds=meta["numFolds"],
+            seed=meta["seed"],
+            collectSubModels=meta.get("collectSubModels", False),
+        )
+
     def _kfold(self, df: DataFrame) -> List[Tuple[DataFrame, DataFrame]]:
         """Deterministic per-row fold assignment on each rank's local shard
         (Spark assigns by rand(seed) — same semantics, shard-local)."""

@@ -167,3 +167,46 @@ def test_rf_single_vector_predict():
     p = model.predictProbability(v)
     assert p.shape == (2,) and abs(p.sum() - 1.0) < 1e-6
     assert model.predictRaw(v).shape == (2,)
+
+
+def test_model_cpu_conversions():
+    import numpy as np
+    from sklearn.datasets import make_blobs
+
+    from spark_rapids_ml_amd import KMeans, PCA
+    from spark_rapids_ml_amd.data import DataFrame
+
+    X, _ = make_blobs(n_samples=200, n_features=5, centers=3, random_state=0)
+    X = X.astype(np.float32)
+    km = KMeans(k=3, maxIter=20, seed=1).fit(DataFrame.from_numpy(X))
+    sk = km.cpu()
+    ours = np.asarray(km.transform(DataFrame.from_numpy(X))["prediction"])
+    assert np.array_equal(sk.predict(X.astype(np.float64)), ours)
+    pca = PCA(k=2).fit(DataFrame.from_numpy(X))
+    sk_pca = pca.cpu()
+    assert sk_pca.transform(X.astype(np.float64)).shape == (200, 2)
+
+
+def test_model_evaluate_and_summary():
+    import numpy as np
+    from sklearn.datasets import make_classification
+
+    from spark_rapids_ml_amd import LinearRegression, LogisticRegression
+    from spark_rapids_ml_amd.data import DataFrame
+
+    rng = np.random.default_rng(0)
+    X = rng.normal(size=(200, 4))
+    y = X @ rng.normal(size=4) + 0.01 * rng.normal(size=200)
+    df = DataFrame.from_numpy(X, y)
+    lr = LinearRegression().fit(df)
+    s = lr.evaluate(df)
+    assert s.rootMeanSquaredError < 0.1
+    assert s.numInstances == 200
+    assert lr.scale == 1.0
+    Xc, yc = make_classification(n_samples=200, n_features=6, random_state=0)
+    dfc = DataFrame.from_numpy(Xc.astype(np.float32), yc.astype(np.float64))
+    lg = LogisticRegression(maxIter=30).fit(dfc)
+    assert lg.hasSummary
+    assert lg.summary.totalIterations >= 1
+    ev = lg.evaluate(dfc)
+    assert ev.accuracy > 0.8

@@ -123,3 +123,18 @@ def test_cv_distributed():
         assert best == 0.0
     # ranks agree on metrics (collectives aligned)
     assert np.allclose(results[0][0], results[1][0])
+
+
+def test_cross_validator_estimator_persistence(tmp_model_path):
+    import os
+
+    lr = LinearRegression()
+    grid = ParamGridBuilder().addGrid(lr.getParam("regParam"), [0.0, 0.1]).build()
+    cv = CrossValidator(lr, grid, RegressionEvaluator(metricName="mae"), numFolds=2, seed=7)
+    cv.save(os.path.join(tmp_model_path, "cv"))
+    loaded = CrossValidator.load(os.path.join(tmp_model_path, "cv"))
+    assert loaded.getOrDefault("numFolds") == 2
+    assert loaded.getEvaluator().getOrDefault("metricName") == "mae"
+    maps = loaded.getEstimatorParamMaps()
+    assert len(maps) == 2
+    assert sorted(v for pm in maps for v in pm.values()) == [0.0, 0.1]
