@@ -198,3 +198,20 @@ def test_kmeans_gpu_float64_inputs():
         DataFrame.from_numpy(X.astype(np.float64))
     )
     assert model.cluster_centers_.dtype == np.float64
+
+
+def test_single_vector_predict_gpu():
+    """predict/predictRaw/predictProbability route 1-row inputs through the
+    device transform path (HIP kernels must handle n=1)."""
+    X, y = make_classification(n_samples=400, n_features=16, random_state=0)
+    X = X.astype(np.float32)
+    df = DataFrame.from_numpy(X, y.astype(np.float64))
+    lg = LogisticRegression(maxIter=30).fit(df)
+    v = X[0]
+    assert lg.predict(v) in (0.0, 1.0)
+    assert abs(lg.predictProbability(v).sum() - 1.0) < 1e-6
+    rf = RandomForestClassifier(numTrees=5, maxDepth=4, seed=1).fit(df)
+    assert rf.predict(v) in (0.0, 1.0)
+    Xb, _ = make_blobs(n_samples=300, n_features=8, centers=3, random_state=0)
+    km = KMeans(k=3, maxIter=15, seed=1).fit(DataFrame.from_numpy(Xb.astype(np.float32)))
+    assert 0 <= km.predict(Xb[0].astype(np.float32)) < 3
