@@ -201,6 +201,8 @@ __global__ __launch_bounds__(256) void kmeans_assign_kernel(
   }
 }
 
+
+
 // ---------------------------------------------------------------------------
 // kmeans_assign_glds: same contract as kmeans_assign_kernel, staging via
 // async global_load_lds DMA (16B) into a 16B-XOR-swizzled linear LDS image.
@@ -228,6 +230,204 @@ __device__ __forceinline__ int kg_swz(int i, int kd) {
   // spread over 8 bank quads (<=4-way conflict) while glds stays lane-linear
   return i * KG_BK + ((((kd >> 2) ^ (i & 7)) << 2) | (kd & 3));
 }
+
+// ---------------------------------------------------------------------------
+// kmeans_assign_256: 256x256 tile via glds, one 512-thread block (8 waves).
+// The 128x128 kernel is HBM-traffic-bound at scale: X re-streams once per
+// centroid tile (k/128 passes) and C once per row tile (n/128 passes) --
+// ~190 GB for 1M x 3000 x k=1000 vs ~39 ms of pure MFMA work. Doubling both
+// tile dims halves BOTH traffic terms (~95 GB). Staging goes through
+// global_load_lds (zero staging registers -- the register-staged 256 variant
+// spilled 384 VGPRs) with the same double-buffered vmcnt state machine as
+// kmeans_assign_glds_kernel; 4 x 256x32 tile images + best[] = 131 KB LDS,
+// one block per CU, still 2 waves/SIMD.
+// ---------------------------------------------------------------------------
+
+constexpr int KG2_BM = 256;
+constexpr int KG2_BN = 256;
+constexpr int KG2_X0 = 0;        // [256*32] floats
+constexpr int KG2_X1 = 8192;
+constexpr int KG2_C0 = 16384;
+constexpr int KG2_C1 = 24576;
+constexpr int KG2_BEST = 32768;  // u64 best[256] = 512 floats
+constexpr int KG2_SCRATCH = 33280;  // 8 doubles = 16 floats
+constexpr int KG2_TOTAL = 33296;
+
+__global__ __launch_bounds__(512) void kmeans_assign_256_kernel(
+    const float* __restrict__ X, const float* __restrict__ C,
+    const float* __restrict__ x_sq, const float* __restrict__ c_sq,
+    int n, int d, int k,
+    int32_t* __restrict__ labels, float* __restrict__ min_dists,
+    double* __restrict__ inertia) {
+  __shared__ __attribute__((aligned(16))) float smem[KG2_TOTAL];
+
+  const int i0 = blockIdx.x * KG2_BM;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = wave >> 1;  // 0..3 -> 64-row band
+  const int wc = wave & 1;   // 0..1 -> 128-col band
+
+  unsigned long long* best = reinterpret_cast<unsigned long long*>(&smem[KG2_BEST]);
+  for (int i = tid; i < KG2_BM; i += blockDim.x) best[i] = ~0ULL;
+
+  const bool full_rows = (i0 + KG2_BM <= n);
+  const int nsteps = (d + KG_BK - 1) / KG_BK;
+  bool full_rows_cols = full_rows;
+
+  auto pair_full = [&](int step) {
+    return full_rows_cols && (step + 1) * KG_BK <= d;
+  };
+
+  // per-wave glds issue: 4 instructions per operand per tile (8 waves x 4
+  // instr x 8 rows x 32 cols = the full 256x32 image)
+  auto issue_glds = [&](const float* __restrict__ src, int base_row, int d0,
+                        int lds_off) {
+#pragma unroll
+    for (int qq = 0; qq < 4; ++qq) {
+      const int q = wave * 4 + qq;
+      const int i = q * 8 + (lane >> 3);
+      const int g = (lane & 7) ^ (i & 7);
+      const float* gp = src + (int64_t)(base_row + i) * d + d0 + (g << 2);
+      auto lp = (__attribute__((address_space(3))) uint32_t*)(&smem[lds_off + q * 256]);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)gp, lp, 16, 0, 0);
+    }
+  };
+  auto issue_scalar = [&](const float* __restrict__ src, int base_row, int lim,
+                          int d0, int lds_off) {
+    for (int e = tid; e < KG2_BM * KG_BK; e += 512) {
+      int i = e >> 5, kd = e & 31;
+      int gr = base_row + i, gd = d0 + kd;
+      smem[lds_off + kg_swz(i, kd)] =
+          (gr < lim && gd < d) ? src[(int64_t)gr * d + gd] : 0.0f;
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  };
+  auto issue_tile = [&](int step, int xoff, int coff, int j0) {
+    const int d0 = step * KG_BK;
+    if (pair_full(step)) {
+      issue_glds(X, i0, d0, xoff);
+      issue_glds(C, j0, d0, coff);
+    } else {
+      issue_scalar(X, i0, n, d0, xoff);
+      issue_scalar(C, j0, k, d0, coff);
+    }
+  };
+
+  for (int j0 = 0; j0 < k; j0 += KG2_BN) {
+    const bool full_cols = (j0 + KG2_BN <= k);
+    full_rows_cols = full_rows && full_cols;
+    f32x16 acc[2][4];
+#pragma unroll
+    for (int m = 0; m < 2; ++m)
+#pragma unroll
+      for (int nn = 0; nn < 4; ++nn)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) acc[m][nn][r] = 0.0f;
+
+    issue_tile(0, KG2_X0, KG2_C0, j0);
+    if (nsteps > 1) issue_tile(1, KG2_X1, KG2_C1, j0);
+    if (pair_full(0)) {
+      if (nsteps > 1 && pair_full(1)) {
+        asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+      } else {
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      }
+    }
+    __builtin_amdgcn_s_barrier();
+
+    for (int step = 0; step < nsteps; ++step) {
+      const int xb = (step & 1) ? KG2_X1 : KG2_X0;
+      const int cb = (step & 1) ? KG2_C1 : KG2_C0;
+      const int xi = wr * 64 + (lane & 31);
+      const int ci = wc * 128 + (lane & 31);
+#pragma unroll 8
+      for (int kk = 0; kk < KG_BK / 2; ++kk) {
+        const int kd = 2 * kk + (lane >> 5);
+        float a0 = smem[xb + kg_swz(xi, kd)];
+        float a1 = smem[xb + kg_swz(xi + 32, kd)];
+        float b0 = smem[cb + kg_swz(ci, kd)];
+        float b1 = smem[cb + kg_swz(ci + 32, kd)];
+        float b2 = smem[cb + kg_swz(ci + 64, kd)];
+        float b3 = smem[cb + kg_swz(ci + 96, kd)];
+        acc[0][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc[0][0], 0, 0, 0);
+        acc[0][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc[0][1], 0, 0, 0);
+        acc[0][2] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b2, acc[0][2], 0, 0, 0);
+        acc[0][3] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b3, acc[0][3], 0, 0, 0);
+        acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc[1][0], 0, 0, 0);
+        acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc[1][1], 0, 0, 0);
+        acc[1][2] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b2, acc[1][2], 0, 0, 0);
+        acc[1][3] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b3, acc[1][3], 0, 0, 0);
+      }
+      __builtin_amdgcn_s_barrier();  // everyone done reading buf[step&1]
+      if (step + 2 < nsteps) {
+        issue_tile(step + 2, xb, cb, j0);
+        if (pair_full(step + 1)) {
+          if (pair_full(step + 2)) {
+            asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+          } else {
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+          }
+        }
+        __builtin_amdgcn_s_barrier();
+      } else if (step + 1 < nsteps) {
+        if (pair_full(step + 1)) {
+          asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        }
+        __builtin_amdgcn_s_barrier();
+      }
+    }
+
+    // epilogue: distances + packed argmin (no glds outstanding)
+#pragma unroll
+    for (int m = 0; m < 2; ++m) {
+#pragma unroll
+      for (int nn = 0; nn < 4; ++nn) {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          int row = wr * 64 + m * 32 + (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
+          int col = wc * 128 + nn * 32 + (lane & 31);
+          int gi = i0 + row, gj = j0 + col;
+          if (gi < n && gj < k) {
+            float dist = x_sq[gi] + c_sq[gj] - 2.0f * acc[m][nn][r];
+            dist = dist < 0.0f ? 0.0f : dist;
+            unsigned long long packed =
+                ((unsigned long long)__float_as_uint(dist) << 32) |
+                (unsigned int)gj;
+            atomicMin(&best[row], packed);
+          }
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  double* block_inertia = reinterpret_cast<double*>(&smem[KG2_SCRATCH]);
+  double partial = 0.0;
+  for (int i = tid; i < KG2_BM; i += blockDim.x) {
+    int gi = i0 + i;
+    if (gi < n) {
+      unsigned long long p = best[i];
+      float dist = __uint_as_float((unsigned int)(p >> 32));
+      labels[gi] = (int32_t)(p & 0xffffffffu);
+      min_dists[gi] = dist;
+      partial += (double)dist;
+    }
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1)
+    partial += __shfl_down(partial, off, 64);
+  if (lane == 0) block_inertia[wave] = partial;
+  __syncthreads();
+  if (tid == 0) {
+    double sum = 0.0;
+#pragma unroll
+    for (int w = 0; w < 8; ++w) sum += block_inertia[w];
+    atomicAdd(inertia, sum);
+  }
+}
+
 
 __global__ __launch_bounds__(256) void kmeans_assign_glds_kernel(
     const float* __restrict__ X, const float* __restrict__ C,
@@ -1354,7 +1554,15 @@ std::vector<torch::Tensor> kmeans_assign(torch::Tensor X, torch::Tensor C,
     // 1-block/CU span 111 ms (single wave/SIMD starves the MFMA pipe).
     // Register staging is the default; glds variants stay selectable.
     static const char* v = getenv("SRML_KMEANS_VARIANT");
-    if (d % 4 == 0 && v && v[0] == 's') {
+    if (v && v[0] == 'b' && d % 4 == 0) {  // big-tile 256x256 glds variant
+      const int grid2 = (int)((n + KG2_BM - 1) / KG2_BM);
+      hipLaunchKernelGGL(kmeans_assign_256_kernel, dim3(grid2), dim3(512), 0,
+                         cur_stream(), X.data_ptr<float>(), C.data_ptr<float>(),
+                         x_sq.data_ptr<float>(), c_sq.data_ptr<float>(), (int)n,
+                         d, k, labels.data_ptr<int32_t>(),
+                         min_dists.data_ptr<float>(),
+                         inertia.data_ptr<double>());
+    } else if (d % 4 == 0 && v && v[0] == 's') {
       const int grid_s = (int)((n + KS_BM - 1) / KS_BM);
       hipLaunchKernelGGL(kmeans_assign_sb3_kernel, dim3(grid_s), dim3(256), 0,
                          cur_stream(), X.data_ptr<float>(), C.data_ptr<float>(),
