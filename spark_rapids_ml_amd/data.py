@@ -175,8 +175,10 @@ class DataFrame:
                 cols[name] = arr.to_numpy(zero_copy_only=False)
         return cls(cols)
 
-    def write_parquet(self, path: str) -> None:
-        """Each rank writes its shard as part-<rank>.parquet."""
+    def write_parquet(self, path: str, row_group_rows: int = 1 << 16) -> None:
+        """Each rank writes its shard as part-<rank>.parquet. row_group_rows
+        sets the row-group granularity (the unit of read_parquet's
+        round-robin sharding)."""
         import pyarrow as pa
         import pyarrow.parquet as pq
 
@@ -200,7 +202,7 @@ class DataFrame:
         pq.write_table(
             table,
             os.path.join(path, f"part-{comm.rank:05d}.parquet"),
-            row_group_size=1 << 16,
+            row_group_size=row_group_rows,
             use_dictionary=False,  # dictionary-encoding floats triples decode time
             compression="snappy",
         )

@@ -60,3 +60,35 @@ def test_to_pandas():
     pdf = df.to_pandas()
     assert len(pdf) == 6
     assert np.allclose(np.stack(pdf["features"].to_list()), X)
+
+
+def _dist_parquet_shard(path):
+    from spark_rapids_ml_amd.data import DataFrame
+    from spark_rapids_ml_amd.parallel.context import get_comm
+
+    comm = get_comm()
+    df = DataFrame.read_parquet(path, vector_cols=["features"])
+    ids = np.asarray(df["id"]) if "id" in df.columns else None
+    return comm.rank, len(df), ids
+
+
+def test_read_parquet_shards_across_ranks(tmp_path):
+    from tests.dist_utils import run_distributed
+
+    from spark_rapids_ml_amd.data import DataFrame
+
+    rng = np.random.default_rng(0)
+    X = rng.normal(size=(5000, 8)).astype(np.float32)
+    ids = np.arange(5000, dtype=np.int64)
+    full = DataFrame({"features": X, "id": ids})
+    out = str(tmp_path / "shard_ds")
+    # row_group_rows small enough to give every rank several groups
+    full.write_parquet(out, row_group_rows=500)
+
+    results = run_distributed(_dist_parquet_shard, world_size=2, args=(out,))
+    counts = {rank: n for rank, n, _ in results}
+    assert sum(counts.values()) == 5000
+    assert min(counts.values()) > 0
+    # shards are disjoint and together cover every row exactly once
+    all_ids = np.concatenate([i for _, _, i in results])
+    assert np.array_equal(np.sort(all_ids), ids)
