@@ -77,3 +77,31 @@ def test_knn_empty_item_shard():
     for r, (idx, dist) in enumerate(results):
         assert np.array_equal(idx, sk_idx[r::2])
         assert np.allclose(dist, sk_dist[r::2], atol=1e-4)
+
+
+def _dist_rf_empty(_):
+    from spark_rapids_ml_amd import RandomForestClassifier
+
+    from spark_rapids_ml_amd.parallel.context import get_comm
+
+    comm = get_comm()
+    rng = np.random.default_rng(0)
+    X = rng.normal(size=(300, 6)).astype(np.float32)
+    y = (X[:, 0] > 0).astype(np.float64)
+    if comm.rank == 0:
+        Xl, yl = X, y
+    else:
+        Xl, yl = X[:0], y[:0]
+    m = RandomForestClassifier(numTrees=4, maxDepth=3, seed=1).fit(DataFrame.from_numpy(Xl, yl))
+    out = m.transform(DataFrame.from_numpy(X))
+    acc = (np.asarray(out["prediction"]) == y).mean()
+    return m.numTrees, acc
+
+
+def test_rf_empty_shard_rank():
+    """Trees assigned to a rank with zero local rows must not break the
+    ensemble merge (an 8-GPU fit on a small dataset hits this)."""
+    results = run_distributed(_dist_rf_empty, world_size=2, args=(None,))
+    for n_trees, acc in results:
+        assert n_trees == 4
+        assert acc > 0.9
