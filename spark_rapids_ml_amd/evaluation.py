@@ -71,10 +71,11 @@ class MulticlassClassificationEvaluator(Evaluator, HasLabelCol, HasPredictionCol
 
     metricName = Param("eval", "metricName", "metric.", TypeConverters.toString)
     probabilityCol = Param("eval", "probabilityCol", "for logLoss.", TypeConverters.toString)
+    metricLabel = Param("eval", "metricLabel", "class for *ByLabel metrics.", TypeConverters.toFloat)
 
     def __init__(self, **kwargs: Any) -> None:
         super().__init__()
-        self._setDefault(metricName="f1", probabilityCol="probability")
+        self._setDefault(metricName="f1", probabilityCol="probability", metricLabel=0.0)
         self._set(**kwargs)
 
     def setMetricName(self, value: str) -> "MulticlassClassificationEvaluator":
@@ -109,7 +110,9 @@ class MulticlassClassificationEvaluator(Evaluator, HasLabelCol, HasPredictionCol
         for items, ll in all_stats:
             m = MulticlassMetrics.from_confusion({tuple(k): v for k, v in items}, ll)
             merged = merged.merge(m)
-        return merged.evaluate(self.getOrDefault("metricName"))
+        return merged.evaluate(
+            self.getOrDefault("metricName"), self.getOrDefault("metricLabel")
+        )
 
 
 class BinaryClassificationEvaluator(Evaluator, HasLabelCol):

@@ -136,8 +136,13 @@ class MulticlassMetrics:
         assert self._log_loss is not None, "log loss was not accumulated"
         return self._log_loss / self._label_count
 
-    def evaluate(self, metric_name: str) -> float:
+    def evaluate(self, metric_name: str, metric_label: float = 0.0) -> float:
         dispatch = {
+            "truePositiveRateByLabel": lambda: self._recall(metric_label),
+            "falsePositiveRateByLabel": lambda: self.false_positive_rate(metric_label),
+            "precisionByLabel": lambda: self._precision(metric_label),
+            "recallByLabel": lambda: self._recall(metric_label),
+            "fMeasureByLabel": lambda: self._f_measure(metric_label),
             "f1": lambda: self.weighted_fmeasure(),
             "accuracy": self.accuracy,
             "weightedPrecision": self.weighted_precision,

@@ -82,3 +82,25 @@ def test_summarizer_buffer_roundtrip():
     back = _SummarizerBuffer.from_list(buf.to_list())
     assert back.total_count == 10
     assert back._curr_mean == [1.0, 2.0]
+
+
+def test_by_label_metrics():
+    from sklearn.metrics import precision_score, recall_score
+
+    from spark_rapids_ml_amd.evaluation import MulticlassClassificationEvaluator
+
+    rng = np.random.default_rng(0)
+    lab = rng.integers(0, 3, size=500).astype(np.float64)
+    pred = lab.copy()
+    flip = rng.random(500) < 0.2
+    pred[flip] = rng.integers(0, 3, size=flip.sum()).astype(np.float64)
+    df = DataFrame({"label": lab, "prediction": pred})
+    for cls in (0.0, 1.0, 2.0):
+        p = MulticlassClassificationEvaluator(
+            metricName="precisionByLabel", metricLabel=cls
+        ).evaluate(df)
+        r = MulticlassClassificationEvaluator(
+            metricName="recallByLabel", metricLabel=cls
+        ).evaluate(df)
+        assert np.isclose(p, precision_score(lab, pred, labels=[cls], average=None)[0])
+        assert np.isclose(r, recall_score(lab, pred, labels=[cls], average=None)[0])
