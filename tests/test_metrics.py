@@ -87,6 +87,7 @@ def test_summarizer_buffer_roundtrip():
 def test_by_label_metrics():
     from sklearn.metrics import precision_score, recall_score
 
+    from spark_rapids_ml_amd.data import DataFrame
     from spark_rapids_ml_amd.evaluation import MulticlassClassificationEvaluator
 
     rng = np.random.default_rng(0)
