@@ -203,6 +203,165 @@ __global__ __launch_bounds__(256) void kmeans_assign_kernel(
 
 
 
+
+
+// ---------------------------------------------------------------------------
+// dbscan_sweep: fused n x n eps-neighborhood pass for DBSCAN, reusing the
+// kmeans_assign tile structure (rows = this rank's slice of the replicated
+// dataset, "centers" = ALL rows, tiled by 128). The torch path materializes
+// [chunk, n] masked int64 label tensors (~8 bytes x n per row per sweep of
+// pure HBM traffic); here the distance never leaves registers and the
+// epilogue reduces straight into a 128-entry LDS accumulator.
+//   mode 0: out[i] = #{ j : d2(i,j) <= eps2 }            (core counting)
+//   mode 1: out[i] = min{ labels[j] : core[j], d2 <= eps2 }  (label sweep /
+//           border assignment; 0x7fffffff when no core neighbor)
+// One kernel serves core detection, the min-label propagation sweeps and
+// the border pass (reference DBSCANMG: adjacency + BFS inside cuML,
+// SURVEY.md §2.3b).
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(256) void dbscan_sweep_kernel(
+    const float* __restrict__ X,       // [n,d] replicated rows
+    const float* __restrict__ x_sq,    // [n]
+    int n, int d,
+    int row0, int n_rows,              // this rank's slice [row0, row0+n_rows)
+    float eps2, int mode,
+    const uint8_t* __restrict__ core,  // [n] (mode 1)
+    const int32_t* __restrict__ labels,// [n] (mode 1)
+    int32_t* __restrict__ out) {       // [n_rows]
+  __shared__ float lds_x[KM_BK][KM_BM + 1];
+  __shared__ float lds_c[KM_BK][KM_BN + 1];
+  __shared__ int acc_row[KM_BM];
+
+  const int i0 = row0 + blockIdx.x * KM_BM;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = wave >> 1;
+  const int wc = wave & 1;
+  const int row_lim = row0 + n_rows;
+
+  for (int i = tid; i < KM_BM; i += blockDim.x)
+    acc_row[i] = (mode == 0) ? 0 : 0x7fffffff;
+
+  const bool full_rows = (i0 + KM_BM <= row_lim);
+  const int nsteps = (d + KM_BK - 1) / KM_BK;
+  float rx[KM_LD], rc[KM_LD];
+
+  for (int j0 = 0; j0 < n; j0 += KM_BN) {
+    const bool full_cols = (j0 + KM_BN <= n);
+    f32x16 acc[2][2];
+#pragma unroll
+    for (int m = 0; m < 2; ++m)
+#pragma unroll
+      for (int nn = 0; nn < 2; ++nn)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) acc[m][nn][r] = 0.0f;
+
+#define DB_LOAD_T(dst, base_row, lim, d0)                                      \
+  do {                                                                         \
+    if (full_rows && full_cols && (d0) + KM_BK <= d) {                         \
+      _Pragma("unroll") for (int q = 0; q < KM_LD; ++q) {                      \
+        int e = q * 256 + tid;                                                 \
+        dst[q] = X[(int64_t)((base_row) + (e >> 6)) * d + (d0) + (e & 63)];    \
+      }                                                                        \
+    } else {                                                                   \
+      _Pragma("unroll") for (int q = 0; q < KM_LD; ++q) {                      \
+        int e = q * 256 + tid;                                                 \
+        int gr = (base_row) + (e >> 6);                                        \
+        int gd = (d0) + (e & 63);                                              \
+        dst[q] = (gr < (lim) && gd < d) ? X[(int64_t)gr * d + gd] : 0.0f;      \
+      }                                                                        \
+    }                                                                          \
+  } while (0)
+
+#define DB_WRITE()                                                             \
+  do {                                                                         \
+    _Pragma("unroll") for (int q = 0; q < KM_LD; ++q) {                        \
+      int e = q * 256 + tid;                                                   \
+      lds_x[e & 63][e >> 6] = rx[q];                                           \
+      lds_c[e & 63][e >> 6] = rc[q];                                           \
+    }                                                                          \
+  } while (0)
+
+    DB_LOAD_T(rx, i0, row_lim, 0);
+    DB_LOAD_T(rc, j0, n, 0);
+    __syncthreads();
+    DB_WRITE();
+    if (nsteps > 1) {
+      DB_LOAD_T(rx, i0, row_lim, KM_BK);
+      DB_LOAD_T(rc, j0, n, KM_BK);
+    }
+    __syncthreads();
+
+    for (int step = 0; step < nsteps; ++step) {
+#pragma unroll 8
+      for (int kk = 0; kk < KM_BK / 2; ++kk) {
+        const int kd = 2 * kk + (lane >> 5);
+        float a0 = lds_x[kd][wr * 64 + (lane & 31)];
+        float a1 = lds_x[kd][wr * 64 + 32 + (lane & 31)];
+        float b0 = lds_c[kd][wc * 64 + (lane & 31)];
+        float b1 = lds_c[kd][wc * 64 + 32 + (lane & 31)];
+        acc[0][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc[0][0], 0, 0, 0);
+        acc[0][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc[0][1], 0, 0, 0);
+        acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc[1][0], 0, 0, 0);
+        acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc[1][1], 0, 0, 0);
+      }
+      __syncthreads();
+      if (step + 1 < nsteps) {
+        DB_WRITE();
+        if (step + 2 < nsteps) {
+          DB_LOAD_T(rx, i0, row_lim, (step + 2) * KM_BK);
+          DB_LOAD_T(rc, j0, n, (step + 2) * KM_BK);
+        }
+        __syncthreads();
+      }
+    }
+#undef DB_LOAD_T
+#undef DB_WRITE
+
+    // epilogue: per-lane only 2 distinct columns -> hoist the per-column
+    // gathers (x_sq / core / labels) out of the 32-row loop
+#pragma unroll
+    for (int nn = 0; nn < 2; ++nn) {
+      const int col = wc * 64 + nn * 32 + (lane & 31);
+      const int gj = j0 + col;
+      if (gj >= n) continue;
+      const float cs = x_sq[gj];
+      int lab_j = 0;
+      bool core_j = true;
+      if (mode == 1) {
+        core_j = core[gj] != 0;
+        if (core_j) lab_j = labels[gj];
+      }
+      if (mode == 1 && !core_j) continue;
+#pragma unroll
+      for (int m = 0; m < 2; ++m) {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          int row = wr * 64 + m * 32 + (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
+          int gi = i0 + row;
+          if (gi < row_lim) {
+            float d2 = x_sq[gi] + cs - 2.0f * acc[m][nn][r];
+            if (d2 <= eps2) {
+              if (mode == 0)
+                atomicAdd(&acc_row[row], 1);
+              else
+                atomicMin(&acc_row[row], lab_j);
+            }
+          }
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  for (int i = tid; i < KM_BM; i += blockDim.x) {
+    int gi = i0 + i;
+    if (gi < row_lim) out[gi - row0] = acc_row[i];
+  }
+}
+
 // ---------------------------------------------------------------------------
 // kmeans_assign_glds: same contract as kmeans_assign_kernel, staging via
 // async global_load_lds DMA (16B) into a 16B-XOR-swizzled linear LDS image.
@@ -1535,6 +1694,31 @@ static inline hipStream_t cur_stream() {
   return c10::hip::getCurrentHIPStream().stream();
 }
 
+
+torch::Tensor dbscan_sweep(torch::Tensor X, torch::Tensor x_sq, int64_t row0,
+                           int64_t n_rows, double eps2, int64_t mode,
+                           torch::Tensor core, torch::Tensor labels) {
+  TORCH_CHECK(X.is_cuda() && x_sq.is_cuda(), "device tensors required");
+  TORCH_CHECK(X.dtype() == torch::kFloat32 && X.is_contiguous());
+  const int64_t n = X.size(0);
+  const int d = (int)X.size(1);
+  auto out = torch::empty({n_rows}, X.options().dtype(torch::kInt32));
+  if (n_rows == 0) return out;
+  TORCH_CHECK(row0 >= 0 && row0 + n_rows <= n, "row slice out of range");
+  if (mode == 1) {
+    TORCH_CHECK(core.dtype() == torch::kUInt8 && core.is_contiguous());
+    TORCH_CHECK(labels.dtype() == torch::kInt32 && labels.is_contiguous());
+  }
+  const int grid = (int)((n_rows + KM_BM - 1) / KM_BM);
+  hipLaunchKernelGGL(dbscan_sweep_kernel, dim3(grid), dim3(256), 0, cur_stream(),
+                     X.data_ptr<float>(), x_sq.data_ptr<float>(), (int)n, d,
+                     (int)row0, (int)n_rows, (float)eps2, (int)mode,
+                     mode == 1 ? core.data_ptr<uint8_t>() : nullptr,
+                     mode == 1 ? labels.data_ptr<int32_t>() : nullptr,
+                     out.data_ptr<int32_t>());
+  return out;
+}
+
 std::vector<torch::Tensor> kmeans_assign(torch::Tensor X, torch::Tensor C,
                                          torch::Tensor x_sq) {
   TORCH_CHECK(X.is_cuda() && C.is_cuda() && x_sq.is_cuda(), "device tensors required");
@@ -1733,5 +1917,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("knn_select", &knn_select, "fused MFMA distance + in-LDS top-k");
   m.def("rf_histogram", &rf_histogram, "LDS-privatized RF split histograms");
   m.def("rf_best_split", &rf_best_split, "fused RF gain scan + block-best reduce");
+  m.def("dbscan_sweep", &dbscan_sweep, "fused eps-neighborhood count / min-core-label pass");
   m.attr("_is_hip") = true;
 }

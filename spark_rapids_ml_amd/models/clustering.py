@@ -555,6 +555,79 @@ class DBSCANModel(_DBSCANParams, Model):
     def _transform_array(self, X: Any) -> np.ndarray:  # pragma: no cover - unused
         raise NotImplementedError("DBSCANModel clusters via transform()")
 
+    def _cluster_hip(self, Xf, x_sq, off, n_local, comm, eps2, min_samples):
+        """GPU path: three uses of the fused dbscan_sweep kernel (core
+        counting, min-label sweeps to fixpoint, border assignment). Exact:
+        every sweep reduces over the FULL eps-adjacency; no capped-graph
+        approximation. The torch path below materializes [chunk, n] masked
+        label tensors — the kernel keeps distances in registers and reduces
+        into a 128-entry LDS accumulator instead."""
+        import torch
+
+        from ..ops.dispatch import hip_ops
+
+        ext = hip_ops()
+        device = Xf.device
+        n = Xf.shape[0]
+        BIG = torch.iinfo(torch.int32).max
+        empty_u8 = torch.empty(0, dtype=torch.uint8, device=device)
+        empty_i32 = torch.empty(0, dtype=torch.int32, device=device)
+
+        counts = ext.dbscan_sweep(Xf, x_sq, off, n_local, eps2, 0, empty_u8, empty_i32)
+        core_local = counts >= min_samples
+        core_full = torch.cat(
+            comm.allgather_rows(core_local.to(torch.uint8)), dim=0
+        ).to(torch.bool)
+        core_u8 = core_full.to(torch.uint8).contiguous()
+
+        labels = torch.arange(n, dtype=torch.int32, device=device)
+        labels[~core_full] = BIG
+        core_ids = torch.nonzero(core_full).flatten()
+
+        def pointer_jump(lab):
+            for _ in range(8):
+                cur = lab[core_ids]
+                tgt = lab[cur.long().clamp(max=n - 1)]
+                tgt = torch.where(cur < BIG, tgt, cur)
+                upd = torch.minimum(cur, tgt)
+                if bool((upd == cur).all()):
+                    break
+                lab[core_ids] = upd
+            return lab
+
+        out = None
+        for _ in range(64):
+            out = ext.dbscan_sweep(Xf, x_sq, off, n_local, eps2, 1, core_u8, labels.contiguous())
+            new_full = torch.full((n,), BIG, dtype=torch.int32, device=device)
+            if n_local:
+                mine = labels[off : off + n_local]
+                new_full[off : off + n_local] = torch.where(
+                    core_local, torch.minimum(mine, out), mine
+                )
+            new_full = comm.allreduce_t(new_full, "min")
+            new_full[~core_full] = BIG
+            new_full = pointer_jump(new_full)
+            if bool((new_full == labels).all()):
+                break
+            labels = new_full
+
+        # border points: `out` was produced by the final (converged) sweep =
+        # min core-neighbor label over the full adjacency
+        final_local = torch.full((n_local,), -1, dtype=torch.int64, device=device)
+        if n_local:
+            border_lab = torch.where(out < BIG, out.to(torch.int64), final_local)
+            final_local = torch.where(
+                core_local, labels[off : off + n_local].to(torch.int64), border_lab
+            )
+
+        all_parts = comm.allgather_obj(as_numpy(final_local))
+        full = np.concatenate(all_parts)
+        uniq = np.unique(full[full >= 0])
+        remap = {int(v): i for i, v in enumerate(uniq)}
+        return np.array(
+            [remap.get(int(v), -1) for v in as_numpy(final_local)], dtype=np.int64
+        )
+
     def _cluster(self, X_local: np.ndarray, comm, pdesc) -> np.ndarray:
         eps = float(self.getOrDefault("eps"))
         min_samples = int(self.getOrDefault("min_samples"))
@@ -568,6 +641,15 @@ class DBSCANModel(_DBSCANParams, Model):
         off = pdesc.row_offset()
         n_local = X_local.shape[0]
         eps2 = eps * eps
+
+        from ..ops.dispatch import use_hip
+
+        if use_hip(Xf):
+            x_sq_full = (Xf * Xf).sum(dim=1)
+            return self._cluster_hip(
+                Xf.contiguous(), x_sq_full.contiguous(), off, n_local, comm,
+                eps2, min_samples,
+            )
 
         # batched adjacency: rows [off, off+n_local) vs all. The masked
         # label tensors are [chunk, n] int64, so the chunk is bounded by

@@ -176,3 +176,39 @@ def test_rf_best_split_matches_ref(ext, classif, C):
         # kernel emits raw (count, sum); torch reference emits (mean, count)
         lk = torch.stack([lk[:, 1] / lk[:, 0].clamp(min=1e-12), lk[:, 0]], dim=1)
     assert torch.allclose(lk, rl[ar[same], f2[same]], rtol=1e-4, atol=1e-3)
+
+
+@pytest.mark.gpu
+def test_dbscan_sweep_matches_torch():
+    from spark_rapids_ml_amd.ops.dispatch import hip_ops
+
+    ext = hip_ops()
+    dev = torch.device("cuda:0")
+    rng = np.random.default_rng(0)
+    n, d = 3000, 37  # non-multiple-of-64 d exercises the tail guards
+    X = torch.from_numpy(rng.normal(size=(n, d)).astype(np.float32)).to(dev)
+    x_sq = (X * X).sum(dim=1)
+    eps2 = float(np.quantile(rng.normal(size=1000) ** 2 * d, 0.02))
+    d2 = torch.cdist(X, X) ** 2
+    # mode 0: neighbor counts
+    empty_u8 = torch.empty(0, dtype=torch.uint8, device=dev)
+    empty_i32 = torch.empty(0, dtype=torch.int32, device=dev)
+    counts = ext.dbscan_sweep(X, x_sq, 0, n, eps2, 0, empty_u8, empty_i32)
+    ref_counts = (d2 <= eps2).sum(dim=1).to(torch.int32)
+    mism = (counts != ref_counts).sum().item()
+    assert mism <= n * 0.003, f"{mism} count mismatches"  # eps-boundary f32 ties
+    # mode 1: min core-neighbor label, on a row slice
+    core = (ref_counts >= 5).to(torch.uint8)
+    labels = torch.arange(n, dtype=torch.int32, device=dev)
+    labels[core == 0] = torch.iinfo(torch.int32).max
+    row0, n_rows = 1000, 1500
+    got = ext.dbscan_sweep(X, x_sq, row0, n_rows, eps2, 1, core, labels)
+    BIG = torch.iinfo(torch.int32).max
+    masked = torch.where(
+        (core[None, :] > 0) & (d2[row0 : row0 + n_rows] <= eps2),
+        labels[None, :].expand(n_rows, n),
+        torch.full((1,), BIG, dtype=torch.int32, device=dev),
+    )
+    ref = masked.min(dim=1).values
+    mism = (got != ref).sum().item()
+    assert mism <= n_rows * 0.003, f"{mism} label mismatches"
