@@ -362,6 +362,84 @@ __global__ __launch_bounds__(256) void dbscan_sweep_kernel(
   }
 }
 
+
+// ---------------------------------------------------------------------------
+// UMAP edge-sampled SGD (reference: cuML UMAP optimize_layout,
+// SURVEY.md §2.3b umap fit). One thread per edge per epoch, Hogwild
+// atomicAdd updates (the torch path's index_add is the same semantics with
+// ~10 kernel launches + intermediate tensors per epoch; here one launch per
+// epoch and the epoch loop lives in the host wrapper). Negative samples use
+// a per-(edge, epoch, trial) wang-hash — distributionally equivalent to the
+// torch generator, not bit-identical (UMAP is stochastic either way).
+// ---------------------------------------------------------------------------
+
+__device__ __forceinline__ unsigned umap_hash(unsigned x) {
+  x = (x ^ 61u) ^ (x >> 16);
+  x *= 9u;
+  x = x ^ (x >> 4);
+  x *= 0x27d4eb2du;
+  x = x ^ (x >> 15);
+  return x;
+}
+
+template <int DIM>
+__global__ __launch_bounds__(256) void umap_sgd_epoch_kernel(
+    float* __restrict__ emb,            // [n_head, DIM] updated in place
+    float* __restrict__ tail_emb,       // [n_vertices, DIM] (== emb when fitting)
+    const int32_t* __restrict__ heads,  // [m]
+    const int32_t* __restrict__ tails,  // [m]
+    const float* __restrict__ eps,      // [m] epochs-per-sample
+    float* __restrict__ next_due,       // [m] state
+    int m, int n_vertices,
+    float a, float b, float alpha, float repulsion,
+    int neg_rate, int move_tail, int epoch, unsigned seed) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= m) return;
+  if (next_due[i] > (float)epoch) return;
+  next_due[i] += eps[i];
+
+  const float clip = 4.0f;
+  const int h = heads[i], t = tails[i];
+  float eh[DIM], diff[DIM];
+  float d2 = 0.0f;
+#pragma unroll
+  for (int c = 0; c < DIM; ++c) {
+    eh[c] = emb[(int64_t)h * DIM + c];
+    diff[c] = eh[c] - tail_emb[(int64_t)t * DIM + c];
+    d2 += diff[c] * diff[c];
+  }
+  float d2c = fmaxf(d2, 1e-12f);
+  float pb = __powf(d2c, b);
+  float gcoef = (-2.0f * a * b * pb / d2c) / (1.0f + a * pb);
+#pragma unroll
+  for (int c = 0; c < DIM; ++c) {
+    float g = fminf(fmaxf(gcoef * diff[c], -clip), clip);
+    atomicAdd(&emb[(int64_t)h * DIM + c], alpha * g);
+    if (move_tail) atomicAdd(&tail_emb[(int64_t)t * DIM + c], -alpha * g);
+  }
+
+  unsigned rng = umap_hash(seed ^ umap_hash((unsigned)i * 2654435761u + (unsigned)epoch));
+  for (int r = 0; r < neg_rate; ++r) {
+    rng = umap_hash(rng + 0x9e3779b9u + (unsigned)r);
+    int j = (int)(rng % (unsigned)n_vertices);
+    d2 = 0.0f;
+#pragma unroll
+    for (int c = 0; c < DIM; ++c) {
+      eh[c] = emb[(int64_t)h * DIM + c];
+      diff[c] = eh[c] - tail_emb[(int64_t)j * DIM + c];
+      d2 += diff[c] * diff[c];
+    }
+    d2c = fmaxf(d2, 1e-12f);
+    pb = __powf(d2c, b);
+    gcoef = (2.0f * repulsion * b) / ((0.001f + d2) * (1.0f + a * pb));
+#pragma unroll
+    for (int c = 0; c < DIM; ++c) {
+      float g = fminf(fmaxf(gcoef * diff[c], -clip), clip);
+      atomicAdd(&emb[(int64_t)h * DIM + c], alpha * g);
+    }
+  }
+}
+
 // ---------------------------------------------------------------------------
 // kmeans_assign_glds: same contract as kmeans_assign_kernel, staging via
 // async global_load_lds DMA (16B) into a 16B-XOR-swizzled linear LDS image.
@@ -1719,6 +1797,42 @@ torch::Tensor dbscan_sweep(torch::Tensor X, torch::Tensor x_sq, int64_t row0,
   return out;
 }
 
+
+torch::Tensor umap_sgd(torch::Tensor emb, torch::Tensor tail_emb,
+                       torch::Tensor heads, torch::Tensor tails,
+                       torch::Tensor eps, int64_t n_epochs, double a, double b,
+                       double lr, double repulsion, int64_t neg_rate,
+                       bool move_tail, int64_t seed) {
+  TORCH_CHECK(emb.is_cuda() && emb.dtype() == torch::kFloat32 && emb.is_contiguous());
+  TORCH_CHECK(heads.dtype() == torch::kInt32 && tails.dtype() == torch::kInt32);
+  TORCH_CHECK(eps.dtype() == torch::kFloat32);
+  const int m = (int)heads.size(0);
+  const int dim = (int)emb.size(1);
+  const int n_vertices = (int)tail_emb.size(0);
+  TORCH_CHECK(dim >= 1 && dim <= 4, "umap_sgd supports dim 1..4");
+  auto next_due = eps.clone();
+  const int grid = (m + 255) / 256;
+  for (int epoch = 1; epoch <= (int)n_epochs; ++epoch) {
+    const float alpha = (float)(lr * (1.0 - (double)epoch / (double)n_epochs));
+#define UMAP_LAUNCH(D)                                                         \
+    hipLaunchKernelGGL(umap_sgd_epoch_kernel<D>, dim3(grid), dim3(256), 0,     \
+                       cur_stream(), emb.data_ptr<float>(),                    \
+                       tail_emb.data_ptr<float>(), heads.data_ptr<int32_t>(),  \
+                       tails.data_ptr<int32_t>(), eps.data_ptr<float>(),       \
+                       next_due.data_ptr<float>(), m, n_vertices, (float)a,    \
+                       (float)b, alpha, (float)repulsion, (int)neg_rate,       \
+                       move_tail ? 1 : 0, epoch, (unsigned)(seed & 0xffffffff))
+    switch (dim) {
+      case 1: UMAP_LAUNCH(1); break;
+      case 2: UMAP_LAUNCH(2); break;
+      case 3: UMAP_LAUNCH(3); break;
+      default: UMAP_LAUNCH(4); break;
+    }
+#undef UMAP_LAUNCH
+  }
+  return emb;
+}
+
 std::vector<torch::Tensor> kmeans_assign(torch::Tensor X, torch::Tensor C,
                                          torch::Tensor x_sq) {
   TORCH_CHECK(X.is_cuda() && C.is_cuda() && x_sq.is_cuda(), "device tensors required");
@@ -1918,5 +2032,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rf_histogram", &rf_histogram, "LDS-privatized RF split histograms");
   m.def("rf_best_split", &rf_best_split, "fused RF gain scan + block-best reduce");
   m.def("dbscan_sweep", &dbscan_sweep, "fused eps-neighborhood count / min-core-label pass");
+  m.def("umap_sgd", &umap_sgd, "edge-sampled UMAP SGD (all epochs, Hogwild)");
   m.attr("_is_hip") = true;
 }

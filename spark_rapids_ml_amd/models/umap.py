@@ -114,10 +114,40 @@ def _optimize_embedding(
 ) -> torch.Tensor:
     """Edge-sampled SGD (umap-learn schedule, vectorized): each edge gets
     epochs_per_sample = max(w)/w; per epoch, due edges attract and spawn
-    negative repulsions. index_add makes the update Hogwild-equivalent."""
+    negative repulsions. index_add makes the update Hogwild-equivalent.
+
+    On GPU the whole schedule runs in the HIP umap_sgd kernel (one launch
+    per epoch, epoch loop in C++; hash-based negative sampling — same
+    distribution as the torch generator, not bit-identical)."""
     device = emb.device
     n_emb = tail_emb if tail_emb is not None else emb
     eps = weights.max() / torch.clamp(weights, min=1e-12)  # epochs per sample
+
+    from ..ops.dispatch import hip_ops, use_hip
+
+    if use_hip(emb) and emb.shape[1] <= 4 and heads.numel():
+        ext = hip_ops()
+        seed = int(gen.initial_seed()) & 0x7FFFFFFF
+        # spectral init can hand over a non-contiguous slice; the kernel
+        # updates in place, so re-alias n_emb AFTER making emb contiguous
+        emb = emb.contiguous()
+        n_emb = tail_emb.contiguous() if tail_emb is not None else emb
+        ext.umap_sgd(
+            emb,
+            n_emb,
+            heads.to(torch.int32).contiguous(),
+            tails.to(torch.int32).contiguous(),
+            eps.to(torch.float32).contiguous(),
+            int(n_epochs),
+            float(a),
+            float(b),
+            float(lr),
+            float(repulsion),
+            int(neg_rate),
+            bool(move_tail),
+            seed,
+        )
+        return emb
     next_due = eps.clone()
     n_vertices = n_emb.shape[0]
     clip = 4.0

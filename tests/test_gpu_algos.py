@@ -215,3 +215,23 @@ def test_single_vector_predict_gpu():
     Xb, _ = make_blobs(n_samples=300, n_features=8, centers=3, random_state=0)
     km = KMeans(k=3, maxIter=15, seed=1).fit(DataFrame.from_numpy(Xb.astype(np.float32)))
     assert 0 <= km.predict(Xb[0].astype(np.float32)) < 3
+
+
+def test_umap_sgd_kernel_quality():
+    """The HIP umap_sgd kernel must produce an embedding of the same quality
+    as the torch SGD path (stochastic — compared by trustworthiness, the
+    reference's own acceptance metric for UMAP)."""
+    from sklearn.datasets import make_blobs
+    from sklearn.manifold import trustworthiness
+
+    from spark_rapids_ml_amd import UMAP
+
+    X, _ = make_blobs(n_samples=2000, n_features=24, centers=6, cluster_std=0.6, random_state=0)
+    X = X.astype(np.float32)
+    model = UMAP(n_neighbors=15, n_epochs=150, random_state=7).fit(DataFrame.from_numpy(X))
+    tw = trustworthiness(X, model.embedding, n_neighbors=10)
+    assert tw > 0.92, tw
+    # transform path (move_tail=False, separate tail embedding)
+    out = model.transform(DataFrame.from_numpy(X[:200]))
+    emb_q = np.asarray(out["embedding"])
+    assert np.isfinite(emb_q).all()
