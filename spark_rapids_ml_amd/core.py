@@ -155,6 +155,9 @@ class Estimator(_NativeParams, HasVerbose):
         return out
 
     def _fit_internal(self, df: DataFrame, param_maps: List[Dict[str, Any]]) -> List["Model"]:
+        from .params import validate_param_bounds
+
+        validate_param_bounds(self)
         comm = get_comm()
         X, y = self._extract_xy(df)
         nnz = X.nnz if _is_sparse(X) else None

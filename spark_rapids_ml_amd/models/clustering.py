@@ -172,7 +172,10 @@ class KMeans(_KMeansParams, Estimator):
         n_iter = 0
         inertia = float("inf")
         dtype_sum = torch.float64
-        for it in range(max_iter):
+        g_counts = torch.zeros(k, dtype=dtype_sum, device=Xt.device)
+        # maxIter=0 (Spark-legal: model = init centers) still reports sizes
+        # and cost from one assignment pass
+        for it in range(max(max_iter, 1)):
             labels, sums, counts, local_inertia = kmeans_assign_reduce(Xt, C, x_sq)
             # ONE fused all-reduce: [k, d+2] = [sums | counts | (inertia in row 0)]
             buf = torch.zeros((k, pdesc.n + 2), dtype=dtype_sum, device=sums.device)
@@ -183,6 +186,8 @@ class KMeans(_KMeansParams, Estimator):
             g_sums = buf[:, : pdesc.n]
             g_counts = buf[:, pdesc.n]
             inertia = float(buf[0, pdesc.n + 1].item())
+            if max_iter == 0:  # stats only; centers stay at init
+                break
             nonempty = g_counts > 0
             C_new = C.clone().to(torch.float64)
             C_new[nonempty] = g_sums[nonempty] / g_counts[nonempty, None]

@@ -160,18 +160,20 @@ class LinearRegression(_LinearRegressionParams, Estimator):
         Xt = to_device_tensor(np.ascontiguousarray(X), ctx.device)
         yt = to_device_tensor(np.ascontiguousarray(y), ctx.device).to(Xt.dtype)
         # fused buffer rows: [0:d]=Gram, d=X^T y, d+1=colsum X, d+2=[ysum, y2sum]
-        buf = torch.zeros((d + 3, d), dtype=torch.float64, device=Xt.device)
+        # (width >= 2 so the label-moment row exists even at d=1)
+        w = max(d, 2)
+        buf = torch.zeros((d + 3, w), dtype=torch.float64, device=Xt.device)
         if Xt.shape[0] > 0:
-            buf[:d] = gram(Xt).to(torch.float64)
-            buf[d] = (Xt.T @ yt).to(torch.float64)
-            buf[d + 1] = Xt.sum(dim=0).to(torch.float64)
+            buf[:d, :d] = gram(Xt).to(torch.float64)
+            buf[d, :d] = (Xt.T @ yt).to(torch.float64)
+            buf[d + 1, :d] = Xt.sum(dim=0).to(torch.float64)
             buf[d + 2, 0] = yt.sum().to(torch.float64)
             buf[d + 2, 1] = (yt * yt).sum().to(torch.float64)
         buf = comm.allreduce_t(buf)
         m = {
-            "G": buf[:d],
-            "Xty": buf[d],
-            "xsum": buf[d + 1],
+            "G": buf[:d, :d],
+            "Xty": buf[d, :d],
+            "xsum": buf[d + 1, :d],
             "ysum": buf[d + 2, 0],
             "y2sum": buf[d + 2, 1],
             "n": pdesc.m,

@@ -630,3 +630,54 @@ class _NativeParams(_NativeClass, Params):
         else:
             raise ValueError("featuresCol or featuresCols must be set")
         return features_col, features_cols
+
+
+# Central numeric-validity table for Spark-standard param names, enforced at
+# fit() entry (reference behavior: cuML validates and raises per estimator;
+# pyspark's ParamValidators reject at set time). (lo, hi, lo_inclusive,
+# hi_inclusive); None = unbounded.
+_PARAM_BOUNDS = {
+    "k": (1, None, True, True),
+    "maxIter": (0, None, True, True),
+    "initSteps": (1, None, True, True),
+    "tol": (0.0, None, True, True),
+    "regParam": (0.0, None, True, True),
+    "elasticNetParam": (0.0, 1.0, True, True),
+    "threshold": (0.0, 1.0, True, True),
+    "numTrees": (1, None, True, True),
+    "maxDepth": (1, None, True, True),
+    "maxBins": (2, 256, True, True),
+    "minInstancesPerNode": (1, None, True, True),
+    "minInfoGain": (0.0, None, True, True),
+    "subsamplingRate": (0.0, 1.0, False, True),
+    "eps": (0.0, None, False, True),
+    "min_samples": (1, None, True, True),
+    "n_neighbors": (1, None, True, True),
+    "n_components": (1, None, True, True),
+    "n_epochs": (0, None, True, True),
+    "numFolds": (2, None, True, True),
+    "negative_sample_rate": (1, None, True, True),
+    "sample_fraction": (0.0, 1.0, False, True),
+}
+
+
+def validate_param_bounds(inst: "Params") -> None:
+    """Raise ValueError for out-of-range numeric params (table above)."""
+    for p in inst.params:
+        if p.name not in _PARAM_BOUNDS or not inst.isDefined(p.name):
+            continue
+        v = inst.getOrDefault(p.name)
+        if v is None or not isinstance(v, (int, float)):
+            continue
+        lo, hi, loi, hii = _PARAM_BOUNDS[p.name]
+        ok = True
+        if lo is not None:
+            ok = ok and (v >= lo if loi else v > lo)
+        if hi is not None:
+            ok = ok and (v <= hi if hii else v < hi)
+        if not ok:
+            lob = "[" if loi else "("
+            hib = "]" if hii else ")"
+            raise ValueError(
+                f"Param {p.name}={v!r} out of range {lob}{lo}, {hi}{hib}"
+            )

@@ -210,3 +210,50 @@ def test_model_evaluate_and_summary():
     assert lg.summary.totalIterations >= 1
     ev = lg.evaluate(dfc)
     assert ev.accuracy > 0.8
+
+
+def test_param_bounds_validation():
+    import numpy as np
+
+    from spark_rapids_ml_amd import KMeans, LinearRegression, LogisticRegression
+    from spark_rapids_ml_amd.data import DataFrame
+
+    X = np.random.rand(50, 3).astype(np.float32)
+    df = DataFrame.from_numpy(X)
+    for bad in (dict(k=0), dict(k=-2), dict(maxIter=-1)):
+        with pytest.raises(ValueError, match="out of range"):
+            KMeans(**bad).fit(df)
+    ydf = DataFrame.from_numpy(X, (X[:, 0] > 0.5).astype(np.float64))
+    with pytest.raises(ValueError, match="out of range"):
+        LogisticRegression(elasticNetParam=1.5).fit(ydf)
+    with pytest.raises(ValueError, match="out of range"):
+        LinearRegression(regParam=-0.1).fit(
+            DataFrame.from_numpy(X, X[:, 0].astype(np.float64))
+        )
+
+
+def test_single_feature_column_fit():
+    import numpy as np
+
+    from spark_rapids_ml_amd import LinearRegression
+    from spark_rapids_ml_amd.data import DataFrame
+
+    rng = np.random.default_rng(0)
+    x = rng.normal(size=(200, 1))
+    y = 3.0 * x[:, 0] + 1.0 + 0.01 * rng.normal(size=200)
+    m = LinearRegression().fit(DataFrame.from_numpy(x, y))
+    assert abs(m.coefficients[0] - 3.0) < 0.05
+    assert abs(m.intercept - 1.0) < 0.05
+
+
+def test_kmeans_zero_iterations():
+    import numpy as np
+
+    from spark_rapids_ml_amd import KMeans
+    from spark_rapids_ml_amd.data import DataFrame
+
+    X = np.random.rand(100, 4).astype(np.float32)
+    m = KMeans(k=3, maxIter=0, seed=1).fit(DataFrame.from_numpy(X))
+    assert m.cluster_centers_.shape == (3, 4)
+    assert m.summary.numIter == 0
+    assert sum(m.summary.clusterSizes) == 100
