@@ -191,3 +191,23 @@ def test_sparse_csr_matches_dense():
     assert (
         np.asarray(out["prediction"]) == np.asarray(out_d["prediction"])
     ).mean() > 0.99
+
+
+def test_noncontiguous_label_values():
+    """Arbitrary float label values (classes {2.0, 5.0, 9.0}) round-trip
+    through the sorted-class index mapping back to original values."""
+    rng = np.random.default_rng(0)
+    X, y_idx = make_classification(
+        n_samples=600, n_features=10, n_informative=6, n_classes=3,
+        n_clusters_per_class=1, random_state=0,
+    )
+    classes = np.array([2.0, 5.0, 9.0])
+    y = classes[y_idx]
+    model = LogisticRegression(maxIter=60).fit(
+        DataFrame.from_numpy(X.astype(np.float32), y)
+    )
+    assert np.array_equal(model.classes_, classes)
+    pred = np.asarray(model.transform(DataFrame.from_numpy(X.astype(np.float32)))["prediction"])
+    assert set(np.unique(pred)).issubset(set(classes))
+    # sklearn reaches 0.813 train accuracy on this dataset; match it
+    assert (pred == y).mean() > 0.79

@@ -104,3 +104,20 @@ def test_pca_input_col_api():
     model = PCA(k=2).setInputCol("feat_vec").setOutputCol("proj").fit(df)
     out = model.transform(df)
     assert np.asarray(out["proj"]).shape == (100, 2)
+
+
+def test_pca_rectangle_fewer_rows_than_cols():
+    """n < d (reference test_fit_rectangle): covariance is rank-deficient but
+    the top components still match sklearn."""
+    from sklearn.decomposition import PCA as SkPCA
+
+    rng = np.random.default_rng(0)
+    X = (rng.normal(size=(60, 200)) @ np.diag(np.linspace(3, 0.1, 200))).astype(np.float32)
+    model = PCA(k=4).fit(DataFrame.from_numpy(X))
+    sk = SkPCA(n_components=4).fit(X)
+    for i in range(4):
+        dot = abs(np.dot(model.components_[i], sk.components_[i]))
+        assert dot > 0.99, (i, dot)
+    assert np.allclose(
+        model.explained_variance_ratio_, sk.explained_variance_ratio_, atol=1e-4
+    )
