@@ -167,6 +167,8 @@ class DataFrame:
                     cols[name] = np.ascontiguousarray(
                         vals[offsets[0] : offsets[-1]].reshape(n, int(widths[0]))
                     )
+                elif n == 0:
+                    cols[name] = np.zeros((0, 0), dtype=vals.dtype)
                 else:
                     cols[name] = np.stack(
                         [vals[offsets[i] : offsets[i + 1]] for i in range(n)]
