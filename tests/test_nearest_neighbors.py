@@ -80,3 +80,11 @@ def test_knn_model_persistence_unsupported():
     model = NearestNeighbors(k=3).fit(DataFrame.from_numpy(X))
     with pytest.raises(NotImplementedError):
         model.save("/tmp/should_not_exist")
+
+
+def test_approx_similarity_join_alias():
+    X = _data(n=80)
+    Q = _data(n=8, seed=1)
+    model = NearestNeighbors(k=3).fit(DataFrame.from_numpy(X))
+    joined = model.approxSimilarityJoin(DataFrame.from_numpy(Q))
+    assert joined.num_rows == 24
