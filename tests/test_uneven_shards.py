@@ -105,3 +105,11 @@ def test_rf_empty_shard_rank():
     for n_trees, acc in results:
         assert n_trees == 4
         assert acc > 0.9
+
+
+def test_three_rank_world():
+    """world_size=3 catches modular-arithmetic assumptions a 2-rank world
+    hides (odd rank counts happen on partially-allocated nodes)."""
+    results = run_distributed(_dist_empty_shard, world_size=3, args=("kmeans",))
+    assert np.allclose(results[0], results[1])
+    assert np.allclose(results[1], results[2])
