@@ -45,6 +45,9 @@ _SUBMODULE_EXPORTS: Dict[str, List[str]] = {
         "MulticlassClassificationEvaluator",
         "BinaryClassificationEvaluator",
     ],
+    "metrics": [],   # filled from the metrics package below
+    "core": [],      # Estimator/Model base aliases
+    "params": [],    # Param surface
 }
 
 
@@ -82,6 +85,23 @@ def install_aliases(alias_root: str = "spark_rapids_ml") -> None:
 
             for n in names:
                 exports[n] = getattr(evaluation, n)
+        if sub == "metrics":
+            from .metrics import MulticlassMetrics, RegressionMetrics
+
+            exports["MulticlassMetrics"] = MulticlassMetrics
+            exports["RegressionMetrics"] = RegressionMetrics
+        if sub == "core":
+            from .core import Estimator, Model
+
+            exports["Estimator"] = exports["_CumlEstimator"] = Estimator
+            exports["Model"] = exports["_CumlModel"] = Model
+        if sub == "params":
+            from .params import Param, Params, TypeConverters, DictTypeConverters
+
+            exports.update(
+                Param=Param, Params=Params, TypeConverters=TypeConverters,
+                DictTypeConverters=DictTypeConverters,
+            )
         m = _make_module(f"{alias_root}.{sub}", exports)
         sys.modules[f"{alias_root}.{sub}"] = m
         setattr(root_mod, sub, m)
