@@ -95,3 +95,21 @@ def test_umap_distributed_model_replicated():
     # fit on gathered data, broadcast: both ranks hold the identical model
     assert np.allclose(emb0, emb1)
     assert out0.shape == (20, 2) and out1.shape == (20, 2)
+
+
+def test_umap_supervised_separates_classes():
+    """Supervised fit (labelCol): cross-label edges shrink by exp(-5), so
+    same-label points embed closer than different-label ones (reference
+    supervised fit, umap.py:1035-1050)."""
+    X, y = _data(n=500)
+    df = DataFrame({"features": X, "label": y.astype(np.float64)})
+    sup = UMAP(n_neighbors=10, n_epochs=100, random_state=1, labelCol="label").fit(df)
+    emb = sup.embedding
+    # mean intra-class distance well below mean inter-class distance
+    intra, inter = [], []
+    rng = np.random.default_rng(0)
+    idx = rng.choice(len(emb), size=(400, 2))
+    for i, j in idx:
+        d = np.linalg.norm(emb[i] - emb[j])
+        (intra if y[i] == y[j] else inter).append(d)
+    assert np.mean(intra) < 0.5 * np.mean(inter)
