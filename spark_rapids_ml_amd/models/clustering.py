@@ -99,8 +99,13 @@ class _KMeansParams(HasFeaturesCol, HasFeaturesCols, HasPredictionCol, HasWeight
             "max_iter": 20,
             "random_state": 1,
             "tol": 1e-4,
-            "metric": "euclidean",
+            # cuML-signature perf knobs accepted for constructor parity
+            # (reference python/README.md: cuML params supplied to ctors);
+            # the HIP kernels batch internally so these are inert
+            "n_init": "auto",
             "oversampling_factor": 2.0,
+            "max_samples_per_batch": 32768,
+            "metric": "euclidean",
             "verbose": False,
         }
 

@@ -51,7 +51,12 @@ class _NNParams(HasFeaturesCol, HasFeaturesCols, HasIDCol):
 
     @classmethod
     def _get_native_params_default(cls) -> Dict[str, Any]:
-        return {"n_neighbors": 5, "verbose": False}
+        return {
+            "n_neighbors": 5,
+            "verbose": False,
+            # cuML-signature knob accepted for ctor parity (inert here)
+            "batch_size": 2000000,
+        }
 
     def getK(self) -> int:
         return self.getOrDefault("k")

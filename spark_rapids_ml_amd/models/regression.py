@@ -111,6 +111,10 @@ class _LinearRegressionParams(HasFeaturesCol, HasFeaturesCols, HasLabelCol, HasP
             "tol": 1e-6,
             "solver": "eig",
             "loss": "squared_loss",
+            # cuML-signature knobs accepted for ctor parity (inert here)
+            "algorithm": "auto",
+            "copy_X": True,
+            "shuffle": True,
             "verbose": False,
         }
 

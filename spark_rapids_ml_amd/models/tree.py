@@ -105,6 +105,10 @@ class _RandomForestParams(HasFeaturesCol, HasFeaturesCols, HasLabelCol, HasPredi
             "bootstrap": True,
             "split_criterion": None,
             "max_batch_size": 4096,
+            # cuML-signature knobs accepted for ctor parity (inert here)
+            "n_streams": 4,
+            "min_samples_split": 2,
+            "max_leaves": -1,
             "verbose": False,
         }
 

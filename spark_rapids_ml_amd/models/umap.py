@@ -235,7 +235,19 @@ class _UMAPParams(HasFeaturesCol, HasFeaturesCols, HasLabelCol, HasOutputCol):
 
     @classmethod
     def _get_native_params_default(cls) -> Dict[str, Any]:
-        return {"n_neighbors": 15, "n_components": 2, "verbose": False}
+        return {
+            "n_neighbors": 15,
+            "n_components": 2,
+            "verbose": False,
+            # cuML-signature knobs accepted for ctor parity (inert here:
+            # euclidean metric; graph build auto-selects brute/nn-descent)
+            "metric": "euclidean",
+            "metric_kwds": None,
+            "transform_queue_size": 4.0,
+            "precomputed_knn": None,
+            "build_algo": "auto",
+            "build_kwds": None,
+        }
 
 
 class UMAP(_UMAPParams, Estimator):

@@ -257,3 +257,18 @@ def test_kmeans_zero_iterations():
     assert m.cluster_centers_.shape == (3, 4)
     assert m.summary.numIter == 0
     assert sum(m.summary.clusterSizes) == 100
+
+
+def test_cuml_signature_ctor_params_accepted():
+    """Reference python/README.md:157-182 — cuML-specific params supply
+    through constructors, appear in cuml_params, have no getters/setters."""
+    from spark_rapids_ml_amd import KMeans, LinearRegression, NearestNeighbors, UMAP
+
+    km = KMeans(k=3, max_samples_per_batch=16384)
+    assert km.cuml_params["max_samples_per_batch"] == 16384
+    assert not hasattr(km, "getMaxSamplesPerBatch") or True
+    with pytest.raises(AttributeError):
+        km.getMaxSamplesPerBatch()  # no Spark getter for cuML-only params
+    assert LinearRegression(algorithm="svd").cuml_params["algorithm"] == "svd"
+    assert NearestNeighbors(batch_size=5).cuml_params["batch_size"] == 5
+    assert UMAP(build_algo="nn_descent").cuml_params["build_algo"] == "nn_descent"
