@@ -73,10 +73,6 @@ def standardize_dataset(
     return Xs, mean.to(X.dtype), std.to(X.dtype)
 
 
-def inertia_dtype(X: torch.Tensor) -> torch.dtype:
-    return torch.float64
-
-
 def as_numpy(t: torch.Tensor) -> np.ndarray:
     return t.detach().cpu().numpy()
 

@@ -48,3 +48,37 @@ def test_random_split_and_sample():
     assert 700 < tr.num_rows < 900
     s = df.sample(0.1, seed=2)
     assert 50 < s.num_rows < 200
+
+
+def test_standardize_dataset_matches_numpy():
+    import torch
+
+    from spark_rapids_ml_amd.parallel.context import PartitionDescriptor, get_comm
+    from spark_rapids_ml_amd.utils import standardize_dataset
+
+    rng = np.random.default_rng(0)
+    X = rng.normal(loc=3.0, scale=2.5, size=(200, 6))
+    comm = get_comm()
+    pdesc = PartitionDescriptor.build(comm, X.shape[0], X.shape[1])
+    Xs, mean, std = standardize_dataset(torch.from_numpy(X), comm, pdesc)
+    assert np.allclose(mean.numpy(), X.mean(axis=0))
+    assert np.allclose(std.numpy(), X.std(axis=0, ddof=1))
+    assert np.allclose(Xs.numpy().mean(axis=0), 0.0, atol=1e-12)
+
+
+def test_torch_ref_logistic_oracle_consistency():
+    """The torch_ref logistic oracle and the production glm pass agree
+    (gradient and loss) on the same weights."""
+    import torch
+
+    from spark_rapids_ml_amd.ops.glm import logistic_grad_loss
+    from spark_rapids_ml_amd.ops.torch_ref import logistic_forward_grad
+
+    rng = np.random.default_rng(0)
+    X = torch.from_numpy(rng.normal(size=(100, 5)))
+    y = torch.from_numpy(rng.integers(0, 3, size=100))
+    W = torch.from_numpy(rng.normal(size=(3, 6)) * 0.1)
+    g1, l1 = logistic_forward_grad(X, y, W, fit_intercept=True)
+    g2, l2 = logistic_grad_loss(X, y, W, fit_intercept=True)
+    assert np.allclose(g1.numpy(), g2.numpy(), atol=1e-10)
+    assert np.isclose(float(l1), float(l2), atol=1e-10)
