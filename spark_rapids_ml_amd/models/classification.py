@@ -69,6 +69,10 @@ class _LogisticRegressionParams(
         "logreg", "family", "auto|binomial|multinomial.", TypeConverters.toString
     )
     threshold = Param("logreg", "threshold", "binary decision threshold.", TypeConverters.toFloat)
+    thresholds = Param(
+        "logreg", "thresholds", "per-class thresholds (unsupported, reference classification.py:689).",
+        TypeConverters.toListFloat,
+    )
     weightCol = Param("logreg", "weightCol", "unsupported on GPU.", TypeConverters.toString)
 
     def __init__(self, **kwargs: Any) -> None:
@@ -96,6 +100,7 @@ class _LogisticRegressionParams(
             "standardization": "standardization",
             "family": "",
             "threshold": "",
+            "thresholds": None,
             "weightCol": None,
         }
 
