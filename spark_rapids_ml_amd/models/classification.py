@@ -74,6 +74,10 @@ class _LogisticRegressionParams(
         TypeConverters.toListFloat,
     )
     weightCol = Param("logreg", "weightCol", "unsupported on GPU.", TypeConverters.toString)
+    lowerBoundsOnCoefficients = Param("logreg", "lowerBoundsOnCoefficients", "unsupported.", TypeConverters.identity)
+    upperBoundsOnCoefficients = Param("logreg", "upperBoundsOnCoefficients", "unsupported.", TypeConverters.identity)
+    lowerBoundsOnIntercepts = Param("logreg", "lowerBoundsOnIntercepts", "unsupported.", TypeConverters.identity)
+    upperBoundsOnIntercepts = Param("logreg", "upperBoundsOnIntercepts", "unsupported.", TypeConverters.identity)
 
     def __init__(self, **kwargs: Any) -> None:
         super().__init__(**kwargs)
@@ -102,6 +106,10 @@ class _LogisticRegressionParams(
             "threshold": "",
             "thresholds": None,
             "weightCol": None,
+            "lowerBoundsOnCoefficients": None,
+            "upperBoundsOnCoefficients": None,
+            "lowerBoundsOnIntercepts": None,
+            "upperBoundsOnIntercepts": None,
         }
 
     @classmethod
