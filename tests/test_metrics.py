@@ -105,3 +105,18 @@ def test_by_label_metrics():
         ).evaluate(df)
         assert np.isclose(p, precision_score(lab, pred, labels=[cls], average=None)[0])
         assert np.isclose(r, recall_score(lab, pred, labels=[cls], average=None)[0])
+
+
+def test_auc_matches_sklearn():
+    from sklearn.metrics import roc_auc_score
+
+    from spark_rapids_ml_amd.data import DataFrame
+    from spark_rapids_ml_amd.evaluation import BinaryClassificationEvaluator
+
+    rng = np.random.default_rng(0)
+    y = rng.integers(0, 2, size=500).astype(np.float64)
+    score = y + rng.normal(scale=1.2, size=500)  # informative but noisy
+    raw = np.stack([-score, score], axis=1)
+    df = DataFrame({"label": y, "rawPrediction": raw})
+    auc = BinaryClassificationEvaluator().evaluate(df)
+    assert np.isclose(auc, roc_auc_score(y, score), atol=1e-9)
