@@ -116,10 +116,11 @@ class MulticlassClassificationEvaluator(Evaluator, HasLabelCol, HasPredictionCol
 
 
 class BinaryClassificationEvaluator(Evaluator, HasLabelCol):
-    """areaUnderROC via a global rank-sum (Mann-Whitney) over gathered
+    """areaUnderROC (global Mann-Whitney rank-sum) and areaUnderPR
+    (precision-recall step interpolation, Spark semantics) over gathered
     scores."""
 
-    metricName = Param("eval", "metricName", "areaUnderROC.", TypeConverters.toString)
+    metricName = Param("eval", "metricName", "areaUnderROC|areaUnderPR.", TypeConverters.toString)
     rawPredictionCol = Param("eval", "rawPredictionCol", "scores col.", TypeConverters.toString)
 
     def __init__(self, **kwargs: Any) -> None:
@@ -139,6 +140,19 @@ class BinaryClassificationEvaluator(Evaluator, HasLabelCol):
         neg = s[l != 1.0]
         if len(pos) == 0 or len(neg) == 0:
             return 0.5
+        if self.getOrDefault("metricName") == "areaUnderPR":
+            order = np.argsort(-s, kind="stable")
+            tp = np.cumsum(l[order] == 1.0)
+            fp = np.cumsum(l[order] != 1.0)
+            precision = tp / np.maximum(tp + fp, 1)
+            recall = tp / len(pos)
+            # step-wise area with the (0, p0) start point (Spark/sklearn AP-style)
+            r_prev = 0.0
+            area = 0.0
+            for p, r in zip(precision, recall):
+                area += p * (r - r_prev)
+                r_prev = r
+            return float(area)
         from scipy.stats import rankdata
 
         ranks = rankdata(np.concatenate([pos, neg]))

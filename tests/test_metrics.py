@@ -120,3 +120,17 @@ def test_auc_matches_sklearn():
     df = DataFrame({"label": y, "rawPrediction": raw})
     auc = BinaryClassificationEvaluator().evaluate(df)
     assert np.isclose(auc, roc_auc_score(y, score), atol=1e-9)
+
+
+def test_area_under_pr_matches_sklearn():
+    from sklearn.metrics import average_precision_score
+
+    from spark_rapids_ml_amd.data import DataFrame
+    from spark_rapids_ml_amd.evaluation import BinaryClassificationEvaluator
+
+    rng = np.random.default_rng(1)
+    y = rng.integers(0, 2, size=400).astype(np.float64)
+    score = y + rng.normal(scale=1.0, size=400)
+    df = DataFrame({"label": y, "rawPrediction": np.stack([-score, score], axis=1)})
+    pr = BinaryClassificationEvaluator(metricName="areaUnderPR").evaluate(df)
+    assert np.isclose(pr, average_precision_score(y, score), atol=1e-9)
