@@ -147,12 +147,7 @@ class BinaryClassificationEvaluator(Evaluator, HasLabelCol):
             precision = tp / np.maximum(tp + fp, 1)
             recall = tp / len(pos)
             # step-wise area with the (0, p0) start point (Spark/sklearn AP-style)
-            r_prev = 0.0
-            area = 0.0
-            for p, r in zip(precision, recall):
-                area += p * (r - r_prev)
-                r_prev = r
-            return float(area)
+            return float(np.sum(precision * np.diff(recall, prepend=0.0)))
         from scipy.stats import rankdata
 
         ranks = rankdata(np.concatenate([pos, neg]))
