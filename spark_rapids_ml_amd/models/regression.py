@@ -263,11 +263,14 @@ class LinearRegression(_LinearRegressionParams, Estimator):
             ).item()
         )
 
+        # total sum of squares around the label mean (for summary r2)
+        tss = float((y2sum - ysum * ysum / max(1, n)).item())
         return {
             "coef_": as_numpy(w).astype(np.float64),
             "intercept_": intercept,
             "n_rows_": n,
             "sse_": max(0.0, sse),
+            "tss_": max(0.0, tss),
         }
 
     def _create_model(self, attrs: Dict[str, Any]) -> "LinearRegressionModel":
@@ -354,6 +357,7 @@ class LinearRegressionModel(_LinearRegressionParams, Model):
         intercept_: float = 0.0,
         n_rows_: int = 0,
         sse_: float = 0.0,
+        tss_: float = 0.0,
         **kwargs: Any,
     ) -> None:
         super().__init__(
@@ -361,6 +365,7 @@ class LinearRegressionModel(_LinearRegressionParams, Model):
             intercept_=float(intercept_),
             n_rows_=int(n_rows_),
             sse_=float(sse_),
+            tss_=float(tss_),
         )
 
     @property
@@ -389,10 +394,12 @@ class LinearRegressionModel(_LinearRegressionParams, Model):
         rootMeanSquaredError/meanSquaredError from the training SSE)."""
         n = max(1, self._model_attributes["n_rows_"])
         sse = self._model_attributes["sse_"]
+        tss = self._model_attributes.get("tss_", 0.0)
         return LinearRegressionTrainingSummary(
             meanSquaredError=sse / n,
             rootMeanSquaredError=float(np.sqrt(sse / n)),
             numInstances=n,
+            r2=(1.0 - sse / tss) if tss > 0 else float("nan"),
         )
 
     def cpu(self):
@@ -483,7 +490,14 @@ class LinearRegressionTrainingSummary:
     """Spark LinearRegressionTrainingSummary equivalent (training-set
     metrics from the fit's sufficient statistics)."""
 
-    def __init__(self, meanSquaredError: float, rootMeanSquaredError: float, numInstances: int):
+    def __init__(
+        self,
+        meanSquaredError: float,
+        rootMeanSquaredError: float,
+        numInstances: int,
+        r2: float = float("nan"),
+    ):
         self.meanSquaredError = meanSquaredError
         self.rootMeanSquaredError = rootMeanSquaredError
         self.numInstances = numInstances
+        self.r2 = r2
