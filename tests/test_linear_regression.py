@@ -140,3 +140,15 @@ def test_ols_distributed_matches_single():
     for coef, icpt in results:
         assert np.allclose(coef, sk.coef_, atol=1e-6)
         assert np.isclose(icpt, sk.intercept_, atol=1e-6)
+
+
+def test_summary_r2_matches_sklearn():
+    from sklearn.linear_model import LinearRegression as SkLR
+
+    rng = np.random.default_rng(0)
+    X = rng.normal(size=(300, 5))
+    y = X @ rng.normal(size=5) + 1.0 + 0.3 * rng.normal(size=300)
+    m = LinearRegression().fit(DataFrame.from_numpy(X, y))
+    sk = SkLR().fit(X, y)
+    # f32 ingest => ~1e-7 agreement with sklearn's f64 score
+    assert abs(m.summary.r2 - sk.score(X, y)) < 1e-5
