@@ -57,3 +57,16 @@ def test_dbscan_distributed_matches_sklearn():
     labels[0::2] = results[0]
     labels[1::2] = results[1]
     assert adjusted_rand_score(labels, sk.labels_) == 1.0
+
+
+def test_dbscan_model_persistence(tmp_model_path):
+    from spark_rapids_ml_amd.models.clustering import DBSCANModel
+
+    model = DBSCAN(eps=0.8, min_samples=7).fit(DataFrame.from_numpy(_data()[0]))
+    model.save(tmp_model_path)
+    loaded = DBSCANModel.load(tmp_model_path)
+    assert loaded.getEps() == 0.8
+    assert loaded.getMinSamples() == 7
+    X, _y = _data()
+    out = loaded.transform(DataFrame.from_numpy(X))
+    assert "prediction" in out.columns
