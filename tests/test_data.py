@@ -92,3 +92,21 @@ def test_read_parquet_shards_across_ranks(tmp_path):
     # shards are disjoint and together cover every row exactly once
     all_ids = np.concatenate([i for _, _, i in results])
     assert np.array_equal(np.sort(all_ids), ids)
+
+
+def _dist_count(_):
+    from spark_rapids_ml_amd.data import DataFrame
+    from spark_rapids_ml_amd.parallel.context import get_comm
+
+    comm = get_comm()
+    n_local = 100 + 50 * comm.rank  # uneven shards
+    df = DataFrame({"x": np.arange(n_local, dtype=np.float64)})
+    return df.num_rows, df.count()
+
+
+def test_count_is_global(tmp_path):
+    from tests.dist_utils import run_distributed
+
+    results = run_distributed(_dist_count, world_size=2, args=(None,))
+    assert results[0] == (100, 250)
+    assert results[1] == (150, 250)
