@@ -70,3 +70,48 @@ def test_dbscan_model_persistence(tmp_model_path):
     X, _y = _data()
     out = loaded.transform(DataFrame.from_numpy(X))
     assert "prediction" in out.columns
+
+
+def test_cluster_hip_logic_with_fake_kernel(monkeypatch):
+    """The GPU clustering path's sweep-loop/fixpoint/border logic, exercised
+    on CPU by faking the dbscan_sweep kernel with its torch reference
+    semantics (the real kernel's numerics are tested on GPU in
+    test_hip_ops)."""
+    import torch
+
+    from spark_rapids_ml_amd.models import clustering as mod
+
+    BIG = torch.iinfo(torch.int32).max
+
+    class FakeExt:
+        @staticmethod
+        def dbscan_sweep(Xf, x_sq, row0, n_rows, eps2, mode, core, labels):
+            d2 = torch.cdist(Xf[row0 : row0 + n_rows], Xf) ** 2
+            if mode == 0:
+                return (d2 <= eps2).sum(dim=1).to(torch.int32)
+            masked = torch.where(
+                (core[None, :] > 0) & (d2 <= eps2),
+                labels[None, :].expand(n_rows, Xf.shape[0]),
+                torch.full((1,), BIG, dtype=torch.int32),
+            )
+            return masked.min(dim=1).values
+
+    import spark_rapids_ml_amd.ops.dispatch as dispatch
+
+    monkeypatch.setattr(dispatch, "hip_ops", lambda: FakeExt)
+
+    X, y = _data(n=400)
+    from sklearn.cluster import DBSCAN as SkDBSCAN
+    from sklearn.metrics import adjusted_rand_score
+
+    model = DBSCAN(eps=0.6, min_samples=5).fit(DataFrame.from_numpy(X))
+    # call the hip path directly (CPU tensors + fake kernel)
+    from spark_rapids_ml_amd.parallel.context import PartitionDescriptor, get_comm
+
+    comm = get_comm()
+    Xf = torch.from_numpy(X.astype(np.float32))
+    labels = model._cluster_hip(
+        Xf, (Xf * Xf).sum(dim=1), 0, len(X), comm, 0.6 * 0.6, 5
+    )
+    sk = SkDBSCAN(eps=0.6, min_samples=5).fit(X)
+    assert adjusted_rand_score(sk.labels_, labels) > 0.95
