@@ -72,6 +72,11 @@ class _FitContext:
 class Estimator(_NativeParams, HasVerbose):
     """Base of all estimators (reference `_CumlEstimator`, core.py:1155)."""
 
+    # Estimators that consume CSR directly override this (LogisticRegression);
+    # everyone else densifies sparse input on ingest, matching Spark's
+    # SparseVector.toArray behavior for algorithms without a sparse path.
+    _supports_sparse = False
+
     def __init__(self) -> None:
         super().__init__()
 
@@ -102,6 +107,10 @@ class Estimator(_NativeParams, HasVerbose):
     def _extract_xy(self, df: DataFrame) -> Tuple[Any, Optional[np.ndarray]]:
         features_col, features_cols = self._get_input_columns()
         X = extract_features(df, features_col, features_cols, self._float32_inputs)
+        if _is_sparse(X) and not self._supports_sparse:
+            X = np.asarray(X.todense()).astype(
+                np.float32 if self._float32_inputs else np.float64
+            )
         y = None
         if self._is_supervised():
             label_col = self.getOrDefault("labelCol")

@@ -132,6 +132,8 @@ class LogisticRegression(_LogisticRegressionParams, Estimator):
     """Distributed logistic regression (reference LogisticRegression,
     classification.py:822)."""
 
+    _supports_sparse = True  # CSR path (reference classification.py:960-966)
+
     def __init__(self, **kwargs: Any) -> None:
         super().__init__()
         self._set_params(**kwargs)

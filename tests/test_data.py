@@ -110,3 +110,22 @@ def test_count_is_global(tmp_path):
     results = run_distributed(_dist_count, world_size=2, args=(None,))
     assert results[0] == (100, 250)
     assert results[1] == (150, 250)
+
+
+def test_sparse_input_densified_for_dense_estimators():
+    """CSR features to estimators without a sparse path densify on ingest
+    (Spark SparseVector.toArray behavior); LogisticRegression keeps CSR."""
+    import scipy.sparse as sp
+
+    from spark_rapids_ml_amd import KMeans, LinearRegression, PCA
+
+    rng = np.random.default_rng(0)
+    X = sp.random(150, 10, density=0.3, format="csr", dtype=np.float32, random_state=0)
+    y = rng.normal(size=150)
+    km = KMeans(k=3, maxIter=5, seed=1).fit(DataFrame.from_numpy(X))
+    assert km.cluster_centers_.shape == (3, 10)
+    assert PCA(k=2).fit(DataFrame.from_numpy(X)).components_.shape == (2, 10)
+    lr = LinearRegression().fit(DataFrame.from_numpy(X, y))
+    # same result as explicit densify
+    lr_d = LinearRegression().fit(DataFrame.from_numpy(np.asarray(X.todense()), y))
+    assert np.allclose(lr.coefficients, lr_d.coefficients, atol=1e-6)
