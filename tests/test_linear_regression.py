@@ -152,3 +152,21 @@ def test_summary_r2_matches_sklearn():
     sk = SkLR().fit(X, y)
     # f32 ingest => ~1e-7 agreement with sklearn's f64 score
     assert abs(m.summary.r2 - sk.score(X, y)) < 1e-5
+
+
+def test_degenerate_columns_and_labels():
+    """Constant features (sigma=0) and constant labels stay finite through
+    every solver path (standardization divides by sigma internally)."""
+    rng = np.random.default_rng(0)
+    X = rng.normal(size=(200, 5))
+    X[:, 2] = 7.0
+    y = X @ np.array([1.0, 2.0, 0.0, 3.0, -1.0]) + 0.5
+    for kw in (dict(regParam=0.1), dict(regParam=0.1, elasticNetParam=0.5, maxIter=20)):
+        m = LinearRegression(**kw).fit(DataFrame.from_numpy(X, y))
+        assert np.isfinite(np.asarray(m.coefficients)).all()
+        assert abs(np.asarray(m.coefficients)[2]) < 1e-6  # constant col gets 0
+    m = LinearRegression(regParam=0.1, elasticNetParam=0.5, maxIter=10).fit(
+        DataFrame.from_numpy(X, np.full(200, 3.0))
+    )
+    assert np.isfinite(np.asarray(m.coefficients)).all()
+    assert abs(m.intercept - 3.0) < 1e-8
