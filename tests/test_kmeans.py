@@ -132,3 +132,14 @@ def test_kmeans_summary():
     assert s.numIter >= 1
     assert sum(s.clusterSizes) == 300
     assert s.trainingCost == model.trainingCost
+
+
+def test_kmeans_degenerate_inputs():
+    # fewer distinct points than k: empty clusters allowed, centers finite
+    X = np.repeat(np.array([[0.0, 0.0], [5.0, 5.0], [9.0, 1.0]], dtype=np.float32), 20, axis=0)
+    m = KMeans(k=5, maxIter=10, seed=1).fit(DataFrame.from_numpy(X))
+    assert np.isfinite(m.cluster_centers_).all()
+    assert sum(m.summary.clusterSizes) == 60
+    # all-identical points
+    m2 = KMeans(k=2, maxIter=5, seed=1).fit(DataFrame.from_numpy(np.zeros((50, 3), np.float32)))
+    assert np.isfinite(m2.cluster_centers_).all()

@@ -88,3 +88,16 @@ def test_approx_similarity_join_alias():
     model = NearestNeighbors(k=3).fit(DataFrame.from_numpy(X))
     joined = model.approxSimilarityJoin(DataFrame.from_numpy(Q))
     assert joined.num_rows == 24
+
+
+def test_knn_k_exceeds_items_pads_with_sentinels():
+    X = _data(n=60, d=4)
+    _, _, knn = NearestNeighbors(k=100).fit(DataFrame.from_numpy(X)).kneighbors(
+        DataFrame.from_numpy(X[:3])
+    )
+    idx = np.asarray(knn["indices"])
+    dist = np.asarray(knn["distances"])
+    assert idx.shape == (3, 100)
+    assert (idx[:, :60] >= 0).all()
+    assert (idx[:, 60:] == -1).all()
+    assert np.isinf(dist[:, 60:]).all()
