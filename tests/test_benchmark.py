@@ -44,11 +44,23 @@ def test_gen_low_rank():
     assert s[0] / s[-1] > 10  # strongly low-rank spectrum
 
 
-@pytest.mark.parametrize("name", ["kmeans", "pca", "linear_regression"])
+@pytest.mark.parametrize(
+    "name",
+    [
+        "kmeans", "pca", "linear_regression", "logistic_regression",
+        "random_forest_classifier", "random_forest_regressor",
+        "nearest_neighbors", "approximate_nearest_neighbors",
+        "dbscan", "umap",
+    ],
+)
 def test_bench_smoke(name):
     argv = ["--num_rows", "2000", "--num_cols", "16"]
     if name == "kmeans":
         argv += ["--k", "8", "--maxIter", "3"]
+    if name == "umap":
+        argv += ["--n_epochs", "20"]
+    if name.startswith("random_forest"):
+        argv += ["--numTrees", "3", "--maxDepth", "3"]
     report = BENCHMARKS[name].run(argv)
     assert report["fit_sec"] > 0
     assert report["num_rows_total"] == 2000
