@@ -138,3 +138,19 @@ def test_cross_validator_estimator_persistence(tmp_model_path):
     maps = loaded.getEstimatorParamMaps()
     assert len(maps) == 2
     assert sorted(v for pm in maps for v in pm.values()) == [0.0, 0.1]
+
+
+def test_cv_smaller_is_better_metric():
+    """CV must pick argmin for loss-like metrics (isLargerBetter False)."""
+    rng = np.random.default_rng(0)
+    X = rng.normal(size=(400, 5))
+    y = X @ rng.normal(size=5) + 0.05 * rng.normal(size=400)
+    df = DataFrame.from_numpy(X, y)
+    lr = LinearRegression()
+    grid = ParamGridBuilder().addGrid(lr.getParam("regParam"), [0.0, 10.0]).build()
+    ev = RegressionEvaluator(metricName="rmse")
+    assert not ev.isLargerBetter()
+    cvm = CrossValidator(lr, grid, ev, numFolds=2, seed=1).fit(df)
+    # heavy ridge on well-posed data must lose; best = regParam 0.0
+    assert cvm.bestModel.getRegParam() == 0.0
+    assert cvm.avgMetrics[0] < cvm.avgMetrics[1]
