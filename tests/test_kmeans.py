@@ -143,3 +143,13 @@ def test_kmeans_degenerate_inputs():
     # all-identical points
     m2 = KMeans(k=2, maxIter=5, seed=1).fit(DataFrame.from_numpy(np.zeros((50, 3), np.float32)))
     assert np.isfinite(m2.cluster_centers_).all()
+
+
+def test_model_copy_with_extra_params():
+    X, _ = _make(n=200)
+    m = KMeans(k=3, maxIter=5, seed=1).fit(DataFrame.from_numpy(X))
+    m2 = m.copy({m.getParam("predictionCol"): "p2"})
+    assert m2.getOrDefault("predictionCol") == "p2"
+    assert m.getOrDefault("predictionCol") == "prediction"  # original untouched
+    assert np.allclose(m2.cluster_centers_, m.cluster_centers_)
+    assert "p2" in m2.transform(DataFrame.from_numpy(X)).columns
