@@ -211,3 +211,30 @@ def test_noncontiguous_label_values():
     assert set(np.unique(pred)).issubset(set(classes))
     # sklearn reaches 0.813 train accuracy on this dataset; match it
     assert (pred == y).mean() > 0.79
+
+
+def test_standardized_penalty_objective():
+    """standardization=True (default): the returned raw coefficients minimize
+    mean logloss + lam/2 * sum((w_j * sigma_j)^2) — Spark's standardized-space
+    L2 penalty (ddof=1 column std, no centering). Verified against a scipy
+    optimum of the same objective."""
+    from scipy.optimize import minimize
+
+    rng = np.random.default_rng(0)
+    X, y = make_classification(n_samples=400, n_features=6, random_state=0)
+    X = (X * np.array([1.0, 5.0, 0.2, 1.0, 3.0, 0.5])).astype(np.float64)  # varied scales
+    lam = 0.05
+    model = LogisticRegression(regParam=lam, maxIter=300, tol=1e-12).fit(
+        DataFrame.from_numpy(X.astype(np.float32), y.astype(np.float64))
+    )
+    sigma = X.std(axis=0, ddof=1)
+
+    def obj(wb):
+        w, b = wb[:6], wb[6]
+        z = X @ w + b
+        t = 2 * y - 1
+        return float(np.mean(np.logaddexp(0, -t * z)) + lam / 2 * np.sum((w * sigma) ** 2))
+
+    ours = obj(np.concatenate([np.asarray(model.coefficients), [model.intercept]]))
+    ref = minimize(obj, np.zeros(7), method="L-BFGS-B", options={"maxiter": 2000}).fun
+    assert ours <= ref * (1 + 1e-5), (ours, ref)
