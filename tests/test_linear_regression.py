@@ -170,3 +170,51 @@ def test_degenerate_columns_and_labels():
     )
     assert np.isfinite(np.asarray(m.coefficients)).all()
     assert abs(m.intercept - 3.0) < 1e-8
+
+
+def test_ridge_standardized_penalty_objective():
+    """standardization=True ridge: raw coefficients minimize
+    1/(2n)||y-Xw-b||^2 + lam/2 * sum((w_j*sigma_j)^2) (the label-sigma
+    factors cancel in raw space). Verified against a scipy optimum."""
+    from scipy.optimize import minimize
+
+    rng = np.random.default_rng(0)
+    X = rng.normal(size=(300, 4)) * np.array([1.0, 4.0, 0.3, 2.0])
+    y = X @ np.array([1.0, -0.5, 2.0, 0.7]) + 1.5 + 0.2 * rng.normal(size=300)
+    lam = 0.3
+    model = LinearRegression(regParam=lam).fit(DataFrame.from_numpy(X, y))
+    sigma = X.std(axis=0, ddof=1)
+    n = len(y)
+
+    def obj(wb):
+        w, b = wb[:4], wb[4]
+        r = y - X @ w - b
+        return float(r @ r / (2 * n) + lam / 2 * np.sum((w * sigma) ** 2))
+
+    ours = obj(np.concatenate([np.asarray(model.coefficients), [model.intercept]]))
+    ref = minimize(obj, np.zeros(5), method="L-BFGS-B", options={"maxiter": 2000}).fun
+    assert ours <= ref * (1 + 1e-6), (ours, ref)
+
+
+def test_elasticnet_matches_sklearn_on_standardized_data():
+    """On unit-variance features the Spark elastic-net objective maps onto
+    sklearn's with alpha' = lam*(a*sy + (1-a)), l1_ratio' = a*sy/(a*sy+(1-a))
+    where sy is the label std (Spark standardizes the label; the L1 term
+    scales linearly in sy, the L2/loss terms quadratically)."""
+    from sklearn.linear_model import ElasticNet as SkEN
+
+    rng = np.random.default_rng(0)
+    X = rng.normal(size=(400, 6))
+    X = (X - X.mean(0)) / X.std(0, ddof=1)  # unit sigma (ddof=1)
+    y = X @ np.array([2.0, -1.0, 0.0, 0.0, 1.5, 0.0]) + 0.5 + 0.1 * rng.normal(size=400)
+    lam, a = 0.1, 0.5
+    model = LinearRegression(regParam=lam, elasticNetParam=a, maxIter=500, tol=1e-12).fit(
+        DataFrame.from_numpy(X, y)
+    )
+    sy = y.std(ddof=1)
+    alpha_sk = lam * (a * sy + (1 - a))
+    l1r_sk = a * sy / (a * sy + (1 - a))
+    sk = SkEN(alpha=alpha_sk, l1_ratio=l1r_sk, max_iter=50000, tol=1e-12).fit(X, y)
+    assert np.allclose(np.asarray(model.coefficients), sk.coef_, atol=2e-4), (
+        np.asarray(model.coefficients), sk.coef_)
+    assert abs(model.intercept - sk.intercept_) < 1e-3
