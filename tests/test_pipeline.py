@@ -68,3 +68,30 @@ def test_pipeline_persistence(tmp_model_path):
     loaded = PipelineModel.load(os.path.join(tmp_model_path, "pm"))
     out = loaded.transform(df)
     assert np.array_equal(np.asarray(out["prediction"]), np.asarray(pm.transform(df)["prediction"]))
+
+
+def _dist_pipeline(_):
+    from spark_rapids_ml_amd import LogisticRegression
+    from spark_rapids_ml_amd.parallel.context import get_comm
+    from spark_rapids_ml_amd.pipeline import Pipeline, VectorAssembler
+
+    comm = get_comm()
+    rng = np.random.default_rng(0)
+    a = rng.normal(size=1000).astype(np.float32)
+    b = rng.normal(size=1000).astype(np.float32)
+    lab = (a + 0.5 * b > 0).astype(np.float64)
+    sl = slice(comm.rank, None, comm.world_size)
+    df = DataFrame({"a": a[sl], "b": b[sl], "label": lab[sl]})
+    pm = Pipeline([VectorAssembler(["a", "b"]), LogisticRegression(maxIter=40)]).fit(df)
+    out = pm.transform(df)
+    return np.asarray(pm.stages[-1].coefficients), (np.asarray(out["prediction"]) == lab[sl]).mean()
+
+
+def test_pipeline_distributed():
+    from tests.dist_utils import run_distributed
+
+    results = run_distributed(_dist_pipeline, world_size=2, args=(None,))
+    coef0, acc0 = results[0]
+    coef1, acc1 = results[1]
+    assert np.allclose(coef0, coef1)  # replicated model
+    assert acc0 > 0.95 and acc1 > 0.95
