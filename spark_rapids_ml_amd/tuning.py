@@ -54,6 +54,7 @@ class CrossValidator(Params):
     seed = Param("cv", "seed", "fold-split seed.", TypeConverters.toInt)
     parallelism = Param("cv", "parallelism", "accepted for API parity; folds run sequentially (SPMD collectives).", TypeConverters.toInt)
     collectSubModels = Param("cv", "collectSubModels", "keep all sub-models.", TypeConverters.toBoolean)
+    foldCol = Param("cv", "foldCol", "user-supplied integer fold column (pyspark CV parity).", TypeConverters.toString)
 
     def __init__(
         self,
@@ -64,10 +65,14 @@ class CrossValidator(Params):
         seed: int = 42,
         parallelism: int = 1,
         collectSubModels: bool = False,
+        foldCol: str = "",
     ) -> None:
         super().__init__()
-        self._setDefault(numFolds=3, seed=42, parallelism=1, collectSubModels=False)
-        self._set(numFolds=numFolds, seed=seed, parallelism=parallelism)
+        self._setDefault(numFolds=3, seed=42, parallelism=1, collectSubModels=False, foldCol="")
+        self._set(
+            numFolds=numFolds, seed=seed, parallelism=parallelism,
+            collectSubModels=collectSubModels, foldCol=foldCol,
+        )
         self._estimator = estimator
         self._est_param_maps = estimatorParamMaps or [{}]
         self._evaluator = evaluator
@@ -164,9 +169,17 @@ class CrossValidator(Params):
         (Spark assigns by rand(seed) — same semantics, shard-local)."""
         comm = get_comm()
         n_folds = self.getOrDefault("numFolds")
-        seed = self.getOrDefault("seed")
-        rng = np.random.default_rng(seed + 1000003 * comm.rank)
-        assign = rng.integers(0, n_folds, size=df.num_rows)
+        fold_col = self.getOrDefault("foldCol")
+        if fold_col:
+            assign = np.asarray(df[fold_col]).astype(np.int64)
+            if assign.size and (assign.min() < 0 or assign.max() >= n_folds):
+                raise ValueError(
+                    f"foldCol values must lie in [0, numFolds={n_folds})"
+                )
+        else:
+            seed = self.getOrDefault("seed")
+            rng = np.random.default_rng(seed + 1000003 * comm.rank)
+            assign = rng.integers(0, n_folds, size=df.num_rows)
         folds = []
         for f in range(n_folds):
             test_idx = np.nonzero(assign == f)[0]

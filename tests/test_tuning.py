@@ -154,3 +154,18 @@ def test_cv_smaller_is_better_metric():
     # heavy ridge on well-posed data must lose; best = regParam 0.0
     assert cvm.bestModel.getRegParam() == 0.0
     assert cvm.avgMetrics[0] < cvm.avgMetrics[1]
+
+
+def test_cv_fold_col():
+    rng = np.random.default_rng(0)
+    X = rng.normal(size=(300, 4))
+    y = X @ rng.normal(size=4)
+    folds = np.arange(300) % 3
+    df = DataFrame({"features": X, "label": y, "fold": folds.astype(np.int64)})
+    lr = LinearRegression()
+    grid = ParamGridBuilder().addGrid(lr.getParam("regParam"), [0.0, 0.1]).build()
+    cvm = CrossValidator(lr, grid, RegressionEvaluator(), numFolds=3, foldCol="fold").fit(df)
+    assert len(cvm.avgMetrics) == 2
+    bad = DataFrame({"features": X, "label": y, "fold": (folds + 5).astype(np.int64)})
+    with pytest.raises(ValueError, match="foldCol"):
+        CrossValidator(lr, grid, RegressionEvaluator(), numFolds=3, foldCol="fold").fit(bad)
