@@ -113,3 +113,37 @@ def test_three_rank_world():
     results = run_distributed(_dist_empty_shard, world_size=3, args=("kmeans",))
     assert np.allclose(results[0], results[1])
     assert np.allclose(results[1], results[2])
+
+
+def test_empty_transform_all_models():
+    """transform of a 0-row shard works for every model family (an 8-GPU
+    transform on a small dataset leaves ranks empty)."""
+    from sklearn.datasets import make_blobs, make_classification
+
+    from spark_rapids_ml_amd import (
+        KMeans,
+        LinearRegression,
+        LogisticRegression,
+        PCA,
+        RandomForestClassifier,
+        UMAP,
+    )
+
+    Xb, _ = make_blobs(n_samples=200, n_features=6, centers=3, random_state=0)
+    Xb = Xb.astype(np.float32)
+    Xc, yc = make_classification(n_samples=200, n_features=6, random_state=0)
+    empty = DataFrame.from_numpy(Xb[:0])
+    models = [
+        KMeans(k=3, maxIter=5, seed=1).fit(DataFrame.from_numpy(Xb)),
+        PCA(k=2).fit(DataFrame.from_numpy(Xb)),
+        LinearRegression().fit(DataFrame.from_numpy(Xc, yc.astype(np.float64))),
+        LogisticRegression(maxIter=20).fit(
+            DataFrame.from_numpy(Xc.astype(np.float32), yc.astype(np.float64))
+        ),
+        RandomForestClassifier(numTrees=3, maxDepth=3).fit(
+            DataFrame.from_numpy(Xc.astype(np.float32), yc.astype(np.float64))
+        ),
+        UMAP(n_neighbors=10, n_epochs=20, random_state=1).fit(DataFrame.from_numpy(Xb)),
+    ]
+    for m in models:
+        assert m.transform(empty).num_rows == 0
