@@ -126,10 +126,13 @@ def test_launch_two_ranks_gloo(tmp_path):
         "c = get_comm()\n"
         "print(f'LAUNCH_RANK {c.rank}/{c.world_size}')\n"
     )
+    import random
+
+    port = str(random.randint(21000, 39000))
     env = dict(os.environ, SRML_BACKEND="gloo")
     r = subprocess.run(
         [sys.executable, "-m", "spark_rapids_ml_amd.launch", "--gpus", "2",
-         "--master-port", "29719", str(script)],
+         "--master-port", port, str(script)],
         env=env, capture_output=True, text=True, timeout=240,
     )
     assert r.returncode == 0, r.stderr
