@@ -47,7 +47,9 @@ def run_distributed(fn: Callable, world_size: int = 2, args: tuple = ()) -> List
     per-rank results ordered by rank. Raises on any rank failure."""
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    port = random.randint(20000, 40000)
+    # keep below the Linux ephemeral range (32768+): an active outbound
+    # connection can hold a port there and the gloo rendezvous bind fails
+    port = random.randint(20000, 32000)
     procs = []
     for r in range(world_size):
         p = ctx.Process(

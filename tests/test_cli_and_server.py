@@ -128,13 +128,17 @@ def test_launch_two_ranks_gloo(tmp_path):
     )
     import random
 
-    port = str(random.randint(21000, 39000))
     env = dict(os.environ, SRML_BACKEND="gloo")
-    r = subprocess.run(
-        [sys.executable, "-m", "spark_rapids_ml_amd.launch", "--gpus", "2",
-         "--master-port", port, str(script)],
-        env=env, capture_output=True, text=True, timeout=240,
-    )
+    r = None
+    for _ in range(2):  # retry once on a port collision
+        port = str(random.randint(21000, 32000))  # below the ephemeral range
+        r = subprocess.run(
+            [sys.executable, "-m", "spark_rapids_ml_amd.launch", "--gpus", "2",
+             "--master-port", port, str(script)],
+            env=env, capture_output=True, text=True, timeout=240,
+        )
+        if r.returncode == 0:
+            break
     assert r.returncode == 0, r.stderr
     assert "LAUNCH_RANK 0/2" in r.stdout
     assert "LAUNCH_RANK 1/2" in r.stdout
