@@ -60,15 +60,17 @@ class ModelRegistry:
 
 
 def _check_path(path: str) -> str:
-    """Optional confinement: with SRML_SERVER_DATA_ROOT set, data/output
-    paths must resolve under it (the service reads/writes as its own user)."""
+    """Filesystem confinement: data/output paths must resolve under
+    SRML_SERVER_DATA_ROOT (defaults to the server's CWD, so a reachable
+    client cannot read/write arbitrary paths as the service user; set the
+    env to widen or relocate the root)."""
     import os
 
-    root = os.environ.get("SRML_SERVER_DATA_ROOT")
-    if root:
-        real = os.path.realpath(path)
-        if not real.startswith(os.path.realpath(root) + os.sep):
-            raise ValueError(f"path {path!r} outside SRML_SERVER_DATA_ROOT")
+    root = os.environ.get("SRML_SERVER_DATA_ROOT") or os.getcwd()
+    real = os.path.realpath(path)
+    root_real = os.path.realpath(root)
+    if real != root_real and not real.startswith(root_real + os.sep):
+        raise ValueError(f"path {path!r} outside SRML_SERVER_DATA_ROOT ({root!r})")
     return path
 
 

@@ -173,6 +173,16 @@ class Estimator(_NativeParams, HasVerbose):
         pdesc = PartitionDescriptor.build(comm, X.shape[0], X.shape[1], nnz=nnz)
         if pdesc.m == 0:
             raise RuntimeError("Dataset is empty across all ranks.")
+        if X.shape[0] == 0 and X.shape[1] != pdesc.n:
+            # empty shard of a variable-width list parquet column reads as
+            # (0,0); realign the local width to the global one so per-rank
+            # buffer shapes agree in the fused collectives
+            if _is_sparse(X):
+                import scipy.sparse as sp
+
+                X = sp.csr_matrix((0, pdesc.n), dtype=X.dtype)
+            else:
+                X = np.zeros((0, pdesc.n), dtype=X.dtype)
         ctx = _FitContext(comm, pdesc, comm.device)
         logger = get_logger(self.__class__)
 

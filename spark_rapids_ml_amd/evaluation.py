@@ -141,13 +141,21 @@ class BinaryClassificationEvaluator(Evaluator, HasLabelCol):
         if len(pos) == 0 or len(neg) == 0:
             return 0.5
         if self.getOrDefault("metricName") == "areaUnderPR":
+            # Spark BinaryClassificationMetrics semantics: PR curve at
+            # DISTINCT score thresholds (ties merged into one confusion
+            # point), a (0, p_first) start point, trapezoidal integration.
             order = np.argsort(-s, kind="stable")
+            so = s[order]
             tp = np.cumsum(l[order] == 1.0)
             fp = np.cumsum(l[order] != 1.0)
+            # last index of each distinct-score run = the curve points
+            last = np.nonzero(np.diff(so, append=np.nan))[0]
+            tp, fp = tp[last], fp[last]
             precision = tp / np.maximum(tp + fp, 1)
             recall = tp / len(pos)
-            # step-wise area with the (0, p0) start point (Spark/sklearn AP-style)
-            return float(np.sum(precision * np.diff(recall, prepend=0.0)))
+            recall = np.concatenate([[0.0], recall])
+            precision = np.concatenate([[precision[0]], precision])
+            return float(np.trapz(precision, recall))
         from scipy.stats import rankdata
 
         ranks = rankdata(np.concatenate([pos, neg]))

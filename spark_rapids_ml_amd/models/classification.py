@@ -178,12 +178,12 @@ class LogisticRegression(_LogisticRegressionParams, Estimator):
         self, X: Any, y: Optional[Any], ctx: _FitContext, params: Dict[str, Any]
     ) -> Dict[str, Any]:
         comm, pdesc = ctx.comm, ctx.pdesc
-        lam = float(self.getOrDefault("regParam"))
-        # reference translates regParam->C=1/(regParam) for cuML
-        # (classification.py:721-744); the native solver here takes λ directly.
-        if "C_override" in params:
-            lam = 1.0 / params["C_override"] if params["C_override"] else 0.0
-        l1r = float(self.getOrDefault("elasticNetParam"))
+        # Per-param-map values land under the native keys 'C'/'l1_ratio'
+        # (mapped from regParam/elasticNetParam); the native solver takes λ
+        # directly, so 'C' carries λ semantics here (reference translates
+        # regParam->C=1/regParam for cuML, classification.py:721-744).
+        lam = float(params.get("C", self.getOrDefault("regParam")))
+        l1r = float(params.get("l1_ratio", self.getOrDefault("elasticNetParam")))
         max_iter = int(params["max_iter"])
         tol = float(params["tol"])
         fit_intercept = bool(params["fit_intercept"])

@@ -786,7 +786,7 @@ class _RandomForestModel(_RandomForestParams, Model):
                     lines.append(f"{pad}Predict: {t['value'][node].tolist()}")
                 else:
                     lines.append(
-                        f"{pad}If (feature {t['feature'][node]} <= "
+                        f"{pad}If (feature {t['feature'][node]} < "
                         f"{t['threshold'][node]:.6g})"
                     )
                     rec(int(t["left"][node]), depth + 1)
@@ -844,7 +844,10 @@ class _RandomForestModel(_RandomForestParams, Model):
                     break
                 f = feature[node].clamp(min=0)
                 xv = Xt.gather(1, f.view(-1, 1)).flatten()
-                go_left = xv <= thr[node]
+                # strict <: training partitions on bin(x) <= b  <=>  x <
+                # edges[b] (=stored threshold), so rows equal to the
+                # threshold went RIGHT during fit
+                go_left = xv < thr[node]
                 nxt = torch.where(go_left, left[node], right[node])
                 node = torch.where(at_leaf, node, nxt)
             v = value[node]

@@ -231,14 +231,19 @@ class PartitionDescriptor:
 
     @classmethod
     def build(cls, comm: Comm, n_local_rows: int, n_cols: int, nnz: Optional[int] = None) -> "PartitionDescriptor":
-        payload = {"rank": comm.rank, "rows": int(n_local_rows)}
+        payload = {"rank": comm.rank, "rows": int(n_local_rows), "cols": int(n_cols)}
         if nnz is not None:
             payload["nnz"] = int(nnz)
         gathered = comm.allgather_obj(payload)
         parts = [(g["rank"], g["rows"]) for g in gathered]
         m = sum(r for _, r in parts)
+        # max over ranks: an EMPTY shard of a variable-width list parquet
+        # column cannot know the feature width locally (it reads as (0,0));
+        # sizing fused all-reduce buffers from a rank-local 0 would deadlock
+        # the collective (buffer shape mismatch across ranks)
+        n_global = max(g.get("cols", 0) for g in gathered)
         total_nnz = sum(g.get("nnz", 0) for g in gathered) if nnz is not None else None
-        return cls(parts_rank_size=parts, m=m, n=int(n_cols), rank=comm.rank, total_nnz=total_nnz)
+        return cls(parts_rank_size=parts, m=m, n=int(n_global), rank=comm.rank, total_nnz=total_nnz)
 
     def row_offset(self) -> int:
         """Global row offset of this rank's first local row."""

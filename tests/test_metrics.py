@@ -122,7 +122,26 @@ def test_auc_matches_sklearn():
     assert np.isclose(auc, roc_auc_score(y, score), atol=1e-9)
 
 
-def test_area_under_pr_matches_sklearn():
+def _spark_pr_auc(y: np.ndarray, score: np.ndarray) -> float:
+    """Independent oracle for Spark BinaryClassificationMetrics.areaUnderPR:
+    PR points at distinct thresholds descending, (0, p_first) start point,
+    trapezoidal rule (reference semantics; Spark BinaryClassificationMetrics)."""
+    pts = []
+    for t in np.unique(score)[::-1]:
+        sel = score >= t
+        tp = float(((y == 1) & sel).sum())
+        fp = float(((y != 1) & sel).sum())
+        prec = tp / max(tp + fp, 1.0)
+        rec = tp / max(float((y == 1).sum()), 1.0)
+        pts.append((rec, prec))
+    pts = [(0.0, pts[0][1])] + pts
+    area = 0.0
+    for (r0, p0), (r1, p1) in zip(pts[:-1], pts[1:]):
+        area += (r1 - r0) * (p0 + p1) / 2.0
+    return area
+
+
+def test_area_under_pr_spark_semantics():
     from sklearn.metrics import average_precision_score
 
     from spark_rapids_ml_amd.data import DataFrame
@@ -133,4 +152,20 @@ def test_area_under_pr_matches_sklearn():
     score = y + rng.normal(scale=1.0, size=400)
     df = DataFrame({"label": y, "rawPrediction": np.stack([-score, score], axis=1)})
     pr = BinaryClassificationEvaluator(metricName="areaUnderPR").evaluate(df)
-    assert np.isclose(pr, average_precision_score(y, score), atol=1e-9)
+    assert np.isclose(pr, _spark_pr_auc(y, score), atol=1e-9)
+    # trapezoid over a fine curve stays near (not equal to) sklearn AP
+    assert abs(pr - average_precision_score(y, score)) < 0.02
+
+
+def test_area_under_pr_tie_heavy():
+    """Ties must collapse to one confusion point (Spark distinct-threshold
+    semantics) before integration."""
+    from spark_rapids_ml_amd.data import DataFrame
+    from spark_rapids_ml_amd.evaluation import BinaryClassificationEvaluator
+
+    rng = np.random.default_rng(7)
+    y = rng.integers(0, 2, size=300).astype(np.float64)
+    score = np.round(y + rng.normal(scale=1.0, size=300), 1)  # heavy ties
+    df = DataFrame({"label": y, "rawPrediction": np.stack([-score, score], axis=1)})
+    pr = BinaryClassificationEvaluator(metricName="areaUnderPR").evaluate(df)
+    assert np.isclose(pr, _spark_pr_auc(y, score), atol=1e-9)

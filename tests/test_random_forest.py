@@ -160,3 +160,21 @@ def test_rf_feature_importances_and_debug_string():
     assert "Tree 0" in s and "feature" in s
     assert model.treeWeights == [1.0] * 10
     assert model.totalNumNodes > 10
+
+
+def test_rf_threshold_equality_consistent_with_training():
+    """Rows whose feature value EQUALS a split threshold must take the same
+    branch at predict time as at fit time (round-1 advisor finding: training
+    partitions on x < edges[b] but inference used <=). Integer-valued,
+    perfectly separable data makes every split land exactly on a data value."""
+    rng = np.random.default_rng(0)
+    X = rng.integers(0, 10, size=(800, 4)).astype(np.float32)
+    y = (X[:, 0] >= 5).astype(np.float64)  # separable on an integer boundary
+    df = DataFrame({"features": X, "label": y})
+    m = RandomForestClassifier(
+        featuresCol="features", labelCol="label", numTrees=1, maxDepth=4,
+        bootstrap=False, seed=3,
+    ).fit(df)
+    pred = np.asarray(m.transform(df)[m.getOrDefault("predictionCol")])
+    acc = float((pred == y).mean())
+    assert acc == 1.0, f"train accuracy {acc} < 1.0 on separable integer data"
