@@ -101,6 +101,26 @@ def main() -> None:
     value = total_rows * args.steps / elapsed
     ms_per_step = elapsed / args.steps * 1000.0
 
+    # Whole-job fit wall-clock (the honest vs-baseline basis):
+    # host→device ingest + k init + 30 Lloyd iterations through the real
+    # estimator path, exactly the job the reference's 82 s covers.
+    from spark_rapids_ml_amd import KMeans as _KMeans
+    from spark_rapids_ml_amd.data import DataFrame as _DF
+
+    Xh = X.cpu().numpy()
+    del X, x_sq
+    if on_gpu:
+        torch.cuda.empty_cache()
+    fit_iters = 30 if on_gpu else 5
+    comm.barrier()
+    t0 = time.perf_counter()
+    _KMeans(k=k, maxIter=fit_iters, initMode="random", seed=5).fit(_DF.from_numpy(Xh))
+    if on_gpu:
+        torch.cuda.synchronize(dev)
+    comm.barrier()
+    fit_wall_s = comm.allreduce_scalar(time.perf_counter() - t0, "max")
+    fit_rows_per_s = total_rows * fit_iters / fit_wall_s
+
     if comm.rank == 0:
         out = {
             "metric": "kmeans_fit_rows_per_s",
@@ -112,7 +132,13 @@ def main() -> None:
             "ms_per_step": ms_per_step,
             "higher_is_better": True,
             "scaling": "weak",
-            "vs_baseline": value / BASELINE_ROWS_PER_S if on_gpu else None,
+            # whole-job ÷ whole-job: full fit() wall-clock (ingest+init+30
+            # Lloyd iters) vs the reference's published 82 s on the same
+            # 1M×3000 k=1000 workload — NOT the steady-state step rate
+            "vs_baseline": fit_rows_per_s / BASELINE_ROWS_PER_S if on_gpu else None,
+            "fit_wall_s": fit_wall_s,
+            "fit_rows_per_s": fit_rows_per_s,
+            "steady_state_vs_baseline": value / BASELINE_ROWS_PER_S if on_gpu else None,
             "dtype": "fp32",
             "data": "synthetic",
             "config": {
@@ -120,6 +146,7 @@ def main() -> None:
                 "rows_per_gpu": rows,
                 "cols": cols,
                 "k": k,
+                "maxIter_full_fit": fit_iters,
                 "global_batch": total_rows,
                 "seq_len": None,
                 "parallelism": f"dp{n_gpus}",

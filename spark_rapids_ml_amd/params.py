@@ -572,6 +572,12 @@ class _NativeParams(_NativeClass, Params):
                 self._sync_to_native(k, v)
             elif k in self._native_params:
                 # native-only param set directly (reference params.py:474-481)
+                if k in ("n_streams", "max_samples_per_batch"):
+                    # cuML-parity ctor knobs with no effect in this engine
+                    # (HIP kernels don't batch/stream the cuML way)
+                    from .config import warn_inert
+
+                    warn_inert(k, type(self).__name__)
                 self._native_params[k] = v
             else:
                 raise ValueError(f"Unsupported param '{k}'.")
