@@ -277,8 +277,27 @@ class Model(_NativeParams, HasVerbose):
         on CPU natively when no GPU is present, so the base conversion is the
         model itself. Subclasses with an exact scikit-learn counterpart
         (KMeans, PCA, linear/logistic regression) return a fitted sklearn
-        object instead."""
+        object instead. For a STOCK-pyspark-loadable artifact use
+        saveAsSparkModel(path)."""
         return self
+
+    def saveAsSparkModel(self, path: str, overwrite: bool = False) -> None:
+        """Write this model in Apache Spark ML's own persistence format
+        (metadata/ + data/ parquet with VectorUDT/MatrixUDT columns) so stock
+        `pyspark.ml.*Model.load(path)` reads it — the reference's
+        Spark-loadable-model capability (reference core.py:268-355 +
+        utils.py:579-809) without a JVM."""
+        from .spark.persist import save_spark_model
+
+        save_spark_model(self, path, overwrite=overwrite)
+
+    @staticmethod
+    def loadFromSparkModel(path: str) -> "Model":
+        """Load a model directory written by stock Spark ML (or by
+        saveAsSparkModel) into the corresponding model of this package."""
+        from .spark.persist import load_spark_model
+
+        return load_spark_model(path)
 
     @staticmethod
     def _combine(models: Sequence["Model"]) -> "Model":
