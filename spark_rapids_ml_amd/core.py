@@ -125,6 +125,14 @@ class Estimator(_NativeParams, HasVerbose):
         if params:
             est = self.copy(params)
             return est.fit(df)
+        from .spark.bridge import is_pyspark_dataframe
+
+        if is_pyspark_dataframe(df):
+            # real pyspark.sql.DataFrame: Arrow export -> per-rank shards
+            # (reference fits via barrier mapInPandas, core.py:742-1013)
+            from .spark.bridge import fit_on_spark
+
+            return fit_on_spark(self, df)
         return self._fit(df)
 
     def _fit(self, df: DataFrame) -> "Model":
@@ -251,7 +259,15 @@ class Model(_NativeParams, HasVerbose):
     def transform(self, df: DataFrame) -> DataFrame:
         """Append output columns to the local shard (reference
         `_CumlModelWithColumns._transform`, core.py:1797: a pandas_udf over
-        the feature struct; here a direct device-batched call)."""
+        the feature struct; here a direct device-batched call). A real
+        pyspark DataFrame round-trips through the Arrow bridge and comes
+        back as a pyspark DataFrame."""
+        from .spark.bridge import is_pyspark_dataframe
+
+        if is_pyspark_dataframe(df):
+            from .spark.bridge import transform_on_spark
+
+            return transform_on_spark(self, df)
         features_col, features_cols = self._get_input_columns()
         X = extract_features(df, features_col, features_cols, self._float32_inputs)
         with annotate(f"srml::{type(self).__name__}.transform"):

@@ -107,8 +107,84 @@ def install_aliases(alias_root: str = "spark_rapids_ml") -> None:
         setattr(root_mod, sub, m)
 
 
+class _PysparkProxyModule(types.ModuleType):
+    """Module proxy over a real pyspark.ml submodule: accelerated class
+    lookups resolve to the spark_rapids_ml_amd classes, everything else
+    falls through to stock pyspark (reference install.py:22-81)."""
+
+    def __init__(self, name: str, orig: types.ModuleType, accel: Dict[str, object]):
+        super().__init__(name, getattr(orig, "__doc__", None))
+        self._srml_orig = orig
+        self._srml_accel = accel
+        self.__file__ = getattr(orig, "__file__", None)
+        self.__package__ = getattr(orig, "__package__", None)
+
+    def __getattr__(self, name: str):
+        accel = self.__dict__["_srml_accel"]
+        if name in accel:
+            return accel[name]
+        return getattr(self.__dict__["_srml_orig"], name)
+
+
+# pyspark.ml submodules the reference proxies (install.py:22-49) and the
+# accelerated class names each one redirects
+_PYSPARK_PROXY: Dict[str, List[str]] = {
+    "feature": ["PCA", "PCAModel"],
+    "clustering": ["KMeans", "KMeansModel"],
+    "classification": [
+        "LogisticRegression",
+        "LogisticRegressionModel",
+        "RandomForestClassifier",
+        "RandomForestClassificationModel",
+    ],
+    "regression": [
+        "LinearRegression",
+        "LinearRegressionModel",
+        "RandomForestRegressor",
+        "RandomForestRegressionModel",
+    ],
+    "tuning": ["CrossValidator", "CrossValidatorModel"],
+}
+
+
+def accelerate_pyspark() -> List[str]:
+    """Install proxy modules over the REAL `pyspark.ml.{feature,clustering,
+    classification,regression,tuning}` so unmodified pyspark scripts resolve
+    the accelerated classes (reference install.py:22-81). Returns the list
+    of proxied module names. Raises ImportError when pyspark is absent."""
+    import importlib
+
+    import pyspark.ml as pml  # raises ImportError without pyspark
+
+    import spark_rapids_ml_amd as root
+
+    proxied: List[str] = []
+    for sub, names in _PYSPARK_PROXY.items():
+        full = f"pyspark.ml.{sub}"
+        orig = sys.modules.get(full) or importlib.import_module(full)
+        if isinstance(orig, _PysparkProxyModule):
+            proxied.append(full)
+            continue
+        accel = {n: getattr(root, n) for n in names if hasattr(root, n)}
+        if sub == "tuning":
+            from .tuning import CrossValidator, CrossValidatorModel
+
+            accel["CrossValidator"] = CrossValidator
+            accel["CrossValidatorModel"] = CrossValidatorModel
+        proxy = _PysparkProxyModule(full, orig, accel)
+        sys.modules[full] = proxy
+        setattr(pml, sub, proxy)
+        proxied.append(full)
+    return proxied
+
+
 def main() -> None:  # console-script parity with reference install.py
     install_aliases()
+    try:
+        proxied = accelerate_pyspark()
+        print(f"pyspark.ml proxies installed: {', '.join(proxied)}")
+    except ImportError:
+        print("pyspark not installed; only spark_rapids_ml aliases active")
     print("spark_rapids_ml -> spark_rapids_ml_amd aliases installed")
 
 
