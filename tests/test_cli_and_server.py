@@ -52,11 +52,14 @@ print("CENTERS", model.cluster_centers_.shape)
 
 
 @pytest.fixture
-def client(tmp_path):
+def client(tmp_path, monkeypatch):
     from starlette.testclient import TestClient
 
     from spark_rapids_ml_amd.connect_server import create_app
 
+    # the server confines data/output paths to SRML_SERVER_DATA_ROOT
+    # (defaulting to its CWD); point it at the test's tmp dir
+    monkeypatch.setenv("SRML_SERVER_DATA_ROOT", str(tmp_path))
     return TestClient(create_app())
 
 
@@ -142,3 +145,13 @@ def test_launch_two_ranks_gloo(tmp_path):
     assert r.returncode == 0, r.stderr
     assert "LAUNCH_RANK 0/2" in r.stdout
     assert "LAUNCH_RANK 1/2" in r.stdout
+
+
+def test_server_rejects_out_of_root_paths(client):
+    # default confinement (advisor finding): without an explicit wider root,
+    # arbitrary filesystem paths must be refused
+    r = client.post(
+        "/fit",
+        json={"estimator": "KMeans", "params": {"k": 2}, "data_path": "/etc/passwd"},
+    )
+    assert r.status_code == 400
