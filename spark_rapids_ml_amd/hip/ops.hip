@@ -1718,6 +1718,217 @@ __global__ __launch_bounds__(256) void rf_best_split_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// rf_partition: counting-sort rows by batch-local node id -> (perm, seg_off).
+// Replaces the per-batch torch sort + nonzero + gather chain (radix sort was
+// ~8% of RF fit kernel time; reference behavior: cuML's batched node trainer
+// keeps per-node row lists, tree.py:384-389).
+// Two kernels: block-privatized LDS count, then a two-pass scatter where each
+// block reserves per-node ranges with ONE global atomic per (block, node)
+// and places its rows via LDS-local cursors (a single global cursor per node
+// would serialize ~n atomics on one word at depth 0).
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(256) void rf_node_count_kernel(
+    const int64_t* __restrict__ node_of_row,  // [n]
+    const int64_t* __restrict__ lut,          // [n_nodes] node -> slot | -1
+    int64_t n, int B,
+    int32_t* __restrict__ counts) {           // [B] pre-zeroed
+  extern __shared__ int32_t lcnt0[];
+  for (int e = threadIdx.x; e < B; e += 256) lcnt0[e] = 0;
+  __syncthreads();
+  const int64_t stride = (int64_t)gridDim.x * 256;
+  for (int64_t i = (int64_t)blockIdx.x * 256 + threadIdx.x; i < n; i += stride) {
+    const int64_t sl = lut[node_of_row[i]];
+    if (sl >= 0) atomicAdd(&lcnt0[sl], 1);
+  }
+  __syncthreads();
+  for (int e = threadIdx.x; e < B; e += 256)
+    if (lcnt0[e]) atomicAdd(&counts[e], lcnt0[e]);
+}
+
+__global__ __launch_bounds__(256) void rf_partition_scatter_kernel(
+    const int64_t* __restrict__ node_of_row,  // [n]
+    const int64_t* __restrict__ lut,          // node -> slot | -1
+    int64_t n, int B,
+    int32_t* __restrict__ cursor,             // [B] global, init = seg_off[:B]
+    int64_t* __restrict__ perm) {             // [>= total]
+  extern __shared__ int32_t smem[];           // lcnt[B] + lbase[B]
+  int32_t* lcnt = smem;
+  int32_t* lbase = smem + B;
+  for (int e = threadIdx.x; e < B; e += 256) lcnt[e] = 0;
+  __syncthreads();
+  const int64_t chunk = (n + gridDim.x - 1) / gridDim.x;
+  const int64_t i0 = (int64_t)blockIdx.x * chunk;
+  const int64_t i1 = min(n, i0 + chunk);
+  for (int64_t i = i0 + threadIdx.x; i < i1; i += 256) {
+    const int64_t sl = lut[node_of_row[i]];
+    if (sl >= 0) atomicAdd(&lcnt[sl], 1);
+  }
+  __syncthreads();
+  for (int e = threadIdx.x; e < B; e += 256) {
+    const int c = lcnt[e];
+    lbase[e] = c ? atomicAdd(&cursor[e], c) : 0;
+    lcnt[e] = 0;
+  }
+  __syncthreads();
+  for (int64_t i = i0 + threadIdx.x; i < i1; i += 256) {
+    const int64_t sl = lut[node_of_row[i]];
+    if (sl >= 0) {
+      const int p = atomicAdd(&lcnt[sl], 1);
+      perm[lbase[sl] + p] = i;
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// rf_reroute: one-pass row-to-child reassignment after a split batch.
+// Replaces the torch lut-gather + nonzero + feature-gather + where chain
+// (the byte index_elementwise gathers were ~17% of RFC fit kernel time).
+// Training rule: bin <= split_bin goes LEFT (== x < edges[split_bin]).
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(256) void rf_reroute_kernel(
+    int64_t* __restrict__ node_of_row,        // [n] in/out
+    const int64_t* __restrict__ lut2,         // node -> split slot | -1
+    const int32_t* __restrict__ sfeat,        // [ns]
+    const int32_t* __restrict__ sbin,         // [ns]
+    const int64_t* __restrict__ lchild,       // [ns]
+    const int64_t* __restrict__ rchild,       // [ns]
+    const uint8_t* __restrict__ Xb,           // [n, d]
+    int64_t n, int64_t d) {
+  const int64_t stride = (int64_t)gridDim.x * 256;
+  for (int64_t i = (int64_t)blockIdx.x * 256 + threadIdx.x; i < n; i += stride) {
+    const int64_t sl = lut2[node_of_row[i]];
+    if (sl < 0) continue;
+    const uint8_t bin = Xb[i * d + sfeat[sl]];
+    node_of_row[i] = (bin <= (uint8_t)sbin[sl]) ? lchild[sl] : rchild[sl];
+  }
+}
+
+// ---------------------------------------------------------------------------
+// knn_merge_topk: running per-query top-k selection fused with the
+// ||q-i||^2 = q_sq + i_sq - 2*dot expansion, consuming a library-GEMM dot
+// block ONCE (the torch path materializes d2 and runs multi-pass topk +
+// cat/gather merges per item chunk). One wave per query row; each lane owns
+// one of 64 candidate slots; tau = current worst kept distance, maintained
+// by wave max-reduce. Insertions serialize per wave but decay to rare as tau
+// shrinks (expected k*ln(c/k) per row on random-order data).
+// Reference: NearestNeighborsMG tiled distance + topk merge (knn.py:763-774).
+// ---------------------------------------------------------------------------
+
+__device__ inline float wave_max_f32(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1)
+    v = fmaxf(v, __shfl_xor(v, off, 64));
+  return v;
+}
+
+__global__ __launch_bounds__(256) void knn_merge_topk_kernel(
+    const float* __restrict__ G,     // [nq, c] dot products Q @ I_chunk^T
+    const float* __restrict__ q_sq,  // [nq]
+    const float* __restrict__ i_sq,  // [c]
+    int64_t nq, int64_t c, int64_t base,
+    float* __restrict__ best_d,      // [nq, 64] slots (+inf live, -inf dead)
+    int64_t* __restrict__ best_i) {  // [nq, 64]
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int64_t row = (int64_t)blockIdx.x * 4 + wave;
+  if (row >= nq) return;
+
+  float cur_d = best_d[row * 64 + lane];
+  int64_t cur_i = best_i[row * 64 + lane];
+  float tau = wave_max_f32(cur_d);
+
+  const float* g = G + row * c;
+  const float qs = q_sq[row];
+
+  for (int64_t j0 = (int64_t)lane * 4; j0 < c; j0 += 256) {
+    float dot4[4], isq4[4];
+    if (j0 + 3 < c) {
+      const float4 gd = *reinterpret_cast<const float4*>(g + j0);
+      const float4 gi = *reinterpret_cast<const float4*>(i_sq + j0);
+      dot4[0] = gd.x; dot4[1] = gd.y; dot4[2] = gd.z; dot4[3] = gd.w;
+      isq4[0] = gi.x; isq4[1] = gi.y; isq4[2] = gi.z; isq4[3] = gi.w;
+    } else {
+      for (int e = 0; e < 4; ++e) {
+        const bool ok = j0 + e < c;
+        dot4[e] = ok ? g[j0 + e] : 0.0f;
+        isq4[e] = ok ? i_sq[j0 + e] : 3.0e38f;  // pushes d2 above any tau
+      }
+    }
+#pragma unroll
+    for (int e = 0; e < 4; ++e) {
+      const float d2 = qs + isq4[e] - 2.0f * dot4[e];
+      unsigned long long m = __ballot(d2 < tau && j0 + e < c);
+      while (m) {
+        const int src = __ffsll((unsigned long long)m) - 1;
+        m &= m - 1;
+        const float v = __shfl(d2, src, 64);
+        if (v >= tau) continue;  // tau shrank since the ballot
+        const int col = __shfl((int)(j0 + e), src, 64);
+        // evict the current worst slot
+        const unsigned long long owners = __ballot(cur_d == tau);
+        const int owner = __ffsll((unsigned long long)owners) - 1;
+        if (lane == owner) { cur_d = v; cur_i = base + col; }
+        tau = wave_max_f32(cur_d);
+      }
+    }
+  }
+
+  best_d[row * 64 + lane] = cur_d;
+  best_i[row * 64 + lane] = cur_i;
+}
+
+// ---------------------------------------------------------------------------
+// gather_dists: d2[i][j] = ||A[i] - B[cand[i][j]]||^2 — the nn-descent /
+// beam-search hot op (per-row private candidate lists make this a gathered
+// row-vs-rows reduction, not a GEMM). One wave per output row: A[i] is
+// staged in registers (d/64 floats per lane), each candidate's B row is read
+// coalesced by the wave and reduced with a 6-step butterfly. The torch path
+// materializes [r, m, d] twice; this reads each gathered row once.
+// Reference: cuVS nn_descent / cagra build (umap.py:359-378).
+// ---------------------------------------------------------------------------
+
+constexpr int GD_DMAX = 2048;  // register-staged A row: d/64 <= 32 VGPRs
+
+__global__ __launch_bounds__(256) void gather_dists_kernel(
+    const float* __restrict__ A,     // [r, d]
+    const float* __restrict__ B,     // [n, d]
+    const int64_t* __restrict__ cand,  // [r, m]
+    int64_t r, int64_t m, int d,
+    float* __restrict__ out) {       // [r, m]
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int64_t row = (int64_t)blockIdx.x * 4 + wave;
+  if (row >= r) return;
+
+  const int nseg = (d + 63) >> 6;  // elements per lane
+  float a_reg[GD_DMAX / 64];
+  const float* arow = A + row * d;
+#pragma unroll 4
+  for (int s = 0; s < nseg; ++s) {
+    const int idx = s * 64 + lane;
+    a_reg[s] = (idx < d) ? arow[idx] : 0.0f;
+  }
+
+  const int64_t* crow = cand + row * m;
+  for (int64_t j = 0; j < m; ++j) {
+    const float* brow = B + crow[j] * d;
+    float acc = 0.0f;
+#pragma unroll 4
+    for (int s = 0; s < nseg; ++s) {
+      const int idx = s * 64 + lane;
+      const float bv = (idx < d) ? brow[idx] : 0.0f;
+      const float diff = a_reg[s] - bv;
+      acc = fmaf(diff, diff, acc);
+    }
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) acc += __shfl_xor(acc, off, 64);
+    if (lane == 0) out[row * m + j] = acc;
+  }
+}
+
+// ---------------------------------------------------------------------------
 // softmax_residual_loss: per-row softmax (C>1) or sigmoid (C==1) residual
 // and summed log-loss. resid = softmax(scores) - onehot(y) (or p - y).
 // One wave per row chunk; memory-bound, fused to one pass.
@@ -2023,6 +2234,80 @@ std::vector<torch::Tensor> rf_best_split(torch::Tensor H, int64_t min_leaf,
   return {gain, feat, bin, lval, rval};
 }
 
+std::vector<torch::Tensor> rf_partition(torch::Tensor node_of_row, torch::Tensor lut,
+                                        int64_t B) {
+  TORCH_CHECK(node_of_row.is_cuda() && node_of_row.dtype() == torch::kInt64 &&
+              node_of_row.is_contiguous());
+  TORCH_CHECK(lut.is_cuda() && lut.dtype() == torch::kInt64 && lut.is_contiguous());
+  TORCH_CHECK(B >= 1 && B <= 8192, "node batch too large for LDS counts");
+  const int64_t n = node_of_row.size(0);
+  auto counts = torch::zeros({B}, node_of_row.options().dtype(torch::kInt32));
+  const size_t lds1 = (size_t)B * 4;
+  const unsigned grid = (unsigned)std::min<int64_t>(1024, (n + 255) / 256 + 1);
+  hipLaunchKernelGGL(rf_node_count_kernel, dim3(grid), dim3(256), lds1, cur_stream(),
+                     node_of_row.data_ptr<int64_t>(), lut.data_ptr<int64_t>(), n, (int)B,
+                     counts.data_ptr<int32_t>());
+  auto seg_off = torch::zeros({B + 1}, node_of_row.options().dtype(torch::kInt64));
+  seg_off.narrow(0, 1, B).copy_(torch::cumsum(counts, 0));
+  auto cursor = seg_off.narrow(0, 0, B).to(torch::kInt32).contiguous();
+  auto perm = torch::empty({n}, node_of_row.options().dtype(torch::kInt64));
+  const size_t lds2 = (size_t)B * 8;
+  hipLaunchKernelGGL(rf_partition_scatter_kernel, dim3(grid), dim3(256), lds2, cur_stream(),
+                     node_of_row.data_ptr<int64_t>(), lut.data_ptr<int64_t>(), n, (int)B,
+                     cursor.data_ptr<int32_t>(), perm.data_ptr<int64_t>());
+  return {perm, seg_off};
+}
+
+void rf_reroute(torch::Tensor node_of_row, torch::Tensor lut2, torch::Tensor sfeat,
+                torch::Tensor sbin, torch::Tensor lchild, torch::Tensor rchild,
+                torch::Tensor Xb) {
+  TORCH_CHECK(node_of_row.is_cuda() && node_of_row.dtype() == torch::kInt64);
+  TORCH_CHECK(lut2.dtype() == torch::kInt64 && sfeat.dtype() == torch::kInt32 &&
+              sbin.dtype() == torch::kInt32 && lchild.dtype() == torch::kInt64 &&
+              rchild.dtype() == torch::kInt64 && Xb.dtype() == torch::kUInt8);
+  const int64_t n = node_of_row.size(0);
+  const int64_t d = Xb.size(1);
+  const unsigned grid = (unsigned)std::min<int64_t>(2048, (n + 255) / 256 + 1);
+  hipLaunchKernelGGL(rf_reroute_kernel, dim3(grid), dim3(256), 0, cur_stream(),
+                     node_of_row.data_ptr<int64_t>(), lut2.data_ptr<int64_t>(),
+                     sfeat.data_ptr<int32_t>(), sbin.data_ptr<int32_t>(),
+                     lchild.data_ptr<int64_t>(), rchild.data_ptr<int64_t>(),
+                     Xb.data_ptr<uint8_t>(), n, d);
+}
+
+void knn_merge_topk(torch::Tensor G, torch::Tensor q_sq, torch::Tensor i_sq,
+                    int64_t base, torch::Tensor best_d, torch::Tensor best_i) {
+  TORCH_CHECK(G.is_cuda() && G.dtype() == torch::kFloat32 && G.is_contiguous());
+  TORCH_CHECK(best_d.size(1) == 64 && best_i.size(1) == 64,
+              "slot buffers must be [nq, 64]");
+  TORCH_CHECK(best_d.is_contiguous() && best_i.is_contiguous());
+  TORCH_CHECK(q_sq.is_contiguous() && i_sq.is_contiguous());
+  const int64_t nq = G.size(0);
+  const int64_t c = G.size(1);
+  TORCH_CHECK(i_sq.size(0) == c && q_sq.size(0) == nq && best_d.size(0) == nq);
+  hipLaunchKernelGGL(knn_merge_topk_kernel, dim3((unsigned)((nq + 3) / 4)), dim3(256), 0,
+                     cur_stream(), G.data_ptr<float>(), q_sq.data_ptr<float>(),
+                     i_sq.data_ptr<float>(), nq, c, base, best_d.data_ptr<float>(),
+                     best_i.data_ptr<int64_t>());
+}
+
+torch::Tensor gather_dists(torch::Tensor A, torch::Tensor B, torch::Tensor cand) {
+  TORCH_CHECK(A.is_cuda() && A.dtype() == torch::kFloat32 && A.is_contiguous());
+  TORCH_CHECK(B.dtype() == torch::kFloat32 && B.is_contiguous());
+  TORCH_CHECK(cand.dtype() == torch::kInt64 && cand.is_contiguous());
+  const int64_t r = cand.size(0);
+  const int64_t m = cand.size(1);
+  const int d = (int)A.size(1);
+  TORCH_CHECK((int)B.size(1) == d && A.size(0) == r);
+  TORCH_CHECK(d <= GD_DMAX, "gather_dists supports d <= 2048");
+  auto out = torch::empty({r, m}, A.options());
+  if (r > 0 && m > 0)
+    hipLaunchKernelGGL(gather_dists_kernel, dim3((unsigned)((r + 3) / 4)), dim3(256), 0,
+                       cur_stream(), A.data_ptr<float>(), B.data_ptr<float>(),
+                       cand.data_ptr<int64_t>(), r, m, d, out.data_ptr<float>());
+  return out;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("kmeans_assign", &kmeans_assign, "fused MFMA distance + argmin");
   m.def("label_accumulate", &label_accumulate, "per-center sum/count scatter");
@@ -2033,5 +2318,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rf_best_split", &rf_best_split, "fused RF gain scan + block-best reduce");
   m.def("dbscan_sweep", &dbscan_sweep, "fused eps-neighborhood count / min-core-label pass");
   m.def("umap_sgd", &umap_sgd, "edge-sampled UMAP SGD (all epochs, Hogwild)");
+  m.def("rf_partition", &rf_partition, "counting-sort rows by node -> (perm, seg_off)");
+  m.def("rf_reroute", &rf_reroute, "one-pass row-to-child reassignment");
+  m.def("knn_merge_topk", &knn_merge_topk, "running top-k merge over a GEMM dot block");
+  m.def("gather_dists", &gather_dists, "gathered row-vs-candidates squared distances");
   m.attr("_is_hip") = true;
 }

@@ -495,16 +495,38 @@ def _dedup_topk(ids: torch.Tensor, dists: torch.Tensor, k: int) -> Tuple[torch.T
     return sid.gather(1, o2), vals
 
 
-def _batched_dists(X: torch.Tensor, rows: torch.Tensor, cand: torch.Tensor) -> torch.Tensor:
-    """||X[rows][:,None] - X[cand]||^2 for cand [m, c] (chunked over m)."""
-    out = torch.empty(cand.shape, dtype=torch.float32, device=X.device)
-    chunk = max(1, (1 << 26) // max(1, cand.shape[1] * X.shape[1]))
+def _gather_dists(A: torch.Tensor, B: torch.Tensor, cand: torch.Tensor) -> torch.Tensor:
+    """d2[i][j] = ||A[i] - B[cand[i][j]]||^2 — the nn-descent / beam-search
+    hot op. GPU: the `gather_dists` HIP kernel (one wave per row, A row in
+    registers, candidate rows read once) replacing the torch broadcast chain
+    that materializes [r, m, d] twice (reference uses
+    cuVS nn_descent, umap.py:359-378)."""
+    from ..ops.dispatch import has_hip_ops, hip_ops, use_hip
+
+    if (
+        A.is_cuda
+        and use_hip(A, B)
+        and has_hip_ops()
+        and A.dtype == torch.float32
+        and A.shape[1] <= 2048
+    ):
+        return hip_ops().gather_dists(
+            A.contiguous(), B.contiguous(), cand.contiguous()
+        )
+    out = torch.empty(cand.shape, dtype=torch.float32, device=A.device)
+    chunk = max(1, (1 << 26) // max(1, cand.shape[1] * A.shape[1]))
     for s in range(0, cand.shape[0], chunk):
         e = min(cand.shape[0], s + chunk)
-        a = X[rows[s:e]][:, None, :]
-        b = X[cand[s:e]]
+        a = A[s:e][:, None, :]
+        b = B[cand[s:e]]
         out[s:e] = ((a - b) ** 2).sum(dim=2)
     return out
+
+
+def _batched_dists(X: torch.Tensor, rows: torch.Tensor, cand: torch.Tensor) -> torch.Tensor:
+    """||X[rows][:,None] - X[cand]||^2 for cand [m, c] (chunked over m).
+    Call sites pass rows = arange(n), so A = X row-aligned with cand."""
+    return _gather_dists(X, X, cand)
 
 
 def _reverse_edges(G: torch.Tensor, cap: int) -> torch.Tensor:
@@ -563,14 +585,7 @@ def _graph_beam_search(
     rowsq = torch.arange(nq, device=device)
 
     def qdists(cand: torch.Tensor) -> torch.Tensor:
-        out = torch.empty(cand.shape, dtype=torch.float32, device=device)
-        chunk = max(1, (1 << 26) // max(1, cand.shape[1] * Qt.shape[1]))
-        for s in range(0, nq, chunk):
-            e = min(nq, s + chunk)
-            a = Qt[s:e][:, None, :]
-            b = Xt[cand[s:e]]
-            out[s:e] = ((a - b) ** 2).sum(dim=2)
-        return out
+        return _gather_dists(Qt, Xt, cand)
 
     bd = qdists(beam)
     expand = max(4, itopk // 8)  # nodes expanded per hop

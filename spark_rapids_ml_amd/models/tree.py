@@ -346,16 +346,24 @@ def _grow_tree(
             # map node id -> local 0..B-1 (rows not in batch -> -1)
             lut = torch.full((tree.n_nodes,), -1, dtype=torch.int64, device=dev)
             lut[batch_t] = torch.arange(B, dtype=torch.int64, device=dev)
-            local = lut[node_of_row]
-            in_batch = local >= 0
-            rows = torch.nonzero(in_batch).flatten()
-            if rows.numel() == 0:
-                continue
-            loc = local[rows]
-            if task == "classification":
-                yb = y_idx[rows]
+            if hip_hist:
+                # counting-sort partition kernel: one pass for counts, one
+                # scatter — replaces sort+nonzero+gathers (rf_partition)
+                perm, seg_off = ext.rf_partition(node_of_row, lut, B)
+                if int(seg_off[-1].item()) == 0:
+                    continue
+                rows = loc = yb = None
             else:
-                yb = yf[rows]
+                local = lut[node_of_row]
+                in_batch = local >= 0
+                rows = torch.nonzero(in_batch).flatten()
+                if rows.numel() == 0:
+                    continue
+                loc = local[rows]
+                if task == "classification":
+                    yb = y_idx[rows]
+                else:
+                    yb = yf[rows]
 
             best_gain = torch.full((B,), -1.0, dtype=torch.float32, device=dev)
             best_feat = torch.full((B,), -1, dtype=torch.int64, device=dev)
@@ -374,7 +382,9 @@ def _grow_tree(
             # reference gets from cuML's per-node sampling.
             if max_features < d:
                 scores = torch.rand((B, d), generator=gen, device=dev)
-                feat_sel = scores.argsort(dim=1)[:, :max_features]  # [B, mf]
+                # partial selection (topk of random scores = uniform subset
+                # without replacement) instead of a full [B, d] argsort
+                feat_sel = scores.topk(max_features, dim=1).indices  # [B, mf]
                 mf = max_features
                 Xb_rows = None  # sampled path reads only selected bytes
             else:
@@ -382,11 +392,6 @@ def _grow_tree(
                 mf = d
                 Xb_rows = None
             if hip_hist:
-                sl, perm_rel = loc.sort()
-                perm = rows[perm_rel].contiguous()
-                seg_off = torch.searchsorted(
-                    sl, torch.arange(B + 1, dtype=torch.int64, device=dev)
-                ).contiguous()
                 fsel32 = (
                     feat_sel.to(torch.int32).contiguous()
                     if feat_sel is not None
@@ -521,11 +526,18 @@ def _grow_tree(
             r_t = torch.from_numpy(r_ids).to(dev)
             lut2 = torch.full((tree.n_nodes,), -1, dtype=torch.int64, device=dev)
             lut2[nid_t] = torch.arange(ns, dtype=torch.int64, device=dev)
-            sl = lut2[node_of_row]
-            mrows = torch.nonzero(sl >= 0).flatten()
-            srel = sl[mrows]
-            go_left = Xb[mrows, f_t[srel]].to(torch.int64) <= b_t[srel]
-            node_of_row[mrows] = torch.where(go_left, l_t[srel], r_t[srel])
+            if hip_hist:
+                # single-pass kernel: gather split byte + child write fused
+                ext.rf_reroute(
+                    node_of_row, lut2,
+                    f_t.to(torch.int32), b_t.to(torch.int32), l_t, r_t, Xb,
+                )
+            else:
+                sl = lut2[node_of_row]
+                mrows = torch.nonzero(sl >= 0).flatten()
+                srel = sl[mrows]
+                go_left = Xb[mrows, f_t[srel]].to(torch.int64) <= b_t[srel]
+                node_of_row[mrows] = torch.where(go_left, l_t[srel], r_t[srel])
         frontier = new_frontier
 
     return tree.to_arrays()

@@ -173,7 +173,7 @@ def test_rf_threshold_equality_consistent_with_training():
     df = DataFrame({"features": X, "label": y})
     m = RandomForestClassifier(
         featuresCol="features", labelCol="label", numTrees=1, maxDepth=4,
-        bootstrap=False, seed=3,
+        bootstrap=False, seed=3, featureSubsetStrategy="all",
     ).fit(df)
     pred = np.asarray(m.transform(df)[m.getOrDefault("predictionCol")])
     acc = float((pred == y).mean())

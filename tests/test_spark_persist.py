@@ -271,7 +271,9 @@ def test_threshold_semantics_preserved_across_spark_roundtrip(tmp_path):
     X = rng.integers(0, 10, size=(600, 4)).astype(np.float32)
     y = (X[:, 0] >= 5).astype(np.float64)
     df = DataFrame.from_numpy(X, y)
-    m = RandomForestClassifier(numTrees=1, maxDepth=4, bootstrap=False, seed=3).fit(df)
+    m = RandomForestClassifier(
+        numTrees=1, maxDepth=4, bootstrap=False, seed=3, featureSubsetStrategy="all"
+    ).fit(df)
     p = str(tmp_path / "rf_thr")
     m.saveAsSparkModel(p)
     m2 = load_spark_model(p)
