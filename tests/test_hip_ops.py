@@ -212,3 +212,87 @@ def test_dbscan_sweep_matches_torch():
     ref = masked.min(dim=1).values
     mism = (got != ref).sum().item()
     assert mism <= n_rows * 0.003, f"{mism} label mismatches"
+
+
+@pytest.mark.parametrize("n,B", [(100000, 1), (50000, 977), (200000, 4096)])
+def test_rf_partition_matches_sort(ext, n, B):
+    g = torch.Generator().manual_seed(3)
+    n_nodes = B * 3 + 5
+    node_of_row = torch.randint(0, n_nodes, (n,), generator=g).cuda()
+    batch = torch.randperm(n_nodes, generator=g)[:B].cuda()
+    lut = torch.full((n_nodes,), -1, dtype=torch.int64).cuda()
+    lut[batch] = torch.arange(B, dtype=torch.int64).cuda()
+    perm, seg_off = ext.rf_partition(node_of_row, lut, B)
+
+    # reference: torch sort path
+    local = lut[node_of_row]
+    rows = torch.nonzero(local >= 0).flatten()
+    loc = local[rows]
+    sl, perm_rel = loc.sort()
+    ref_seg = torch.searchsorted(sl, torch.arange(B + 1, dtype=torch.int64).cuda())
+    assert torch.equal(seg_off.cpu(), ref_seg.cpu())
+    total = int(seg_off[-1].item())
+    assert total == rows.numel()
+    # each segment must hold exactly the rows of its node (order-free)
+    got = perm[:total]
+    assert torch.equal(lut[node_of_row[got]].cpu().sort().values,
+                       sl.cpu())
+    for b in [0, B // 2, B - 1]:
+        s, e = int(seg_off[b]), int(seg_off[b + 1])
+        seg_rows = got[s:e]
+        assert (lut[node_of_row[seg_rows]] == b).all()
+    # perm must be a permutation of the in-batch rows
+    assert torch.equal(got.sort().values.cpu(), rows.sort().values.cpu())
+
+
+def test_rf_reroute_matches_torch(ext):
+    g = torch.Generator().manual_seed(5)
+    n, d, ns = 100000, 32, 37
+    n_nodes = 200
+    Xb = torch.randint(0, 64, (n, d), generator=g, dtype=torch.uint8).cuda()
+    node_of_row = torch.randint(0, n_nodes, (n,), generator=g).cuda()
+    split_nodes = torch.randperm(n_nodes, generator=g)[:ns].cuda()
+    lut2 = torch.full((n_nodes,), -1, dtype=torch.int64).cuda()
+    lut2[split_nodes] = torch.arange(ns, dtype=torch.int64).cuda()
+    f_t = torch.randint(0, d, (ns,), generator=g, dtype=torch.int32).cuda()
+    b_t = torch.randint(0, 64, (ns,), generator=g, dtype=torch.int32).cuda()
+    l_t = torch.arange(n_nodes, n_nodes + ns, dtype=torch.int64).cuda()
+    r_t = l_t + ns
+
+    ref = node_of_row.clone()
+    sl = lut2[ref]
+    mrows = torch.nonzero(sl >= 0).flatten()
+    srel = sl[mrows]
+    go_left = Xb[mrows, f_t[srel].long()].to(torch.int64) <= b_t[srel].long()
+    ref[mrows] = torch.where(go_left, l_t[srel], r_t[srel])
+
+    got = node_of_row.clone()
+    ext.rf_reroute(got, lut2, f_t, b_t, l_t, r_t, Xb)
+    assert torch.equal(got.cpu(), ref.cpu())
+
+
+@pytest.mark.parametrize("nq,ni,d,k", [(500, 20000, 64, 64), (333, 7777, 100, 17), (64, 500, 768, 64)])
+def test_knn_gemm_select_matches_ref(ext, nq, ni, d, k):
+    from spark_rapids_ml_amd.ops.knn import _knn_topk_gemm_select
+
+    Q = _rand(nq, d, seed=7)
+    I = _rand(ni, d, seed=8)
+    dd, ii = _knn_topk_gemm_select(Q, I, k, chunk_elems=1 << 22)  # force chunking
+    rd, ri = torch_ref.knn_topk(Q, I, k)
+    assert torch.allclose(dd.cpu(), rd.cpu(), rtol=1e-3, atol=1e-2)
+    # indices can differ on ties; distances of chosen indices must match
+    same = (ii.cpu() == ri.cpu()).float().mean().item()
+    assert same > 0.99
+
+
+@pytest.mark.parametrize("r,n,d,m", [(10000, 10000, 256, 40), (5000, 7000, 100, 16), (1000, 1000, 2048, 8)])
+def test_gather_dists_matches_torch(ext, r, n, d, m):
+    g = torch.Generator().manual_seed(9)
+    A = _rand(r, d, seed=10)
+    B = _rand(n, d, seed=11)
+    cand = torch.randint(0, n, (r, m), generator=g).cuda()
+    got = ext.gather_dists(A, B, cand)
+    a = A[:, None, :]
+    b = B[cand]
+    ref = ((a - b) ** 2).sum(dim=2)
+    assert torch.allclose(got, ref, rtol=1e-3, atol=1e-2)
