@@ -1842,9 +1842,12 @@ __global__ __launch_bounds__(256) void knn_merge_topk_kernel(
   const float* g = G + row * c;
   const float qs = q_sq[row];
 
+  // float4 loads are legal only when every row base is 16B-aligned, i.e.
+  // c % 4 == 0 (g = G + row*c); otherwise scalar loads
+  const bool vec4 = (c & 3) == 0;
   for (int64_t j0 = (int64_t)lane * 4; j0 < c; j0 += 256) {
     float dot4[4], isq4[4];
-    if (j0 + 3 < c) {
+    if (vec4 && j0 + 3 < c) {
       const float4 gd = *reinterpret_cast<const float4*>(g + j0);
       const float4 gi = *reinterpret_cast<const float4*>(i_sq + j0);
       dot4[0] = gd.x; dot4[1] = gd.y; dot4[2] = gd.z; dot4[3] = gd.w;

@@ -333,6 +333,21 @@ class UMAP(_UMAPParams, Estimator):
     def _fit_single(
         self, X: np.ndarray, device: torch.device, y: Optional[np.ndarray] = None
     ) -> np.ndarray:
+        import os as _os
+        import time as _time
+
+        _timing = _os.environ.get("SRML_UMAP_TIMING") == "1"
+
+        def _mark(label: str, t0: float) -> float:
+            if _timing:
+                if device.type == "cuda":
+                    torch.cuda.synchronize(device)
+                t = _time.perf_counter()
+                print(f"[umap-timing] {label}: {t - t0:.3f}s", flush=True)
+                return t
+            return t0
+
+        _t0 = _time.perf_counter()
         n = X.shape[0]
         k = int(float(self.getOrDefault("n_neighbors")))
         k = min(k, max(2, n - 1))
@@ -354,6 +369,7 @@ class UMAP(_UMAPParams, Estimator):
             d = torch.cat(
                 [torch.zeros(n, 1, device=Xt.device), torch.sqrt(vals[:, :k])], dim=1
             )
+        _t0 = _mark("knn_graph", _t0)
         knn_d, knn_i = as_numpy(d)[:, 1:], as_numpy(i)[:, 1:]  # drop self
 
         rows, cols, vals = _fuzzy_simplicial_set(
@@ -371,10 +387,12 @@ class UMAP(_UMAPParams, Estimator):
             vals = vals.copy()
             vals[diff] *= far
         a, b = self._ab()
+        _t0 = _mark("fuzzy_set", _t0)
 
         n_epochs = int(self.getOrDefault("n_epochs")) or (500 if n <= 10000 else 200)
         emb = self._init_embedding(X, rows, cols, vals, dim, seed)
         emb_t = torch.from_numpy(emb).to(device)
+        _t0 = _mark("init_embedding", _t0)
         gen = torch.Generator(device=device)
         gen.manual_seed(seed)
         emb_t = _optimize_embedding(
@@ -390,6 +408,7 @@ class UMAP(_UMAPParams, Estimator):
             float(self.getOrDefault("repulsion_strength")),
             gen,
         )
+        _mark("sgd", _t0)
         return as_numpy(emb_t)
 
     def _init_embedding(self, X, rows, cols, vals, dim, seed) -> np.ndarray:

@@ -65,7 +65,9 @@ def _knn_topk_gemm_select(
         best_d[:, k:] = float("-inf")
     best_i = torch.full((nq, 64), -1, dtype=torch.int64, device=dev)
 
-    ichunk = max(256, min(ni, chunk_elems // max(1, nq)))
+    # multiple of 64 so every non-final chunk keeps the kernel's float4 path
+    # (row base 16B-aligned requires chunk width % 4 == 0)
+    ichunk = max(256, (min(ni, chunk_elems // max(1, nq)) // 64) * 64)
     G = torch.empty((nq, min(ichunk, ni)), dtype=torch.float32, device=dev)
     for s in range(0, ni, ichunk):
         e = min(ni, s + ichunk)
