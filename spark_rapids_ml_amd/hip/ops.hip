@@ -1845,7 +1845,13 @@ __global__ __launch_bounds__(256) void knn_merge_topk_kernel(
   // float4 loads are legal only when every row base is 16B-aligned, i.e.
   // c % 4 == 0 (g = G + row*c); otherwise scalar loads
   const bool vec4 = (c & 3) == 0;
-  for (int64_t j0 = (int64_t)lane * 4; j0 < c; j0 += 256) {
+  // UNIFORM trip count: every lane of the wave must reach every ballot /
+  // shuffle below even when its own j0 has run past c (a per-lane `j0 < c`
+  // loop bound left part of the wave outside the evict sequence whenever
+  // c % 256 != 0, corrupting tau)
+  const int64_t n_iter = (c + 255) / 256;
+  for (int64_t it = 0; it < n_iter; ++it) {
+    const int64_t j0 = it * 256 + (int64_t)lane * 4;
     float dot4[4], isq4[4];
     if (vec4 && j0 + 3 < c) {
       const float4 gd = *reinterpret_cast<const float4*>(g + j0);

@@ -113,3 +113,31 @@ def test_umap_supervised_separates_classes():
         d = np.linalg.norm(emb[i] - emb[j])
         (intra if y[i] == y[j] else inter).append(d)
     assert np.mean(intra) < 0.5 * np.mean(inter)
+
+
+def test_fuzzy_simplicial_set_torch_matches_numpy():
+    """The device-resident fuzzy-set path must reproduce the numpy/scipy
+    reference (same smooth-knn bisection, same symmetrization algebra)."""
+    import torch
+
+    from spark_rapids_ml_amd.models.umap import (
+        _fuzzy_simplicial_set,
+        _fuzzy_simplicial_set_t,
+    )
+
+    rng = np.random.default_rng(0)
+    n, k = 500, 15
+    knn_d = np.sort(np.abs(rng.normal(size=(n, k))), axis=1).astype(np.float64)
+    knn_i = np.stack([rng.permutation(n)[:k] for _ in range(n)]).astype(np.int64)
+
+    r1, c1, v1 = _fuzzy_simplicial_set(knn_d.astype(np.float32), knn_i, 1.0, 1.0)
+    r2, c2, v2 = _fuzzy_simplicial_set_t(
+        torch.from_numpy(knn_d.astype(np.float32)), torch.from_numpy(knn_i), 1.0, 1.0
+    )
+    # compare as sparse maps (orderings differ)
+    m1 = {(int(r), int(c)): float(v) for r, c, v in zip(r1, c1, v1)}
+    m2 = {(int(r), int(c)): float(v) for r, c, v in zip(
+        r2.numpy(), c2.numpy(), v2.numpy())}
+    assert set(m1) == set(m2)
+    for key in m1:
+        assert abs(m1[key] - m2[key]) < 1e-5, key
