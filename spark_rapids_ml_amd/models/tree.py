@@ -323,9 +323,14 @@ def _grow_tree(
     node_of_row = torch.zeros(n, dtype=torch.int64, device=dev)
 
     # HIP histogram path (LDS-privatized, rows segment-sorted by node)
+    import os as _os
+
     from ..ops.dispatch import has_hip_ops, use_hip
 
     hip_hist = Xb.is_cuda and use_hip(Xb) and has_hip_ops()
+    _rf_debug = _os.environ.get("SRML_RF_DEBUG") == "1"
+    if _rf_debug:
+        print(f"[rf-debug] hip_hist={hip_hist} n={n} d={d} device={dev}", flush=True)
     if hip_hist:
         from ..ops.dispatch import hip_ops
 
@@ -538,6 +543,12 @@ def _grow_tree(
                 srel = sl[mrows]
                 go_left = Xb[mrows, f_t[srel]].to(torch.int64) <= b_t[srel]
                 node_of_row[mrows] = torch.where(go_left, l_t[srel], r_t[srel])
+        if _rf_debug:
+            print(
+                f"[rf-debug] depth={depth} frontier={len(frontier)} -> "
+                f"{len(new_frontier)} nodes={tree.n_nodes}",
+                flush=True,
+            )
         frontier = new_frontier
 
     return tree.to_arrays()
