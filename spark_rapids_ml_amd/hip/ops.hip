@@ -1537,12 +1537,14 @@ __global__ __launch_bounds__(256) void knn_select_kernel(
 // ---------------------------------------------------------------------------
 
 __global__ __launch_bounds__(256) void rf_histogram_kernel(
-    const uint8_t* __restrict__ Xb,   // [n, d] binned
-    const int64_t* __restrict__ perm, // [m] rows sorted by node
+    const uint8_t* __restrict__ Xb,   // [n_phys, d] binned
+    const int64_t* __restrict__ perm, // [m] VIRTUAL rows sorted by node
     const int64_t* __restrict__ seg_off,  // [B+1]
     const int32_t* __restrict__ feat_sel, // [B, mf] or nullptr (identity)
-    const int32_t* __restrict__ y_cls,    // [n] class ids (classification)
-    const float* __restrict__ y_reg,      // [n] targets (regression)
+    const int32_t* __restrict__ y_cls,    // [n_phys] class ids (classification)
+    const float* __restrict__ y_reg,      // [n_phys] targets (regression)
+    const int32_t* __restrict__ sample,   // [vn] virtual->physical, or nullptr
+    int64_t n_phys,                       // modulo map when sample==nullptr
     int d, int mf, int f0, int FC, int n_bins, int C, int split,
     float* __restrict__ out) {            // [B, FC, n_bins, C]
   extern __shared__ __attribute__((aligned(16))) float lhist[];  // [FC][nb][C]
@@ -1564,7 +1566,9 @@ __global__ __launch_bounds__(256) void rf_histogram_kernel(
 
   const bool classif = (y_cls != nullptr);
   for (int64_t r = rs + tid; r < re; r += 256) {
-    const int64_t row = perm[r];
+    const int64_t vrow = perm[r];
+    const int64_t row = sample ? (int64_t)sample[vrow]
+                               : (vrow >= n_phys ? vrow % n_phys : vrow);
     const uint8_t* xrow = Xb + row * (int64_t)d;
     const int yc = classif ? y_cls[row] : 0;
     const float yv = classif ? 1.0f : y_reg[row];
@@ -1788,19 +1792,23 @@ __global__ __launch_bounds__(256) void rf_partition_scatter_kernel(
 // ---------------------------------------------------------------------------
 
 __global__ __launch_bounds__(256) void rf_reroute_kernel(
-    int64_t* __restrict__ node_of_row,        // [n] in/out
+    int64_t* __restrict__ node_of_row,        // [vn] in/out (virtual rows)
     const int64_t* __restrict__ lut2,         // node -> split slot | -1
     const int32_t* __restrict__ sfeat,        // [ns]
     const int32_t* __restrict__ sbin,         // [ns]
     const int64_t* __restrict__ lchild,       // [ns]
     const int64_t* __restrict__ rchild,       // [ns]
-    const uint8_t* __restrict__ Xb,           // [n, d]
-    int64_t n, int64_t d) {
+    const uint8_t* __restrict__ Xb,           // [n_phys, d]
+    const int32_t* __restrict__ sample,       // [vn] or nullptr
+    int64_t n_phys,
+    int64_t vn, int64_t d) {
   const int64_t stride = (int64_t)gridDim.x * 256;
-  for (int64_t i = (int64_t)blockIdx.x * 256 + threadIdx.x; i < n; i += stride) {
+  for (int64_t i = (int64_t)blockIdx.x * 256 + threadIdx.x; i < vn; i += stride) {
     const int64_t sl = lut2[node_of_row[i]];
     if (sl < 0) continue;
-    const uint8_t bin = Xb[i * d + sfeat[sl]];
+    const int64_t row = sample ? (int64_t)sample[i]
+                               : (i >= n_phys ? i % n_phys : i);
+    const uint8_t bin = Xb[row * d + sfeat[sl]];
     node_of_row[i] = (bin <= (uint8_t)sbin[sl]) ? lchild[sl] : rchild[sl];
   }
 }
@@ -2197,14 +2205,18 @@ std::vector<torch::Tensor> softmax_residual_loss(torch::Tensor scores, torch::Te
 
 torch::Tensor rf_histogram(torch::Tensor Xb, torch::Tensor perm, torch::Tensor seg_off,
                            torch::Tensor feat_sel, torch::Tensor y, int64_t f0,
-                           int64_t FC, int64_t n_bins, int64_t n_classes) {
+                           int64_t FC, int64_t n_bins, int64_t n_classes,
+                           torch::Tensor sample) {
   TORCH_CHECK(Xb.is_cuda() && Xb.dtype() == torch::kUInt8 && Xb.is_contiguous());
   TORCH_CHECK(perm.dtype() == torch::kInt64 && seg_off.dtype() == torch::kInt64);
   const int d = (int)Xb.size(1);
+  const int64_t n_phys = Xb.size(0);
   const int B = (int)seg_off.size(0) - 1;
   const bool classif = n_classes > 0;
   const int C = classif ? (int)n_classes : 2;
   const int mf = feat_sel.numel() > 0 ? (int)feat_sel.size(1) : 0;
+  const bool has_sample = sample.numel() > 0;
+  if (has_sample) TORCH_CHECK(sample.dtype() == torch::kInt32 && sample.is_contiguous());
   auto out = torch::zeros({(int64_t)B, FC, n_bins, (int64_t)C}, Xb.options().dtype(torch::kFloat32));
   const size_t lds = (size_t)FC * n_bins * C * 4;
   TORCH_CHECK(lds <= 160 * 1024, "feature chunk too large for LDS");
@@ -2215,6 +2227,7 @@ torch::Tensor rf_histogram(torch::Tensor Xb, torch::Tensor perm, torch::Tensor s
                      mf > 0 ? feat_sel.data_ptr<int32_t>() : nullptr,
                      classif ? y.data_ptr<int32_t>() : nullptr,
                      classif ? nullptr : y.data_ptr<float>(),
+                     has_sample ? sample.data_ptr<int32_t>() : nullptr, n_phys,
                      d, mf, (int)f0, (int)FC, (int)n_bins, C, split,
                      out.data_ptr<float>());
   return out;
@@ -2269,19 +2282,24 @@ std::vector<torch::Tensor> rf_partition(torch::Tensor node_of_row, torch::Tensor
 
 void rf_reroute(torch::Tensor node_of_row, torch::Tensor lut2, torch::Tensor sfeat,
                 torch::Tensor sbin, torch::Tensor lchild, torch::Tensor rchild,
-                torch::Tensor Xb) {
+                torch::Tensor Xb, torch::Tensor sample) {
   TORCH_CHECK(node_of_row.is_cuda() && node_of_row.dtype() == torch::kInt64);
   TORCH_CHECK(lut2.dtype() == torch::kInt64 && sfeat.dtype() == torch::kInt32 &&
               sbin.dtype() == torch::kInt32 && lchild.dtype() == torch::kInt64 &&
               rchild.dtype() == torch::kInt64 && Xb.dtype() == torch::kUInt8);
-  const int64_t n = node_of_row.size(0);
+  const int64_t vn = node_of_row.size(0);
+  const int64_t n_phys = Xb.size(0);
   const int64_t d = Xb.size(1);
-  const unsigned grid = (unsigned)std::min<int64_t>(2048, (n + 255) / 256 + 1);
+  const bool has_sample = sample.numel() > 0;
+  if (has_sample) TORCH_CHECK(sample.dtype() == torch::kInt32 && sample.is_contiguous());
+  const unsigned grid = (unsigned)std::min<int64_t>(2048, (vn + 255) / 256 + 1);
   hipLaunchKernelGGL(rf_reroute_kernel, dim3(grid), dim3(256), 0, cur_stream(),
                      node_of_row.data_ptr<int64_t>(), lut2.data_ptr<int64_t>(),
                      sfeat.data_ptr<int32_t>(), sbin.data_ptr<int32_t>(),
                      lchild.data_ptr<int64_t>(), rchild.data_ptr<int64_t>(),
-                     Xb.data_ptr<uint8_t>(), n, d);
+                     Xb.data_ptr<uint8_t>(),
+                     has_sample ? sample.data_ptr<int32_t>() : nullptr, n_phys,
+                     vn, d);
 }
 
 void knn_merge_topk(torch::Tensor G, torch::Tensor q_sq, torch::Tensor i_sq,

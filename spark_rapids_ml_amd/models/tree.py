@@ -220,10 +220,13 @@ def _bin_data(X: torch.Tensor, edges: torch.Tensor) -> torch.Tensor:
 
 
 class _Tree:
-    """Flat-array binary tree with BULK growth (one numpy append per node
-    batch — the per-node Python loop was the depth-13 bottleneck)."""
+    """Flat-array node ARENA with BULK growth (one numpy append per node
+    batch — the per-node Python loop was the depth-13 bottleneck). Round 2:
+    one arena holds ALL of a worker's trees (cuML-style batched node
+    processing, reference tree.py:384-389); `tree_of` records which tree
+    owns each node and `split_by_tree` extracts the per-tree arrays."""
 
-    __slots__ = ("feature", "threshold", "left", "right", "is_leaf", "gain", "value", "n_nodes")
+    __slots__ = ("feature", "threshold", "left", "right", "is_leaf", "gain", "value", "tree_of", "n_nodes")
 
     def __init__(self, value_width: int):
         self.feature = [np.empty(0, np.int32)]
@@ -233,9 +236,10 @@ class _Tree:
         self.is_leaf = [np.empty(0, bool)]
         self.gain = [np.empty(0, np.float32)]
         self.value = [np.empty((0, value_width), np.float32)]
+        self.tree_of = [np.empty(0, np.int32)]
         self.n_nodes = 0
 
-    def add_leaves(self, values: np.ndarray) -> int:
+    def add_leaves(self, values: np.ndarray, tree_ids: Optional[np.ndarray] = None) -> int:
         """Append len(values) leaf nodes; returns the first new node id."""
         m = values.shape[0]
         first = self.n_nodes
@@ -247,6 +251,9 @@ class _Tree:
         self.is_leaf.append(np.ones(m, bool))
         self.gain.append(np.zeros(m, np.float32))
         self.value.append(values.astype(np.float32))
+        self.tree_of.append(
+            tree_ids.astype(np.int32) if tree_ids is not None else np.zeros(m, np.int32)
+        )
         return first
 
     def _consolidate(self) -> None:
@@ -258,6 +265,7 @@ class _Tree:
             self.is_leaf = [np.concatenate(self.is_leaf)]
             self.gain = [np.concatenate(self.gain)]
             self.value = [np.concatenate(self.value, axis=0)]
+            self.tree_of = [np.concatenate(self.tree_of)]
 
     def set_splits(
         self,
@@ -289,9 +297,36 @@ class _Tree:
             "value": self.value[0],
         }
 
+    def split_by_tree(self, n_trees: int) -> List[Dict[str, np.ndarray]]:
+        """Extract each tree's arrays from the arena with node ids remapped
+        to 0..m-1 (ascending arena order keeps every tree's root at 0:
+        roots are created before any child)."""
+        self._consolidate()
+        tree_of = self.tree_of[0]
+        out: List[Dict[str, np.ndarray]] = []
+        remap = np.full(self.n_nodes + 1, -1, np.int32)
+        for t in range(n_trees):
+            ids = np.nonzero(tree_of == t)[0]
+            remap[:] = -1
+            remap[ids] = np.arange(len(ids), dtype=np.int32)
+            left = self.left[0][ids]
+            right = self.right[0][ids]
+            out.append(
+                {
+                    "feature": self.feature[0][ids],
+                    "threshold": self.threshold[0][ids],
+                    "left": np.where(left >= 0, remap[left], -1).astype(np.int32),
+                    "right": np.where(right >= 0, remap[right], -1).astype(np.int32),
+                    "is_leaf": self.is_leaf[0][ids],
+                    "gain": self.gain[0][ids],
+                    "value": self.value[0][ids],
+                }
+            )
+        return out
 
-def _grow_tree(
-    Xb: torch.Tensor,  # [n,d] uint8 binned
+
+def _grow_forest(
+    Xb: torch.Tensor,  # [n,d] uint8 binned (shared across trees)
     y: torch.Tensor,  # [n] int64 class idx (classification) or f32 target
     edges: torch.Tensor,  # [d, nb-1]
     task: str,
@@ -302,25 +337,60 @@ def _grow_tree(
     min_gain: float,
     max_features: int,
     gen: torch.Generator,
+    n_trees: int = 1,
+    sample: Optional[torch.Tensor] = None,  # int32 [n_trees*per] bootstrap map
     node_batch: int = 4096,
     feat_chunk: int = 256,
-) -> Dict[str, np.ndarray]:
+) -> List[Dict[str, np.ndarray]]:
+    """Grow ALL of this worker's trees at once in one node arena over a
+    VIRTUAL row space [n_trees × per] (cuML-style batched node processing —
+    reference tree.py:384-389). Virtual row v belongs to tree v // per and
+    maps to physical row sample[v] (bootstrap) or v % n. Per-tree loops cost
+    trees × depth × batches host round-trips; this costs depth × batches."""
     n, d = Xb.shape
     dev = Xb.device
     value_width = n_classes if task == "classification" else 2  # (mean, count)
     tree = _Tree(value_width)
 
+    per = (sample.numel() // n_trees) if sample is not None else n
+    vn = n_trees * per
+    t_of_v = torch.arange(vn, device=dev) // max(1, per)
+    phys_all = sample.to(torch.int64) if sample is not None else None
+
     if task == "classification":
         y_idx = y.to(torch.int64)
-        root_counts = torch.bincount(y_idx, minlength=n_classes).to(torch.float32)
-        root_val = as_numpy(root_counts)
+        if phys_all is not None:
+            yp = y_idx[phys_all]
+            root_counts = torch.bincount(
+                t_of_v * n_classes + yp, minlength=n_trees * n_classes
+            ).view(n_trees, n_classes).to(torch.float32)
+            root_vals = as_numpy(root_counts)
+        else:
+            counts = torch.bincount(y_idx, minlength=n_classes).to(torch.float32)
+            root_vals = np.tile(as_numpy(counts), (n_trees, 1))
+        yf = None
     else:
         yf = y.to(torch.float32)
-        root_val = np.array([float(yf.mean().item()) if n else 0.0, float(n)], dtype=np.float32)
+        if phys_all is not None:
+            sums = torch.zeros(n_trees, dtype=torch.float64, device=dev)
+            sums.scatter_add_(0, t_of_v, yf[phys_all].to(torch.float64))
+            means = as_numpy((sums / max(1, per)).to(torch.float32))
+        else:
+            means = np.full(n_trees, float(yf.mean().item()) if n else 0.0, np.float32)
+        root_vals = np.stack(
+            [means, np.full(n_trees, float(per), np.float32)], axis=1
+        )
+        y_idx = None
 
-    root = tree.add_leaves(root_val[None, :])
+    tree.add_leaves(root_vals, tree_ids=np.arange(n_trees, dtype=np.int32))
     edges_np = as_numpy(edges)
-    node_of_row = torch.zeros(n, dtype=torch.int64, device=dev)
+    # virtual row v starts at its tree's root (arena ids 0..n_trees-1)
+    node_of_row = t_of_v.clone()
+    sample_t = (
+        sample.contiguous()
+        if sample is not None
+        else torch.empty(0, dtype=torch.int32, device=dev)
+    )
 
     # HIP histogram path (LDS-privatized, rows segment-sorted by node)
     import os as _os
@@ -338,7 +408,7 @@ def _grow_tree(
         y32 = y_idx.to(torch.int32) if task == "classification" else yf
         C_ch = n_classes if task == "classification" else 2
         fc_kernel = max(1, min(512, (150 * 1024 // 4) // max(1, n_bins * C_ch)))
-    frontier = [root]  # node ids whose rows are still splittable
+    frontier = list(range(n_trees))  # arena node ids still splittable
 
     for depth in range(max_depth):
         if not frontier:
@@ -365,10 +435,12 @@ def _grow_tree(
                 if rows.numel() == 0:
                     continue
                 loc = local[rows]
+                # virtual -> physical rows for feature/label gathers
+                prows = phys_all[rows] if phys_all is not None else rows % n
                 if task == "classification":
-                    yb = y_idx[rows]
+                    yb = y_idx[prows]
                 else:
-                    yb = yf[rows]
+                    yb = yf[prows]
 
             best_gain = torch.full((B,), -1.0, dtype=torch.float32, device=dev)
             best_feat = torch.full((B,), -1, dtype=torch.int64, device=dev)
@@ -405,7 +477,7 @@ def _grow_tree(
                 fc_step = fc_kernel
             else:
                 if feat_sel is None:
-                    Xb_rows = Xb[rows]
+                    Xb_rows = Xb[prows]
                 fc_step = feat_chunk
             for f0 in range(0, mf, fc_step):
                 f1 = min(mf, f0 + fc_step)
@@ -414,6 +486,7 @@ def _grow_tree(
                     H = ext.rf_histogram(
                         Xb, perm, seg_off, fsel32, y32, f0, F, n_bins,
                         n_classes if task == "classification" else 0,
+                        sample_t,
                     )
                     if H.shape[3] <= 16:
                         # fused gain scan + per-node best (kernel)
@@ -447,7 +520,7 @@ def _grow_tree(
                 elif task == "classification":
                     if feat_sel is not None:
                         sel = feat_sel[loc][:, f0:f1]
-                        bins = Xb[rows[:, None], sel].to(torch.int64)
+                        bins = Xb[prows[:, None], sel].to(torch.int64)
                     else:
                         bins = Xb_rows[:, f0:f1].to(torch.int64)
                     base = (loc[:, None] * F + torch.arange(F, device=dev)[None, :]) * n_bins + bins
@@ -459,7 +532,7 @@ def _grow_tree(
                 else:
                     if feat_sel is not None:
                         sel = feat_sel[loc][:, f0:f1]
-                        bins = Xb[rows[:, None], sel].to(torch.int64)
+                        bins = Xb[prows[:, None], sel].to(torch.int64)
                     else:
                         bins = Xb_rows[:, f0:f1].to(torch.int64)
                     base = (loc[:, None] * F + torch.arange(F, device=dev)[None, :]) * n_bins + bins
@@ -505,7 +578,10 @@ def _grow_tree(
             children = np.empty((2 * ns, lv.shape[1]), dtype=np.float32)
             children[0::2] = lv[idxs]
             children[1::2] = rv[idxs]
-            first = tree.add_leaves(children)
+            # children inherit their parent's tree
+            tree._consolidate()
+            ptree = tree.tree_of[0][batch_np[idxs]]
+            first = tree.add_leaves(children, tree_ids=np.repeat(ptree, 2))
             l_ids = first + 2 * np.arange(ns)
             r_ids = l_ids + 1
             fsel = bf[idxs].astype(np.int64)
@@ -536,12 +612,14 @@ def _grow_tree(
                 ext.rf_reroute(
                     node_of_row, lut2,
                     f_t.to(torch.int32), b_t.to(torch.int32), l_t, r_t, Xb,
+                    sample_t,
                 )
             else:
                 sl = lut2[node_of_row]
                 mrows = torch.nonzero(sl >= 0).flatten()
                 srel = sl[mrows]
-                go_left = Xb[mrows, f_t[srel]].to(torch.int64) <= b_t[srel]
+                pm = phys_all[mrows] if phys_all is not None else mrows % n
+                go_left = Xb[pm, f_t[srel]].to(torch.int64) <= b_t[srel]
                 node_of_row[mrows] = torch.where(go_left, l_t[srel], r_t[srel])
         if _rf_debug:
             print(
@@ -551,7 +629,7 @@ def _grow_tree(
             )
         frontier = new_frontier
 
-    return tree.to_arrays()
+    return tree.split_by_tree(n_trees)
 
 
 def _best_split_class(H: torch.Tensor, min_leaf: int):
@@ -665,8 +743,24 @@ class _RandomForestEstimator(_RandomForestParams, Estimator):
         bootstrap = bool(params["bootstrap"])
         max_samples = float(params["max_samples"])
 
+        import os as _os
+        import time as _time
+
+        _dbg = _os.environ.get("SRML_RF_DEBUG") == "1"
+
+        def _phase(label: str, t0: float) -> float:
+            if _dbg:
+                if ctx.device.type == "cuda":
+                    torch.cuda.synchronize(ctx.device)
+                t = _time.perf_counter()
+                print(f"[rf-phase] {label}: {t - t0:.3f}s", flush=True)
+                return t
+            return t0
+
+        _t = _time.perf_counter()
         Xt = ctx.device_tensor(np.asarray(X, dtype=np.float32))
         n, d = Xt.shape
+        _t = _phase("ingest_h2d", _t)
 
         if self._task == "classification":
             y_local = np.asarray(y)
@@ -682,7 +776,9 @@ class _RandomForestEstimator(_RandomForestParams, Estimator):
             yt = to_device_tensor(np.asarray(y, dtype=np.float32), ctx.device)
 
         edges = _compute_bin_edges(Xt, n_bins, comm, seed=seed)
+        _t = _phase("bin_edges", _t)
         Xb = _bin_data(Xt, edges)
+        _t = _phase("bin_data", _t)
         max_features = _resolve_max_features(str(params["max_features"]), d, self._task)
 
         counts = _estimators_per_worker(n_estimators, comm.world_size)
@@ -690,33 +786,56 @@ class _RandomForestEstimator(_RandomForestParams, Estimator):
         tree_id0 = sum(counts[: comm.rank])
 
         trees: List[Dict[str, np.ndarray]] = []
-        for t in range(my_trees):
-            tseed = seed + 1315423911 * (tree_id0 + t + 1)
+        if my_trees > 0 and n > 0:
+            # all of this worker's trees grow together in ONE arena over a
+            # virtual row space (bootstrap = int32 sample map, no gathered
+            # [n,d] copies per tree); seed offset by tree_id0 so ranks draw
+            # disjoint streams
             gen = torch.Generator(device=ctx.device)
-            gen.manual_seed(tseed & 0x7FFFFFFFFFFF)
-            if bootstrap and n > 0:
+            gen.manual_seed((seed + 1315423911 * (tree_id0 + 1)) & 0x7FFFFFFFFFFF)
+            if bootstrap:
                 n_draw = max(1, int(round(max_samples * n)))
-                rows = torch.randint(0, n, (n_draw,), generator=gen, device=ctx.device)
-                Xb_t, y_t = Xb[rows], yt[rows]
-            else:
-                Xb_t, y_t = Xb, yt
-            trees.append(
-                _grow_tree(
-                    Xb_t,
-                    y_t,
-                    edges,
-                    self._task,
-                    n_classes,
-                    n_bins,
-                    max_depth,
-                    min_leaf,
-                    min_gain,
-                    max_features,
-                    gen,
-                    node_batch=int(params.get("max_batch_size", 4096)),
+                sample = torch.randint(
+                    0, n, (my_trees * n_draw,), generator=gen,
+                    device=ctx.device, dtype=torch.int32,
                 )
+            else:
+                sample = None
+            trees = _grow_forest(
+                Xb,
+                yt,
+                edges,
+                self._task,
+                n_classes,
+                n_bins,
+                max_depth,
+                min_leaf,
+                min_gain,
+                max_features,
+                gen,
+                n_trees=my_trees,
+                sample=sample,
+                node_batch=int(params.get("max_batch_size", 4096)),
             )
+        elif my_trees > 0:
+            # empty local shard: contribute degenerate single-leaf trees so
+            # the merged forest still has numTrees members
+            width = n_classes if self._task == "classification" else 2
+            trees = [
+                {
+                    "feature": np.full(1, -1, np.int32),
+                    "threshold": np.zeros(1, np.float32),
+                    "left": np.full(1, -1, np.int32),
+                    "right": np.full(1, -1, np.int32),
+                    "is_leaf": np.ones(1, bool),
+                    "gain": np.zeros(1, np.float32),
+                    "value": np.zeros((1, width), np.float32),
+                }
+                for _ in range(my_trees)
+            ]
+        _t = _phase("grow_trees_total", _t)
 
+        _t = _phase("grow_trees", _t)
         # merge sub-forests (reference allGathers treelite bytes, tree.py:424-447)
         blobs = comm.allgather_obj(pickle.dumps(trees, protocol=4))
         forest: List[Dict[str, np.ndarray]] = []

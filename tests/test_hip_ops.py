@@ -122,7 +122,8 @@ def test_rf_histogram_matches_ref(ext, task, nc):
     seg_off = torch.searchsorted(sl, torch.arange(B + 1).cuda()).contiguous()
     if task == "classification":
         y = torch.randint(0, nc, (n,), generator=g, dtype=torch.int32).cuda()
-        H = ext.rf_histogram(Xb, perm, seg_off, feat_sel.to(torch.int32), y, 0, mf, nb, nc)
+        H = ext.rf_histogram(Xb, perm, seg_off, feat_sel.to(torch.int32), y, 0, mf, nb, nc,
+                             torch.empty(0, dtype=torch.int32).cuda())
         ref = torch.zeros(B, mf, nb, nc).cuda()
         for b in range(B):
             m = loc == b
@@ -138,7 +139,8 @@ def test_rf_histogram_matches_ref(ext, task, nc):
                 )
     else:
         y = torch.randn(n, generator=g).cuda()
-        H = ext.rf_histogram(Xb, perm, seg_off, feat_sel.to(torch.int32), y, 0, mf, nb, 0)
+        H = ext.rf_histogram(Xb, perm, seg_off, feat_sel.to(torch.int32), y, 0, mf, nb, 0,
+                             torch.empty(0, dtype=torch.int32).cuda())
         ref = torch.zeros(B, mf, nb, 2).cuda()
         for b in range(B):
             m = loc == b
@@ -267,7 +269,8 @@ def test_rf_reroute_matches_torch(ext):
     ref[mrows] = torch.where(go_left, l_t[srel], r_t[srel])
 
     got = node_of_row.clone()
-    ext.rf_reroute(got, lut2, f_t, b_t, l_t, r_t, Xb)
+    ext.rf_reroute(got, lut2, f_t, b_t, l_t, r_t, Xb,
+                   torch.empty(0, dtype=torch.int32).cuda())
     assert torch.equal(got.cpu(), ref.cpu())
 
 
@@ -296,3 +299,27 @@ def test_gather_dists_matches_torch(ext, r, n, d, m):
     b = B[cand]
     ref = ((a - b) ** 2).sum(dim=2)
     assert torch.allclose(got, ref, rtol=1e-3, atol=1e-2)
+
+
+def test_rf_histogram_virtual_rows(ext):
+    """Histogram over a VIRTUAL row space (forest arena): sample maps
+    virtual->physical; with sample empty, phys = v %% n_phys."""
+    g = torch.Generator().manual_seed(11)
+    n, d, nb, nc, T = 5000, 8, 16, 3, 4
+    Xb = torch.randint(0, nb, (n, d), generator=g, dtype=torch.uint8).cuda()
+    y = torch.randint(0, nc, (n,), generator=g, dtype=torch.int32).cuda()
+    vn = T * n
+    sample = torch.randint(0, n, (vn,), generator=g, dtype=torch.int32).cuda()
+    node_of_row = torch.arange(vn).cuda() // n  # node t = tree t root
+    lut = torch.arange(T, dtype=torch.int64).cuda()
+    perm, seg_off = ext.rf_partition(node_of_row, lut, T)
+    feat_sel = torch.empty((0, 0), dtype=torch.int32).cuda()
+    H = ext.rf_histogram(Xb, perm, seg_off, feat_sel, y, 0, d, nb, nc, sample)
+    # reference per tree
+    for t in range(T):
+        rows = sample[t * n : (t + 1) * n].long()
+        for f in [0, d - 1]:
+            ref = torch.zeros(nb, nc).cuda()
+            idx = Xb[rows, f].long() * nc + y[rows].long()
+            ref.view(-1).scatter_add_(0, idx, torch.ones_like(idx, dtype=torch.float32))
+            assert torch.allclose(H[t, f], ref), (t, f)
