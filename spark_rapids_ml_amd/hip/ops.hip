@@ -1537,7 +1537,11 @@ __global__ __launch_bounds__(256) void knn_select_kernel(
 // ---------------------------------------------------------------------------
 
 __global__ __launch_bounds__(256) void rf_histogram_kernel(
-    const uint8_t* __restrict__ Xb,   // [n_phys, d] binned
+    const uint8_t* __restrict__ Xcm,  // [d, n_phys] binned, COLUMN-major:
+                                      // per-feature columns keep a segment's
+                                      // row gathers inside dense cache lines
+                                      // (row-major fetched ~47 64B lines per
+                                      // row to use 54 sampled bytes)
     const int64_t* __restrict__ perm, // [m] VIRTUAL rows sorted by node
     const int64_t* __restrict__ seg_off,  // [B+1]
     const int32_t* __restrict__ feat_sel, // [B, mf] or nullptr (identity)
@@ -1569,12 +1573,11 @@ __global__ __launch_bounds__(256) void rf_histogram_kernel(
     const int64_t vrow = perm[r];
     const int64_t row = sample ? (int64_t)sample[vrow]
                                : (vrow >= n_phys ? vrow % n_phys : vrow);
-    const uint8_t* xrow = Xb + row * (int64_t)d;
     const int yc = classif ? y_cls[row] : 0;
     const float yv = classif ? 1.0f : y_reg[row];
     for (int q = 0; q < FC; ++q) {
       const int f = feat_sel ? feat_sel[(int64_t)b * mf + f0 + q] : (f0 + q);
-      const int bin = xrow[f];
+      const int bin = Xcm[(int64_t)f * n_phys + row];
       if (classif) {
         atomicAdd(&lhist[(q * n_bins + bin) * C + yc], 1.0f);
       } else {
@@ -1798,7 +1801,7 @@ __global__ __launch_bounds__(256) void rf_reroute_kernel(
     const int32_t* __restrict__ sbin,         // [ns]
     const int64_t* __restrict__ lchild,       // [ns]
     const int64_t* __restrict__ rchild,       // [ns]
-    const uint8_t* __restrict__ Xb,           // [n_phys, d]
+    const uint8_t* __restrict__ Xcm,          // [d, n_phys] column-major
     const int32_t* __restrict__ sample,       // [vn] or nullptr
     int64_t n_phys,
     int64_t vn, int64_t d) {
@@ -1808,7 +1811,7 @@ __global__ __launch_bounds__(256) void rf_reroute_kernel(
     if (sl < 0) continue;
     const int64_t row = sample ? (int64_t)sample[i]
                                : (i >= n_phys ? i % n_phys : i);
-    const uint8_t bin = Xb[row * d + sfeat[sl]];
+    const uint8_t bin = Xcm[(int64_t)sfeat[sl] * n_phys + row];
     node_of_row[i] = (bin <= (uint8_t)sbin[sl]) ? lchild[sl] : rchild[sl];
   }
 }
@@ -2207,10 +2210,11 @@ torch::Tensor rf_histogram(torch::Tensor Xb, torch::Tensor perm, torch::Tensor s
                            torch::Tensor feat_sel, torch::Tensor y, int64_t f0,
                            int64_t FC, int64_t n_bins, int64_t n_classes,
                            torch::Tensor sample) {
+  // Xb here is the COLUMN-major binned matrix [d, n_phys]
   TORCH_CHECK(Xb.is_cuda() && Xb.dtype() == torch::kUInt8 && Xb.is_contiguous());
   TORCH_CHECK(perm.dtype() == torch::kInt64 && seg_off.dtype() == torch::kInt64);
-  const int d = (int)Xb.size(1);
-  const int64_t n_phys = Xb.size(0);
+  const int d = (int)Xb.size(0);
+  const int64_t n_phys = Xb.size(1);
   const int B = (int)seg_off.size(0) - 1;
   const bool classif = n_classes > 0;
   const int C = classif ? (int)n_classes : 2;
@@ -2288,8 +2292,8 @@ void rf_reroute(torch::Tensor node_of_row, torch::Tensor lut2, torch::Tensor sfe
               sbin.dtype() == torch::kInt32 && lchild.dtype() == torch::kInt64 &&
               rchild.dtype() == torch::kInt64 && Xb.dtype() == torch::kUInt8);
   const int64_t vn = node_of_row.size(0);
-  const int64_t n_phys = Xb.size(0);
-  const int64_t d = Xb.size(1);
+  const int64_t n_phys = Xb.size(1);  // [d, n_phys] column-major
+  const int64_t d = Xb.size(0);
   const bool has_sample = sample.numel() > 0;
   if (has_sample) TORCH_CHECK(sample.dtype() == torch::kInt32 && sample.is_contiguous());
   const unsigned grid = (unsigned)std::min<int64_t>(2048, (vn + 255) / 256 + 1);

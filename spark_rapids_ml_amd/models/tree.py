@@ -408,6 +408,10 @@ def _grow_forest(
         y32 = y_idx.to(torch.int32) if task == "classification" else yf
         C_ch = n_classes if task == "classification" else 2
         fc_kernel = max(1, min(512, (150 * 1024 // 4) // max(1, n_bins * C_ch)))
+        # column-major binned matrix for the histogram/reroute kernels: a
+        # node segment's row gathers stay inside dense per-feature cache
+        # lines (row-major fetched a ~47-line row to read ~54 sampled bytes)
+        Xcm = Xb.T.contiguous()
     frontier = list(range(n_trees))  # arena node ids still splittable
 
     for depth in range(max_depth):
@@ -425,8 +429,8 @@ def _grow_forest(
                 # counting-sort partition kernel: one pass for counts, one
                 # scatter — replaces sort+nonzero+gathers (rf_partition)
                 perm, seg_off = ext.rf_partition(node_of_row, lut, B)
-                if int(seg_off[-1].item()) == 0:
-                    continue
+                # no host sync here: an all-empty batch flows through as
+                # no-op kernels and an empty split set
                 rows = loc = yb = None
             else:
                 local = lut[node_of_row]
@@ -484,7 +488,7 @@ def _grow_forest(
                 F = f1 - f0
                 if hip_hist:
                     H = ext.rf_histogram(
-                        Xb, perm, seg_off, fsel32, y32, f0, F, n_bins,
+                        Xcm, perm, seg_off, fsel32, y32, f0, F, n_bins,
                         n_classes if task == "classification" else 0,
                         sample_t,
                     )
@@ -564,11 +568,25 @@ def _grow_forest(
             # materialize splits: fully vectorized (one bulk leaf append +
             # one set_splits per batch; the per-node loop was the depth-13
             # wall-clock bottleneck)
-            bg = as_numpy(best_gain)
-            bf = as_numpy(best_feat)
-            bb = as_numpy(best_bin)
-            lv = as_numpy(best_lval)
-            rv = as_numpy(best_rval)
+            # ONE packed D2H instead of five device syncs (values fit f32
+            # exactly: features < 2^24, bins < 256)
+            pack = torch.cat(
+                [
+                    best_gain[:, None],
+                    best_feat.to(torch.float32)[:, None],
+                    best_bin.to(torch.float32)[:, None],
+                    best_lval,
+                    best_rval,
+                ],
+                dim=1,
+            )
+            packed = as_numpy(pack)
+            w = best_lval.shape[1]
+            bg = packed[:, 0]
+            bf = packed[:, 1].astype(np.int64)
+            bb = packed[:, 2].astype(np.int64)
+            lv = packed[:, 3 : 3 + w]
+            rv = packed[:, 3 + w :]
             batch_np = np.asarray(batch, dtype=np.int64)
             valid = (bg > min_gain) & (bf >= 0)
             idxs = np.nonzero(valid)[0]
@@ -611,7 +629,7 @@ def _grow_forest(
                 # single-pass kernel: gather split byte + child write fused
                 ext.rf_reroute(
                     node_of_row, lut2,
-                    f_t.to(torch.int32), b_t.to(torch.int32), l_t, r_t, Xb,
+                    f_t.to(torch.int32), b_t.to(torch.int32), l_t, r_t, Xcm,
                     sample_t,
                 )
             else:

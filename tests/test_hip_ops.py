@@ -122,7 +122,7 @@ def test_rf_histogram_matches_ref(ext, task, nc):
     seg_off = torch.searchsorted(sl, torch.arange(B + 1).cuda()).contiguous()
     if task == "classification":
         y = torch.randint(0, nc, (n,), generator=g, dtype=torch.int32).cuda()
-        H = ext.rf_histogram(Xb, perm, seg_off, feat_sel.to(torch.int32), y, 0, mf, nb, nc,
+        H = ext.rf_histogram(Xb.T.contiguous(), perm, seg_off, feat_sel.to(torch.int32), y, 0, mf, nb, nc,
                              torch.empty(0, dtype=torch.int32).cuda())
         ref = torch.zeros(B, mf, nb, nc).cuda()
         for b in range(B):
@@ -139,7 +139,7 @@ def test_rf_histogram_matches_ref(ext, task, nc):
                 )
     else:
         y = torch.randn(n, generator=g).cuda()
-        H = ext.rf_histogram(Xb, perm, seg_off, feat_sel.to(torch.int32), y, 0, mf, nb, 0,
+        H = ext.rf_histogram(Xb.T.contiguous(), perm, seg_off, feat_sel.to(torch.int32), y, 0, mf, nb, 0,
                              torch.empty(0, dtype=torch.int32).cuda())
         ref = torch.zeros(B, mf, nb, 2).cuda()
         for b in range(B):
@@ -269,7 +269,7 @@ def test_rf_reroute_matches_torch(ext):
     ref[mrows] = torch.where(go_left, l_t[srel], r_t[srel])
 
     got = node_of_row.clone()
-    ext.rf_reroute(got, lut2, f_t, b_t, l_t, r_t, Xb,
+    ext.rf_reroute(got, lut2, f_t, b_t, l_t, r_t, Xb.T.contiguous(),
                    torch.empty(0, dtype=torch.int32).cuda())
     assert torch.equal(got.cpu(), ref.cpu())
 
@@ -314,7 +314,7 @@ def test_rf_histogram_virtual_rows(ext):
     lut = torch.arange(T, dtype=torch.int64).cuda()
     perm, seg_off = ext.rf_partition(node_of_row, lut, T)
     feat_sel = torch.empty((0, 0), dtype=torch.int32).cuda()
-    H = ext.rf_histogram(Xb, perm, seg_off, feat_sel, y, 0, d, nb, nc, sample)
+    H = ext.rf_histogram(Xb.T.contiguous(), perm, seg_off, feat_sel, y, 0, d, nb, nc, sample)
     # reference per tree
     for t in range(T):
         rows = sample[t * n : (t + 1) * n].long()
