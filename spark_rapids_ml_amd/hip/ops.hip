@@ -400,11 +400,17 @@ __global__ __launch_bounds__(256) void umap_sgd_epoch_kernel(
 
   const float clip = 4.0f;
   const int h = heads[i], t = tails[i];
-  float eh[DIM], diff[DIM];
+  // head update accumulates in REGISTERS across the positive edge and all
+  // negative samples (sequential-local semantics, like umap-learn's CPU
+  // loop), committed with ONE atomic per component at the end — the
+  // per-sample version fired 2+2*neg_rate head atomics per due edge and the
+  // atomic pipe (not bandwidth) bounded the epoch.
+  float eh[DIM], diff[DIM], upd[DIM];
   float d2 = 0.0f;
 #pragma unroll
   for (int c = 0; c < DIM; ++c) {
     eh[c] = emb[(int64_t)h * DIM + c];
+    upd[c] = 0.0f;
     diff[c] = eh[c] - tail_emb[(int64_t)t * DIM + c];
     d2 += diff[c] * diff[c];
   }
@@ -414,7 +420,7 @@ __global__ __launch_bounds__(256) void umap_sgd_epoch_kernel(
 #pragma unroll
   for (int c = 0; c < DIM; ++c) {
     float g = fminf(fmaxf(gcoef * diff[c], -clip), clip);
-    atomicAdd(&emb[(int64_t)h * DIM + c], alpha * g);
+    upd[c] += g;
     if (move_tail) atomicAdd(&tail_emb[(int64_t)t * DIM + c], -alpha * g);
   }
 
@@ -425,8 +431,7 @@ __global__ __launch_bounds__(256) void umap_sgd_epoch_kernel(
     d2 = 0.0f;
 #pragma unroll
     for (int c = 0; c < DIM; ++c) {
-      eh[c] = emb[(int64_t)h * DIM + c];
-      diff[c] = eh[c] - tail_emb[(int64_t)j * DIM + c];
+      diff[c] = (eh[c] + alpha * upd[c]) - tail_emb[(int64_t)j * DIM + c];
       d2 += diff[c] * diff[c];
     }
     d2c = fmaxf(d2, 1e-12f);
@@ -435,9 +440,12 @@ __global__ __launch_bounds__(256) void umap_sgd_epoch_kernel(
 #pragma unroll
     for (int c = 0; c < DIM; ++c) {
       float g = fminf(fmaxf(gcoef * diff[c], -clip), clip);
-      atomicAdd(&emb[(int64_t)h * DIM + c], alpha * g);
+      upd[c] += g;
     }
   }
+#pragma unroll
+  for (int c = 0; c < DIM; ++c)
+    if (upd[c] != 0.0f) atomicAdd(&emb[(int64_t)h * DIM + c], alpha * upd[c]);
 }
 
 // ---------------------------------------------------------------------------
