@@ -412,7 +412,9 @@ def _grow_forest(
         ext = hip_ops()
         y32 = y_idx.to(torch.int32) if task == "classification" else yf
         C_ch = n_classes if task == "classification" else 2
-        fc_kernel = max(1, min(512, (150 * 1024 // 4) // max(1, n_bins * C_ch)))
+        # ~30 KB LDS per block -> ~5 blocks/CU: the histogram is
+        # LDS-atomic-latency bound, and a 150 KB chunk left only 4 waves/CU
+        fc_kernel = max(1, min(512, (30 * 1024 // 4) // max(1, n_bins * C_ch)))
         # column-major binned matrix for the histogram/reroute kernels: a
         # node segment's row gathers stay inside dense per-feature cache
         # lines (row-major fetched a ~47-line row to read ~54 sampled bytes)
@@ -468,9 +470,13 @@ def _grow_forest(
             # reference gets from cuML's per-node sampling.
             if max_features < d:
                 scores = torch.rand((B, d), generator=gen, device=dev)
-                # partial selection (topk of random scores = uniform subset
-                # without replacement) instead of a full [B, d] argsort
-                feat_sel = scores.topk(max_features, dim=1).indices  # [B, mf]
+                # uniform subset without replacement: topk of random scores
+                # for small mf; torch.topk degrades badly past k~256 on ROCm
+                # (389 ms for one [10,3000] k=1000 call) — radix argsort wins
+                if max_features <= 256:
+                    feat_sel = scores.topk(max_features, dim=1).indices
+                else:
+                    feat_sel = scores.argsort(dim=1)[:, :max_features]
                 mf = max_features
                 Xb_rows = None  # sampled path reads only selected bytes
             else:
