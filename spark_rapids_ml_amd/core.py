@@ -282,10 +282,24 @@ class Model(_NativeParams, HasVerbose):
     def _transformEvaluate(self, df: DataFrame, evaluator) -> List[float]:
         """Single-pass transform+evaluate (reference
         `_CumlModel._transformEvaluate` / `_transform_evaluate_internal`,
-        core.py:1572-1693: one job evaluates the model; multi-model variants
-        merge per-rank sufficient statistics)."""
-        out = self.transform(df)
-        return [evaluator.evaluate(out)]
+        core.py:1572-1693: ONE job evaluates every combined model — features
+        are extracted once and each model's outputs are computed from the
+        same arrays, the `pred.model_index` design without the re-read)."""
+        models: Sequence[Model] = getattr(self, "_combined", None) or [self]
+        features_col, features_cols = self._get_input_columns()
+        X = extract_features(df, features_col, features_cols, self._float32_inputs)
+        metrics: List[float] = []
+        for m in models:
+            with annotate(f"srml::{type(m).__name__}.transformEvaluate"):
+                out = m._transform_array(X)
+            res = df
+            if isinstance(out, dict):
+                for name, col in out.items():
+                    res = res.with_column(name, col)
+            else:
+                res = df.with_column(m._out_col_name(), out)
+            metrics.append(evaluator.evaluate(res))
+        return metrics
 
     def cpu(self):
         """CPU-model conversion (reference `cpu()` builds the Spark JVM model

@@ -195,12 +195,19 @@ class CrossValidator(Params):
         metrics = np.zeros((len(epm), n_folds))
         collect = bool(self.getOrDefault("collectSubModels"))
         sub = [[None] * len(epm) for _ in range(n_folds)] if collect else None
+        from .core import Model
+
         for fold, (train, test) in enumerate(self._kfold(df)):
-            # single data pass over all param maps (reference tuning.py:123-130)
-            for idx, model in self._estimator.fitMultiple(train, epm):
-                out = model.transform(test)
-                metrics[idx, fold] = self._evaluator.evaluate(out)
-                if collect:
+            # single data pass over all param maps (reference tuning.py:123-130):
+            # fitMultiple trains every map against one load, _combine +
+            # _transformEvaluate scores them all in ONE feature extraction
+            # (reference one-job multi-model evaluate, core.py:1572-1693)
+            models = [m for _, m in self._estimator.fitMultiple(train, epm)]
+            combined = Model._combine(models)
+            fold_metrics = combined._transformEvaluate(test, self._evaluator)
+            metrics[:, fold] = fold_metrics
+            if collect:
+                for idx, model in enumerate(models):
                     sub[fold][idx] = model
         avg = metrics.mean(axis=1)
         std = metrics.std(axis=1)

@@ -169,3 +169,35 @@ def test_cv_fold_col():
     bad = DataFrame({"features": X, "label": y, "fold": (folds + 5).astype(np.int64)})
     with pytest.raises(ValueError, match="foldCol"):
         CrossValidator(lr, grid, RegressionEvaluator(), numFolds=3, foldCol="fold").fit(bad)
+
+
+def test_transform_evaluate_single_pass_counts_extractions(monkeypatch):
+    """Combined-model _transformEvaluate must extract features ONCE for all
+    models (reference one-job multi-model evaluate, core.py:1572-1693)."""
+    import spark_rapids_ml_amd.core as core_mod
+    from spark_rapids_ml_amd import LogisticRegression
+    from spark_rapids_ml_amd.core import Model
+    from spark_rapids_ml_amd.data import DataFrame
+    from spark_rapids_ml_amd.evaluation import MulticlassClassificationEvaluator
+
+    X, y = make_classification(n_samples=300, n_features=6, random_state=0)
+    df = DataFrame.from_numpy(X.astype(np.float64), y.astype(np.float64))
+    est = LogisticRegression(maxIter=30)
+    maps = [{est.regParam: 0.0}, {est.regParam: 1.0}, {est.regParam: 10.0}]
+    models = [m for _, m in est.fitMultiple(df, maps)]
+
+    calls = {"n": 0}
+    orig = core_mod.extract_features
+
+    def counting(*a, **k):
+        calls["n"] += 1
+        return orig(*a, **k)
+
+    monkeypatch.setattr(core_mod, "extract_features", counting)
+    combined = Model._combine(models)
+    ev = MulticlassClassificationEvaluator(metricName="accuracy")
+    ms = combined._transformEvaluate(df, ev)
+    assert len(ms) == 3
+    assert calls["n"] == 1, f"expected 1 feature extraction, saw {calls['n']}"
+    # heavier regularization should not beat the unregularized fit here
+    assert ms[0] >= ms[2] - 1e-9
