@@ -195,6 +195,7 @@ class LogisticRegression(_LogisticRegressionParams, Estimator):
         from ..data import _is_sparse
 
         sparse = _is_sparse(X)
+        stream_cap: Optional[int] = None  # set when the dense shard streams
         # enable_sparse_data_optim (reference params.py:45-66): None=auto,
         # False=densify, True=require sparse input
         opt = (
@@ -225,7 +226,17 @@ class LogisticRegression(_LogisticRegressionParams, Estimator):
                 size=Xcsr.shape,
             ).to(ctx.device)
         else:
-            Xt = ctx.device_tensor(X)
+            from ..streaming import should_stream, stream_cap_bytes
+
+            Xnp = np.ascontiguousarray(X)
+            if should_stream(Xnp.nbytes, ctx.device):
+                # shard exceeds the device-data cap: keep X host-resident and
+                # stream chunks per L-BFGS iteration (grad accumulation on
+                # device; reference capacity mechanism utils.py:403-522)
+                stream_cap = stream_cap_bytes(ctx.device)
+                Xt = None
+            else:
+                Xt = ctx.device_tensor(X)
 
         # classes: global sorted unique labels (reference allgathers classes_)
         local_classes = np.unique(np.asarray(y)) if len(np.asarray(y)) else np.array([])
@@ -259,10 +270,24 @@ class LogisticRegression(_LogisticRegressionParams, Estimator):
         C_out = n_classes if multinomial else 1
 
         # feature scale (no centering: Spark's sparsity-preserving scaling)
-        sigma = self._column_std(Xt, comm, n)
+        if stream_cap is not None:
+            from ..streaming import iter_device_chunks
+
+            sbuf = torch.zeros((2, d), dtype=torch.float64, device=ctx.device)
+            for _s, _e, Xc in iter_device_chunks(Xnp, ctx.device, stream_cap):
+                sbuf[0] += Xc.sum(dim=0).to(torch.float64)
+                sbuf[1] += (Xc * Xc).sum(dim=0).to(torch.float64)
+            sbuf = comm.allreduce_t(sbuf)
+            smean = sbuf[0] / n
+            svar = torch.clamp((sbuf[1] - n * smean * smean) / max(1, n - 1), min=0.0)
+            sigma = torch.sqrt(svar)
+        else:
+            sigma = self._column_std(Xt, comm, n)
         sig_safe = torch.where(sigma > 0, sigma, torch.ones_like(sigma))
         XsT = None
-        if sparse:
+        if stream_cap is not None:
+            Xs = None
+        elif sparse:
             inv = (1.0 / sig_safe).to(torch.float32)
             vals = Xt.values() * inv[Xt.col_indices()]
             Xs = torch.sparse_csr_tensor(
@@ -287,11 +312,29 @@ class LogisticRegression(_LogisticRegressionParams, Estimator):
         l2 = lam * (1.0 - l1r)
         l1 = lam * l1r
 
+        def _streamed_grad_loss(W: torch.Tensor):
+            from ..streaming import iter_device_chunks
+
+            inv = (1.0 / sig_safe).to(torch.float32)
+            gacc = None
+            lacc = torch.zeros((), dtype=torch.float64, device=ctx.device)
+            for _s, _e, Xc in iter_device_chunks(
+                Xnp, ctx.device, stream_cap, dtype=torch.float32
+            ):
+                Xcs = Xc * inv[None, :]
+                g, l = logistic_grad_loss(Xcs, y_t[_s:_e], W, fit_intercept)
+                gacc = g.to(torch.float64) if gacc is None else gacc + g.to(torch.float64)
+                lacc += l.to(torch.float64)
+            return gacc, lacc
+
         def closure(wv: torch.Tensor) -> Tuple[float, torch.Tensor]:
             W = wv.to(ctx.device).view(C_out, ncol).to(
-                torch.float32 if sparse else Xs.dtype
+                torch.float32 if (sparse or stream_cap is not None) else Xs.dtype
             )
-            grad, loss = logistic_grad_loss(Xs, y_t, W, fit_intercept, XT=XsT)
+            if stream_cap is not None:
+                grad, loss = _streamed_grad_loss(W)
+            else:
+                grad, loss = logistic_grad_loss(Xs, y_t, W, fit_intercept, XT=XsT)
             buf = torch.zeros(C_out * ncol + 1, dtype=torch.float64, device=grad.device)
             buf[:-1] = grad.to(torch.float64).flatten()
             buf[-1] = loss.to(torch.float64)

@@ -90,13 +90,28 @@ class PCA(_PCAParams, Estimator):
         if k > min(pdesc.m, d):
             raise ValueError(f"k={k} > min(n_rows={pdesc.m}, n_cols={d})")
 
-        Xt = ctx.device_tensor(X)
+        from ..streaming import should_stream, stream_cap_bytes, streamed_moments
+
+        Xnp = np.asarray(X) if not hasattr(X, "toarray") else X
         # fused partials: [d+1, d] buffer = [Gram (d×d) ; colsum (1×d)]
-        buf = torch.zeros((d + 1, d), dtype=torch.float64, device=Xt.device)
-        if Xt.shape[0] > 0:
-            g = gram(Xt)  # f32 MFMA on GPU
-            buf[:d] = g.to(torch.float64)
-            buf[d] = Xt.sum(dim=0).to(torch.float64)
+        if (
+            isinstance(Xnp, np.ndarray)
+            and should_stream(Xnp.nbytes, ctx.device)
+            and Xnp.shape[0] > 0
+        ):
+            # shard exceeds the device-data cap: stream chunks through a
+            # pinned buffer, accumulating the same partials on device
+            mom = streamed_moments(Xnp, None, ctx.device, stream_cap_bytes(ctx.device))
+            buf = torch.zeros((d + 1, d), dtype=torch.float64, device=ctx.device)
+            buf[:d] = mom["G"]
+            buf[d] = mom["xsum"]
+        else:
+            Xt = ctx.device_tensor(X)
+            buf = torch.zeros((d + 1, d), dtype=torch.float64, device=Xt.device)
+            if Xt.shape[0] > 0:
+                g = gram(Xt)  # f32 MFMA on GPU
+                buf[:d] = g.to(torch.float64)
+                buf[d] = Xt.sum(dim=0).to(torch.float64)
         buf = comm.allreduce_t(buf)
         mean = buf[d] / pdesc.m
         cov = cov_from_gram(buf[:d], mean, pdesc.m, ddof=1)

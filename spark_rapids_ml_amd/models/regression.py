@@ -161,18 +161,37 @@ class LinearRegression(_LinearRegressionParams, Estimator):
             return ctx.cache["moments"]
         comm, pdesc = ctx.comm, ctx.pdesc
         d = pdesc.n
-        Xt = to_device_tensor(np.ascontiguousarray(X), ctx.device)
-        yt = to_device_tensor(np.ascontiguousarray(y), ctx.device).to(Xt.dtype)
+        from ..streaming import should_stream, stream_cap_bytes, streamed_moments
+
+        Xnp = np.asarray(X) if not hasattr(X, "toarray") else X
         # fused buffer rows: [0:d]=Gram, d=X^T y, d+1=colsum X, d+2=[ysum, y2sum]
         # (width >= 2 so the label-moment row exists even at d=1)
         w = max(d, 2)
-        buf = torch.zeros((d + 3, w), dtype=torch.float64, device=Xt.device)
-        if Xt.shape[0] > 0:
-            buf[:d, :d] = gram(Xt).to(torch.float64)
-            buf[d, :d] = (Xt.T @ yt).to(torch.float64)
-            buf[d + 1, :d] = Xt.sum(dim=0).to(torch.float64)
-            buf[d + 2, 0] = yt.sum().to(torch.float64)
-            buf[d + 2, 1] = (yt * yt).sum().to(torch.float64)
+        if (
+            isinstance(Xnp, np.ndarray)
+            and should_stream(Xnp.nbytes, ctx.device)
+            and Xnp.shape[0] > 0
+        ):
+            # shard exceeds the device-data cap: pinned-chunk streaming pass
+            mom = streamed_moments(
+                Xnp, np.asarray(y), ctx.device, stream_cap_bytes(ctx.device)
+            )
+            buf = torch.zeros((d + 3, w), dtype=torch.float64, device=ctx.device)
+            buf[:d, :d] = mom["G"]
+            buf[d, :d] = mom["Xty"]
+            buf[d + 1, :d] = mom["xsum"]
+            buf[d + 2, 0] = mom["ysum"]
+            buf[d + 2, 1] = mom["y2sum"]
+        else:
+            Xt = to_device_tensor(np.ascontiguousarray(X), ctx.device)
+            yt = to_device_tensor(np.ascontiguousarray(y), ctx.device).to(Xt.dtype)
+            buf = torch.zeros((d + 3, w), dtype=torch.float64, device=Xt.device)
+            if Xt.shape[0] > 0:
+                buf[:d, :d] = gram(Xt).to(torch.float64)
+                buf[d, :d] = (Xt.T @ yt).to(torch.float64)
+                buf[d + 1, :d] = Xt.sum(dim=0).to(torch.float64)
+                buf[d + 2, 0] = yt.sum().to(torch.float64)
+                buf[d + 2, 1] = (yt * yt).sum().to(torch.float64)
         buf = comm.allreduce_t(buf)
         m = {
             "G": buf[:d, :d],
