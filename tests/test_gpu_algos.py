@@ -235,3 +235,26 @@ def test_umap_sgd_kernel_quality():
     out = model.transform(DataFrame.from_numpy(X[:200]))
     emb_q = np.asarray(out["embedding"])
     assert np.isfinite(emb_q).all()
+
+
+@pytest.mark.gpu
+def test_streaming_fit_on_gpu(monkeypatch):
+    """The pinned-chunk streaming path on device must match in-memory."""
+    import numpy as np
+
+    from sklearn.datasets import make_regression
+
+    from spark_rapids_ml_amd import LinearRegression
+    from spark_rapids_ml_amd.data import DataFrame
+
+    X, y = make_regression(n_samples=200_000, n_features=64, noise=2.0, random_state=0)
+    X = X.astype(np.float32)
+    df = DataFrame.from_numpy(X, y)
+    monkeypatch.setenv("SRML_STREAM_CAP_BYTES", str(8 << 20))  # 8 MB cap, 51 MB data
+    m_stream = LinearRegression().fit(df)
+    monkeypatch.delenv("SRML_STREAM_CAP_BYTES")
+    m_mem = LinearRegression().fit(df)
+    np.testing.assert_allclose(
+        np.asarray(m_stream.coefficients), np.asarray(m_mem.coefficients),
+        rtol=1e-3, atol=1e-4,
+    )
