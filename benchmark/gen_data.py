@@ -60,39 +60,68 @@ def gen_low_rank_matrix(
 
 def gen_regression(
     num_rows: int, num_cols: int, n_informative: Optional[int] = None, noise: float = 1.0,
-    bias: float = 0.5, seed: int = 0, dtype: str = "float32",
-) -> Tuple[np.ndarray, np.ndarray]:
+    bias: float = 0.5, effective_rank: Optional[int] = None, tail_strength: float = 0.5,
+    seed: int = 0, dtype: str = "float32", return_coef: bool = False,
+):
+    """sklearn.make_regression semantics, rank-sharded (reference
+    RegressionDataGen, gen_data_distributed.py:84-968): shared ground-truth
+    coefficients from the GLOBAL seed, per-rank feature/noise draws; optional
+    low-rank X via shared singular profile + right factors."""
     comm = get_comm()
     rng = np.random.default_rng(seed)
     ninf = n_informative or max(1, num_cols // 10)
     w = np.zeros(num_cols)
     w[rng.choice(num_cols, ninf, replace=False)] = rng.normal(scale=10.0, size=ninf)
+    if effective_rank is not None:
+        sing = np.exp(-((np.arange(num_cols) / effective_rank) ** 2)) * (
+            1 - tail_strength
+        ) + tail_strength / (1 + np.arange(num_cols))
+        V = np.linalg.qr(rng.normal(size=(num_cols, num_cols)))[0]
     rng_local = np.random.default_rng(seed + 104729 * (comm.rank + 1))
     n_local = _shard(num_rows, comm.rank, comm.world_size)
     X = rng_local.normal(size=(n_local, num_cols))
+    if effective_rank is not None:
+        X = (X * sing[None, :]) @ V.T
     y = X @ w + bias + noise * rng_local.normal(size=n_local)
+    if return_coef:
+        return X.astype(dtype), y.astype(np.float64), w
     return X.astype(dtype), y.astype(np.float64)
 
 
 def gen_classification(
     num_rows: int, num_cols: int, n_classes: int = 2, n_informative: Optional[int] = None,
+    n_redundant: Optional[int] = None, n_clusters_per_class: int = 1,
+    class_sep: float = 1.0, flip_y: float = 0.01,
     seed: int = 0, dtype: str = "float32",
 ) -> Tuple[np.ndarray, np.ndarray]:
+    """sklearn.make_classification semantics, rank-sharded (reference
+    ClassificationDataGen): shared hypercube-ish centroids per (class,
+    cluster) and a shared redundant-feature mixing matrix come from the
+    GLOBAL seed; rows, cluster choices and label noise are per-rank."""
     comm = get_comm()
     rng = np.random.default_rng(seed)
     ninf = n_informative or max(2, num_cols // 10)
-    # centroid separation chosen so classes overlap (Bayes error ~10-20%):
+    nred = n_redundant if n_redundant is not None else 0
+    assert ninf + nred <= num_cols, "n_informative + n_redundant > num_cols"
     # per-dim signal shrinks with sqrt(ninf) so the problem stays hard as
-    # dimensionality grows (a solver must actually iterate).
-    centroids = rng.normal(scale=2.0 / np.sqrt(ninf), size=(n_classes, ninf))
+    # dimensionality grows (a solver must actually iterate)
+    centroids = rng.normal(
+        scale=2.0 * class_sep / np.sqrt(ninf),
+        size=(n_classes, n_clusters_per_class, ninf),
+    )
+    mix = rng.normal(size=(ninf, nred)) if nred else None
     rng_local = np.random.default_rng(seed + 104729 * (comm.rank + 1))
     n_local = _shard(num_rows, comm.rank, comm.world_size)
     y = rng_local.integers(0, n_classes, n_local)
+    clust = rng_local.integers(0, n_clusters_per_class, n_local)
     X = rng_local.normal(size=(n_local, num_cols))
-    X[:, :ninf] += centroids[y]
-    # 1% label noise
-    flip = rng_local.random(n_local) < 0.01
-    y[flip] = rng_local.integers(0, n_classes, int(flip.sum()))
+    X[:, :ninf] += centroids[y, clust]
+    if mix is not None:
+        # redundant features: linear combinations of the informative block
+        X[:, ninf : ninf + nred] = X[:, :ninf] @ mix / np.sqrt(ninf)
+    if flip_y > 0:
+        flip = rng_local.random(n_local) < flip_y
+        y[flip] = rng_local.integers(0, n_classes, int(flip.sum()))
     return X.astype(dtype), y.astype(np.float64)
 
 

@@ -78,3 +78,50 @@ def test_gen_data_cli(tmp_path):
 
     df = DataFrame.read_parquet(str(tmp_path / "blobs"), vector_cols=["features"])
     assert df.num_rows == 500
+
+
+def test_gen_classification_redundant_and_clusters():
+    X, y = gen_data.gen_classification(
+        3000, 20, n_classes=2, n_informative=6, n_redundant=4,
+        n_clusters_per_class=2, seed=7,
+    )
+    assert X.shape == (3000, 20)
+    # redundant block must be an exact linear function of the informative one
+    A, res, *_ = np.linalg.lstsq(X[:, :6], X[:, 6:10], rcond=None)
+    recon = X[:, :6] @ A
+    assert np.allclose(recon, X[:, 6:10], atol=1e-4)
+    # remaining features are (mostly) independent noise
+    B, *_ = np.linalg.lstsq(X[:, :6], X[:, 10:], rcond=None)
+    resid = X[:, 10:] - X[:, :6] @ B
+    assert resid.std() > 0.8
+
+
+def test_gen_regression_effective_rank_and_coef():
+    X, y, w = gen_data.gen_regression(
+        2000, 24, n_informative=6, noise=0.0, effective_rank=4, seed=9,
+        return_coef=True,
+    )
+    # low-rank spectrum: top-4 singular values carry most of the energy
+    s = np.linalg.svd(np.asarray(X, np.float64), compute_uv=False)
+    assert s[:4].sum() / s.sum() > 0.5
+    # exact relation with zero noise
+    np.testing.assert_allclose(X @ w + 0.5, y, rtol=1e-4, atol=1e-4)
+
+
+def _dist_gen_union(kind):
+    from benchmark import gen_data
+
+    fn = getattr(gen_data, f"gen_{kind}")
+    X, y = fn(4000, 12, seed=21)
+    return X.sum(axis=0), X.shape[0]
+
+
+def test_generators_shard_across_ranks():
+    """2-rank generation must cover the global row count with disjoint,
+    statistically consistent shards (reference gen_data_distributed's
+    mapInPandas sharding)."""
+    from .dist_utils import run_distributed
+
+    results = run_distributed(_dist_gen_union, world_size=2, args=("classification",))
+    assert sum(r[1] for r in results) == 4000
+    assert all(r[1] > 0 for r in results)

@@ -46,3 +46,50 @@ def test_logreg_objective_history_small():
     hist = model.objectiveHistory
     assert len(hist) == model._model_attributes["n_iter_"] + 1
     assert hist[-1] <= hist[0]
+
+
+@pytest.mark.gpu
+def test_large_logreg_objective_10m_gpu():
+    """tests_large analog (reference test_large_logistic_regression.py:39-60):
+    at 10M x 64 the GPU fit's final objective must descend monotonically and
+    beat a subsample-estimated objective bound."""
+    from benchmark import gen_data
+
+    X, y = gen_data.gen_classification(
+        10_000_000, 64, n_classes=2, n_informative=32, seed=0
+    )
+    lam = 1e-6
+    model = LogisticRegression(
+        regParam=lam, standardization=False, maxIter=60, tol=1e-9
+    ).fit(DataFrame.from_numpy(X, y))
+    hist = model.objectiveHistory
+    assert len(hist) >= 3
+    assert np.all(np.diff(hist) <= 1e-12)
+
+    # objective sanity on a held subsample with the fitted coefficients
+    w = np.asarray(model.coefficients, np.float64).ravel()
+    b = float(np.asarray(model.intercept_).ravel()[0])
+    sub = slice(0, 500_000)
+    z = X[sub].astype(np.float64) @ w + b
+    t = 2 * y[sub] - 1
+    sub_obj = float(np.mean(np.logaddexp(0, -t * z)))
+    assert abs(sub_obj - hist[-1]) < 0.02
+    # accuracy must be far above chance on informative data
+    acc = float(((z > 0) == (y[sub] > 0)).mean())
+    assert acc > 0.8
+
+
+@pytest.mark.gpu
+def test_large_linreg_exact_recovery_20m_gpu():
+    """20M x 32 linear regression recovers the planted coefficients."""
+    from benchmark import gen_data
+
+    from spark_rapids_ml_amd import LinearRegression
+
+    X, y, w = gen_data.gen_regression(
+        20_000_000, 32, n_informative=16, noise=0.5, seed=1, return_coef=True
+    )
+    m = LinearRegression().fit(DataFrame.from_numpy(X, y))
+    got = np.asarray(m.coefficients, np.float64)
+    np.testing.assert_allclose(got, w, atol=2e-3)
+    assert np.isclose(m.intercept, 0.5, atol=2e-3)
