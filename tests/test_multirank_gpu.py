@@ -1,0 +1,110 @@
+"""Multi-rank fits WITH DEVICE TENSORS on a single GPU box (VERDICT r01
+next-step #5): two gloo ranks share cuda:0, so every collective call site
+(fused Lloyd all-reduce, logreg grad, kNN merge/allgather, DBSCAN
+replication, PartitionDescriptor, bin-edge gather) runs with device-built
+buffers through the world_size=2 wiring — the exact code path an 8-GPU RCCL
+run takes, minus the backend (which is torchrun-selected; see
+docs/SCALE_CHECKLIST.md for the 8-GPU contract)."""
+
+import numpy as np
+import pytest
+
+from .dist_utils import run_distributed
+
+pytestmark = pytest.mark.gpu
+
+
+def _dist_kmeans_gpu(_):
+    import torch
+
+    from spark_rapids_ml_amd import KMeans
+    from spark_rapids_ml_amd.data import DataFrame
+    from spark_rapids_ml_amd.parallel.context import get_comm
+
+    comm = get_comm()
+    assert comm.device.type == "cuda", f"rank {comm.rank} not on GPU"
+    rng = np.random.default_rng(comm.rank)
+    X = rng.normal(size=(20000, 64)).astype(np.float32)
+    m = KMeans(k=8, maxIter=10, seed=1).fit(DataFrame.from_numpy(X))
+    return np.sort(np.asarray(m.cluster_centers_), axis=0)
+
+
+def test_kmeans_two_ranks_device(monkeypatch):
+    res = run_distributed(_dist_kmeans_gpu, world_size=2)
+    np.testing.assert_allclose(res[0], res[1], rtol=1e-4, atol=1e-4)
+
+
+def _dist_logreg_gpu(_):
+    from sklearn.datasets import make_classification
+
+    from spark_rapids_ml_amd import LogisticRegression
+    from spark_rapids_ml_amd.data import DataFrame
+    from spark_rapids_ml_amd.parallel.context import get_comm
+
+    comm = get_comm()
+    X, y = make_classification(n_samples=40000, n_features=32, random_state=0)
+    sl = slice(comm.rank * 20000, (comm.rank + 1) * 20000)
+    m = LogisticRegression(maxIter=40).fit(
+        DataFrame.from_numpy(X[sl].astype(np.float64), y[sl].astype(np.float64))
+    )
+    return np.asarray(m.coefficients)
+
+
+def test_logreg_two_ranks_device():
+    res = run_distributed(_dist_logreg_gpu, world_size=2)
+    np.testing.assert_allclose(res[0], res[1], rtol=1e-6)
+
+
+def _dist_knn_gpu(_):
+    from spark_rapids_ml_amd import NearestNeighbors
+    from spark_rapids_ml_amd.data import DataFrame
+    from spark_rapids_ml_amd.parallel.context import get_comm
+
+    comm = get_comm()
+    rng = np.random.default_rng(3)
+    items = rng.normal(size=(4000, 32)).astype(np.float32)
+    queries = rng.normal(size=(100, 32)).astype(np.float32)
+    sl = slice(comm.rank * 2000, (comm.rank + 1) * 2000)
+    nn = NearestNeighbors(k=8).fit(DataFrame.from_numpy(items[sl]))
+    _, _, knn_df = nn.kneighbors(DataFrame.from_numpy(queries))
+    return np.asarray(knn_df["distances"])
+
+
+def test_knn_merge_two_ranks_device():
+    res = run_distributed(_dist_knn_gpu, world_size=2)
+    np.testing.assert_allclose(res[0], res[1], rtol=1e-4)
+    # parity vs single-process exact result
+    import torch
+
+    from spark_rapids_ml_amd.ops import torch_ref
+
+    rng = np.random.default_rng(3)
+    items = torch.from_numpy(rng.normal(size=(4000, 32)).astype(np.float32)).cuda()
+    queries = torch.from_numpy(rng.normal(size=(100, 32)).astype(np.float32)).cuda()
+    rd, _ = torch_ref.knn_topk(queries, items, 8)
+    np.testing.assert_allclose(res[0], rd.cpu().numpy(), rtol=1e-3, atol=1e-3)
+
+
+def _dist_dbscan_gpu(_):
+    from spark_rapids_ml_amd import DBSCAN
+    from spark_rapids_ml_amd.data import DataFrame
+    from spark_rapids_ml_amd.parallel.context import get_comm
+
+    comm = get_comm()
+    rng = np.random.default_rng(0)
+    blob_a = rng.normal(0.0, 0.2, size=(500, 8))
+    blob_b = rng.normal(5.0, 0.2, size=(500, 8))
+    X = np.concatenate([blob_a, blob_b]).astype(np.float32)
+    sl = slice(comm.rank * 500, (comm.rank + 1) * 500)
+    model = DBSCAN(eps=1.0, min_samples=5).fit(DataFrame.from_numpy(X[sl]))
+    out = model.transform(DataFrame.from_numpy(X[sl]))
+    return np.asarray(out["prediction"])
+
+
+def test_dbscan_two_ranks_device():
+    res = run_distributed(_dist_dbscan_gpu, world_size=2)
+    # rank 0 holds blob A rows, rank 1 blob B rows; each shard must be one
+    # cluster with consistent global labels across the replicated compute
+    assert len(set(res[0].tolist())) == 1
+    assert len(set(res[1].tolist())) == 1
+    assert res[0][0] != res[1][0]
