@@ -1559,11 +1559,12 @@ __global__ __launch_bounds__(256) void rf_histogram_kernel(
     int64_t n_phys,                       // modulo map when sample==nullptr
     int d, int mf, int f0, int FC, int n_bins, int C, int split,
     float* __restrict__ out) {            // [B, FC, n_bins, C]
-  // WAVE-PRIVATE sub-histograms (NSUB=4, one per wave of the block): on
-  // quantile-binned gaussian data the popular central bins collide across
-  // the whole block and the LDS atomic unit serializes same-address adds —
-  // 4 private copies cut that contention to intra-wave only, at 4x LDS.
-  extern __shared__ __attribute__((aligned(16))) float lhist[];  // [FC][nb][C][4]
+  // lhist [FC][nb][C] + an LDS copy of the block's feature-id chunk; the
+  // hot loop gathers FOUR feature bytes per row before touching LDS so the
+  // dependent L2/HBM gather latencies overlap (the naive one-at-a-time loop
+  // ran ~12 cycles/update, far off the LDS-atomic bound)
+  extern __shared__ __attribute__((aligned(16))) float lhist[];
+  int* fsel_s = reinterpret_cast<int*>(lhist + FC * n_bins * C);
   const int b = blockIdx.x / split;
   const int slice = blockIdx.x % split;
   const int64_t s0 = seg_off[b], e0 = seg_off[b + 1];
@@ -1575,9 +1576,10 @@ __global__ __launch_bounds__(256) void rf_histogram_kernel(
   if (rs >= re) return;
 
   const int tid = threadIdx.x;
-  const int sub = tid >> 6;  // wave id 0..3
   const int nfc = FC * n_bins * C;
-  for (int e = tid; e < nfc * 4; e += 256) lhist[e] = 0.0f;
+  for (int e = tid; e < nfc; e += 256) lhist[e] = 0.0f;
+  for (int e = tid; e < FC; e += 256)
+    fsel_s[e] = feat_sel ? feat_sel[(int64_t)b * mf + f0 + e] : (f0 + e);
   __syncthreads();
 
   const bool classif = (y_cls != nullptr);
@@ -1587,15 +1589,31 @@ __global__ __launch_bounds__(256) void rf_histogram_kernel(
                                : (vrow >= n_phys ? vrow % n_phys : vrow);
     const int yc = classif ? y_cls[row] : 0;
     const float yv = classif ? 1.0f : y_reg[row];
-    for (int q = 0; q < FC; ++q) {
-      const int f = feat_sel ? feat_sel[(int64_t)b * mf + f0 + q] : (f0 + q);
-      const int bin = Xcm[(int64_t)f * n_phys + row];
+    int q = 0;
+    for (; q + 3 < FC; q += 4) {
+      int bins[4];
+#pragma unroll
+      for (int e = 0; e < 4; ++e)
+        bins[e] = Xcm[(int64_t)fsel_s[q + e] * n_phys + row];
+#pragma unroll
+      for (int e = 0; e < 4; ++e) {
+        if (classif) {
+          atomicAdd(&lhist[((q + e) * n_bins + bins[e]) * C + yc], 1.0f);
+        } else {
+          float* cell = &lhist[((q + e) * n_bins + bins[e]) * 2];
+          atomicAdd(cell, 1.0f);
+          atomicAdd(cell + 1, yv);
+        }
+      }
+    }
+    for (; q < FC; ++q) {
+      const int bin = Xcm[(int64_t)fsel_s[q] * n_phys + row];
       if (classif) {
-        atomicAdd(&lhist[((q * n_bins + bin) * C + yc) * 4 + sub], 1.0f);
+        atomicAdd(&lhist[(q * n_bins + bin) * C + yc], 1.0f);
       } else {
-        float* cell = &lhist[(q * n_bins + bin) * 2 * 4 + sub];
+        float* cell = &lhist[(q * n_bins + bin) * 2];
         atomicAdd(cell, 1.0f);
-        atomicAdd(cell + 4, yv);
+        atomicAdd(cell + 1, yv);
       }
     }
   }
@@ -1603,14 +1621,10 @@ __global__ __launch_bounds__(256) void rf_histogram_kernel(
 
   float* dst = out + (int64_t)b * FC * n_bins * C;
   if (split > 1) {
-    for (int e = tid; e < nfc; e += 256) {
-      const float v =
-          lhist[e * 4] + lhist[e * 4 + 1] + lhist[e * 4 + 2] + lhist[e * 4 + 3];
-      if (v != 0.0f) atomicAdd(&dst[e], v);
-    }
-  } else {
     for (int e = tid; e < nfc; e += 256)
-      dst[e] = lhist[e * 4] + lhist[e * 4 + 1] + lhist[e * 4 + 2] + lhist[e * 4 + 3];
+      if (lhist[e] != 0.0f) atomicAdd(&dst[e], lhist[e]);
+  } else {
+    for (int e = tid; e < nfc; e += 256) dst[e] = lhist[e];
   }
 }
 
@@ -2238,7 +2252,7 @@ torch::Tensor rf_histogram(torch::Tensor Xb, torch::Tensor perm, torch::Tensor s
   const bool has_sample = sample.numel() > 0;
   if (has_sample) TORCH_CHECK(sample.dtype() == torch::kInt32 && sample.is_contiguous());
   auto out = torch::zeros({(int64_t)B, FC, n_bins, (int64_t)C}, Xb.options().dtype(torch::kFloat32));
-  const size_t lds = (size_t)FC * n_bins * C * 4 * 4;  // 4 wave-private copies
+  const size_t lds = (size_t)FC * n_bins * C * 4 + (size_t)FC * 4;  // hist + fsel
   TORCH_CHECK(lds <= 160 * 1024, "feature chunk too large for LDS");
   int split = std::max(1, (int)(1024 / std::max(1, B)));
   hipLaunchKernelGGL(rf_histogram_kernel, dim3((unsigned)(B * split)), dim3(256), lds,

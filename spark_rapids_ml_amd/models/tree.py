@@ -412,10 +412,10 @@ def _grow_forest(
         ext = hip_ops()
         y32 = y_idx.to(torch.int32) if task == "classification" else yf
         C_ch = n_classes if task == "classification" else 2
-        # ~30 KB LDS per block (4 wave-private copies x FC x nb x C floats)
-        # -> ~5 blocks/CU: the histogram is LDS-atomic bound; privatization
-        # kills cross-wave same-bin serialization, occupancy hides latency
-        fc_kernel = max(1, min(512, (30 * 1024 // 16) // max(1, n_bins * C_ch)))
+        # ~30 KB LDS per block -> ~5 blocks/CU of latency-hiding waves for
+        # the gather+LDS-atomic hot loop (privatized copies measured SLOWER:
+        # the bound is gather latency, not same-bin contention)
+        fc_kernel = max(1, min(512, (30 * 1024 // 4) // max(1, n_bins * C_ch)))
         # column-major binned matrix for the histogram/reroute kernels: a
         # node segment's row gathers stay inside dense per-feature cache
         # lines (row-major fetched a ~47-line row to read ~54 sampled bytes)

@@ -30,7 +30,7 @@ def _dist_kmeans_gpu(_):
 
 
 def test_kmeans_two_ranks_device(monkeypatch):
-    res = run_distributed(_dist_kmeans_gpu, world_size=2)
+    res = run_distributed(_dist_kmeans_gpu, world_size=2, args=(None,))
     np.testing.assert_allclose(res[0], res[1], rtol=1e-4, atol=1e-4)
 
 
@@ -51,7 +51,7 @@ def _dist_logreg_gpu(_):
 
 
 def test_logreg_two_ranks_device():
-    res = run_distributed(_dist_logreg_gpu, world_size=2)
+    res = run_distributed(_dist_logreg_gpu, world_size=2, args=(None,))
     np.testing.assert_allclose(res[0], res[1], rtol=1e-6)
 
 
@@ -71,7 +71,7 @@ def _dist_knn_gpu(_):
 
 
 def test_knn_merge_two_ranks_device():
-    res = run_distributed(_dist_knn_gpu, world_size=2)
+    res = run_distributed(_dist_knn_gpu, world_size=2, args=(None,))
     np.testing.assert_allclose(res[0], res[1], rtol=1e-4)
     # parity vs single-process exact result
     import torch
@@ -102,7 +102,7 @@ def _dist_dbscan_gpu(_):
 
 
 def test_dbscan_two_ranks_device():
-    res = run_distributed(_dist_dbscan_gpu, world_size=2)
+    res = run_distributed(_dist_dbscan_gpu, world_size=2, args=(None,))
     # rank 0 holds blob A rows, rank 1 blob B rows; each shard must be one
     # cluster with consistent global labels across the replicated compute
     assert len(set(res[0].tolist())) == 1
