@@ -829,6 +829,11 @@ class _RandomForestEstimator(_RandomForestParams, Estimator):
                     0, n, (my_trees * n_draw,), generator=gen,
                     device=ctx.device, dtype=torch.int32,
                 )
+                # sort each tree's draw: a bootstrap multiset is unordered,
+                # and ascending physical rows turn the histogram's random
+                # per-row byte gathers into near-sequential column reads
+                # (random gathers fetched a 64B line per byte)
+                sample = sample.view(my_trees, n_draw).sort(dim=1).values.reshape(-1)
             else:
                 sample = None
             trees = _grow_forest(
