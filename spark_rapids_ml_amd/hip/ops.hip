@@ -1979,6 +1979,57 @@ __global__ __launch_bounds__(256) void gather_dists_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// fil_predict: forest inference (the reference's FIL predict kernel,
+// tree.py:709-721 / cuML FIL). All trees live flattened in one node arena
+// (per-tree root offsets); one THREAD walks every tree for its row,
+// accumulating leaf values (class-count vote or regression mean). Routing
+// rule is the training one: x < threshold goes LEFT.
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(256) void fil_predict_kernel(
+    const float* __restrict__ X,        // [n, d]
+    const int32_t* __restrict__ feat,   // arena [total_nodes]
+    const float* __restrict__ thr,      // arena
+    const int32_t* __restrict__ left,   // arena (tree-local ids)
+    const int32_t* __restrict__ right,  // arena
+    const float* __restrict__ value,    // arena [total_nodes, vw]
+    const int64_t* __restrict__ roots,  // [T] arena offset of each tree root
+    int64_t n, int d, int T, int vw, int classif,
+    float* __restrict__ out) {          // [n, vw] summed votes / means
+  const int64_t stride = (int64_t)gridDim.x * 256;
+  float acc[16];  // vw <= 16 (host falls back to torch past that)
+  for (int64_t i = (int64_t)blockIdx.x * 256 + threadIdx.x; i < n; i += stride) {
+    const float* xrow = X + i * d;
+    for (int c = 0; c < vw; ++c) acc[c] = 0.0f;
+    for (int t = 0; t < T; ++t) {
+      const int64_t base = roots[t];
+      int node = 0;
+      int f = feat[base + node];
+      while (f >= 0) {
+        node = (xrow[f] < thr[base + node]) ? left[base + node]
+                                            : right[base + node];
+        f = feat[base + node];
+      }
+      const float* v = value + (base + node) * vw;
+      if (classif) {
+        // normalized class-count vote (matches the torch traversal)
+        float s = 0.0f;
+        for (int c = 0; c < vw; ++c) s += v[c];
+        s = fmaxf(s, 1e-12f);
+        for (int c = 0; c < vw; ++c) acc[c] += v[c] / s;
+      } else {
+        acc[0] += v[0];  // (mean, count) leaves: mean is v[0]
+      }
+    }
+    if (classif) {
+      for (int c = 0; c < vw; ++c) out[i * vw + c] = acc[c];
+    } else {
+      out[i * vw + 0] = acc[0];
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // softmax_residual_loss: per-row softmax (C>1) or sigmoid (C==1) residual
 // and summed log-loss. resid = softmax(scores) - onehot(y) (or p - y).
 // One wave per row chunk; memory-bound, fused to one pass.
@@ -2369,7 +2420,33 @@ torch::Tensor gather_dists(torch::Tensor A, torch::Tensor B, torch::Tensor cand)
   return out;
 }
 
+torch::Tensor fil_predict(torch::Tensor X, torch::Tensor feat, torch::Tensor thr,
+                          torch::Tensor left, torch::Tensor right, torch::Tensor value,
+                          torch::Tensor roots, bool classif) {
+  TORCH_CHECK(X.is_cuda() && X.dtype() == torch::kFloat32 && X.is_contiguous());
+  TORCH_CHECK(feat.dtype() == torch::kInt32 && left.dtype() == torch::kInt32 &&
+              right.dtype() == torch::kInt32 && thr.dtype() == torch::kFloat32 &&
+              value.dtype() == torch::kFloat32 && roots.dtype() == torch::kInt64);
+  const int64_t n = X.size(0);
+  const int d = (int)X.size(1);
+  const int T = (int)roots.size(0);
+  const int vw = (int)value.size(1);
+  TORCH_CHECK(vw <= 16, "fil_predict supports value width <= 16");
+  auto out = torch::zeros({n, (int64_t)vw}, X.options());
+  if (n > 0 && T > 0) {
+    const unsigned grid = (unsigned)std::min<int64_t>(4096, (n + 255) / 256 + 1);
+    hipLaunchKernelGGL(fil_predict_kernel, dim3(grid), dim3(256), 0, cur_stream(),
+                       X.data_ptr<float>(), feat.data_ptr<int32_t>(),
+                       thr.data_ptr<float>(), left.data_ptr<int32_t>(),
+                       right.data_ptr<int32_t>(), value.data_ptr<float>(),
+                       roots.data_ptr<int64_t>(), n, d, T, vw, classif ? 1 : 0,
+                       out.data_ptr<float>());
+  }
+  return out;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("fil_predict", &fil_predict, "forest inference (FIL-style traversal)");
   m.def("kmeans_assign", &kmeans_assign, "fused MFMA distance + argmin");
   m.def("label_accumulate", &label_accumulate, "per-center sum/count scatter");
   m.def("gram_f32", &gram_f32, "A^T A via MFMA f32");

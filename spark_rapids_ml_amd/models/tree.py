@@ -994,15 +994,65 @@ class _RandomForestModel(_RandomForestParams, Model):
     def setPredictionCol(self, value: str):
         return self._set_params(predictionCol=value)
 
+    def _device_arena(self, device) -> Dict[str, torch.Tensor]:
+        """Flatten all trees into one device node arena for fil_predict
+        (cached on the model; rebuilt only on device change)."""
+        cached = getattr(self, "_fil_arena", None)
+        if cached is not None and cached["feat"].device == device:
+            return cached
+        feats, thrs, lefts, rights, vals, roots = [], [], [], [], [], []
+        off = 0
+        for t in self.trees:
+            m = t["feature"].shape[0]
+            roots.append(off)
+            # leaves marked by feature = -1 (the kernel's stop test)
+            f = np.where(t["is_leaf"], -1, t["feature"]).astype(np.int32)
+            feats.append(f)
+            thrs.append(t["threshold"].astype(np.float32))
+            lefts.append(t["left"].astype(np.int32))
+            rights.append(t["right"].astype(np.int32))
+            vals.append(np.atleast_2d(t["value"]).astype(np.float32))
+            off += m
+        arena = {
+            "feat": torch.from_numpy(np.concatenate(feats)).to(device),
+            "thr": torch.from_numpy(np.concatenate(thrs)).to(device),
+            "left": torch.from_numpy(np.concatenate(lefts)).to(device),
+            "right": torch.from_numpy(np.concatenate(rights)).to(device),
+            "value": torch.from_numpy(np.concatenate(vals, axis=0)).contiguous().to(device),
+            "roots": torch.tensor(roots, dtype=torch.int64, device=device),
+        }
+        self._fil_arena = arena
+        return arena
+
     def _predict_raw(self, X: Any) -> torch.Tensor:
-        """Vectorized forest traversal (the FIL-analog inference path,
-        reference tree.py:682-721): iterate depth levels with gathers.
-        Returns [n, C] vote/probability sums (classification) or [n] mean."""
+        """Forest inference. GPU: the fil_predict HIP kernel — one thread
+        walks every tree for its row over a flattened node arena (the
+        reference's FIL predict, tree.py:709-721). CPU fallback: vectorized
+        level-wise torch traversal. Returns [n, C] vote sums
+        (classification) or [n] mean."""
+        from ..ops.dispatch import has_hip_ops, hip_ops, use_hip
         from ..parallel.context import get_comm
 
         device = get_comm().device
         Xt = to_device_tensor(np.ascontiguousarray(X, dtype=np.float32), device)
         n = Xt.shape[0]
+        vw = self.numClasses if self._task == "classification" else 2
+        if (
+            Xt.is_cuda
+            and use_hip(Xt)
+            and has_hip_ops()
+            and 0 < vw <= 16
+            and n > 0
+            and self.numTrees > 0
+        ):
+            a = self._device_arena(device)
+            out = hip_ops().fil_predict(
+                Xt.contiguous(), a["feat"], a["thr"], a["left"], a["right"],
+                a["value"], a["roots"], self._task == "classification",
+            )
+            if self._task == "classification":
+                return out
+            return out[:, 0]
         if self._task == "classification":
             acc = torch.zeros((n, self.numClasses), dtype=torch.float32, device=device)
         else:

@@ -323,3 +323,59 @@ def test_rf_histogram_virtual_rows(ext):
             idx = Xb[rows, f].long() * nc + y[rows].long()
             ref.view(-1).scatter_add_(0, idx, torch.ones_like(idx, dtype=torch.float32))
             assert torch.allclose(H[t, f], ref), (t, f)
+
+
+@pytest.mark.parametrize("task", ["classification", "regression"])
+def test_fil_predict_matches_torch_traversal(ext, task):
+    """fil_predict kernel vs the level-wise torch traversal on a real fit."""
+    import numpy as np
+
+    from sklearn.datasets import make_classification, make_regression
+
+    from spark_rapids_ml_amd import RandomForestClassifier, RandomForestRegressor
+    from spark_rapids_ml_amd.data import DataFrame
+
+    if task == "classification":
+        X, y = make_classification(n_samples=3000, n_features=12, n_classes=3,
+                                   n_informative=6, n_clusters_per_class=1,
+                                   random_state=0)
+        est = RandomForestClassifier(numTrees=7, maxDepth=6, seed=1)
+    else:
+        X, y = make_regression(n_samples=3000, n_features=12, noise=3.0, random_state=0)
+        est = RandomForestRegressor(numTrees=7, maxDepth=6, seed=1)
+    X = X.astype(np.float32)
+    m = est.fit(DataFrame.from_numpy(X, y.astype(np.float64)))
+
+    Xt = torch.from_numpy(X).cuda()
+    got = m._predict_raw(X)  # GPU -> fil kernel path
+
+    # torch reference traversal on device
+    device = Xt.device
+    n = Xt.shape[0]
+    if task == "classification":
+        ref = torch.zeros((n, m.numClasses), dtype=torch.float32, device=device)
+    else:
+        ref = torch.zeros(n, dtype=torch.float32, device=device)
+    for t in m.trees:
+        feature = torch.from_numpy(t["feature"]).to(device, torch.int64)
+        thr = torch.from_numpy(t["threshold"]).to(device)
+        left = torch.from_numpy(t["left"]).to(device, torch.int64)
+        right = torch.from_numpy(t["right"]).to(device, torch.int64)
+        leaf = torch.from_numpy(t["is_leaf"]).to(device)
+        value = torch.from_numpy(t["value"]).to(device)
+        node = torch.zeros(n, dtype=torch.int64, device=device)
+        while True:
+            at_leaf = leaf[node]
+            if bool(at_leaf.all()):
+                break
+            f = feature[node].clamp(min=0)
+            xv = Xt.gather(1, f.view(-1, 1)).flatten()
+            go_left = xv < thr[node]
+            nxt = torch.where(go_left, left[node], right[node])
+            node = torch.where(at_leaf, node, nxt)
+        v = value[node]
+        if task == "classification":
+            ref += v / torch.clamp(v.sum(dim=1, keepdim=True), min=1e-12)
+        else:
+            ref += v[:, 0]
+    assert torch.allclose(got, ref, rtol=1e-4, atol=1e-4)
