@@ -404,6 +404,28 @@ def _grow_forest(
 
     hip_hist = Xb.is_cuda and use_hip(Xb) and has_hip_ops()
     _rf_debug = _os.environ.get("SRML_RF_DEBUG") == "1"
+    _rf_t = _os.environ.get("SRML_RF_DEBUG") == "2"
+    _acc: Dict[str, float] = {}
+
+    def _tick(label, t0):
+        if not _rf_t:
+            return 0.0
+        import time as _tm
+
+        if Xb.is_cuda:
+            torch.cuda.synchronize()
+        t = _tm.perf_counter()
+        _acc[label] = _acc.get(label, 0.0) + (t - t0)
+        return t
+
+    def _now():
+        if not _rf_t:
+            return 0.0
+        import time as _tm
+
+        if Xb.is_cuda:
+            torch.cuda.synchronize()
+        return _tm.perf_counter()
     if _rf_debug:
         print(f"[rf-debug] hip_hist={hip_hist} n={n} d={d} device={dev}", flush=True)
     if hip_hist:
@@ -434,9 +456,11 @@ def _grow_forest(
             lut = torch.full((tree.n_nodes,), -1, dtype=torch.int64, device=dev)
             lut[batch_t] = torch.arange(B, dtype=torch.int64, device=dev)
             if hip_hist:
+                _t0 = _now()
                 # counting-sort partition kernel: one pass for counts, one
                 # scatter — replaces sort+nonzero+gathers (rf_partition)
                 perm, seg_off = ext.rf_partition(node_of_row, lut, B)
+                _tick("partition", _t0)
                 # no host sync here: an all-empty batch flows through as
                 # no-op kernels and an empty split set
                 rows = loc = yb = None
@@ -469,6 +493,7 @@ def _grow_forest(
             # sampled features (a gather per node), not all d then masked —
             # at sqrt(d) sampling this is the d/sqrt(d)-fold work saver the
             # reference gets from cuML's per-node sampling.
+            _t0 = _now()
             if max_features < d:
                 scores = torch.rand((B, d), generator=gen, device=dev)
                 # uniform subset without replacement: topk of random scores
@@ -478,6 +503,7 @@ def _grow_forest(
                     feat_sel = scores.topk(max_features, dim=1).indices
                 else:
                     feat_sel = scores.argsort(dim=1)[:, :max_features]
+                _tick("feat_sel", _t0)
                 mf = max_features
                 Xb_rows = None  # sampled path reads only selected bytes
             else:
@@ -495,6 +521,7 @@ def _grow_forest(
                 if feat_sel is None:
                     Xb_rows = Xb[prows]
                 fc_step = feat_chunk
+            _t0 = _now()
             for f0 in range(0, mf, fc_step):
                 f1 = min(mf, f0 + fc_step)
                 F = f1 - f0
@@ -577,6 +604,8 @@ def _grow_forest(
                 best_lval = torch.where(upd[:, None], lval[ar, fidx], best_lval)
                 best_rval = torch.where(upd[:, None], rval[ar, fidx], best_rval)
 
+            _tick("hist_scan", _t0)
+            _t0 = _now()
             # materialize splits: fully vectorized (one bulk leaf append +
             # one set_splits per batch; the per-node loop was the depth-13
             # wall-clock bottleneck)
@@ -629,6 +658,8 @@ def _grow_forest(
                     rcount = rv[idxs].sum(axis=1)
                 new_frontier.extend(l_ids[lcount >= 2 * min_leaf].tolist())
                 new_frontier.extend(r_ids[rcount >= 2 * min_leaf].tolist())
+            _tick("materialize", _t0)
+            _t0 = _now()
             # reroute rows of split nodes (touch only the split feature byte)
             nid_t = torch.from_numpy(batch_np[idxs]).to(dev)
             f_t = torch.from_numpy(fsel).to(dev)
@@ -651,6 +682,9 @@ def _grow_forest(
                 pm = phys_all[mrows] if phys_all is not None else mrows % n
                 go_left = Xb[pm, f_t[srel]].to(torch.int64) <= b_t[srel]
                 node_of_row[mrows] = torch.where(go_left, l_t[srel], r_t[srel])
+            _tick("reroute", _t0)
+        if _rf_t:
+            pass
         if _rf_debug:
             print(
                 f"[rf-debug] depth={depth} frontier={len(frontier)} -> "
@@ -658,6 +692,8 @@ def _grow_forest(
                 flush=True,
             )
         frontier = new_frontier
+    if _rf_t:
+        print(f"[rf-times] {sorted(_acc.items(), key=lambda kv: -kv[1])}", flush=True)
 
     return tree.split_by_tree(n_trees)
 
