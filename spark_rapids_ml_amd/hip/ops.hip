@@ -1569,14 +1569,21 @@ __global__ __launch_bounds__(256) void rf_histogram_kernel(
   const int slice = blockIdx.x % split;
   const int64_t s0 = seg_off[b], e0 = seg_off[b + 1];
   const int64_t len = e0 - s0;
-  if (len == 0) return;
+  const int tid = threadIdx.x;
+  const int nfc = FC * n_bins * C;
+  if (len == 0) {
+    // split==1 gets an UNINITIALIZED out buffer (no host zero-fill): an
+    // empty segment must still define its cells
+    if (split == 1) {
+      float* dst0 = out + (int64_t)b * nfc;
+      for (int e = tid; e < nfc; e += 256) dst0[e] = 0.0f;
+    }
+    return;
+  }
   const int64_t chunk = (len + split - 1) / split;
   const int64_t rs = s0 + slice * chunk;
   const int64_t re = min(e0, rs + chunk);
   if (rs >= re) return;
-
-  const int tid = threadIdx.x;
-  const int nfc = FC * n_bins * C;
   for (int e = tid; e < nfc; e += 256) lhist[e] = 0.0f;
   for (int e = tid; e < FC; e += 256)
     fsel_s[e] = feat_sel ? feat_sel[(int64_t)b * mf + f0 + e] : (f0 + e);
@@ -2302,10 +2309,13 @@ torch::Tensor rf_histogram(torch::Tensor Xb, torch::Tensor perm, torch::Tensor s
   const int mf = feat_sel.numel() > 0 ? (int)feat_sel.size(1) : 0;
   const bool has_sample = sample.numel() > 0;
   if (has_sample) TORCH_CHECK(sample.dtype() == torch::kInt32 && sample.is_contiguous());
-  auto out = torch::zeros({(int64_t)B, FC, n_bins, (int64_t)C}, Xb.options().dtype(torch::kFloat32));
   const size_t lds = (size_t)FC * n_bins * C * 4 + (size_t)FC * 4;  // hist + fsel
   TORCH_CHECK(lds <= 160 * 1024, "feature chunk too large for LDS");
   int split = std::max(1, (int)(1024 / std::max(1, B)));
+  // split==1 writes every cell exactly once -> skip the zero-fill kernel
+  auto out = split > 1
+      ? torch::zeros({(int64_t)B, FC, n_bins, (int64_t)C}, Xb.options().dtype(torch::kFloat32))
+      : torch::empty({(int64_t)B, FC, n_bins, (int64_t)C}, Xb.options().dtype(torch::kFloat32));
   hipLaunchKernelGGL(rf_histogram_kernel, dim3((unsigned)(B * split)), dim3(256), lds,
                      cur_stream(), Xb.data_ptr<uint8_t>(), perm.data_ptr<int64_t>(),
                      seg_off.data_ptr<int64_t>(),

@@ -868,8 +868,15 @@ class _RandomForestEstimator(_RandomForestParams, Estimator):
                 # sort each tree's draw: a bootstrap multiset is unordered,
                 # and ascending physical rows turn the histogram's random
                 # per-row byte gathers into near-sequential column reads
-                # (random gathers fetched a 64B line per byte)
-                sample = sample.view(my_trees, n_draw).sort(dim=1).values.reshape(-1)
+                # (random gathers fetched a 64B line per byte). ONE flat
+                # radix sort over tree-offset keys — the segmented
+                # [T, n] sort.values cost 390 ms at 10x1M
+                t_of = torch.arange(
+                    my_trees, device=ctx.device
+                ).repeat_interleave(n_draw)
+                keys = t_of * n + sample.to(torch.int64)
+                keys, _ = torch.sort(keys)
+                sample = (keys % n).to(torch.int32)
             else:
                 sample = None
             trees = _grow_forest(
