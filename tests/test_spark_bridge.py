@@ -300,3 +300,30 @@ def test_spark_to_local_shards_across_ranks():
     # both ranks hold NON-overlapping, non-empty shards covering all rows
     assert all(r[1] > 0 for r in results)
     np.testing.assert_allclose(sum(r[0] for r in results), X.sum(axis=0), atol=1e-9)
+
+
+def test_collect_as_arrow_fallback(stub_pyspark):
+    """Spark 3.x has no toArrow(); the bridge must use _collect_as_arrow."""
+    import pyarrow as pa
+
+    from spark_rapids_ml_amd.spark.bridge import spark_to_local
+
+    table, X = _blob_table()
+
+    class Spark3DF(stub_pyspark):
+        def __init__(self, t):
+            super().__init__(t)
+            self.used_collect_as_arrow = False
+
+        toArrow = property()  # hasattr(..., "toArrow") -> False via raising
+
+        def _collect_as_arrow(self):
+            self.used_collect_as_arrow = True
+            return self._table.to_batches()
+
+    # property() without fget raises AttributeError on access -> hasattr False
+    sdf = Spark3DF(table)
+    assert not hasattr(sdf, "toArrow")
+    local = spark_to_local(sdf, shard=False)
+    assert sdf.used_collect_as_arrow
+    np.testing.assert_allclose(np.asarray(local["features"]), X)
