@@ -293,3 +293,45 @@ def test_overwrite_semantics(tmp_path):
         m.saveAsSparkModel(p)
     m.saveAsSparkModel(p, overwrite=True)
     assert spark_model_class(p).endswith("KMeansModel")
+
+
+def test_load_sparse_vector_model_written_by_stock_spark(tmp_path):
+    """Stock Spark may persist SPARSE vectors/matrices (type=0). Hand-craft
+    such a directory (as Spark would write it) and load it."""
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+
+    from spark_rapids_ml_amd.spark.persist import (
+        LINREG_CLS,
+        _vector_struct_type,
+        _write_metadata,
+    )
+
+    p = str(tmp_path / "sparse_lr")
+    os.makedirs(os.path.join(p, "data"))
+    _write_metadata(p, LINREG_CLS, "LinearRegression_stock", {}, {})
+
+    # sparse coefficients: size=6, indices [1, 4], values [2.0, -3.0]
+    sparse_vec = pa.StructArray.from_arrays(
+        [
+            pa.array([0], type=pa.int8()),
+            pa.array([6], type=pa.int32()),
+            pa.array([[1, 4]], type=pa.list_(pa.int32())),
+            pa.array([[2.0, -3.0]], type=pa.list_(pa.float64())),
+        ],
+        fields=list(_vector_struct_type()),
+    )
+    table = pa.table(
+        {
+            "intercept": pa.array([0.5], type=pa.float64()),
+            "coefficients": sparse_vec,
+            "scale": pa.array([1.0], type=pa.float64()),
+        }
+    )
+    pq.write_table(table, os.path.join(p, "data", "part-00000.snappy.parquet"))
+
+    m = load_spark_model(p)
+    np.testing.assert_allclose(
+        np.asarray(m.coefficients), [0.0, 2.0, 0.0, 0.0, -3.0, 0.0]
+    )
+    assert np.isclose(m.intercept, 0.5)
