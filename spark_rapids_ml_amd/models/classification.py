@@ -293,8 +293,15 @@ class LogisticRegression(_LogisticRegressionParams, Estimator):
             Xs = torch.sparse_csr_tensor(
                 Xt.crow_indices(), Xt.col_indices(), vals, size=Xt.shape
             )
-            # pre-transpose once for the per-iteration XᵀR SpMM
-            XsT = Xs.t().to_sparse_csr()
+            from ..ops.dispatch import has_hip_ops, use_hip as _use_hip
+
+            if _use_hip(vals) and has_hip_ops() and C_out <= 32:
+                # own csr_grad kernel scatters by column — no transposed
+                # CSR needed (the 400M-pair transpose sort cost ~0.5 s)
+                XsT = None
+            else:
+                # pre-transpose once for the per-iteration XᵀR SpMM
+                XsT = Xs.t().to_sparse_csr()
         else:
             Xs = Xt / sig_safe[None, :].to(Xt.dtype)
 
@@ -392,10 +399,25 @@ class LogisticRegression(_LogisticRegressionParams, Estimator):
         buf = torch.zeros((2, Xt.shape[1]), dtype=torch.float64, device=Xt.device)
         if Xt.shape[0] > 0:
             if Xt.layout == torch.sparse_csr:
-                vals = Xt.values().to(torch.float64)
-                cols = Xt.col_indices()
-                buf[0].index_add_(0, cols, vals)
-                buf[1].index_add_(0, cols, vals * vals)
+                from ..ops.dispatch import has_hip_ops, hip_ops, use_hip as _uh
+
+                vals32 = Xt.values()
+                if (
+                    _uh(vals32)
+                    and has_hip_ops()
+                    and vals32.dtype == torch.float32
+                    and Xt.shape[1] * 16 <= 160 * 1024
+                ):
+                    # one-pass LDS-privatized column moments (torch's f64
+                    # index_add over 400M nnz was the sparse-fit wall)
+                    buf += hip_ops().csr_col_moments(
+                        Xt.col_indices(), vals32, Xt.shape[1]
+                    )
+                else:
+                    vals = vals32.to(torch.float64)
+                    cols = Xt.col_indices()
+                    buf[0].index_add_(0, cols, vals)
+                    buf[1].index_add_(0, cols, vals * vals)
             else:
                 # chunked, f32 partial sums accumulated in f64 (the f64-upcast
                 # reduce kernel is ~40x slower; per-chunk f32 sums match the

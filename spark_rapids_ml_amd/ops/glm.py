@@ -37,7 +37,17 @@ def logistic_grad_loss(
     d = X.shape[1]
     C = W.shape[0]
     coef = W[:, :d]
-    if sparse:
+    hip_csr = sparse and use_hip(W) and C <= 32 and W.dtype == torch.float32
+    if hip_csr:
+        # own CSR kernels: thread-per-row forward, column-scatter gradient
+        # (no pre-transposed CSR needed; rocSPARSE csrmm + the 400M-pair
+        # transpose sort were the sparse-logreg wall)
+        ext = hip_ops()
+        scores = ext.csr_fwd(
+            X.crow_indices(), X.col_indices(), X.values(),
+            coef.T.contiguous(),
+        )
+    elif sparse:
         scores = torch.sparse.mm(X, coef.T.contiguous())
     else:
         scores = X @ coef.T
@@ -53,7 +63,12 @@ def logistic_grad_loss(
     else:
         resid, loss = torch_ref.softmax_residual(scores, y_idx)
 
-    if sparse:
+    if hip_csr:
+        grad_coef = hip_ops().csr_grad(
+            X.crow_indices(), X.col_indices(), X.values(),
+            resid.to(torch.float32).contiguous(), d,
+        )
+    elif sparse:
         assert XT is not None, "sparse path requires pre-transposed CSR"
         grad_coef = torch.sparse.mm(XT, resid).T
     else:
