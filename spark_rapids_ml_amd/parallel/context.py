@@ -163,10 +163,16 @@ def init_comm(backend: Optional[str] = None, timeout_s: int = 300) -> Comm:
 
     Reads torchrun env (RANK/WORLD_SIZE/LOCAL_RANK/MASTER_ADDR/MASTER_PORT).
     Without env: single-process mode, using cuda:0 when a GPU is present.
+    SRML_COMM_TIMEOUT_S bounds how long a collective blocks on a dead peer
+    (the failure-detection analog of the reference's barrier-stage
+    all-or-nothing semantics + NCCL abort, cuml_context.py:162-167).
     """
     global _COMM
     if _COMM is not None:
         return _COMM
+    env_to = os.environ.get("SRML_COMM_TIMEOUT_S")
+    if env_to:
+        timeout_s = int(env_to)
 
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))

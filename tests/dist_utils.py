@@ -58,10 +58,29 @@ def run_distributed(fn: Callable, world_size: int = 2, args: tuple = ()) -> List
         )
         p.start()
         procs.append(p)
+    import queue as _queue
+    import time as _time
+
     results: dict = {}
     errors = []
-    for _ in range(world_size):
-        rank, status, payload = q.get(timeout=300)
+    got = 0
+    deadline = _time.monotonic() + 300
+    while got < world_size:
+        try:
+            rank, status, payload = q.get(timeout=10)
+        except _queue.Empty:
+            # a rank that died without reporting (crash/_exit) would block
+            # the queue forever — detect it and fail instead of hanging
+            dead = [r for r, p in enumerate(procs) if not p.is_alive()]
+            missing = [r for r in dead if r not in results]
+            if missing and got >= world_size - len(missing):
+                errors.append(f"rank(s) {missing} exited without reporting")
+                break
+            if _time.monotonic() > deadline:
+                errors.append("timed out waiting for rank results")
+                break
+            continue
+        got += 1
         if status == "ok":
             results[rank] = pickle.loads(payload)
         else:

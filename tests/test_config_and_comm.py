@@ -82,3 +82,37 @@ def test_torch_ref_logistic_oracle_consistency():
     g2, l2 = logistic_grad_loss(X, y, W, fit_intercept=True)
     assert np.allclose(g1.numpy(), g2.numpy(), atol=1e-10)
     assert np.isclose(float(l1), float(l2), atol=1e-10)
+
+
+def _dist_dead_rank(_):
+    """Rank 1 dies before the collective; rank 0 must fail FAST (bounded by
+    SRML_COMM_TIMEOUT_S), not hang — the reference's all-or-nothing barrier
+    + NCCL-abort semantics (cuml_context.py:162-167)."""
+    import os
+    import time
+
+    import torch
+
+    from spark_rapids_ml_amd.parallel.context import get_comm
+
+    comm = get_comm()
+    if comm.rank == 1:
+        os._exit(0)  # simulate a dead worker (no clean shutdown)
+    t0 = time.perf_counter()
+    try:
+        comm.allreduce(torch.ones(4))
+        return ("no-error", time.perf_counter() - t0)
+    except Exception as e:
+        return ("error", time.perf_counter() - t0, type(e).__name__)
+
+
+def test_dead_rank_fails_fast(monkeypatch):
+    import pytest as _pytest
+
+    from .dist_utils import run_distributed
+
+    monkeypatch.setenv("SRML_COMM_TIMEOUT_S", "8")
+    with _pytest.raises(RuntimeError):
+        # rank 1 exits without reporting; the harness reports the missing
+        # rank (and rank 0's collective error) as a failure
+        run_distributed(_dist_dead_rank, world_size=2, args=(None,))
