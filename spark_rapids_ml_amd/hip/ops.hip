@@ -1564,7 +1564,9 @@ __global__ __launch_bounds__(256) void rf_histogram_kernel(
   // dependent L2/HBM gather latencies overlap (the naive one-at-a-time loop
   // ran ~12 cycles/update, far off the LDS-atomic bound)
   extern __shared__ __attribute__((aligned(16))) float lhist[];
-  int* fsel_s = reinterpret_cast<int*>(lhist + FC * n_bins * C);
+  // per-feature COLUMN BASE OFFSETS (f * n_phys precomputed once — the
+  // per-element int64 multiply cost ~3 issue slots in the hot loop)
+  int64_t* foff_s = reinterpret_cast<int64_t*>(lhist + FC * n_bins * C);
   const int b = blockIdx.x / split;
   const int slice = blockIdx.x % split;
   const int64_t s0 = seg_off[b], e0 = seg_off[b + 1];
@@ -1585,8 +1587,10 @@ __global__ __launch_bounds__(256) void rf_histogram_kernel(
   const int64_t re = min(e0, rs + chunk);
   if (rs >= re) return;
   for (int e = tid; e < nfc; e += 256) lhist[e] = 0.0f;
-  for (int e = tid; e < FC; e += 256)
-    fsel_s[e] = feat_sel ? feat_sel[(int64_t)b * mf + f0 + e] : (f0 + e);
+  for (int e = tid; e < FC; e += 256) {
+    const int f = feat_sel ? feat_sel[(int64_t)b * mf + f0 + e] : (f0 + e);
+    foff_s[e] = (int64_t)f * n_phys;
+  }
   __syncthreads();
 
   const bool classif = (y_cls != nullptr);
@@ -1601,7 +1605,7 @@ __global__ __launch_bounds__(256) void rf_histogram_kernel(
       int bins[4];
 #pragma unroll
       for (int e = 0; e < 4; ++e)
-        bins[e] = Xcm[(int64_t)fsel_s[q + e] * n_phys + row];
+        bins[e] = Xcm[foff_s[q + e] + row];
 #pragma unroll
       for (int e = 0; e < 4; ++e) {
         if (classif) {
@@ -1614,7 +1618,7 @@ __global__ __launch_bounds__(256) void rf_histogram_kernel(
       }
     }
     for (; q < FC; ++q) {
-      const int bin = Xcm[(int64_t)fsel_s[q] * n_phys + row];
+      const int bin = Xcm[foff_s[q] + row];
       if (classif) {
         atomicAdd(&lhist[(q * n_bins + bin) * C + yc], 1.0f);
       } else {
@@ -2377,7 +2381,7 @@ torch::Tensor rf_histogram(torch::Tensor Xb, torch::Tensor perm, torch::Tensor s
   const int mf = feat_sel.numel() > 0 ? (int)feat_sel.size(1) : 0;
   const bool has_sample = sample.numel() > 0;
   if (has_sample) TORCH_CHECK(sample.dtype() == torch::kInt32 && sample.is_contiguous());
-  const size_t lds = (size_t)FC * n_bins * C * 4 + (size_t)FC * 4;  // hist + fsel
+  const size_t lds = (size_t)FC * n_bins * C * 4 + (size_t)FC * 8;  // hist + col offsets
   TORCH_CHECK(lds <= 160 * 1024, "feature chunk too large for LDS");
   int split = std::max(1, (int)(1024 / std::max(1, B)));
   // split==1 writes every cell exactly once -> skip the zero-fill kernel
