@@ -1565,8 +1565,10 @@ __global__ __launch_bounds__(256) void rf_histogram_kernel(
   // ran ~12 cycles/update, far off the LDS-atomic bound)
   extern __shared__ __attribute__((aligned(16))) float lhist[];
   // per-feature COLUMN BASE OFFSETS (f * n_phys precomputed once — the
-  // per-element int64 multiply cost ~3 issue slots in the hot loop)
-  int64_t* foff_s = reinterpret_cast<int64_t*>(lhist + FC * n_bins * C);
+  // per-element int64 multiply cost ~3 issue slots in the hot loop);
+  // region rounded up to 8B alignment (odd nfc with odd bins*classes)
+  int64_t* foff_s =
+      reinterpret_cast<int64_t*>(lhist + ((FC * n_bins * C + 1) & ~1));
   const int b = blockIdx.x / split;
   const int slice = blockIdx.x % split;
   const int64_t s0 = seg_off[b], e0 = seg_off[b + 1];
@@ -2381,7 +2383,7 @@ torch::Tensor rf_histogram(torch::Tensor Xb, torch::Tensor perm, torch::Tensor s
   const int mf = feat_sel.numel() > 0 ? (int)feat_sel.size(1) : 0;
   const bool has_sample = sample.numel() > 0;
   if (has_sample) TORCH_CHECK(sample.dtype() == torch::kInt32 && sample.is_contiguous());
-  const size_t lds = (size_t)FC * n_bins * C * 4 + (size_t)FC * 8;  // hist + col offsets
+  const size_t lds = ((size_t)FC * n_bins * C + 1 & ~1ull) * 4 + (size_t)FC * 8;  // hist + col offsets
   TORCH_CHECK(lds <= 160 * 1024, "feature chunk too large for LDS");
   int split = std::max(1, (int)(1024 / std::max(1, B)));
   // split==1 writes every cell exactly once -> skip the zero-fill kernel
