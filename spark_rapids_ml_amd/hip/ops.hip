@@ -1603,13 +1603,13 @@ __global__ __launch_bounds__(256) void rf_histogram_kernel(
     const int yc = classif ? y_cls[row] : 0;
     const float yv = classif ? 1.0f : y_reg[row];
     int q = 0;
-    for (; q + 3 < FC; q += 4) {
-      int bins[4];
+    for (; q + 7 < FC; q += 8) {
+      int bins[8];
 #pragma unroll
-      for (int e = 0; e < 4; ++e)
+      for (int e = 0; e < 8; ++e)
         bins[e] = Xcm[foff_s[q + e] + row];
 #pragma unroll
-      for (int e = 0; e < 4; ++e) {
+      for (int e = 0; e < 8; ++e) {
         if (classif) {
           atomicAdd(&lhist[((q + e) * n_bins + bins[e]) * C + yc], 1.0f);
         } else {
