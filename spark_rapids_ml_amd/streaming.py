@@ -59,18 +59,27 @@ def iter_device_chunks(
     itemsize = X.dtype.itemsize
     rows = max(1, int((cap_bytes // 4) // max(1, d * itemsize)))
     rows = min(rows, n)
-    pin: Optional[torch.Tensor] = None
     use_pin = device.type == "cuda" and torch.cuda.is_available()
     if use_pin:
         tdtype = torch.from_numpy(np.empty(0, dtype=X.dtype)).dtype
-        pin = torch.empty((rows, d), dtype=tdtype, pin_memory=True)
-    for s in range(0, n, rows):
+        # TWO pinned buffers alternated, each guarded by an event recorded
+        # after its async H2D: the host must not rewrite a pinned buffer the
+        # DMA may still be reading
+        pins = [torch.empty((rows, d), dtype=tdtype, pin_memory=True) for _ in range(2)]
+        evs = [torch.cuda.Event() for _ in range(2)]
+        recorded = [False, False]
+    for it, s in enumerate(range(0, n, rows)):
         e = min(n, s + rows)
         chunk = np.ascontiguousarray(X[s:e])
-        if use_pin and pin is not None:
-            pv = pin[: e - s]
+        if use_pin:
+            b = it & 1
+            if recorded[b]:
+                evs[b].synchronize()
+            pv = pins[b][: e - s]
             pv.copy_(torch.from_numpy(chunk))
             t = pv.to(device, non_blocking=True)
+            evs[b].record()
+            recorded[b] = True
         else:
             t = torch.from_numpy(chunk).to(device)
         if dtype is not None:
