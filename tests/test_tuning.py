@@ -201,3 +201,38 @@ def test_transform_evaluate_single_pass_counts_extractions(monkeypatch):
     assert calls["n"] == 1, f"expected 1 feature extraction, saw {calls['n']}"
     # heavier regularization should not beat the unregularized fit here
     assert ms[0] >= ms[2] - 1e-9
+
+
+def test_fit_multiple_grids_vary_for_all_estimators():
+    """Grid values must actually reach every estimator's fit (the round-1
+    advisor found logreg ignoring its grid; pin the whole family)."""
+    from spark_rapids_ml_amd import KMeans, RandomForestClassifier
+
+    rng = np.random.default_rng(0)
+    X = rng.normal(size=(400, 8)).astype(np.float32)
+    df = DataFrame.from_numpy(X)
+    km = KMeans(maxIter=5, seed=1)
+    models = [m for _, m in km.fitMultiple(df, [{km.k: 2}, {km.k: 5}])]
+    assert models[0].cluster_centers_.shape[0] == 2
+    assert models[1].cluster_centers_.shape[0] == 5
+
+    y = (X[:, 0] > 0).astype(np.float64)
+    dfc = DataFrame.from_numpy(X, y)
+    rf = RandomForestClassifier(maxDepth=3, seed=1)
+    models = [m for _, m in rf.fitMultiple(dfc, [{rf.numTrees: 3}, {rf.numTrees: 7}])]
+    assert models[0].numTrees == 3
+    assert models[1].numTrees == 7
+
+    lr_X, lr_y = X.astype(np.float64), X[:, 0] * 2.0 + 1.0
+    from spark_rapids_ml_amd import LinearRegression
+
+    lr = LinearRegression(maxIter=20)
+    models = [
+        m for _, m in lr.fitMultiple(
+            DataFrame.from_numpy(lr_X, lr_y),
+            [{lr.regParam: 0.0}, {lr.regParam: 100.0}],
+        )
+    ]
+    c0 = np.linalg.norm(np.asarray(models[0].coefficients))
+    c1 = np.linalg.norm(np.asarray(models[1].coefficients))
+    assert c1 < c0  # heavy ridge shrinks
