@@ -1558,6 +1558,9 @@ __global__ __launch_bounds__(256) void rf_histogram_kernel(
     const int32_t* __restrict__ sample,   // [vn] virtual->physical, or nullptr
     int64_t n_phys,                       // modulo map when sample==nullptr
     int d, int mf, int f0, int FC, int n_bins, int C, int split,
+    float y_inv_scale,                    // regression: 1/max|y| for the
+                                          // packed-u64 path; 0 = disabled
+    float y_scale,                        //             max|y|
     float* __restrict__ out) {            // [B, FC, n_bins, C]
   // lhist [FC][nb][C] + an LDS copy of the block's feature-id chunk; the
   // hot loop gathers FOUR feature bytes per row before touching LDS so the
@@ -1602,6 +1605,16 @@ __global__ __launch_bounds__(256) void rf_histogram_kernel(
                                : (vrow >= n_phys ? vrow % n_phys : vrow);
     const int yc = classif ? y_cls[row] : 0;
     const float yv = classif ? 1.0f : y_reg[row];
+    // regression fast path: ONE u64 LDS atomic per (row, feature) — count
+    // in bits 41+, biased Q14 fixed-point sum of y/max|y| below (each
+    // contribution positive, so carries never cross the field boundary;
+    // count <= 2^23 rows per node enforced host-side)
+    const bool packed = !classif && y_inv_scale > 0.0f;
+    unsigned long long yq = 0;
+    if (packed) {
+      const long long fix = llroundf(yv * y_inv_scale * 16384.0f) + (1ll << 15);
+      yq = (1ull << 41) + (unsigned long long)fix;
+    }
     int q = 0;
     for (; q + 7 < FC; q += 8) {
       int bins[8];
@@ -1612,6 +1625,10 @@ __global__ __launch_bounds__(256) void rf_histogram_kernel(
       for (int e = 0; e < 8; ++e) {
         if (classif) {
           atomicAdd(&lhist[((q + e) * n_bins + bins[e]) * C + yc], 1.0f);
+        } else if (packed) {
+          atomicAdd(reinterpret_cast<unsigned long long*>(lhist) +
+                        (q + e) * n_bins + bins[e],
+                    yq);
         } else {
           float* cell = &lhist[((q + e) * n_bins + bins[e]) * 2];
           atomicAdd(cell, 1.0f);
@@ -1623,6 +1640,9 @@ __global__ __launch_bounds__(256) void rf_histogram_kernel(
       const int bin = Xcm[foff_s[q] + row];
       if (classif) {
         atomicAdd(&lhist[(q * n_bins + bin) * C + yc], 1.0f);
+      } else if (packed) {
+        atomicAdd(reinterpret_cast<unsigned long long*>(lhist) + q * n_bins + bin,
+                  yq);
       } else {
         float* cell = &lhist[(q * n_bins + bin) * 2];
         atomicAdd(cell, 1.0f);
@@ -1633,11 +1653,34 @@ __global__ __launch_bounds__(256) void rf_histogram_kernel(
   __syncthreads();
 
   float* dst = out + (int64_t)b * FC * n_bins * C;
-  if (split > 1) {
-    for (int e = tid; e < nfc; e += 256)
-      if (lhist[e] != 0.0f) atomicAdd(&dst[e], lhist[e]);
+  const bool packed_flush = (y_cls == nullptr) && y_inv_scale > 0.0f;
+  if (!packed_flush) {
+    if (split > 1) {
+      for (int e = tid; e < nfc; e += 256)
+        if (lhist[e] != 0.0f) atomicAdd(&dst[e], lhist[e]);
+    } else {
+      for (int e = tid; e < nfc; e += 256) dst[e] = lhist[e];
+    }
   } else {
-    for (int e = tid; e < nfc; e += 256) dst[e] = lhist[e];
+    // decode packed (count, biased Q14 sum) -> the [.., 2] float layout
+    const unsigned long long* ph =
+        reinterpret_cast<const unsigned long long*>(lhist);
+    for (int e2 = tid; e2 < FC * n_bins; e2 += 256) {
+      const unsigned long long v = ph[e2];
+      const long long cnt = (long long)(v >> 41);
+      const long long sq =
+          (long long)(v & ((1ull << 41) - 1)) - cnt * (1ll << 15);
+      const float sum = (float)sq * (y_scale / 16384.0f);
+      if (split > 1) {
+        if (cnt) {
+          atomicAdd(&dst[e2 * 2], (float)cnt);
+          atomicAdd(&dst[e2 * 2 + 1], sum);
+        }
+      } else {
+        dst[e2 * 2] = (float)cnt;
+        dst[e2 * 2 + 1] = sum;
+      }
+    }
   }
 }
 
@@ -2383,6 +2426,16 @@ torch::Tensor rf_histogram(torch::Tensor Xb, torch::Tensor perm, torch::Tensor s
   const int mf = feat_sel.numel() > 0 ? (int)feat_sel.size(1) : 0;
   const bool has_sample = sample.numel() > 0;
   if (has_sample) TORCH_CHECK(sample.dtype() == torch::kInt32 && sample.is_contiguous());
+  // regression packed-u64 path: needs max|y| and a per-node row-count cap
+  float y_scale = 0.0f, y_inv_scale = 0.0f;
+  if (!classif) {
+    const int64_t vn = has_sample ? sample.numel() : perm.numel();
+    if (vn < (1ll << 23)) {
+      y_scale = y.abs().max().item<float>();
+      if (y_scale <= 0.0f) y_scale = 1.0f;
+      y_inv_scale = 1.0f / y_scale;
+    }
+  }
   const size_t lds = ((size_t)FC * n_bins * C + 1 & ~1ull) * 4 + (size_t)FC * 8;  // hist + col offsets
   TORCH_CHECK(lds <= 160 * 1024, "feature chunk too large for LDS");
   int split = std::max(1, (int)(1024 / std::max(1, B)));
@@ -2398,6 +2451,7 @@ torch::Tensor rf_histogram(torch::Tensor Xb, torch::Tensor perm, torch::Tensor s
                      classif ? nullptr : y.data_ptr<float>(),
                      has_sample ? sample.data_ptr<int32_t>() : nullptr, n_phys,
                      d, mf, (int)f0, (int)FC, (int)n_bins, C, split,
+                     y_inv_scale, y_scale,
                      out.data_ptr<float>());
   return out;
 }
