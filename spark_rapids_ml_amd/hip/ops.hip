@@ -1685,6 +1685,113 @@ __global__ __launch_bounds__(256) void rf_histogram_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// rf_histogram_fw: feature-wide variant — LANES run over (sorted) sampled
+// features of ONE row, the loop runs over rows. The row-lane kernel above
+// issued one address-divergent gather per (row, feature): each wave-load
+// touched up to 64 distinct cache lines and the LSU serialized them
+// (~28 cyc/update, invariant under unrolls/LDS/atomic-count experiments).
+// Here a wave-load reads 32 SORTED features of one row — a short span of
+// the row-major row (avg gap d/mf) — so the per-update issue cost drops by
+// the lane width. Two rows per wave (lane groups of 32). Regression uses
+// the packed single-u64-atomic cells.
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(256) void rf_histogram_fw_kernel(
+    const uint8_t* __restrict__ Xrm,  // [n_phys, d] ROW-major
+    const int64_t* __restrict__ perm,
+    const int64_t* __restrict__ seg_off,  // [B+1]
+    const int32_t* __restrict__ feat_sel, // [B, mf] SORTED per node, or null
+    const int32_t* __restrict__ y_cls,
+    const float* __restrict__ y_reg,
+    const int32_t* __restrict__ sample,
+    int64_t n_phys,
+    int d, int mf, int f0, int FC, int n_bins, int C, int split,
+    float y_inv_scale, float y_scale,
+    float* __restrict__ out) {            // [B, FC, n_bins, C]
+  extern __shared__ __attribute__((aligned(16))) float lhist[];
+  int* fsel_s = reinterpret_cast<int*>(
+      lhist + (((int64_t)FC * n_bins * C + 1) & ~1ll));
+  const int b = blockIdx.x / split;
+  const int slice = blockIdx.x % split;
+  const int64_t s0 = seg_off[b], e0 = seg_off[b + 1];
+  const int64_t len = e0 - s0;
+  const int tid = threadIdx.x;
+  const int nfc = FC * n_bins * C;
+  const bool classif = (y_cls != nullptr);
+  const bool packed = !classif && y_inv_scale > 0.0f;
+  if (len == 0) {
+    if (split == 1) {
+      float* dst0 = out + (int64_t)b * nfc;
+      for (int e = tid; e < nfc; e += 256) dst0[e] = 0.0f;
+    }
+    return;
+  }
+  const int64_t chunk = (len + split - 1) / split;
+  const int64_t rs = s0 + slice * chunk;
+  const int64_t re = min(e0, rs + chunk);
+  if (rs >= re) return;
+
+  for (int e = tid; e < nfc; e += 256) lhist[e] = 0.0f;
+  for (int e = tid; e < FC; e += 256)
+    fsel_s[e] = feat_sel ? feat_sel[(int64_t)b * mf + f0 + e] : (f0 + e);
+  __syncthreads();
+
+  // 8 rows per block iteration: 4 waves x 2 lane-groups of 32
+  const int group = tid >> 5;       // 0..7
+  const int fq = tid & 31;          // feature slot within the chunk
+  const int f = (fq < FC) ? fsel_s[fq] : -1;
+  for (int64_t r0 = rs + group; r0 < re; r0 += 8) {
+    const int64_t vrow = perm[r0];
+    const int64_t row = sample ? (int64_t)sample[vrow]
+                               : (vrow >= n_phys ? vrow % n_phys : vrow);
+    if (f < 0) continue;
+    const int bin = Xrm[row * d + f];
+    if (classif) {
+      atomicAdd(&lhist[(fq * n_bins + bin) * C + y_cls[row]], 1.0f);
+    } else if (packed) {
+      const long long fix =
+          llroundf(y_reg[row] * y_inv_scale * 16384.0f) + (1ll << 15);
+      atomicAdd(reinterpret_cast<unsigned long long*>(lhist) + fq * n_bins + bin,
+                (1ull << 41) + (unsigned long long)fix);
+    } else {
+      float* cell = &lhist[(fq * n_bins + bin) * 2];
+      atomicAdd(cell, 1.0f);
+      atomicAdd(cell + 1, y_reg[row]);
+    }
+  }
+  __syncthreads();
+
+  float* dst = out + (int64_t)b * nfc;
+  if (!packed) {
+    if (split > 1) {
+      for (int e = tid; e < nfc; e += 256)
+        if (lhist[e] != 0.0f) atomicAdd(&dst[e], lhist[e]);
+    } else {
+      for (int e = tid; e < nfc; e += 256) dst[e] = lhist[e];
+    }
+  } else {
+    const unsigned long long* ph =
+        reinterpret_cast<const unsigned long long*>(lhist);
+    for (int e2 = tid; e2 < FC * n_bins; e2 += 256) {
+      const unsigned long long v = ph[e2];
+      const long long cnt = (long long)(v >> 41);
+      const long long sq =
+          (long long)(v & ((1ull << 41) - 1)) - cnt * (1ll << 15);
+      const float sum = (float)sq * (y_scale / 16384.0f);
+      if (split > 1) {
+        if (cnt) {
+          atomicAdd(&dst[e2 * 2], (float)cnt);
+          atomicAdd(&dst[e2 * 2 + 1], sum);
+        }
+      } else {
+        dst[e2 * 2] = (float)cnt;
+        dst[e2 * 2 + 1] = sum;
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // rf_best_split: fused split-finding scan over rf_histogram output.
 // One block per node: threads own features, scan bins accumulating left
 // stats in registers, compute the gain (gini for classification; the
@@ -2456,6 +2563,49 @@ torch::Tensor rf_histogram(torch::Tensor Xb, torch::Tensor perm, torch::Tensor s
   return out;
 }
 
+torch::Tensor rf_histogram_fw(torch::Tensor Xrm, torch::Tensor perm, torch::Tensor seg_off,
+                              torch::Tensor feat_sel, torch::Tensor y, int64_t f0,
+                              int64_t FC, int64_t n_bins, int64_t n_classes,
+                              torch::Tensor sample) {
+  // Xrm is ROW-major [n_phys, d]
+  TORCH_CHECK(Xrm.is_cuda() && Xrm.dtype() == torch::kUInt8 && Xrm.is_contiguous());
+  TORCH_CHECK(perm.dtype() == torch::kInt64 && seg_off.dtype() == torch::kInt64);
+  TORCH_CHECK(FC <= 32, "rf_histogram_fw: FC <= 32 (one lane group)");
+  const int d = (int)Xrm.size(1);
+  const int64_t n_phys = Xrm.size(0);
+  const int B = (int)seg_off.size(0) - 1;
+  const bool classif = n_classes > 0;
+  const int C = classif ? (int)n_classes : 2;
+  const int mf = feat_sel.numel() > 0 ? (int)feat_sel.size(1) : 0;
+  const bool has_sample = sample.numel() > 0;
+  float y_scale = 0.0f, y_inv_scale = 0.0f;
+  if (!classif) {
+    const int64_t vn = has_sample ? sample.numel() : perm.numel();
+    if (vn < (1ll << 23)) {
+      y_scale = y.abs().max().item<float>();
+      if (y_scale <= 0.0f) y_scale = 1.0f;
+      y_inv_scale = 1.0f / y_scale;
+    }
+  }
+  const size_t lds = (((size_t)FC * n_bins * C + 1) & ~1ull) * 4 + (size_t)FC * 4;
+  TORCH_CHECK(lds <= 160 * 1024, "feature chunk too large for LDS");
+  int split = std::max(1, (int)(1024 / std::max(1, B)));
+  auto out = split > 1
+      ? torch::zeros({(int64_t)B, FC, n_bins, (int64_t)C}, Xrm.options().dtype(torch::kFloat32))
+      : torch::empty({(int64_t)B, FC, n_bins, (int64_t)C}, Xrm.options().dtype(torch::kFloat32));
+  hipLaunchKernelGGL(rf_histogram_fw_kernel, dim3((unsigned)(B * split)), dim3(256), lds,
+                     cur_stream(), Xrm.data_ptr<uint8_t>(), perm.data_ptr<int64_t>(),
+                     seg_off.data_ptr<int64_t>(),
+                     mf > 0 ? feat_sel.data_ptr<int32_t>() : nullptr,
+                     classif ? y.data_ptr<int32_t>() : nullptr,
+                     classif ? nullptr : y.data_ptr<float>(),
+                     has_sample ? sample.data_ptr<int32_t>() : nullptr, n_phys,
+                     d, mf, (int)f0, (int)FC, (int)n_bins, C, split,
+                     y_inv_scale, y_scale,
+                     out.data_ptr<float>());
+  return out;
+}
+
 std::vector<torch::Tensor> rf_best_split(torch::Tensor H, int64_t min_leaf,
                                           bool classif) {
   TORCH_CHECK(H.is_cuda() && H.dtype() == torch::kFloat32 && H.is_contiguous());
@@ -2659,6 +2809,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("softmax_residual_loss", &softmax_residual_loss, "fused softmax residual + loss");
   m.def("knn_select", &knn_select, "fused MFMA distance + in-LDS top-k");
   m.def("rf_histogram", &rf_histogram, "LDS-privatized RF split histograms");
+  m.def("rf_histogram_fw", &rf_histogram_fw, "feature-wide RF histograms (lanes over sorted features)");
   m.def("rf_best_split", &rf_best_split, "fused RF gain scan + block-best reduce");
   m.def("dbscan_sweep", &dbscan_sweep, "fused eps-neighborhood count / min-core-label pass");
   m.def("umap_sgd", &umap_sgd, "edge-sampled UMAP SGD (all epochs, Hogwild)");

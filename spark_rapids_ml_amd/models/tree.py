@@ -434,10 +434,13 @@ def _grow_forest(
         ext = hip_ops()
         y32 = y_idx.to(torch.int32) if task == "classification" else yf
         C_ch = n_classes if task == "classification" else 2
-        # ~30 KB LDS per block -> ~5 blocks/CU of latency-hiding waves for
-        # the gather+LDS-atomic hot loop (privatized copies measured SLOWER:
-        # the bound is gather latency, not same-bin contention)
-        fc_kernel = max(1, min(512, (30 * 1024 // 4) // max(1, n_bins * C_ch)))
+        # feature-wide kernel: 32 features per chunk (one lane group);
+        # SRML_RF_HIST=old selects the row-lane kernel for A/B
+        _hist_old = _os.environ.get("SRML_RF_HIST") == "old"
+        if _hist_old:
+            fc_kernel = max(1, min(512, (30 * 1024 // 4) // max(1, n_bins * C_ch)))
+        else:
+            fc_kernel = max(1, min(32, (150 * 1024 // 4) // max(1, n_bins * C_ch)))
         # column-major binned matrix for the histogram/reroute kernels: a
         # node segment's row gathers stay inside dense per-feature cache
         # lines (row-major fetched a ~47-line row to read ~54 sampled bytes)
@@ -503,6 +506,11 @@ def _grow_forest(
                     feat_sel = scores.topk(max_features, dim=1).indices
                 else:
                     feat_sel = scores.argsort(dim=1)[:, :max_features]
+                if hip_hist and not _hist_old:
+                    # feature-wide kernel wants SORTED per-node subsets: a
+                    # wave-load then spans consecutive row bytes (avg gap
+                    # d/mf) instead of the whole row
+                    feat_sel = feat_sel.sort(dim=1).values
                 _tick("feat_sel", _t0)
                 mf = max_features
                 Xb_rows = None  # sampled path reads only selected bytes
@@ -526,8 +534,10 @@ def _grow_forest(
                 f1 = min(mf, f0 + fc_step)
                 F = f1 - f0
                 if hip_hist:
-                    H = ext.rf_histogram(
-                        Xcm, perm, seg_off, fsel32, y32, f0, F, n_bins,
+                    _hist_fn = ext.rf_histogram if _hist_old else ext.rf_histogram_fw
+                    _hist_x = Xcm if _hist_old else Xb
+                    H = _hist_fn(
+                        _hist_x, perm, seg_off, fsel32, y32, f0, F, n_bins,
                         n_classes if task == "classification" else 0,
                         sample_t,
                     )
