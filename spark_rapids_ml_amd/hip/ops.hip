@@ -1736,27 +1736,47 @@ __global__ __launch_bounds__(256) void rf_histogram_fw_kernel(
     fsel_s[e] = feat_sel ? feat_sel[(int64_t)b * mf + f0 + e] : (f0 + e);
   __syncthreads();
 
-  // 8 rows per block iteration: 4 waves x 2 lane-groups of 32
+  // 32 rows per block iteration: 8 lane-groups x 4 INDEPENDENT row chains
+  // per group — the perm->sample->row->gather dependency is ~3 serial
+  // memory latencies; one chain per group left waves issue-stalled (PMC:
+  // 71% SQ_WAIT_INST_ANY), four chains quadruple the loads in flight
   const int group = tid >> 5;       // 0..7
   const int fq = tid & 31;          // feature slot within the chunk
   const int f = (fq < FC) ? fsel_s[fq] : -1;
-  for (int64_t r0 = rs + group; r0 < re; r0 += 8) {
-    const int64_t vrow = perm[r0];
-    const int64_t row = sample ? (int64_t)sample[vrow]
-                               : (vrow >= n_phys ? vrow % n_phys : vrow);
-    if (f < 0) continue;
-    const int bin = Xrm[row * d + f];
-    if (classif) {
-      atomicAdd(&lhist[(fq * n_bins + bin) * C + y_cls[row]], 1.0f);
-    } else if (packed) {
-      const long long fix =
-          llroundf(y_reg[row] * y_inv_scale * 16384.0f) + (1ll << 15);
-      atomicAdd(reinterpret_cast<unsigned long long*>(lhist) + fq * n_bins + bin,
-                (1ull << 41) + (unsigned long long)fix);
-    } else {
-      float* cell = &lhist[(fq * n_bins + bin) * 2];
-      atomicAdd(cell, 1.0f);
-      atomicAdd(cell + 1, y_reg[row]);
+  for (int64_t base = rs; base < re; base += 32) {
+    int64_t rowv[4];
+    int binv[4];
+#pragma unroll
+    for (int g = 0; g < 4; ++g) {
+      const int64_t r = base + group + 8 * g;
+      if (r < re) {
+        const int64_t vrow = perm[r];
+        rowv[g] = sample ? (int64_t)sample[vrow]
+                         : (vrow >= n_phys ? vrow % n_phys : vrow);
+      } else {
+        rowv[g] = -1;
+      }
+    }
+#pragma unroll
+    for (int g = 0; g < 4; ++g)
+      binv[g] = (rowv[g] >= 0 && f >= 0) ? Xrm[rowv[g] * d + f] : -1;
+#pragma unroll
+    for (int g = 0; g < 4; ++g) {
+      if (binv[g] < 0) continue;
+      const int64_t row = rowv[g];
+      const int bin = binv[g];
+      if (classif) {
+        atomicAdd(&lhist[(fq * n_bins + bin) * C + y_cls[row]], 1.0f);
+      } else if (packed) {
+        const long long fix =
+            llroundf(y_reg[row] * y_inv_scale * 16384.0f) + (1ll << 15);
+        atomicAdd(reinterpret_cast<unsigned long long*>(lhist) + fq * n_bins + bin,
+                  (1ull << 41) + (unsigned long long)fix);
+      } else {
+        float* cell = &lhist[(fq * n_bins + bin) * 2];
+        atomicAdd(cell, 1.0f);
+        atomicAdd(cell + 1, y_reg[row]);
+      }
     }
   }
   __syncthreads();
