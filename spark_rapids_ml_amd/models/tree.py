@@ -434,9 +434,12 @@ def _grow_forest(
         ext = hip_ops()
         y32 = y_idx.to(torch.int32) if task == "classification" else yf
         C_ch = n_classes if task == "classification" else 2
-        # feature-wide kernel: 32 features per chunk (one lane group);
-        # SRML_RF_HIST=old selects the row-lane kernel for A/B
-        _hist_old = _os.environ.get("SRML_RF_HIST") == "old"
+        # Measured dispatch (A/B in profiles/README.md): the row-lane
+        # kernel wins for classification (sqrt-sampled features, sorted
+        # bootstrap = dense segments), the feature-wide kernel wins for
+        # regression (1/3-sampled features); SRML_RF_HIST=old|fw overrides
+        _mode = _os.environ.get("SRML_RF_HIST")
+        _hist_old = (_mode == "old") or (_mode != "fw" and task == "classification")
         if _hist_old:
             fc_kernel = max(1, min(512, (30 * 1024 // 4) // max(1, n_bins * C_ch)))
         else:
