@@ -487,6 +487,31 @@ class _DBSCANParams(HasFeaturesCol, HasFeaturesCols, HasPredictionCol, HasIDCol)
             max_mbytes_per_batch=None, algorithm="brute",
         )
 
+    # reference DBSCAN getters/setters (clustering.py:733+ surface)
+    def getEps(self) -> float:
+        return self.getOrDefault("eps")
+
+    def getMinSamples(self) -> int:
+        return self.getOrDefault("min_samples")
+
+    def getMetric(self) -> str:
+        return self.getOrDefault("metric")
+
+    def getAlgorithm(self) -> str:
+        return self.getOrDefault("algorithm")
+
+    def getMaxMbytesPerBatch(self):
+        return self.getOrDefault("max_mbytes_per_batch")
+
+    def setMetric(self, value: str):
+        return self._set_params(metric=value)
+
+    def setAlgorithm(self, value: str):
+        return self._set_params(algorithm=value)
+
+    def setMaxMbytesPerBatch(self, value):
+        return self._set_params(max_mbytes_per_batch=value)
+
     @classmethod
     def _param_mapping(cls) -> Dict[str, Optional[str]]:
         return {

@@ -127,6 +127,18 @@ class LinearRegression(_LinearRegressionParams, Estimator):
         super().__init__()
         self._set_params(**kwargs)
 
+    def setTol(self, value: float) -> "LinearRegression":
+        return self._set_params(tol=value)
+
+    def setSolver(self, value: str) -> "LinearRegression":
+        return self._set_params(solver=value)
+
+    def setLoss(self, value: str) -> "LinearRegression":
+        # reference supports squaredError only on GPU (regression.py:240)
+        if value != "squaredError":
+            raise ValueError("Only loss='squaredError' is supported.")
+        return self._set_params(loss=value)
+
     def setRegParam(self, value: float) -> "LinearRegression":
         return self._set_params(regParam=value)
 
