@@ -565,18 +565,19 @@ class UMAPModel(_UMAPParams, Model):
         k = int(float(self.getOrDefault("n_neighbors")))
         k = min(k, raw.shape[0])
         d, i = knn_topk(Xt, raw, k)
-        knn_d, knn_i = as_numpy(d), as_numpy(i)
-        sigma, rho = _smooth_knn(knn_d, float(self.getOrDefault("local_connectivity")))
-        w = np.exp(-np.maximum(knn_d - rho[:, None], 0.0) / sigma[:, None]).astype(np.float32)
-        w = w / np.clip(w.sum(axis=1, keepdims=True), 1e-12, None)
-        wt = torch.from_numpy(w).to(device)
+        # device-resident smooth-knn + membership weights (the numpy hop
+        # cost seconds at 1M transform rows)
+        sigma, rho = _smooth_knn_t(d, float(self.getOrDefault("local_connectivity")))
+        wt = torch.exp(
+            -torch.clamp(d.to(torch.float64) - rho[:, None], min=0.0) / sigma[:, None]
+        ).to(torch.float32)
+        wt = wt / torch.clamp(wt.sum(dim=1, keepdim=True), min=1e-12)
         new_emb = (wt[:, :, None] * emb[i]).sum(dim=1)
 
         # few refinement epochs against the training embedding
         heads = torch.arange(n_new, device=device).repeat_interleave(k)
-        tails = i.flatten() if isinstance(i, np.ndarray) else i.flatten()
-        tails = torch.from_numpy(knn_i.flatten()).to(device)
-        weights = torch.from_numpy(w.flatten()).to(device)
+        tails = i.reshape(-1).to(device)
+        weights = wt.reshape(-1)
         gen = torch.Generator(device=device)
         gen.manual_seed(int(self.getOrDefault("random_state")))
         new_emb = _optimize_embedding(
