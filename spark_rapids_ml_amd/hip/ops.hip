@@ -1938,6 +1938,53 @@ __global__ __launch_bounds__(256) void rf_best_split_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// kmeans_argmin_kn: fused epilogue over a library-GEMM dot block.
+// dots is [k, n] (C @ X^T): per row i the k dot values sit at stride n, so
+// the 256 threads of a block read coalesced spans per center. d2 =
+// c_sq[kk] - 2*dot (x_sq[i] added once at the end). One pass over the
+// [k, n] block; c_sq staged in LDS. The all-in-one MFMA assign kernel runs
+// at 84 TF; hipBLASLt runs the same dot block at 141 TF, so GEMM + this
+// bandwidth-bound pass wins (measured; SRML_KMEANS_VARIANT=fused keeps the
+// old path).
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(256) void kmeans_argmin_kn_kernel(
+    const float* __restrict__ dots,  // [k, n]
+    const float* __restrict__ x_sq,  // [n]
+    const float* __restrict__ c_sq,  // [k]
+    int64_t n, int k,
+    int32_t* __restrict__ labels,    // [n]
+    float* __restrict__ min_d,       // [n] squared distance to winner
+    double* __restrict__ inertia) {  // [1] accumulated
+  extern __shared__ float csq_s[];
+  for (int e = threadIdx.x; e < k; e += 256) csq_s[e] = c_sq[e];
+  __syncthreads();
+  double local_sum = 0.0;
+  const int64_t stride = (int64_t)gridDim.x * 256;
+  for (int64_t i = (int64_t)blockIdx.x * 256 + threadIdx.x; i < n; i += stride) {
+    float best = 3.0e38f;
+    int bi = 0;
+    for (int kk = 0; kk < k; ++kk) {
+      const float v = csq_s[kk] - 2.0f * dots[(int64_t)kk * n + i];
+      if (v < best) { best = v; bi = kk; }
+    }
+    const float d2 = fmaxf(best + x_sq[i], 0.0f);
+    labels[i] = bi;
+    min_d[i] = d2;
+    local_sum += (double)d2;
+  }
+  // block-reduce the inertia partial, one global atomic per block
+  __shared__ double red_s[256];
+  red_s[threadIdx.x] = local_sum;
+  __syncthreads();
+  for (int off = 128; off > 0; off >>= 1) {
+    if (threadIdx.x < off) red_s[threadIdx.x] += red_s[threadIdx.x + off];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) atomicAdd(inertia, red_s[0]);
+}
+
+// ---------------------------------------------------------------------------
 // rf_partition: counting-sort rows by batch-local node id -> (perm, seg_off).
 // Replaces the per-batch torch sort + nonzero + gather chain (radix sort was
 // ~8% of RF fit kernel time; reference behavior: cuML's batched node trainer
@@ -2649,6 +2696,24 @@ std::vector<torch::Tensor> rf_best_split(torch::Tensor H, int64_t min_leaf,
   return {gain, feat, bin, lval, rval};
 }
 
+std::vector<torch::Tensor> kmeans_argmin_kn(torch::Tensor dots, torch::Tensor x_sq,
+                                            torch::Tensor c_sq) {
+  TORCH_CHECK(dots.is_cuda() && dots.dtype() == torch::kFloat32 && dots.is_contiguous());
+  TORCH_CHECK(x_sq.is_contiguous() && c_sq.is_contiguous());
+  const int k = (int)dots.size(0);
+  const int64_t n = dots.size(1);
+  TORCH_CHECK((size_t)k * 4 <= 64 * 1024, "k too large for LDS stage");
+  auto labels = torch::empty({n}, dots.options().dtype(torch::kInt32));
+  auto min_d = torch::empty({n}, dots.options());
+  auto inertia = torch::zeros({1}, dots.options().dtype(torch::kFloat64));
+  const unsigned grid = (unsigned)std::min<int64_t>(4096, (n + 255) / 256 + 1);
+  hipLaunchKernelGGL(kmeans_argmin_kn_kernel, dim3(grid), dim3(256), (size_t)k * 4,
+                     cur_stream(), dots.data_ptr<float>(), x_sq.data_ptr<float>(),
+                     c_sq.data_ptr<float>(), n, k, labels.data_ptr<int32_t>(),
+                     min_d.data_ptr<float>(), inertia.data_ptr<double>());
+  return {labels, min_d, inertia};
+}
+
 std::vector<torch::Tensor> rf_partition(torch::Tensor node_of_row, torch::Tensor lut,
                                         int64_t B) {
   TORCH_CHECK(node_of_row.is_cuda() && node_of_row.dtype() == torch::kInt64 &&
@@ -2824,6 +2889,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("csr_grad", &csr_grad, "CSR grad = (A^T R)^T by column scatter");
   m.def("csr_col_moments", &csr_col_moments, "per-column sum/sumsq of CSR values");
   m.def("kmeans_assign", &kmeans_assign, "fused MFMA distance + argmin");
+  m.def("kmeans_argmin_kn", &kmeans_argmin_kn, "argmin epilogue over a [k,n] GEMM dot block");
   m.def("label_accumulate", &label_accumulate, "per-center sum/count scatter");
   m.def("gram_f32", &gram_f32, "A^T A via MFMA f32");
   m.def("softmax_residual_loss", &softmax_residual_loss, "fused softmax residual + loss");

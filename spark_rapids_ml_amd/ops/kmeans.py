@@ -22,17 +22,58 @@ def kmeans_assign_reduce(
     C: torch.Tensor,
     x_sq: Optional[torch.Tensor] = None,
 ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor, float]:
-    """Returns (labels i32 [n], sums [k,d], counts [k], inertia)."""
+    """Returns (labels i32 [n], sums [k,d], counts [k], inertia).
+
+    Default GPU path: hipBLASLt GEMM for the C @ Xᵀ dot block (141 TF on the
+    1M×3000×k=1000 headline shape) + the own `kmeans_argmin_kn` epilogue
+    (one bandwidth-bound pass) + `label_accumulate`. The all-in-one MFMA
+    `kmeans_assign` kernel (84 TF, profiles/README.md ladder) stays selected
+    via SRML_KMEANS_VARIANT=fused."""
+    import os
+
     if use_hip(X) and X.dtype == torch.float32:
         ext = hip_ops()
         if x_sq is None:
             x_sq = _xsq(X)
-        labels, _min_d, inertia = ext.kmeans_assign(
-            X.contiguous(), C.contiguous(), x_sq.contiguous()
-        )
+        n, k = X.shape[0], C.shape[0]
+        fused = os.environ.get("SRML_KMEANS_VARIANT") == "fused"
+        if not fused and k * 4 <= 64 * 1024 and n > 0:
+            labels, inertia = _assign_gemm(ext, X, C, x_sq, n, k)
+        else:
+            labels, _min_d, inertia = ext.kmeans_assign(
+                X.contiguous(), C.contiguous(), x_sq.contiguous()
+            )
         sums, counts = ext.label_accumulate(X.contiguous(), labels, C.shape[0])
         return labels, sums.to(torch.float64), counts.to(torch.float64), float(inertia.item())
     return torch_ref.kmeans_assign_reduce(X, C, x_sq)
+
+
+def _assign_gemm(
+    ext, X: torch.Tensor, C: torch.Tensor, x_sq: torch.Tensor, n: int, k: int,
+    max_dots_bytes: int = 8 << 30,
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    """GEMM + argmin-epilogue assignment, row-chunked so the [k, chunk] dot
+    block stays under `max_dots_bytes`."""
+    c_sq = (C * C).sum(dim=1).contiguous()
+    Ct = C.contiguous()
+    rchunk = max(256, int(max_dots_bytes // max(1, k * 4)))
+    if rchunk >= n:
+        dots = torch.mm(Ct, X.T)
+        labels, _min_d, inertia = ext.kmeans_argmin_kn(dots, x_sq, c_sq)
+        return labels, inertia
+    labels = torch.empty(n, dtype=torch.int32, device=X.device)
+    inertia = torch.zeros(1, dtype=torch.float64, device=X.device)
+    dots = torch.empty((k, min(rchunk, n)), dtype=torch.float32, device=X.device)
+    for s in range(0, n, rchunk):
+        e = min(n, s + rchunk)
+        dv = dots if e - s == dots.shape[1] else torch.empty(
+            (k, e - s), dtype=torch.float32, device=X.device
+        )
+        torch.mm(Ct, X[s:e].T, out=dv)
+        lb, _md, it = ext.kmeans_argmin_kn(dv, x_sq[s:e].contiguous(), c_sq)
+        labels[s:e] = lb
+        inertia += it
+    return labels, inertia
 
 
 def kmeans_predict(X: torch.Tensor, C: torch.Tensor) -> torch.Tensor:

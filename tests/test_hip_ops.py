@@ -409,3 +409,25 @@ def test_csr_glm_kernels_match_torch(ext):
     dense = torch.from_numpy(np.asarray(Xc.todense(), np.float64)).cuda()
     assert torch.allclose(mom[0], dense.sum(0), rtol=1e-6, atol=1e-6)
     assert torch.allclose(mom[1], (dense * dense).sum(0), rtol=1e-6, atol=1e-6)
+
+
+@pytest.mark.parametrize("n,d,k", [(50000, 128, 100), (12345, 64, 37), (8000, 3000, 1000)])
+def test_kmeans_gemm_argmin_matches_fused(ext, n, d, k):
+    """GEMM + kmeans_argmin_kn epilogue vs the all-in-one assign kernel."""
+    X = _rand(n, d, seed=21)
+    C = _rand(k, d, seed=22)
+    x_sq = (X * X).sum(dim=1)
+    from spark_rapids_ml_amd.ops.kmeans import _assign_gemm
+
+    labels_g, inertia_g = _assign_gemm(ext, X, C, x_sq, n, k,
+                                       max_dots_bytes=16 << 20)  # force chunking
+    labels_f, _md, inertia_f = ext.kmeans_assign(X, C, x_sq)
+    mism = (labels_g != labels_f)
+    if bool(mism.any()):
+        # ties may break differently; winning distances must agree
+        d_g = ((X[mism] - C[labels_g[mism].long()]) ** 2).sum(1)
+        d_f = ((X[mism] - C[labels_f[mism].long()]) ** 2).sum(1)
+        assert torch.allclose(d_g, d_f, rtol=1e-3, atol=1e-2)
+    assert abs(float(inertia_g.item()) - float(inertia_f.item())) <= (
+        1e-4 * max(1.0, abs(float(inertia_f.item())))
+    )
