@@ -77,8 +77,18 @@ def _assign_gemm(
 
 
 def kmeans_predict(X: torch.Tensor, C: torch.Tensor) -> torch.Tensor:
+    import os
+
     if use_hip(X) and X.dtype == torch.float32:
         ext = hip_ops()
+        n, k = X.shape[0], C.shape[0]
+        if (
+            os.environ.get("SRML_KMEANS_VARIANT") != "fused"
+            and k * 4 <= 64 * 1024
+            and n > 0
+        ):
+            labels, _ = _assign_gemm(ext, X.contiguous(), C, _xsq(X), n, k)
+            return labels
         labels, _, _ = ext.kmeans_assign(X.contiguous(), C.contiguous(), _xsq(X))
         return labels
     return torch_ref.kmeans_predict(X, C)
