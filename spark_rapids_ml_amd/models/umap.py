@@ -172,6 +172,43 @@ def _fuzzy_simplicial_set_t(
     return uk // n, uk % n, sym[keep].to(torch.float32)
 
 
+def _spectral_init_lobpcg(
+    rows: torch.Tensor, cols: torch.Tensor, vals: torch.Tensor,
+    n: int, dim: int, seed: int, iters: int = 60,
+) -> Optional[torch.Tensor]:
+    """Spectral embedding init at ANY size: smallest nontrivial eigenvectors
+    of the normalized Laplacian via torch.lobpcg with sparse matmul (scipy
+    eigsh needed the graph on the host and sigma-shifted factorization;
+    this runs on the membership graph where it already lives). Returns
+    [n, dim] f32 or None on failure (caller falls back to random)."""
+    try:
+        dev = rows.device
+        W = torch.sparse_coo_tensor(
+            torch.stack([rows, cols]), vals.to(torch.float32), (n, n)
+        ).coalesce()
+        deg = torch.sparse.sum(W, dim=1).to_dense()
+        dinv = 1.0 / torch.sqrt(torch.clamp(deg, min=1e-12))
+        # L = I - D^-1/2 W D^-1/2 as a scaled sparse operator
+        sv = W.values() * dinv[W.indices()[0]] * dinv[W.indices()[1]]
+        Wn = torch.sparse_coo_tensor(W.indices(), sv, (n, n)).coalesce()
+        eye_idx = torch.arange(n, device=dev)
+        L = torch.sparse_coo_tensor(
+            torch.cat([Wn.indices(), torch.stack([eye_idx, eye_idx])], dim=1),
+            torch.cat([-Wn.values(), torch.ones(n, device=dev)]),
+            (n, n),
+        ).coalesce()
+        g = torch.Generator(device=dev)
+        g.manual_seed(seed & 0x7FFFFFFF)
+        X0 = torch.randn(n, dim + 1, generator=g, device=dev)
+        vals_e, vecs = torch.lobpcg(L, k=dim + 1, X=X0, largest=False, niter=iters)
+        order = torch.argsort(vals_e)
+        emb = vecs[:, order[1 : dim + 1]]
+        scale = 10.0 / torch.clamp(emb.abs().max(), min=1e-12)
+        return (emb * scale).to(torch.float32)
+    except Exception:
+        return None
+
+
 def _optimize_embedding(
     emb: torch.Tensor,
     heads: torch.Tensor,
@@ -466,13 +503,23 @@ class UMAP(_UMAPParams, Estimator):
         _t0 = _mark("fuzzy_set", _t0)
 
         n_epochs = int(self.getOrDefault("n_epochs")) or (500 if n <= 10000 else 200)
-        emb = self._init_embedding(
-            X, as_numpy(rows_t), as_numpy(cols_t), as_numpy(vals_t), dim, seed
-        ) if (self.getOrDefault("init") == "spectral" and dim + 1 < n <= 50000) else None
+        emb = None
+        if self.getOrDefault("init") == "spectral" and dim + 1 < n:
+            if n <= 50000:
+                emb = self._init_embedding(
+                    X, as_numpy(rows_t), as_numpy(cols_t), as_numpy(vals_t), dim, seed
+                )
+            else:
+                # device LOBPCG spectral init (the scipy eigsh path caps at
+                # 50k; the reference's cuML default is spectral at any size)
+                emb = _spectral_init_lobpcg(
+                    rows_t.to(device), cols_t.to(device), vals_t.to(device),
+                    n, dim, seed,
+                )
         if emb is None:
             rng = np.random.default_rng(seed)
             emb = rng.uniform(-10, 10, size=(n, dim)).astype(np.float32)
-        emb_t = torch.from_numpy(emb).to(device)
+        emb_t = torch.from_numpy(emb).to(device) if isinstance(emb, np.ndarray) else emb.to(device)
         _t0 = _mark("init_embedding", _t0)
         gen = torch.Generator(device=device)
         gen.manual_seed(seed)

@@ -141,3 +141,44 @@ def test_fuzzy_simplicial_set_torch_matches_numpy():
     assert set(m1) == set(m2)
     for key in m1:
         assert abs(m1[key] - m2[key]) < 1e-5, key
+
+
+def test_spectral_init_lobpcg_matches_scipy_subspace():
+    """LOBPCG spectral init must span the same eigen-subspace as the scipy
+    path on a small graph (sign/rotation-invariant comparison)."""
+    import scipy.sparse as sp
+    import torch
+    from scipy.sparse.linalg import eigsh
+
+    from spark_rapids_ml_amd.models.umap import _spectral_init_lobpcg
+
+    rng = np.random.default_rng(0)
+    n, k = 600, 10
+    # symmetric kNN-ish graph
+    rows = np.repeat(np.arange(n), k)
+    cols = rng.integers(0, n, n * k)
+    vals = rng.random(n * k)
+    W = sp.coo_matrix((vals, (rows, cols)), shape=(n, n))
+    W = (W + W.T).tocoo()
+
+    emb = _spectral_init_lobpcg(
+        torch.from_numpy(W.row.astype(np.int64)),
+        torch.from_numpy(W.col.astype(np.int64)),
+        torch.from_numpy(W.data),
+        n, 2, seed=1, iters=200,
+    )
+    assert emb is not None and emb.shape == (n, 2)
+
+    deg = np.asarray(W.sum(axis=1)).flatten()
+    dinv = 1.0 / np.sqrt(np.maximum(deg, 1e-12))
+    L = sp.eye(n) - sp.diags(dinv) @ W.tocsr() @ sp.diags(dinv)
+    w, v = eigsh(L, k=3, sigma=0.0, which="LM")
+    ref = v[:, 1:3]
+
+    # subspace alignment: projection of emb onto ref explains ~all variance
+    E = emb.numpy()
+    E = E / np.linalg.norm(E, axis=0, keepdims=True)
+    R = ref / np.linalg.norm(ref, axis=0, keepdims=True)
+    proj = R @ (R.T @ E)
+    frac = (proj * E).sum() / 2.0
+    assert frac > 0.97, frac
