@@ -174,7 +174,7 @@ def _fuzzy_simplicial_set_t(
 
 def _spectral_init_lobpcg(
     rows: torch.Tensor, cols: torch.Tensor, vals: torch.Tensor,
-    n: int, dim: int, seed: int, iters: int = 60,
+    n: int, dim: int, seed: int, iters: int = 30,
 ) -> Optional[torch.Tensor]:
     """Spectral embedding init at ANY size: smallest nontrivial eigenvectors
     of the normalized Laplacian via torch.lobpcg with sparse matmul (scipy
