@@ -244,6 +244,15 @@ def _optimize_embedding(
         # updates in place, so re-alias n_emb AFTER making emb contiguous
         emb = emb.contiguous()
         n_emb = tail_emb.contiguous() if tail_emb is not None else emb
+        # sort edges by epochs-per-sample: only ~10-20% of edges are due in
+        # any epoch, and with symmetrization-order edges nearly EVERY wave
+        # held a due lane — the whole grid ran the body at ~15% lane
+        # utilization. Sorted, the due lanes cluster into few waves.
+        # (Hogwild is order-free; the hash RNG keys off the sorted index.)
+        order = torch.argsort(eps)
+        heads = heads[order]
+        tails = tails[order]
+        eps = eps[order]
         ext.umap_sgd(
             emb,
             n_emb,
