@@ -2588,7 +2588,7 @@ std::vector<torch::Tensor> softmax_residual_loss(torch::Tensor scores, torch::Te
 torch::Tensor rf_histogram(torch::Tensor Xb, torch::Tensor perm, torch::Tensor seg_off,
                            torch::Tensor feat_sel, torch::Tensor y, int64_t f0,
                            int64_t FC, int64_t n_bins, int64_t n_classes,
-                           torch::Tensor sample) {
+                           torch::Tensor sample, double y_max) {
   // Xb here is the COLUMN-major binned matrix [d, n_phys]
   TORCH_CHECK(Xb.is_cuda() && Xb.dtype() == torch::kUInt8 && Xb.is_contiguous());
   TORCH_CHECK(perm.dtype() == torch::kInt64 && seg_off.dtype() == torch::kInt64);
@@ -2600,13 +2600,14 @@ torch::Tensor rf_histogram(torch::Tensor Xb, torch::Tensor perm, torch::Tensor s
   const int mf = feat_sel.numel() > 0 ? (int)feat_sel.size(1) : 0;
   const bool has_sample = sample.numel() > 0;
   if (has_sample) TORCH_CHECK(sample.dtype() == torch::kInt32 && sample.is_contiguous());
-  // regression packed-u64 path: needs max|y| and a per-node row-count cap
+  // regression packed-u64 path: caller supplies max|y| (computed once per
+  // forest fit — a per-call .item() would sync every chunk) and the
+  // per-node row-count cap gates it
   float y_scale = 0.0f, y_inv_scale = 0.0f;
-  if (!classif) {
+  if (!classif && y_max > 0.0) {
     const int64_t vn = has_sample ? sample.numel() : perm.numel();
     if (vn < (1ll << 23)) {
-      y_scale = y.abs().max().item<float>();
-      if (y_scale <= 0.0f) y_scale = 1.0f;
+      y_scale = (float)y_max;
       y_inv_scale = 1.0f / y_scale;
     }
   }
@@ -2633,7 +2634,7 @@ torch::Tensor rf_histogram(torch::Tensor Xb, torch::Tensor perm, torch::Tensor s
 torch::Tensor rf_histogram_fw(torch::Tensor Xrm, torch::Tensor perm, torch::Tensor seg_off,
                               torch::Tensor feat_sel, torch::Tensor y, int64_t f0,
                               int64_t FC, int64_t n_bins, int64_t n_classes,
-                              torch::Tensor sample) {
+                              torch::Tensor sample, double y_max) {
   // Xrm is ROW-major [n_phys, d]
   TORCH_CHECK(Xrm.is_cuda() && Xrm.dtype() == torch::kUInt8 && Xrm.is_contiguous());
   TORCH_CHECK(perm.dtype() == torch::kInt64 && seg_off.dtype() == torch::kInt64);
@@ -2646,11 +2647,10 @@ torch::Tensor rf_histogram_fw(torch::Tensor Xrm, torch::Tensor perm, torch::Tens
   const int mf = feat_sel.numel() > 0 ? (int)feat_sel.size(1) : 0;
   const bool has_sample = sample.numel() > 0;
   float y_scale = 0.0f, y_inv_scale = 0.0f;
-  if (!classif) {
+  if (!classif && y_max > 0.0) {
     const int64_t vn = has_sample ? sample.numel() : perm.numel();
     if (vn < (1ll << 23)) {
-      y_scale = y.abs().max().item<float>();
-      if (y_scale <= 0.0f) y_scale = 1.0f;
+      y_scale = (float)y_max;
       y_inv_scale = 1.0f / y_scale;
     }
   }

@@ -433,6 +433,12 @@ def _grow_forest(
 
         ext = hip_ops()
         y32 = y_idx.to(torch.int32) if task == "classification" else yf
+        # max|y| for the packed regression histogram cells, ONE sync per fit
+        y_max = (
+            float(yf.abs().max().item()) if task == "regression" and n else 0.0
+        )
+        if y_max <= 0.0 and task == "regression":
+            y_max = 1.0
         C_ch = n_classes if task == "classification" else 2
         # Measured dispatch (A/B in profiles/README.md): the row-lane
         # kernel wins for classification (sqrt-sampled features, sorted
@@ -542,7 +548,7 @@ def _grow_forest(
                     H = _hist_fn(
                         _hist_x, perm, seg_off, fsel32, y32, f0, F, n_bins,
                         n_classes if task == "classification" else 0,
-                        sample_t,
+                        sample_t, y_max if task == "regression" else 0.0,
                     )
                     if H.shape[3] <= 16:
                         # fused gain scan + per-node best (kernel)

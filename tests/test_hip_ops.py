@@ -123,7 +123,7 @@ def test_rf_histogram_matches_ref(ext, task, nc):
     if task == "classification":
         y = torch.randint(0, nc, (n,), generator=g, dtype=torch.int32).cuda()
         H = ext.rf_histogram(Xb.T.contiguous(), perm, seg_off, feat_sel.to(torch.int32), y, 0, mf, nb, nc,
-                             torch.empty(0, dtype=torch.int32).cuda())
+                             torch.empty(0, dtype=torch.int32).cuda(), 0.0)
         ref = torch.zeros(B, mf, nb, nc).cuda()
         for b in range(B):
             m = loc == b
@@ -140,7 +140,8 @@ def test_rf_histogram_matches_ref(ext, task, nc):
     else:
         y = torch.randn(n, generator=g).cuda()
         H = ext.rf_histogram(Xb.T.contiguous(), perm, seg_off, feat_sel.to(torch.int32), y, 0, mf, nb, 0,
-                             torch.empty(0, dtype=torch.int32).cuda())
+                             torch.empty(0, dtype=torch.int32).cuda(),
+                             float(y.abs().max().item()))
         ref = torch.zeros(B, mf, nb, 2).cuda()
         for b in range(B):
             m = loc == b
@@ -314,7 +315,7 @@ def test_rf_histogram_virtual_rows(ext):
     lut = torch.arange(T, dtype=torch.int64).cuda()
     perm, seg_off = ext.rf_partition(node_of_row, lut, T)
     feat_sel = torch.empty((0, 0), dtype=torch.int32).cuda()
-    H = ext.rf_histogram(Xb.T.contiguous(), perm, seg_off, feat_sel, y, 0, d, nb, nc, sample)
+    H = ext.rf_histogram(Xb.T.contiguous(), perm, seg_off, feat_sel, y, 0, d, nb, nc, sample, 0.0)
     # reference per tree
     for t in range(T):
         rows = sample[t * n : (t + 1) * n].long()
