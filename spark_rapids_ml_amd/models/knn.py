@@ -246,9 +246,16 @@ class NearestNeighborsModel(_NNModelBase):
         def search(Qt, X, k):
             from ..parallel.context import get_comm
 
-            Xt = to_device_tensor(
-                np.ascontiguousarray(X, dtype=np.float32), get_comm().device
-            )
+            # cache the device-resident item shard across kneighbors calls
+            # (re-uploading 30 GB of items dominated a 10M-item search)
+            cached = getattr(self, "_item_dev", None)
+            if cached is None or cached[0] != id(X):
+                Xt = to_device_tensor(
+                    np.ascontiguousarray(X, dtype=np.float32), get_comm().device
+                )
+                self._item_dev = (id(X), Xt)
+            else:
+                Xt = cached[1]
             return knn_topk(Qt, Xt, k)
 
         return self._kneighbors_impl(query_df, search)
