@@ -875,45 +875,57 @@ class _RandomForestEstimator(_RandomForestParams, Estimator):
             # all of this worker's trees grow together in ONE arena over a
             # virtual row space (bootstrap = int32 sample map, no gathered
             # [n,d] copies per tree); seed offset by tree_id0 so ranks draw
-            # disjoint streams
-            gen = torch.Generator(device=ctx.device)
-            gen.manual_seed((seed + 1315423911 * (tree_id0 + 1)) & 0x7FFFFFFFFFFF)
-            if bootstrap:
-                n_draw = max(1, int(round(max_samples * n)))
-                sample = torch.randint(
-                    0, n, (my_trees * n_draw,), generator=gen,
-                    device=ctx.device, dtype=torch.int32,
+            # disjoint streams. The virtual row space costs ~13 bytes/row
+            # (node map + sample), so cap it at ~5e8 rows and grow LARGE
+            # forests in sequential tree groups (500 trees × 20M rows would
+            # otherwise need >100 GB of arena state).
+            group_cap = max(1, int(5e8) // max(1, n))
+            t0g = 0
+            while t0g < my_trees:
+                g_trees = min(group_cap, my_trees - t0g)
+                gen = torch.Generator(device=ctx.device)
+                gen.manual_seed(
+                    (seed + 1315423911 * (tree_id0 + t0g + 1)) & 0x7FFFFFFFFFFF
                 )
-                # sort each tree's draw: a bootstrap multiset is unordered,
-                # and ascending physical rows turn the histogram's random
-                # per-row byte gathers into near-sequential column reads
-                # (random gathers fetched a 64B line per byte). ONE flat
-                # radix sort over tree-offset keys — the segmented
-                # [T, n] sort.values cost 390 ms at 10x1M
-                t_of = torch.arange(
-                    my_trees, device=ctx.device
-                ).repeat_interleave(n_draw)
-                keys = t_of * n + sample.to(torch.int64)
-                keys, _ = torch.sort(keys)
-                sample = (keys % n).to(torch.int32)
-            else:
-                sample = None
-            trees = _grow_forest(
-                Xb,
-                yt,
-                edges,
-                self._task,
-                n_classes,
-                n_bins,
-                max_depth,
-                min_leaf,
-                min_gain,
-                max_features,
-                gen,
-                n_trees=my_trees,
-                sample=sample,
-                node_batch=int(params.get("max_batch_size", 4096)),
-            )
+                if bootstrap:
+                    n_draw = max(1, int(round(max_samples * n)))
+                    sample = torch.randint(
+                        0, n, (g_trees * n_draw,), generator=gen,
+                        device=ctx.device, dtype=torch.int32,
+                    )
+                    # sort each tree's draw: a bootstrap multiset is
+                    # unordered, and ascending physical rows turn the
+                    # histogram's random per-row byte gathers into
+                    # near-sequential column reads. ONE flat radix sort over
+                    # tree-offset keys (the segmented [T, n] sort.values
+                    # cost 390 ms at 10x1M)
+                    t_of = torch.arange(
+                        g_trees, device=ctx.device
+                    ).repeat_interleave(n_draw)
+                    keys = t_of * n + sample.to(torch.int64)
+                    keys, _ = torch.sort(keys)
+                    sample = (keys % n).to(torch.int32)
+                else:
+                    sample = None
+                trees.extend(
+                    _grow_forest(
+                        Xb,
+                        yt,
+                        edges,
+                        self._task,
+                        n_classes,
+                        n_bins,
+                        max_depth,
+                        min_leaf,
+                        min_gain,
+                        max_features,
+                        gen,
+                        n_trees=g_trees,
+                        sample=sample,
+                        node_batch=int(params.get("max_batch_size", 4096)),
+                    )
+                )
+                t0g += g_trees
         elif my_trees > 0:
             # empty local shard: contribute degenerate single-leaf trees so
             # the merged forest still has numTrees members
