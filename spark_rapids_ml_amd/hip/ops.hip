@@ -2719,7 +2719,7 @@ std::vector<torch::Tensor> rf_partition(torch::Tensor node_of_row, torch::Tensor
   TORCH_CHECK(node_of_row.is_cuda() && node_of_row.dtype() == torch::kInt64 &&
               node_of_row.is_contiguous());
   TORCH_CHECK(lut.is_cuda() && lut.dtype() == torch::kInt64 && lut.is_contiguous());
-  TORCH_CHECK(B >= 1 && B <= 8192, "node batch too large for LDS counts");
+  TORCH_CHECK(B >= 1 && B <= 16384, "node batch too large for LDS counts");
   const int64_t n = node_of_row.size(0);
   auto counts = torch::zeros({B}, node_of_row.options().dtype(torch::kInt32));
   const size_t lds1 = (size_t)B * 4;

@@ -104,7 +104,7 @@ class _RandomForestParams(HasFeaturesCol, HasFeaturesCols, HasLabelCol, HasPredi
             "random_state": 0,
             "bootstrap": True,
             "split_criterion": None,
-            "max_batch_size": 8192,
+            "max_batch_size": 16384,
             # cuML-signature knobs accepted for ctor parity (inert here)
             "n_streams": 4,
             "min_samples_split": 2,
@@ -344,7 +344,7 @@ def _grow_forest(
     gen: torch.Generator,
     n_trees: int = 1,
     sample: Optional[torch.Tensor] = None,  # int32 [n_trees*per] bootstrap map
-    node_batch: int = 8192,
+    node_batch: int = 16384,
     feat_chunk: int = 256,
 ) -> List[Dict[str, np.ndarray]]:
     """Grow ALL of this worker's trees at once in one node arena over a
