@@ -343,6 +343,14 @@ class _ANNModelParams(_ANNParams):
 class ApproximateNearestNeighborsModel(_ANNModelParams, _NNModelBase):
     """ANN model: local IVF index per rank (reference knn.py:1217)."""
 
+    def approxSimilarityJoin(
+        self, query_df: DataFrame, distCol: str = "distCol"
+    ) -> DataFrame:
+        """k-ANN join of query rows against the fitted items (reference
+        approxSimilarityJoin, knn.py:1694-1723): exploded
+        (item_id, query_id, distance) rows from this model's kneighbors."""
+        return self.exactNearestNeighborsJoin(query_df, distCol=distCol)
+
     def _default_nlist(self, n: int) -> int:
         return max(1, min(1024, int(math.sqrt(max(1, n)))))
 

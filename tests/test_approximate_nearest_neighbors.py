@@ -118,3 +118,14 @@ def test_ann_distributed_recall():
             hits += len(set(row_pred) & set(row_true))
             total += len(row_true)
     assert hits / total > 0.85, f"recall {hits/total}"
+
+
+def test_approx_similarity_join():
+    rng = np.random.default_rng(0)
+    X = rng.normal(size=(2000, 32)).astype(np.float32)
+    Q = rng.normal(size=(50, 32)).astype(np.float32)
+    model = ApproximateNearestNeighbors(k=4).fit(DataFrame.from_numpy(X))
+    joined = model.approxSimilarityJoin(DataFrame.from_numpy(Q), distCol="d")
+    assert joined.num_rows == 50 * 4
+    assert "d" in joined.columns
+    assert float(np.asarray(joined["d"]).min()) >= 0.0
