@@ -387,6 +387,10 @@ class KMeansModel(_KMeansParams, Model):
     def cluster_centers_(self) -> np.ndarray:
         return self._model_attributes["cluster_centers_"]
 
+    @property
+    def numFeatures(self) -> int:
+        return int(self.cluster_centers_.shape[1])
+
     def clusterCenters(self) -> List[np.ndarray]:
         return list(self.cluster_centers_)
 
@@ -578,6 +582,10 @@ class DBSCANModel(_DBSCANParams, Model):
 
     def __init__(self, n_cols: int = 0, **kwargs: Any) -> None:
         super().__init__(n_cols=n_cols)
+
+    @property
+    def numFeatures(self) -> int:
+        return int(self._model_attributes.get("n_cols", 0))
 
     def transform(self, df: DataFrame) -> DataFrame:
         features_col, features_cols = self._get_input_columns()

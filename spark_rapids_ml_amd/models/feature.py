@@ -81,6 +81,9 @@ class PCA(_PCAParams, Estimator):
     def setOutputCol(self, value: str) -> "PCA":
         return self._set_params(outputCol=value)
 
+    def setInputCols(self, value) -> "PCA":
+        return self._set_params(featuresCols=list(value))
+
     def _fit_array(
         self, X: Any, y: Optional[Any], ctx: _FitContext, params: Dict[str, Any]
     ) -> Dict[str, Any]:
@@ -141,6 +144,9 @@ class PCA(_PCAParams, Estimator):
 
 class PCAModel(_PCAParams, Model):
     """Fitted PCA model (reference PCAModel, feature.py:291)."""
+
+    def setInputCols(self, value) -> "PCAModel":
+        return self._set_params(featuresCols=list(value))
 
     def __init__(
         self,

@@ -61,6 +61,20 @@ class _NNParams(HasFeaturesCol, HasFeaturesCols, HasIDCol):
     def getK(self) -> int:
         return self.getOrDefault("k")
 
+    # reference kNN exposes inputCol aliases over featuresCol (knn.py:203+)
+    def setInputCol(self, value):
+        if isinstance(value, (list, tuple)):
+            return self._set_params(featuresCols=list(value))
+        return self._set_params(featuresCol=value)
+
+    def setInputCols(self, value):
+        return self._set_params(featuresCols=list(value))
+
+    def getInputCol(self):
+        if self.isSet("featuresCols"):
+            return self.getOrDefault("featuresCols")
+        return self.getOrDefault("featuresCol")
+
 
 class _NNModelBase(_NNParams, Model):
     """Shared kneighbors/join plumbing (reference _NNModelBase, knn.py:203+)."""
@@ -293,9 +307,18 @@ class _ANNParams(_NNParams):
         "ann", "algoParams", "algorithm tuning dict.", DictTypeConverters._toDict
     )
 
+    metric = Param(
+        "ann", "metric", "distance metric (euclidean/sqeuclidean).", TypeConverters.toString
+    )
+
     def __init__(self, **kwargs: Any) -> None:
         super().__init__(**kwargs)
-        self._setDefault(algorithm="ivfflat", algoParams=None)
+        self._setDefault(algorithm="ivfflat", algoParams=None, metric="euclidean")
+
+    def setMetric(self, value: str):
+        if value not in ("euclidean", "l2", "sqeuclidean"):
+            raise ValueError("metric must be euclidean/l2/sqeuclidean")
+        return self._set_params(metric=value)
 
 
 class ApproximateNearestNeighbors(_ANNParams, Estimator):

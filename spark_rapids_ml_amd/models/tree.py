@@ -1158,6 +1158,24 @@ class _RandomForestModel(_RandomForestParams, Model):
                 acc += v[:, 0]
         return acc
 
+    def predictLeaf(self, value) -> np.ndarray:
+        """Leaf ordinal per tree for one feature vector (reference
+        tree.py:613-617 delegates to the Spark model; here a host
+        traversal — leaves numbered left-to-right per tree)."""
+        x = np.asarray(value, dtype=np.float32).ravel()
+        out = np.zeros(self.numTrees, dtype=np.float64)
+        for ti, t in enumerate(self.trees):
+            # leaf ordinals by ascending node id among leaves
+            leaf_ord = np.cumsum(t["is_leaf"]) - 1
+            node = 0
+            while not t["is_leaf"][node]:
+                f = int(t["feature"][node])
+                node = int(
+                    t["left"][node] if x[f] < t["threshold"][node] else t["right"][node]
+                )
+            out[ti] = float(leaf_ord[node])
+        return out
+
     def predict(self, value) -> float:
         """Single-vector prediction (pyspark RandomForest*Model.predict)."""
         out = self._transform_array(np.asarray(value, dtype=np.float32).reshape(1, -1))

@@ -19,6 +19,7 @@ import torch
 from ..core import Estimator, Model, _FitContext
 from ..data import to_device_tensor
 from ..params import (
+    DictTypeConverters,
     HasFeaturesCol,
     HasFeaturesCols,
     HasLabelCol,
@@ -324,10 +325,34 @@ class _UMAPParams(HasFeaturesCol, HasFeaturesCols, HasLabelCol, HasOutputCol):
     sample_fraction = Param(
         "umap", "sample_fraction", "fit subsample fraction (reference umap.py:923-945).", TypeConverters.toFloat
     )
+    metric = Param("umap", "metric", "distance metric (euclidean/l2).", TypeConverters.toString)
+    metric_kwds = Param(
+        "umap", "metric_kwds", "metric kwargs (accepted; euclidean needs none).",
+        DictTypeConverters._toDict,
+    )
+    build_algo = Param(
+        "umap", "build_algo",
+        "kNN graph build: auto|brute_force_knn|nn_descent (reference umap.py:359-378).",
+        TypeConverters.toString,
+    )
+    build_kwds = Param(
+        "umap", "build_kwds", "graph-build kwargs (nn_descent knobs).",
+        DictTypeConverters._toDict,
+    )
+    transform_queue_size = Param(
+        "umap", "transform_queue_size",
+        "transform search-queue multiplier (inert: transform is exact kNN here).",
+        TypeConverters.toFloat,
+    )
 
     def __init__(self, **kwargs: Any) -> None:
         super().__init__(**kwargs)
         self._setDefault(
+            metric="euclidean",
+            metric_kwds=None,
+            build_algo="auto",
+            build_kwds=None,
+            transform_queue_size=4.0,
             n_neighbors=15.0,
             n_components=2,
             n_epochs=0,
@@ -352,6 +377,8 @@ class _UMAPParams(HasFeaturesCol, HasFeaturesCols, HasLabelCol, HasOutputCol):
             "n_neighbors", "n_components", "n_epochs", "learning_rate", "init",
             "min_dist", "spread", "set_op_mix_ratio", "local_connectivity",
             "repulsion_strength", "negative_sample_rate", "a", "b", "random_state",
+            "metric", "metric_kwds", "build_algo", "build_kwds",
+            "transform_queue_size",
         ]}
 
     @classmethod
@@ -475,7 +502,9 @@ class UMAP(_UMAPParams, Estimator):
         dim = int(self.getOrDefault("n_components"))
         seed = int(self.getOrDefault("random_state"))
         Xt = to_device_tensor(X, device)
-        if n <= 50000:
+        build_algo = str(self.getOrDefault("build_algo") or "auto").lower()
+        brute = build_algo == "brute_force_knn" or (build_algo == "auto" and n <= 50000)
+        if brute:
             d, i = knn_topk(Xt, Xt, k + 1)
         else:
             # nn-descent graph (reference switches off brute force above 50k
