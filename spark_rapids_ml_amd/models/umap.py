@@ -382,6 +382,17 @@ class _UMAPParams(HasFeaturesCol, HasFeaturesCols, HasLabelCol, HasOutputCol):
         ]}
 
     @classmethod
+    def _param_value_mapping(cls):
+        # euclidean distances only (reference passes other metrics to cuML;
+        # unsupported values must error, not silently compute euclidean)
+        return {
+            "metric": lambda v: v if v in ("euclidean", "l2") else None,
+            "build_algo": lambda v: v
+            if v in ("auto", "brute_force_knn", "nn_descent")
+            else None,
+        }
+
+    @classmethod
     def _get_native_params_default(cls) -> Dict[str, Any]:
         return {
             "n_neighbors": 15,
