@@ -327,3 +327,32 @@ def test_collect_as_arrow_fallback(stub_pyspark):
     local = spark_to_local(sdf, shard=False)
     assert sdf.used_collect_as_arrow
     np.testing.assert_allclose(np.asarray(local["features"]), X)
+
+
+def test_no_code_change_flow_proxy_plus_bridge(stub_pyspark):
+    """The complete reference user story: accelerate_pyspark() proxies the
+    import, the class fits a (stub) pyspark DataFrame through the Arrow
+    bridge, and the model saves in stock Spark format."""
+    import importlib
+
+    from spark_rapids_ml_amd.install import accelerate_pyspark
+
+    accelerate_pyspark()
+    clustering = importlib.import_module("pyspark.ml.clustering")
+
+    table, X = _blob_table()
+    sdf = stub_pyspark(table)
+    model = clustering.KMeans(k=3, maxIter=10, seed=1).fit(sdf)
+    assert np.asarray(model.cluster_centers_).shape == (3, 6)
+
+    import tempfile
+
+    with tempfile.TemporaryDirectory() as td:
+        p = f"{td}/km"
+        model.saveAsSparkModel(p)
+        from spark_rapids_ml_amd.spark import spark_model_class
+
+        assert spark_model_class(p) == "org.apache.spark.ml.clustering.KMeansModel"
+
+    out = model.transform(sdf)
+    assert type(out).__module__.startswith("pyspark.")
