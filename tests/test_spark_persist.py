@@ -335,3 +335,47 @@ def test_load_sparse_vector_model_written_by_stock_spark(tmp_path):
         np.asarray(m.coefficients), [0.0, 2.0, 0.0, 0.0, -3.0, 0.0]
     )
     assert np.isclose(m.intercept, 0.5)
+
+
+@pytest.mark.parametrize("shape", [(60, 1), (60, 2), (200, 33)])
+def test_spark_format_edge_shapes(tmp_path, shape):
+    """Tiny/odd shapes must survive the Spark-format round trip."""
+    n, d = shape
+    rng = np.random.default_rng(0)
+    X = rng.normal(size=(n, d)).astype(np.float32)
+    y = (X[:, 0] > 0).astype(np.float64)
+
+    km = KMeans(k=2, maxIter=5, seed=1).fit(DataFrame.from_numpy(X))
+    p1 = str(tmp_path / "km")
+    km.saveAsSparkModel(p1)
+    km2 = load_spark_model(p1)
+    np.testing.assert_allclose(km2.cluster_centers_, km.cluster_centers_, rtol=1e-6)
+
+    lr = LinearRegression().fit(DataFrame.from_numpy(X, y))
+    p2 = str(tmp_path / "lr")
+    lr.saveAsSparkModel(p2)
+    lr2 = load_spark_model(p2)
+    np.testing.assert_allclose(
+        np.asarray(lr2.coefficients), np.asarray(lr.coefficients), rtol=1e-5, atol=1e-7
+    )
+
+
+def test_spark_format_deep_forest_roundtrip(tmp_path):
+    """A deeper forest (many nodes, both tasks) round-trips bit-equal
+    predictions through the Spark node-data format."""
+    from sklearn.datasets import make_classification
+
+    X, y = make_classification(n_samples=2000, n_features=10, n_informative=6,
+                               n_classes=4, n_clusters_per_class=1, random_state=1)
+    df = DataFrame.from_numpy(X.astype(np.float32), y.astype(np.float64))
+    m = RandomForestClassifier(numTrees=9, maxDepth=9, seed=2).fit(df)
+    p = str(tmp_path / "deep")
+    m.saveAsSparkModel(p)
+    m2 = load_spark_model(p)
+    pred1 = np.asarray(m.transform(df)[m.getOrDefault("predictionCol")])
+    pred2 = np.asarray(m2.transform(df)[m2.getOrDefault("predictionCol")])
+    np.testing.assert_array_equal(pred1, pred2)
+    # probability parity too (vote normalization must survive)
+    pr1 = np.asarray(m.transform(df)[m.getOrDefault("probabilityCol")])
+    pr2 = np.asarray(m2.transform(df)[m2.getOrDefault("probabilityCol")])
+    np.testing.assert_allclose(pr1, pr2, atol=1e-6)
