@@ -624,6 +624,11 @@ class DBSCANModel(_DBSCANParams, Model):
         core_full = torch.cat(
             comm.allgather_rows(core_local.to(torch.uint8)), dim=0
         ).to(torch.bool)
+        if not bool(core_full.any()):
+            # no core points anywhere -> every point is noise; the
+            # min-label propagation sweeps (2 more O(N^2) passes) are
+            # vacuous
+            return np.full(n_local, -1, dtype=np.int64)
         core_u8 = core_full.to(torch.uint8).contiguous()
 
         labels = torch.arange(n, dtype=torch.int32, device=device)

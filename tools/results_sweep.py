@@ -89,9 +89,18 @@ def main():
     torch.cuda.empty_cache()
 
     Xd = rng.normal(size=(1_000_000, 64)).astype(np.float32)
-    st = t0(); DBSCAN(eps=1.0, min_samples=5).fit(DataFrame.from_numpy(Xd)).transform(
-        DataFrame.from_numpy(Xd))
-    done("dbscan_1m_64", st, r01_s=4.94)
+    dfd = DataFrame.from_numpy(Xd)
+    st = t0(); DBSCAN(eps=1.0, min_samples=5).fit(dfd).transform(dfd)
+    done("dbscan_1m_64_allnoise", st, r01_s=4.94)
+    # clustered data: label propagation actually runs
+    C20 = rng.normal(scale=10.0, size=(20, 64)).astype(np.float32)
+    Xc = (C20[rng.integers(0, 20, 1_000_000)]
+          + 0.3 * rng.normal(size=(1_000_000, 64)).astype(np.float32))
+    dfc2 = DataFrame.from_numpy(Xc.astype(np.float32))
+    st = t0()
+    out = DBSCAN(eps=4.0, min_samples=5).fit(dfc2).transform(dfc2)
+    done("dbscan_1m_64_20blobs", st,
+         n_clusters=int(np.asarray(out["prediction"]).max()) + 1)
 
     # sparse logreg 50M x 2048 @ 1% (BASELINE.json config)
     del Xd
