@@ -646,8 +646,11 @@ class DBSCANModel(_DBSCANParams, Model):
                 lab[core_ids] = upd
             return lab
 
+        import os as _os
+
+        _dbg = _os.environ.get("SRML_DBSCAN_DEBUG") == "1"
         out = None
-        for _ in range(64):
+        for _sweep in range(64):
             out = ext.dbscan_sweep(Xf, x_sq, off, n_local, eps2, 1, core_u8, labels.contiguous())
             new_full = torch.full((n,), BIG, dtype=torch.int32, device=device)
             if n_local:
@@ -659,6 +662,8 @@ class DBSCANModel(_DBSCANParams, Model):
             new_full[~core_full] = BIG
             new_full = pointer_jump(new_full)
             if bool((new_full == labels).all()):
+                if _dbg:
+                    print(f"[dbscan] converged after {_sweep + 1} label sweeps", flush=True)
                 break
             labels = new_full
 
@@ -789,6 +794,8 @@ class DBSCANModel(_DBSCANParams, Model):
             new_full[~core_full] = BIG
             new_full = pointer_jump(new_full)
             if bool((new_full == labels).all()):
+                if _dbg:
+                    print(f"[dbscan] converged after {_sweep + 1} label sweeps", flush=True)
                 break
             labels = new_full
 
@@ -819,6 +826,8 @@ class DBSCANModel(_DBSCANParams, Model):
             new_full[~core_full] = BIG
             new_full = pointer_jump(new_full)
             if bool((new_full == labels).all()):
+                if _dbg:
+                    print(f"[dbscan] converged after {_sweep + 1} label sweeps", flush=True)
                 break
             labels = new_full
 
