@@ -247,6 +247,14 @@ __global__ __launch_bounds__(256) void dbscan_sweep_kernel(
   const bool full_rows = (i0 + KM_BM <= row_lim);
   const int nsteps = (d + KM_BK - 1) / KM_BK;
   float rx[KM_LD], rc[KM_LD];
+  // per-lane per-fragment-row accumulator carried across ALL column tiles:
+  // dense data fires an LDS atomic per in-eps PAIR otherwise (~5e10 atomics
+  // on 20-blob 1M x 64 — the label sweep cost 2.6x the count pass)
+  int racc[2][16];
+#pragma unroll
+  for (int m = 0; m < 2; ++m)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) racc[m][r] = (mode == 0) ? 0 : 0x7fffffff;
 
   for (int j0 = 0; j0 < n; j0 += KM_BN) {
     const bool full_cols = (j0 + KM_BN <= n);
@@ -345,9 +353,9 @@ __global__ __launch_bounds__(256) void dbscan_sweep_kernel(
             float d2 = x_sq[gi] + cs - 2.0f * acc[m][nn][r];
             if (d2 <= eps2) {
               if (mode == 0)
-                atomicAdd(&acc_row[row], 1);
+                ++racc[m][r];
               else
-                atomicMin(&acc_row[row], lab_j);
+                racc[m][r] = min(racc[m][r], lab_j);
             }
           }
         }
@@ -355,6 +363,20 @@ __global__ __launch_bounds__(256) void dbscan_sweep_kernel(
     }
     __syncthreads();
   }
+
+  // one LDS atomic per fragment row for the whole kernel
+#pragma unroll
+  for (int m = 0; m < 2; ++m)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      int row = wr * 64 + m * 32 + (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
+      if (mode == 0) {
+        if (racc[m][r]) atomicAdd(&acc_row[row], racc[m][r]);
+      } else if (racc[m][r] != 0x7fffffff) {
+        atomicMin(&acc_row[row], racc[m][r]);
+      }
+    }
+  __syncthreads();
 
   for (int i = tid; i < KM_BM; i += blockDim.x) {
     int gi = i0 + i;

@@ -794,8 +794,6 @@ class DBSCANModel(_DBSCANParams, Model):
             new_full[~core_full] = BIG
             new_full = pointer_jump(new_full)
             if bool((new_full == labels).all()):
-                if _dbg:
-                    print(f"[dbscan] converged after {_sweep + 1} label sweeps", flush=True)
                 break
             labels = new_full
 
@@ -826,8 +824,6 @@ class DBSCANModel(_DBSCANParams, Model):
             new_full[~core_full] = BIG
             new_full = pointer_jump(new_full)
             if bool((new_full == labels).all()):
-                if _dbg:
-                    print(f"[dbscan] converged after {_sweep + 1} label sweeps", flush=True)
                 break
             labels = new_full
 
