@@ -218,6 +218,13 @@ __global__ __launch_bounds__(256) void kmeans_assign_kernel(
 // One kernel serves core detection, the min-label propagation sweeps and
 // the border pass (reference DBSCANMG: adjacency + BFS inside cuML,
 // SURVEY.md §2.3b).
+//
+// Ball-cover pruning (reference algorithm="rbc", clustering.py:686-695):
+// when tile_off != nullptr the column loop walks only the CSR list of
+// admissible 128-column tiles for this 128-row block — tiles whose bounding
+// ball can contain an eps-neighbor of the row block's ball (triangle
+// inequality, computed host-side after a coarse-kmeans row permutation).
+// The per-pair eps test below is unchanged, so results are exact.
 // ---------------------------------------------------------------------------
 
 __global__ __launch_bounds__(256) void dbscan_sweep_kernel(
@@ -228,6 +235,8 @@ __global__ __launch_bounds__(256) void dbscan_sweep_kernel(
     float eps2, int mode,
     const uint8_t* __restrict__ core,  // [n] (mode 1)
     const int32_t* __restrict__ labels,// [n] (mode 1)
+    const int32_t* __restrict__ tile_idx, // CSR col-tile ids (nullptr = dense)
+    const int32_t* __restrict__ tile_off, // [n_row_blocks+1] (nullptr = dense)
     int32_t* __restrict__ out) {       // [n_rows]
   __shared__ float lds_x[KM_BK][KM_BM + 1];
   __shared__ float lds_c[KM_BK][KM_BN + 1];
@@ -256,7 +265,11 @@ __global__ __launch_bounds__(256) void dbscan_sweep_kernel(
 #pragma unroll
     for (int r = 0; r < 16; ++r) racc[m][r] = (mode == 0) ? 0 : 0x7fffffff;
 
-  for (int j0 = 0; j0 < n; j0 += KM_BN) {
+  const bool pruned = (tile_off != nullptr);
+  const int t_beg = pruned ? tile_off[blockIdx.x] : 0;
+  const int t_end = pruned ? tile_off[blockIdx.x + 1] : (n + KM_BN - 1) / KM_BN;
+  for (int t = t_beg; t < t_end; ++t) {
+    const int j0 = (pruned ? tile_idx[t] : t) * KM_BN;
     const bool full_cols = (j0 + KM_BN <= n);
     f32x16 acc[2][2];
 #pragma unroll
@@ -2423,7 +2436,8 @@ static inline hipStream_t cur_stream() {
 
 torch::Tensor dbscan_sweep(torch::Tensor X, torch::Tensor x_sq, int64_t row0,
                            int64_t n_rows, double eps2, int64_t mode,
-                           torch::Tensor core, torch::Tensor labels) {
+                           torch::Tensor core, torch::Tensor labels,
+                           torch::Tensor tile_idx, torch::Tensor tile_off) {
   TORCH_CHECK(X.is_cuda() && x_sq.is_cuda(), "device tensors required");
   TORCH_CHECK(X.dtype() == torch::kFloat32 && X.is_contiguous());
   const int64_t n = X.size(0);
@@ -2436,11 +2450,19 @@ torch::Tensor dbscan_sweep(torch::Tensor X, torch::Tensor x_sq, int64_t row0,
     TORCH_CHECK(labels.dtype() == torch::kInt32 && labels.is_contiguous());
   }
   const int grid = (int)((n_rows + KM_BM - 1) / KM_BM);
+  const bool pruned = tile_off.numel() > 0;
+  if (pruned) {
+    TORCH_CHECK(tile_idx.dtype() == torch::kInt32 && tile_idx.is_contiguous());
+    TORCH_CHECK(tile_off.dtype() == torch::kInt32 && tile_off.is_contiguous());
+    TORCH_CHECK(tile_off.numel() == grid + 1, "tile_off must have one entry per row block + 1");
+  }
   hipLaunchKernelGGL(dbscan_sweep_kernel, dim3(grid), dim3(256), 0, cur_stream(),
                      X.data_ptr<float>(), x_sq.data_ptr<float>(), (int)n, d,
                      (int)row0, (int)n_rows, (float)eps2, (int)mode,
                      mode == 1 ? core.data_ptr<uint8_t>() : nullptr,
                      mode == 1 ? labels.data_ptr<int32_t>() : nullptr,
+                     pruned ? tile_idx.data_ptr<int32_t>() : nullptr,
+                     pruned ? tile_off.data_ptr<int32_t>() : nullptr,
                      out.data_ptr<int32_t>());
   return out;
 }

@@ -85,8 +85,19 @@ def test_cluster_hip_logic_with_fake_kernel(monkeypatch):
 
     class FakeExt:
         @staticmethod
-        def dbscan_sweep(Xf, x_sq, row0, n_rows, eps2, mode, core, labels):
+        def dbscan_sweep(Xf, x_sq, row0, n_rows, eps2, mode, core, labels,
+                         tile_idx, tile_off):
+            n = Xf.shape[0]
             d2 = torch.cdist(Xf[row0 : row0 + n_rows], Xf) ** 2
+            if tile_off.numel():
+                # honor the pruned tile list: columns outside the admissible
+                # tiles are invisible, exactly as in the HIP kernel — wrong
+                # lists would therefore produce wrong labels below
+                allowed = torch.zeros((n_rows, n), dtype=torch.bool)
+                for b in range((n_rows + 127) // 128):
+                    for c in tile_idx[tile_off[b] : tile_off[b + 1]].tolist():
+                        allowed[b * 128 : (b + 1) * 128, c * 128 : (c + 1) * 128] = True
+                d2 = torch.where(allowed, d2, torch.full_like(d2, float("inf")))
             if mode == 0:
                 return (d2 <= eps2).sum(dim=1).to(torch.int32)
             masked = torch.where(
@@ -104,14 +115,54 @@ def test_cluster_hip_logic_with_fake_kernel(monkeypatch):
     from sklearn.cluster import DBSCAN as SkDBSCAN
     from sklearn.metrics import adjusted_rand_score
 
-    model = DBSCAN(eps=0.6, min_samples=5).fit(DataFrame.from_numpy(X))
-    # call the hip path directly (CPU tensors + fake kernel)
-    from spark_rapids_ml_amd.parallel.context import PartitionDescriptor, get_comm
+    from spark_rapids_ml_amd.parallel.context import get_comm
 
     comm = get_comm()
     Xf = torch.from_numpy(X.astype(np.float32))
-    labels = model._cluster_hip(
-        Xf, (Xf * Xf).sum(dim=1), 0, len(X), comm, 0.6 * 0.6, 5
-    )
     sk = SkDBSCAN(eps=0.6, min_samples=5).fit(X)
-    assert adjusted_rand_score(sk.labels_, labels) > 0.95
+    for algo in ("brute", "rbc"):
+        model = DBSCAN(eps=0.6, min_samples=5, algorithm=algo).fit(
+            DataFrame.from_numpy(X)
+        )
+        # call the hip path directly (CPU tensors + fake kernel)
+        labels = model._cluster_hip(
+            Xf, (Xf * Xf).sum(dim=1), 0, len(X), comm, 0.6 * 0.6, 5
+        )
+        assert adjusted_rand_score(sk.labels_, labels) > 0.95, algo
+        assert np.array_equal(labels == -1, sk.labels_ == -1), algo
+
+
+def test_dbscan_rbc_tile_lists_conservative():
+    """Every pair within eps must land in an admissible (row-block,
+    column-tile) pair — the triangle-inequality bound may keep extra tiles
+    but must never drop a real neighbor."""
+    import torch
+
+    from spark_rapids_ml_amd.models.clustering import (
+        _coarse_assign,
+        _tile_lists,
+        _tile_stats,
+    )
+
+    rng = np.random.default_rng(3)
+    X, _ = _data(n=2000, seed=3)
+    Xf = torch.from_numpy(X.astype(np.float32))
+    perm = torch.argsort(_coarse_assign(Xf, 16, iters=2), stable=True)
+    Xp = Xf[perm].contiguous()
+    eps = 1.0
+    tile_idx, tile_off, kept = _tile_lists(Xp, 0, len(X), eps)
+    assert 0.0 < kept <= 1.0
+    nb = (len(X) + 127) // 128
+    assert tile_off.shape[0] == nb + 1
+    admissible = set()
+    for b in range(nb):
+        for c in tile_idx[tile_off[b] : tile_off[b + 1]].tolist():
+            admissible.add((b, int(c)))
+    d2 = torch.cdist(Xp, Xp) ** 2
+    ii, jj = torch.nonzero(d2 <= eps * eps, as_tuple=True)
+    for i, j in zip(ii.tolist(), jj.tolist()):
+        assert (i // 128, j // 128) in admissible
+    # on clustered data the bound actually prunes
+    c, r = _tile_stats(Xp)
+    assert c.shape == (nb, X.shape[1]) and r.shape == (nb,)
+    assert kept < 1.0

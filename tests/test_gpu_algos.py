@@ -108,6 +108,32 @@ def test_dbscan_gpu_matches_sklearn():
     assert adjusted_rand_score(np.asarray(out["prediction"]), sk.labels_) == 1.0
 
 
+def test_dbscan_gpu_rbc_matches_brute():
+    """algorithm="rbc" (coarse-permuted ball-cover tile pruning) must give
+    exactly the same clustering as the dense sweep — the pruned HIP kernel
+    runs the same per-pair eps test over a conservative tile superset."""
+    from sklearn.cluster import DBSCAN as SkDBSCAN
+    from sklearn.metrics import adjusted_rand_score
+
+    from spark_rapids_ml_amd import DBSCAN
+
+    X, _ = make_blobs(
+        n_samples=30000, n_features=16, centers=12, cluster_std=0.5, random_state=1
+    )
+    X = X.astype(np.float32)
+    df = DataFrame.from_numpy(X)
+    brute = np.asarray(
+        DBSCAN(eps=1.5, min_samples=5, algorithm="brute").fit(df).transform(df)["prediction"]
+    )
+    rbc = np.asarray(
+        DBSCAN(eps=1.5, min_samples=5, algorithm="rbc").fit(df).transform(df)["prediction"]
+    )
+    assert adjusted_rand_score(brute, rbc) == 1.0
+    assert np.array_equal(brute == -1, rbc == -1)
+    sk = SkDBSCAN(eps=1.5, min_samples=5).fit(X)
+    assert adjusted_rand_score(rbc, sk.labels_) == 1.0
+
+
 def test_ann_gpu_recall():
     from sklearn.neighbors import NearestNeighbors as SkNN
 

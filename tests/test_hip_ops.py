@@ -196,7 +196,9 @@ def test_dbscan_sweep_matches_torch():
     # mode 0: neighbor counts
     empty_u8 = torch.empty(0, dtype=torch.uint8, device=dev)
     empty_i32 = torch.empty(0, dtype=torch.int32, device=dev)
-    counts = ext.dbscan_sweep(X, x_sq, 0, n, eps2, 0, empty_u8, empty_i32)
+    counts = ext.dbscan_sweep(
+        X, x_sq, 0, n, eps2, 0, empty_u8, empty_i32, empty_i32, empty_i32
+    )
     ref_counts = (d2 <= eps2).sum(dim=1).to(torch.int32)
     mism = (counts != ref_counts).sum().item()
     assert mism <= n * 0.003, f"{mism} count mismatches"  # eps-boundary f32 ties
@@ -205,7 +207,9 @@ def test_dbscan_sweep_matches_torch():
     labels = torch.arange(n, dtype=torch.int32, device=dev)
     labels[core == 0] = torch.iinfo(torch.int32).max
     row0, n_rows = 1000, 1500
-    got = ext.dbscan_sweep(X, x_sq, row0, n_rows, eps2, 1, core, labels)
+    got = ext.dbscan_sweep(
+        X, x_sq, row0, n_rows, eps2, 1, core, labels, empty_i32, empty_i32
+    )
     BIG = torch.iinfo(torch.int32).max
     masked = torch.where(
         (core[None, :] > 0) & (d2[row0 : row0 + n_rows] <= eps2),

@@ -101,6 +101,10 @@ def main():
     out = DBSCAN(eps=4.0, min_samples=5).fit(dfc2).transform(dfc2)
     done("dbscan_1m_64_20blobs", st,
          n_clusters=int(np.asarray(out["prediction"]).max()) + 1)
+    st = t0()
+    out = DBSCAN(eps=4.0, min_samples=5, algorithm="rbc").fit(dfc2).transform(dfc2)
+    done("dbscan_1m_64_20blobs_rbc", st,
+         n_clusters=int(np.asarray(out["prediction"]).max()) + 1)
 
     # sparse logreg 50M x 2048 @ 1% (BASELINE.json config)
     del Xd
