@@ -59,6 +59,69 @@ def test_dbscan_distributed_matches_sklearn():
     assert adjusted_rand_score(labels, sk.labels_) == 1.0
 
 
+class _FakeSweepExt:
+    """torch reference semantics of the dbscan_sweep kernel, honoring the
+    pruned tile list (shared by the single- and multi-rank fake tests)."""
+
+    @staticmethod
+    def dbscan_sweep(Xf, x_sq, row0, n_rows, eps2, mode, core, labels,
+                     tile_idx, tile_off):
+        import torch
+
+        BIG = torch.iinfo(torch.int32).max
+        n = Xf.shape[0]
+        d2 = torch.cdist(Xf[row0 : row0 + n_rows], Xf) ** 2
+        if tile_off.numel():
+            allowed = torch.zeros((n_rows, n), dtype=torch.bool)
+            for b in range((n_rows + 127) // 128):
+                for c in tile_idx[tile_off[b] : tile_off[b + 1]].tolist():
+                    allowed[b * 128 : (b + 1) * 128, c * 128 : (c + 1) * 128] = True
+            d2 = torch.where(allowed, d2, torch.full_like(d2, float("inf")))
+        if mode == 0:
+            return (d2 <= eps2).sum(dim=1).to(torch.int32)
+        masked = torch.where(
+            (core[None, :] > 0) & (d2 <= eps2),
+            labels[None, :].expand(n_rows, Xf.shape[0]),
+            torch.full((1,), BIG, dtype=torch.int32),
+        )
+        return masked.min(dim=1).values
+
+
+def _dist_dbscan_rbc(seed: int):
+    """Exercise the rbc path's DISTRIBUTED logic (perm broadcast from rank
+    0, balanced permuted slices, allgather + unpermute) on 2 gloo ranks with
+    the fake kernel."""
+    import torch
+
+    import spark_rapids_ml_amd.ops.dispatch as dispatch
+    from spark_rapids_ml_amd.parallel.context import get_comm
+
+    dispatch.hip_ops = lambda: _FakeSweepExt
+    comm = get_comm()
+    # n must clear the rbc n>=4096 gate in _cluster_hip
+    X, _ = _data(n=5000, seed=seed)
+    n = len(X)
+    base, rem = divmod(n, comm.world_size)
+    n_local = base + (1 if comm.rank < rem else 0)
+    off = comm.rank * base + min(comm.rank, rem)
+    Xf = torch.from_numpy(X.astype(np.float32))  # replicated, original order
+    model = DBSCAN(eps=1.0, min_samples=5, algorithm="rbc").fit(
+        DataFrame.from_numpy(X[off : off + n_local])
+    )
+    return model._cluster_hip(
+        Xf, (Xf * Xf).sum(dim=1), off, n_local, comm, 1.0, 5
+    )
+
+
+def test_dbscan_distributed_rbc_matches_sklearn():
+    results = run_distributed(_dist_dbscan_rbc, world_size=2, args=(7,))
+    X, _ = _data(n=5000, seed=7)
+    labels = np.concatenate(results)  # contiguous rank slices, original order
+    sk = SkDBSCAN(eps=1.0, min_samples=5).fit(X)
+    assert adjusted_rand_score(labels, sk.labels_) == 1.0
+    assert np.array_equal(labels == -1, sk.labels_ == -1)
+
+
 def test_dbscan_model_persistence(tmp_model_path):
     from spark_rapids_ml_amd.models.clustering import DBSCANModel
 
