@@ -127,13 +127,14 @@ class Comm:
         return [o[: counts[r]].to(t.device) for r, o in enumerate(out)]
 
     def broadcast(self, t: torch.Tensor, src: int = 0) -> torch.Tensor:
-        if self.initialized:
-            if t.is_contiguous():
-                dist.broadcast(t, src=src)
-            else:
-                ct = t.contiguous()
-                dist.broadcast(ct, src=src)
-                t.copy_(ct)
+        """In-place broadcast, staged on the backend's device (gloo wants
+        CPU buffers even when t lives on the GPU, same as allreduce_t)."""
+        if not self.initialized:
+            return t
+        ct = self.to_coll(t).contiguous()
+        dist.broadcast(ct, src=src)
+        if ct.data_ptr() != t.data_ptr():
+            t.copy_(ct.to(t.device))
         return t
 
     def broadcast_obj(self, obj: Any, src: int = 0) -> Any:

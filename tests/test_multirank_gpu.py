@@ -108,3 +108,37 @@ def test_dbscan_two_ranks_device():
     assert len(set(res[0].tolist())) == 1
     assert len(set(res[1].tolist())) == 1
     assert res[0][0] != res[1][0]
+
+
+def _dist_dbscan_rbc_gpu(_):
+    from spark_rapids_ml_amd import DBSCAN
+    from spark_rapids_ml_amd.data import DataFrame
+    from spark_rapids_ml_amd.parallel.context import get_comm
+
+    comm = get_comm()
+    rng = np.random.default_rng(0)
+    # n=6000 clears the rbc n>=4096 gate; 4 well-separated blobs
+    C = rng.normal(scale=12.0, size=(4, 16)).astype(np.float32)
+    X = (C[rng.integers(0, 4, 6000)]
+         + 0.3 * rng.normal(size=(6000, 16)).astype(np.float32))
+    sl = slice(comm.rank * 3000, (comm.rank + 1) * 3000)
+    out = {}
+    for algo in ("brute", "rbc"):
+        model = DBSCAN(eps=3.0, min_samples=5, algorithm=algo).fit(
+            DataFrame.from_numpy(X[sl])
+        )
+        out[algo] = np.asarray(
+            model.transform(DataFrame.from_numpy(X[sl]))["prediction"]
+        )
+    return out
+
+
+def test_dbscan_rbc_two_ranks_device():
+    """The rbc path's distributed pieces (perm broadcast, balanced permuted
+    slices, allreduce-min label sweeps, unpermute) with device tensors."""
+    res = run_distributed(_dist_dbscan_rbc_gpu, world_size=2, args=(None,))
+    for r in range(2):
+        assert np.array_equal(res[r]["brute"], res[r]["rbc"]), r
+    # 4 clusters globally, consistent labels across ranks
+    all_brute = np.concatenate([res[0]["brute"], res[1]["brute"]])
+    assert len(set(all_brute.tolist()) - {-1}) == 4
