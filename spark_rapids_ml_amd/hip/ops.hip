@@ -1252,6 +1252,57 @@ __global__ __launch_bounds__(256) void segment_sum_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// label_accumulate_lds: direct one-pass scatter for small k*d. Each block
+// owns an LDS-private [k,d] f32 accumulator (+ k counts), streams its row
+// range in NATURAL order — X read once, coalesced, no sort and no permuted
+// gather — and atomically merges into global once at the end (same
+// privatization pattern as csr_grad). Used when k*d*4 <= 120 KB: on the
+// BASELINE k=200 d=128 config the sort+segment path's permuted row gather
+// ran at ~0.4 TB/s (128 ms for 51 GB); this pass is a plain streaming read.
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(512) void label_accumulate_lds_kernel(
+    const float* __restrict__ X, const int32_t* __restrict__ labels,
+    int64_t n, int d, int k, int64_t rows_per_block,
+    float* __restrict__ sums, float* __restrict__ counts) {
+  extern __shared__ float lacc[];  // [k*d] sums then [k] counts
+  float* lcnt = lacc + (size_t)k * d;
+  const int kd = k * d;
+  for (int i = threadIdx.x; i < kd + k; i += blockDim.x) lacc[i] = 0.0f;
+  __syncthreads();
+  const int64_t rs = (int64_t)blockIdx.x * rows_per_block;
+  const int64_t re = min(n, rs + rows_per_block);
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int nw = (int)(blockDim.x >> 6);
+  if ((d & 1) == 0) {
+    // float2 row reads: d=128 -> one 512 B wave access covers the row
+    for (int64_t r = rs + wave; r < re; r += nw) {
+      const float2* row = reinterpret_cast<const float2*>(X + r * (int64_t)d);
+      float* dst = lacc + (size_t)labels[r] * d;
+      for (int c2 = lane; c2 < (d >> 1); c2 += 64) {
+        float2 v = row[c2];
+        atomicAdd(&dst[2 * c2], v.x);
+        atomicAdd(&dst[2 * c2 + 1], v.y);
+      }
+      if (lane == 0) atomicAdd(&lcnt[labels[r]], 1.0f);
+    }
+  } else {
+    for (int64_t r = rs + wave; r < re; r += nw) {
+      const float* row = X + r * (int64_t)d;
+      float* dst = lacc + (size_t)labels[r] * d;
+      for (int c = lane; c < d; c += 64) atomicAdd(&dst[c], row[c]);
+      if (lane == 0) atomicAdd(&lcnt[labels[r]], 1.0f);
+    }
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < kd; i += blockDim.x)
+    if (lacc[i] != 0.0f) atomicAdd(&sums[i], lacc[i]);
+  for (int i = threadIdx.x; i < k; i += blockDim.x)
+    if (lcnt[i] != 0.0f) atomicAdd(&counts[i], lcnt[i]);
+}
+
+// ---------------------------------------------------------------------------
 // gram_f32: C[d,d] = A^T A for A [n,d] row-major (TN MFMA GEMM).
 // Upper blocks only would halve work; full matrix kept for simplicity —
 // the symmetric skip is a planned optimization.
@@ -2510,8 +2561,22 @@ std::vector<torch::Tensor> kmeans_assign(torch::Tensor X, torch::Tensor C,
   TORCH_CHECK(X.is_contiguous() && C.is_contiguous());
   const int64_t n = X.size(0);
   const int d = (int)X.size(1);
-  const int k = (int)C.size(0);
+  const int k_real = (int)C.size(0);
   auto c_sq = (C * C).sum(1);
+  // Pad centers to a KM_BN multiple: a partial center tile routes BOTH the
+  // X and C stages of that tile through the guarded per-element load path
+  // (the de-pipelining trap in profiles/README.md) — at k=200 that is half
+  // of all tiles and cost ~2x (209 ms vs ~100 ms on 100M x 128). Pad rows
+  // carry c_sq = FLT_MAX so they can never win the argmin.
+  const int k = (int)((k_real + KM_BN - 1) / KM_BN * KM_BN);
+  if (k != k_real) {
+    auto Cp = torch::zeros({(int64_t)k, (int64_t)d}, C.options());
+    Cp.narrow(0, 0, k_real).copy_(C);
+    auto cp = torch::full({(int64_t)k}, 3.0e38f, c_sq.options());
+    cp.narrow(0, 0, k_real).copy_(c_sq);
+    C = Cp.contiguous();
+    c_sq = cp.contiguous();
+  }
   auto labels = torch::empty({n}, X.options().dtype(torch::kInt32));
   auto min_dists = torch::empty({n}, X.options());
   auto inertia = torch::zeros({1}, X.options().dtype(torch::kFloat64));
@@ -2565,7 +2630,20 @@ std::vector<torch::Tensor> label_accumulate(torch::Tensor X, torch::Tensor label
   const int d = (int)X.size(1);
   auto sums = torch::zeros({k, d}, X.options());
   auto counts = torch::zeros({k}, X.options());
-  if (n > 0) {
+  if (n == 0) return {sums, counts};
+  const int64_t kd4 = k * (int64_t)d * 4;
+  if (kd4 + k * 4 <= 120 * 1024 && labels.dtype() == torch::kInt32) {
+    const unsigned grid = (unsigned)std::min<int64_t>(1024, (n + 511) / 512 + 1);
+    const int64_t rows_per_block = (n + grid - 1) / grid;
+    const size_t lds = (size_t)kd4 + (size_t)k * 4;
+    hipLaunchKernelGGL(label_accumulate_lds_kernel, dim3(grid), dim3(512), lds,
+                       cur_stream(), X.data_ptr<float>(),
+                       labels.data_ptr<int32_t>(), n, d, (int)k,
+                       rows_per_block, sums.data_ptr<float>(),
+                       counts.data_ptr<float>());
+    return {sums, counts};
+  }
+  {
     auto sorted = labels.to(torch::kInt64).sort();
     auto perm = std::get<1>(sorted).contiguous();
     auto sl = std::get<0>(sorted).contiguous();

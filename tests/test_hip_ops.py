@@ -42,7 +42,15 @@ def test_kmeans_assign_matches_ref(ext, n, d, k):
     assert abs(float(inertia.item()) - ref_inertia) / max(1.0, abs(ref_inertia)) < 1e-4
 
 
-@pytest.mark.parametrize("n,d,k", [(4096, 128, 100), (1000, 65, 7)])
+@pytest.mark.parametrize(
+    "n,d,k",
+    [
+        (4096, 128, 100),    # LDS-privatized scatter path (k*d*4 <= 120KB, d even)
+        (1000, 65, 7),       # LDS path, odd d (scalar loads)
+        (100000, 128, 200),  # BASELINE km100m shape (LDS path)
+        (2048, 3000, 40),    # k*d*4 > 120KB -> sort+segment path
+    ],
+)
 def test_label_accumulate_matches_ref(ext, n, d, k):
     X = _rand(n, d)
     labels = torch.randint(0, k, (n,), dtype=torch.int32).cuda()
