@@ -1276,8 +1276,39 @@ __global__ __launch_bounds__(512) void label_accumulate_lds_kernel(
   const int wave = threadIdx.x >> 6;
   const int nw = (int)(blockDim.x >> 6);
   if ((d & 1) == 0) {
-    // float2 row reads: d=128 -> one 512 B wave access covers the row
-    for (int64_t r = rs + wave; r < re; r += nw) {
+    // float2 row reads (d=128 -> one 512 B wave access covers the row),
+    // 4 rows in flight so the row/label loads overlap the LDS atomics
+    int64_t r = rs + wave;
+    for (; r + 3 * (int64_t)nw < re; r += 4 * (int64_t)nw) {
+      const int l0 = labels[r], l1 = labels[r + nw];
+      const int l2 = labels[r + 2 * nw], l3 = labels[r + 3 * nw];
+      const float2* r0 = reinterpret_cast<const float2*>(X + r * (int64_t)d);
+      const float2* r1 = reinterpret_cast<const float2*>(X + (r + nw) * (int64_t)d);
+      const float2* r2 = reinterpret_cast<const float2*>(X + (r + 2 * nw) * (int64_t)d);
+      const float2* r3 = reinterpret_cast<const float2*>(X + (r + 3 * nw) * (int64_t)d);
+      float* d0 = lacc + (size_t)l0 * d;
+      float* d1 = lacc + (size_t)l1 * d;
+      float* d2 = lacc + (size_t)l2 * d;
+      float* d3 = lacc + (size_t)l3 * d;
+      for (int c2 = lane; c2 < (d >> 1); c2 += 64) {
+        float2 v0 = r0[c2], v1 = r1[c2], v2 = r2[c2], v3 = r3[c2];
+        atomicAdd(&d0[2 * c2], v0.x);
+        atomicAdd(&d0[2 * c2 + 1], v0.y);
+        atomicAdd(&d1[2 * c2], v1.x);
+        atomicAdd(&d1[2 * c2 + 1], v1.y);
+        atomicAdd(&d2[2 * c2], v2.x);
+        atomicAdd(&d2[2 * c2 + 1], v2.y);
+        atomicAdd(&d3[2 * c2], v3.x);
+        atomicAdd(&d3[2 * c2 + 1], v3.y);
+      }
+      if (lane == 0) {
+        atomicAdd(&lcnt[l0], 1.0f);
+        atomicAdd(&lcnt[l1], 1.0f);
+        atomicAdd(&lcnt[l2], 1.0f);
+        atomicAdd(&lcnt[l3], 1.0f);
+      }
+    }
+    for (; r < re; r += nw) {
       const float2* row = reinterpret_cast<const float2*>(X + r * (int64_t)d);
       float* dst = lacc + (size_t)labels[r] * d;
       for (int c2 = lane; c2 < (d >> 1); c2 += 64) {
@@ -2071,6 +2102,53 @@ __global__ __launch_bounds__(256) void kmeans_argmin_kn_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// kmeans_argmin_nk: argmin epilogue over a ROW-major [m, k] dot block
+// (dots = X_chunk @ C^T — the tall-skinny GEMM hipBLASLt runs at full rate,
+// unlike the [k, n] layout whose skinny-m GEMM stalls at small k). One wave
+// per row: lanes stride k contiguously (coalesced), then a shuffle
+// argmin-reduce; block accumulates inertia with one atomic.
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(256) void kmeans_argmin_nk_kernel(
+    const float* __restrict__ dots,  // [m, k] row-major
+    const float* __restrict__ x_sq,  // [m]
+    const float* __restrict__ c_sq,  // [k]
+    int64_t m, int k,
+    int32_t* __restrict__ labels,    // [m]
+    float* __restrict__ min_d,       // [m]
+    double* __restrict__ inertia) {  // [1]
+  __shared__ double block_in[4];
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int64_t row = (int64_t)blockIdx.x * 4 + wave;
+  float best = 3.0e38f;
+  int bj = 0;
+  if (row < m) {
+    const float* dr = dots + row * (int64_t)k;
+    for (int j = lane; j < k; j += 64) {
+      const float v = c_sq[j] - 2.0f * dr[j];
+      if (v < best) { best = v; bj = j; }
+    }
+  }
+  for (int off = 32; off > 0; off >>= 1) {
+    const float ob = __shfl_down(best, off, 64);
+    const int oj = __shfl_down(bj, off, 64);
+    if (ob < best) { best = ob; bj = oj; }
+  }
+  double local = 0.0;
+  if (row < m && lane == 0) {
+    const float d2 = fmaxf(best + x_sq[row], 0.0f);
+    labels[row] = bj;
+    min_d[row] = d2;
+    local = (double)d2;
+  }
+  if (lane == 0) block_in[wave] = local;
+  __syncthreads();
+  if (threadIdx.x == 0)
+    atomicAdd(inertia, block_in[0] + block_in[1] + block_in[2] + block_in[3]);
+}
+
+// ---------------------------------------------------------------------------
 // rf_partition: counting-sort rows by batch-local node id -> (perm, seg_off).
 // Replaces the per-batch torch sort + nonzero + gather chain (radix sort was
 // ~8% of RF fit kernel time; reference behavior: cuML's batched node trainer
@@ -2852,6 +2930,25 @@ std::vector<torch::Tensor> kmeans_argmin_kn(torch::Tensor dots, torch::Tensor x_
   return {labels, min_d, inertia};
 }
 
+std::vector<torch::Tensor> kmeans_argmin_nk(torch::Tensor dots, torch::Tensor x_sq,
+                                            torch::Tensor c_sq) {
+  TORCH_CHECK(dots.is_cuda() && dots.dtype() == torch::kFloat32 && dots.is_contiguous());
+  TORCH_CHECK(x_sq.is_contiguous() && c_sq.is_contiguous());
+  const int64_t m = dots.size(0);
+  const int k = (int)dots.size(1);
+  auto labels = torch::empty({m}, dots.options().dtype(torch::kInt32));
+  auto min_d = torch::empty({m}, dots.options());
+  auto inertia = torch::zeros({1}, dots.options().dtype(torch::kFloat64));
+  if (m > 0) {
+    const unsigned grid = (unsigned)((m + 3) / 4);
+    hipLaunchKernelGGL(kmeans_argmin_nk_kernel, dim3(grid), dim3(256), 0,
+                       cur_stream(), dots.data_ptr<float>(), x_sq.data_ptr<float>(),
+                       c_sq.data_ptr<float>(), m, k, labels.data_ptr<int32_t>(),
+                       min_d.data_ptr<float>(), inertia.data_ptr<double>());
+  }
+  return {labels, min_d, inertia};
+}
+
 std::vector<torch::Tensor> rf_partition(torch::Tensor node_of_row, torch::Tensor lut,
                                         int64_t B) {
   TORCH_CHECK(node_of_row.is_cuda() && node_of_row.dtype() == torch::kInt64 &&
@@ -3031,6 +3128,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("csr_col_moments", &csr_col_moments, "per-column sum/sumsq of CSR values");
   m.def("kmeans_assign", &kmeans_assign, "fused MFMA distance + argmin");
   m.def("kmeans_argmin_kn", &kmeans_argmin_kn, "argmin epilogue over a [k,n] GEMM dot block");
+  m.def("kmeans_argmin_nk", &kmeans_argmin_nk, "argmin epilogue over a row-major [m,k] GEMM dot block");
   m.def("label_accumulate", &label_accumulate, "per-center sum/count scatter");
   m.def("gram_f32", &gram_f32, "A^T A via MFMA f32");
   m.def("softmax_residual_loss", &softmax_residual_loss, "fused softmax residual + loss");

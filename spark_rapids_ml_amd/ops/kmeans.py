@@ -37,12 +37,17 @@ def kmeans_assign_reduce(
             x_sq = _xsq(X)
         n, k = X.shape[0], C.shape[0]
         fused = os.environ.get("SRML_KMEANS_VARIANT") == "fused"
-        # GEMM+epilogue wins when the dot block is compute-shaped (large k:
-        # 47.6 vs 74.5 ms at k=1000); at small k the [k, n] dots write/read
-        # traffic dominates and the all-in-one kernel wins (261 vs 280 ms at
-        # k=200 on 100M x 128)
+        # Dispatch by k (measured on 100M x 128 / 1M x 3000, profiles/README):
+        # - large k: [k, n] GEMM + kmeans_argmin_kn (47.6 vs 74.5 ms at
+        #   k=1000 — the dot block is compute-shaped there)
+        # - small k: X @ C^T tall-skinny GEMM + kmeans_argmin_nk wave-per-row
+        #   epilogue (the fused kernel's 2-k-step pipeline never reaches
+        #   steady state at small d, and the [k, n] skinny-m GEMM stalls:
+        #   205 / 280 ms vs 66 ms at k=200)
         if not fused and 384 <= k and k * 4 <= 64 * 1024 and n > 0:
             labels, inertia = _assign_gemm(ext, X, C, x_sq, n, k)
+        elif not fused and k < 384 and n > 0:
+            labels, inertia = _assign_gemm_nk(ext, X, C, x_sq, n, k)
         else:
             labels, _min_d, inertia = ext.kmeans_assign(
                 X.contiguous(), C.contiguous(), x_sq.contiguous()
@@ -80,19 +85,46 @@ def _assign_gemm(
     return labels, inertia
 
 
+def _assign_gemm_nk(
+    ext, X: torch.Tensor, C: torch.Tensor, x_sq: torch.Tensor, n: int, k: int,
+    max_dots_bytes: int = 8 << 30,
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Tall-skinny GEMM (X_chunk @ Cᵀ, row-major [chunk, k]) + wave-per-row
+    argmin epilogue — the small-k assignment path."""
+    c_sq = (C * C).sum(dim=1).contiguous()
+    CT = C.t().contiguous()  # [d, k] so the GEMM B operand is plain N-layout
+    rchunk = max(256, int(max_dots_bytes // max(1, k * 4)))
+    if rchunk >= n:
+        dots = torch.mm(X, CT)
+        labels, _min_d, inertia = ext.kmeans_argmin_nk(dots, x_sq, c_sq)
+        return labels, inertia
+    labels = torch.empty(n, dtype=torch.int32, device=X.device)
+    inertia = torch.zeros(1, dtype=torch.float64, device=X.device)
+    dots = torch.empty((min(rchunk, n), k), dtype=torch.float32, device=X.device)
+    for s in range(0, n, rchunk):
+        e = min(n, s + rchunk)
+        dv = dots if e - s == dots.shape[0] else torch.empty(
+            (e - s, k), dtype=torch.float32, device=X.device
+        )
+        torch.mm(X[s:e], CT, out=dv)
+        lb, _md, it = ext.kmeans_argmin_nk(dv, x_sq[s:e].contiguous(), c_sq)
+        labels[s:e] = lb
+        inertia += it
+    return labels, inertia
+
+
 def kmeans_predict(X: torch.Tensor, C: torch.Tensor) -> torch.Tensor:
     import os
 
     if use_hip(X) and X.dtype == torch.float32:
         ext = hip_ops()
         n, k = X.shape[0], C.shape[0]
-        if (
-            os.environ.get("SRML_KMEANS_VARIANT") != "fused"
-            and 384 <= k
-            and k * 4 <= 64 * 1024
-            and n > 0
-        ):
+        fused = os.environ.get("SRML_KMEANS_VARIANT") == "fused"
+        if not fused and 384 <= k and k * 4 <= 64 * 1024 and n > 0:
             labels, _ = _assign_gemm(ext, X.contiguous(), C, _xsq(X), n, k)
+            return labels
+        if not fused and k < 384 and n > 0:
+            labels, _ = _assign_gemm_nk(ext, X.contiguous(), C, _xsq(X), n, k)
             return labels
         labels, _, _ = ext.kmeans_assign(X.contiguous(), C.contiguous(), _xsq(X))
         return labels

@@ -444,3 +444,25 @@ def test_kmeans_gemm_argmin_matches_fused(ext, n, d, k):
     assert abs(float(inertia_g.item()) - float(inertia_f.item())) <= (
         1e-4 * max(1.0, abs(float(inertia_f.item())))
     )
+
+
+@pytest.mark.parametrize("n,d,k", [(50000, 128, 200), (12345, 64, 37), (7003, 37, 5)])
+def test_kmeans_gemm_argmin_nk_matches_fused(ext, n, d, k):
+    """Tall-skinny GEMM + kmeans_argmin_nk (the small-k path) vs the
+    all-in-one assign kernel."""
+    X = _rand(n, d, seed=31)
+    C = _rand(k, d, seed=32)
+    x_sq = (X * X).sum(dim=1)
+    from spark_rapids_ml_amd.ops.kmeans import _assign_gemm_nk
+
+    labels_g, inertia_g = _assign_gemm_nk(ext, X, C, x_sq, n, k,
+                                          max_dots_bytes=4 << 20)  # force chunking
+    labels_f, _md, inertia_f = ext.kmeans_assign(X, C, x_sq)
+    mism = (labels_g != labels_f)
+    if bool(mism.any()):
+        d_g = ((X[mism] - C[labels_g[mism].long()]) ** 2).sum(1)
+        d_f = ((X[mism] - C[labels_f[mism].long()]) ** 2).sum(1)
+        assert torch.allclose(d_g, d_f, rtol=1e-3, atol=1e-2)
+    assert abs(float(inertia_g.item()) - float(inertia_f.item())) <= (
+        1e-4 * max(1.0, abs(float(inertia_f.item())))
+    )
