@@ -1277,35 +1277,33 @@ __global__ __launch_bounds__(512) void label_accumulate_lds_kernel(
   const int nw = (int)(blockDim.x >> 6);
   if ((d & 1) == 0) {
     // float2 row reads (d=128 -> one 512 B wave access covers the row),
-    // 4 rows in flight so the row/label loads overlap the LDS atomics
+    // 8 rows in flight so enough bytes are outstanding per CU to stream
+    // at bandwidth (1 LDS-limited block/CU = 8 waves; the row/label loads
+    // overlap the LDS atomics)
     int64_t r = rs + wave;
-    for (; r + 3 * (int64_t)nw < re; r += 4 * (int64_t)nw) {
-      const int l0 = labels[r], l1 = labels[r + nw];
-      const int l2 = labels[r + 2 * nw], l3 = labels[r + 3 * nw];
-      const float2* r0 = reinterpret_cast<const float2*>(X + r * (int64_t)d);
-      const float2* r1 = reinterpret_cast<const float2*>(X + (r + nw) * (int64_t)d);
-      const float2* r2 = reinterpret_cast<const float2*>(X + (r + 2 * nw) * (int64_t)d);
-      const float2* r3 = reinterpret_cast<const float2*>(X + (r + 3 * nw) * (int64_t)d);
-      float* d0 = lacc + (size_t)l0 * d;
-      float* d1 = lacc + (size_t)l1 * d;
-      float* d2 = lacc + (size_t)l2 * d;
-      float* d3 = lacc + (size_t)l3 * d;
+    for (; r + 7 * (int64_t)nw < re; r += 8 * (int64_t)nw) {
+      int lab[8];
+      const float2* rp[8];
+      float* dp[8];
+#pragma unroll
+      for (int q = 0; q < 8; ++q) {
+        lab[q] = labels[r + q * (int64_t)nw];
+        rp[q] = reinterpret_cast<const float2*>(X + (r + q * (int64_t)nw) * (int64_t)d);
+        dp[q] = lacc + (size_t)lab[q] * d;
+      }
       for (int c2 = lane; c2 < (d >> 1); c2 += 64) {
-        float2 v0 = r0[c2], v1 = r1[c2], v2 = r2[c2], v3 = r3[c2];
-        atomicAdd(&d0[2 * c2], v0.x);
-        atomicAdd(&d0[2 * c2 + 1], v0.y);
-        atomicAdd(&d1[2 * c2], v1.x);
-        atomicAdd(&d1[2 * c2 + 1], v1.y);
-        atomicAdd(&d2[2 * c2], v2.x);
-        atomicAdd(&d2[2 * c2 + 1], v2.y);
-        atomicAdd(&d3[2 * c2], v3.x);
-        atomicAdd(&d3[2 * c2 + 1], v3.y);
+        float2 v[8];
+#pragma unroll
+        for (int q = 0; q < 8; ++q) v[q] = rp[q][c2];
+#pragma unroll
+        for (int q = 0; q < 8; ++q) {
+          atomicAdd(&dp[q][2 * c2], v[q].x);
+          atomicAdd(&dp[q][2 * c2 + 1], v[q].y);
+        }
       }
       if (lane == 0) {
-        atomicAdd(&lcnt[l0], 1.0f);
-        atomicAdd(&lcnt[l1], 1.0f);
-        atomicAdd(&lcnt[l2], 1.0f);
-        atomicAdd(&lcnt[l3], 1.0f);
+#pragma unroll
+        for (int q = 0; q < 8; ++q) atomicAdd(&lcnt[lab[q]], 1.0f);
       }
     }
     for (; r < re; r += nw) {
@@ -2120,27 +2118,31 @@ __global__ __launch_bounds__(256) void kmeans_argmin_nk_kernel(
   __shared__ double block_in[4];
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
-  const int64_t row = (int64_t)blockIdx.x * 4 + wave;
-  float best = 3.0e38f;
-  int bj = 0;
-  if (row < m) {
+  // grid-stride over rows: the grid is capped (<=4096 blocks) so the
+  // inertia merge is one atomic per BLOCK, not per row — a block-per-row
+  // launch serializes millions of f64 atomics on one word
+  const int64_t wstride = (int64_t)gridDim.x * 4;
+  double local = 0.0;
+  for (int64_t row = (int64_t)blockIdx.x * 4 + wave; row < m; row += wstride) {
     const float* dr = dots + row * (int64_t)k;
+    float best = 3.0e38f;
+    int bj = 0;
     for (int j = lane; j < k; j += 64) {
       const float v = c_sq[j] - 2.0f * dr[j];
       if (v < best) { best = v; bj = j; }
     }
-  }
-  for (int off = 32; off > 0; off >>= 1) {
-    const float ob = __shfl_down(best, off, 64);
-    const int oj = __shfl_down(bj, off, 64);
-    if (ob < best) { best = ob; bj = oj; }
-  }
-  double local = 0.0;
-  if (row < m && lane == 0) {
-    const float d2 = fmaxf(best + x_sq[row], 0.0f);
-    labels[row] = bj;
-    min_d[row] = d2;
-    local = (double)d2;
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) {
+      const float ob = __shfl_down(best, off, 64);
+      const int oj = __shfl_down(bj, off, 64);
+      if (ob < best) { best = ob; bj = oj; }
+    }
+    if (lane == 0) {
+      const float d2 = fmaxf(best + x_sq[row], 0.0f);
+      labels[row] = bj;
+      min_d[row] = d2;
+      local += (double)d2;
+    }
   }
   if (lane == 0) block_in[wave] = local;
   __syncthreads();
@@ -2940,7 +2942,7 @@ std::vector<torch::Tensor> kmeans_argmin_nk(torch::Tensor dots, torch::Tensor x_
   auto min_d = torch::empty({m}, dots.options());
   auto inertia = torch::zeros({1}, dots.options().dtype(torch::kFloat64));
   if (m > 0) {
-    const unsigned grid = (unsigned)((m + 3) / 4);
+    const unsigned grid = (unsigned)std::min<int64_t>(4096, (m + 3) / 4);
     hipLaunchKernelGGL(kmeans_argmin_nk_kernel, dim3(grid), dim3(256), 0,
                        cur_stream(), dots.data_ptr<float>(), x_sq.data_ptr<float>(),
                        c_sq.data_ptr<float>(), m, k, labels.data_ptr<int32_t>(),
