@@ -2713,7 +2713,7 @@ std::vector<torch::Tensor> label_accumulate(torch::Tensor X, torch::Tensor label
   if (n == 0) return {sums, counts};
   const int64_t kd4 = k * (int64_t)d * 4;
   if (kd4 + k * 4 <= 120 * 1024 && labels.dtype() == torch::kInt32) {
-    const unsigned grid = (unsigned)std::min<int64_t>(1024, (n + 511) / 512 + 1);
+    const unsigned grid = (unsigned)std::min<int64_t>(512, (n + 511) / 512 + 1);
     const int64_t rows_per_block = (n + grid - 1) / grid;
     const size_t lds = (size_t)kd4 + (size_t)k * 4;
     hipLaunchKernelGGL(label_accumulate_lds_kernel, dim3(grid), dim3(512), lds,
@@ -2942,7 +2942,7 @@ std::vector<torch::Tensor> kmeans_argmin_nk(torch::Tensor dots, torch::Tensor x_
   auto min_d = torch::empty({m}, dots.options());
   auto inertia = torch::zeros({1}, dots.options().dtype(torch::kFloat64));
   if (m > 0) {
-    const unsigned grid = (unsigned)std::min<int64_t>(4096, (m + 3) / 4);
+    const unsigned grid = (unsigned)std::min<int64_t>(16384, (m + 3) / 4);
     hipLaunchKernelGGL(kmeans_argmin_nk_kernel, dim3(grid), dim3(256), 0,
                        cur_stream(), dots.data_ptr<float>(), x_sq.data_ptr<float>(),
                        c_sq.data_ptr<float>(), m, k, labels.data_ptr<int32_t>(),
