@@ -216,13 +216,15 @@ class BenchDBSCAN(BenchmarkBase):
     def add_arguments(self, ap):
         ap.add_argument("--eps", type=float, default=3.0)
         ap.add_argument("--min_samples", type=int, default=5)
+        ap.add_argument("--algorithm", default="brute", choices=["brute", "rbc"])
 
     def make_data(self, args):
         X, _ = gen_data.gen_blobs(args.num_rows, args.num_cols, cluster_std=0.5, seed=args.seed)
         return DataFrame.from_numpy(X)
 
     def make_estimator(self, args):
-        return DBSCAN(eps=args.eps, min_samples=args.min_samples)
+        return DBSCAN(eps=args.eps, min_samples=args.min_samples,
+                      algorithm=args.algorithm)
 
 
 class BenchUMAP(BenchmarkBase):
