@@ -153,3 +153,64 @@ def test_model_copy_with_extra_params():
     assert m.getOrDefault("predictionCol") == "prediction"  # original untouched
     assert np.allclose(m2.cluster_centers_, m.cluster_centers_)
     assert "p2" in m2.transform(DataFrame.from_numpy(X)).columns
+
+
+def test_assign_dispatch_gates(monkeypatch):
+    """The k-gated assignment dispatch (measured on MI355X, ops/kmeans.py):
+    [k,n] GEMM + argmin_kn for 384 <= k (fits LDS), tall-skinny GEMM +
+    argmin_nk for k < 384, fused kernel only via SRML_KMEANS_VARIANT."""
+    import torch
+
+    import spark_rapids_ml_amd.ops.kmeans as km
+
+    calls = []
+
+    class FakeExt:
+        @staticmethod
+        def kmeans_assign(X, C, x_sq):
+            calls.append("fused")
+            lab, _s, _c, inertia = km.torch_ref.kmeans_assign_reduce(X, C, x_sq)
+            return lab.to(torch.int32), torch.zeros_like(x_sq), torch.tensor([inertia])
+
+        @staticmethod
+        def kmeans_argmin_kn(dots, x_sq, c_sq):
+            calls.append("kn")
+            d2 = x_sq[None, :] + c_sq[:, None] - 2.0 * dots
+            md, lab = d2.min(dim=0)
+            return lab.to(torch.int32), md, md.clamp(min=0).double().sum().reshape(1)
+
+        @staticmethod
+        def kmeans_argmin_nk(dots, x_sq, c_sq):
+            calls.append("nk")
+            d2 = x_sq[:, None] + c_sq[None, :] - 2.0 * dots
+            md, lab = d2.min(dim=1)
+            return lab.to(torch.int32), md, md.clamp(min=0).double().sum().reshape(1)
+
+        @staticmethod
+        def label_accumulate(X, labels, k):
+            sums = torch.zeros(k, X.shape[1])
+            sums.index_add_(0, labels.long(), X)
+            cnt = torch.bincount(labels.long(), minlength=k).float()
+            return sums, cnt
+
+    monkeypatch.setattr(km, "hip_ops", lambda: FakeExt)
+    monkeypatch.setattr(km, "use_hip", lambda X: True)
+
+    g = torch.Generator().manual_seed(0)
+    X = torch.randn(500, 16, generator=g)
+    for k, expect in [(8, "nk"), (383, "nk"), (384, "kn"), (512, "kn")]:
+        calls.clear()
+        C = torch.randn(k, 16, generator=g)
+        labels, sums, counts, inertia = km.kmeans_assign_reduce(X, C)
+        assert calls[0] == expect, (k, calls)
+        ref_lab, ref_sums, ref_cnt, ref_in = km.torch_ref.kmeans_assign_reduce(
+            X, C, (X * X).sum(dim=1)
+        )
+        assert (labels.long() == ref_lab.long()).float().mean() > 0.99
+        assert abs(float(inertia) - float(ref_in)) < 1e-3 * max(1.0, abs(float(ref_in)))
+
+    # fused stays selectable
+    monkeypatch.setenv("SRML_KMEANS_VARIANT", "fused")
+    calls.clear()
+    km.kmeans_assign_reduce(X, torch.randn(384, 16, generator=g))
+    assert calls[0] == "fused"
