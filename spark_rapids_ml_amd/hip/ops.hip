@@ -1275,37 +1275,67 @@ __global__ __launch_bounds__(512) void label_accumulate_lds_kernel(
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int nw = (int)(blockDim.x >> 6);
-  if ((d & 1) == 0) {
-    // float2 row reads (d=128 -> one 512 B wave access covers the row),
-    // 8 rows in flight so enough bytes are outstanding per CU to stream
-    // at bandwidth (1 LDS-limited block/CU = 8 waves; the row/label loads
-    // overlap the LDS atomics)
+  if ((d & 1) == 0 && d <= 128) {
+    // float2 row reads (d=128 -> one 512 B wave access covers the row).
+    // Two-deep software pipeline over 8-row groups: the ISA of the naive
+    // form showed 8 SERIALIZED uniform label loads (staggered vmcnt(7..0))
+    // and no cross-iteration overlap — ~2 full HBM latencies per group per
+    // wave. Here the 8 labels arrive in ONE 8-lane load (broadcast by
+    // shuffles) and the NEXT group's label+row loads are issued before the
+    // current group's waits, so every address (a function of r alone)
+    // pipelines.
+    const int d2i = d >> 1;
+    const bool act = lane < d2i;
+    const int64_t gs = 8 * (int64_t)nw;
     int64_t r = rs + wave;
-    for (; r + 7 * (int64_t)nw < re; r += 8 * (int64_t)nw) {
-      int lab[8];
-      const float2* rp[8];
-      float* dp[8];
+    int labv = (lane < 8 && r + (int64_t)lane * nw < re)
+                   ? labels[r + (int64_t)lane * nw] : 0;
+    float2 va[8], vb[8];
+#pragma unroll
+    for (int q = 0; q < 8; ++q) {
+      const int64_t rr = r + (int64_t)q * nw;
+      va[q] = (act && rr < re)
+                  ? reinterpret_cast<const float2*>(X + rr * (int64_t)d)[lane]
+                  : float2{0.0f, 0.0f};
+    }
+    for (; r + 7 * (int64_t)nw < re;) {
+      const int64_t rn = r + gs;
+      int labn = 0;
+      if (rn + 7 * (int64_t)nw < re) {
+        labn = (lane < 8) ? labels[rn + (int64_t)lane * nw] : 0;
+#pragma unroll
+        for (int q = 0; q < 8; ++q)
+          vb[q] = act ? reinterpret_cast<const float2*>(
+                            X + (rn + (int64_t)q * nw) * (int64_t)d)[lane]
+                      : float2{0.0f, 0.0f};
+      }
 #pragma unroll
       for (int q = 0; q < 8; ++q) {
-        lab[q] = labels[r + q * (int64_t)nw];
-        rp[q] = reinterpret_cast<const float2*>(X + (r + q * (int64_t)nw) * (int64_t)d);
-        dp[q] = lacc + (size_t)lab[q] * d;
-      }
-      for (int c2 = lane; c2 < (d >> 1); c2 += 64) {
-        float2 v[8];
-#pragma unroll
-        for (int q = 0; q < 8; ++q) v[q] = rp[q][c2];
-#pragma unroll
-        for (int q = 0; q < 8; ++q) {
-          atomicAdd(&dp[q][2 * c2], v[q].x);
-          atomicAdd(&dp[q][2 * c2 + 1], v[q].y);
+        const int lq = __shfl(labv, q, 64);
+        float* dst = lacc + (size_t)lq * d;
+        if (act) {
+          atomicAdd(&dst[2 * lane], va[q].x);
+          atomicAdd(&dst[2 * lane + 1], va[q].y);
         }
+        if (lane == 0) atomicAdd(&lcnt[lq], 1.0f);
       }
-      if (lane == 0) {
+      r = rn;
+      labv = labn;
 #pragma unroll
-        for (int q = 0; q < 8; ++q) atomicAdd(&lcnt[lab[q]], 1.0f);
-      }
+      for (int q = 0; q < 8; ++q) va[q] = vb[q];
     }
+    for (; r < re; r += nw) {
+      const float2* row = reinterpret_cast<const float2*>(X + r * (int64_t)d);
+      float* dst = lacc + (size_t)labels[r] * d;
+      for (int c2 = lane; c2 < d2i; c2 += 64) {
+        float2 v = row[c2];
+        atomicAdd(&dst[2 * c2], v.x);
+        atomicAdd(&dst[2 * c2 + 1], v.y);
+      }
+      if (lane == 0) atomicAdd(&lcnt[labels[r]], 1.0f);
+    }
+  } else if ((d & 1) == 0) {
+    int64_t r = rs + wave;
     for (; r < re; r += nw) {
       const float2* row = reinterpret_cast<const float2*>(X + r * (int64_t)d);
       float* dst = lacc + (size_t)labels[r] * d;
