@@ -17,7 +17,6 @@ Run: python -m spark_rapids_ml_amd.connect_server --port 8571
 
 import argparse
 import importlib
-import json
 import uuid
 from typing import Any, Dict, Optional
 

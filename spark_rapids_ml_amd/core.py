@@ -22,7 +22,6 @@ from __future__ import annotations
 
 import json
 import os
-import pickle
 from abc import abstractmethod
 from typing import Any, Dict, Iterator, List, Optional, Sequence, Tuple, Union
 

@@ -11,7 +11,6 @@ from __future__ import annotations
 
 import os
 import shutil
-import sys
 
 HERE = os.path.dirname(os.path.abspath(__file__))
 

@@ -5,7 +5,6 @@ so ranks only exchange tiny dictionaries."""
 
 from __future__ import annotations
 
-import math
 from typing import Dict, Optional, Tuple
 
 

@@ -30,14 +30,13 @@ from ..params import (
     HasIDCol,
     HasPredictionCol,
     Param,
-    Params,
     TypeConverters,
     HasWeightCol,
 )
 from ..data import to_device_tensor
 from ..ops import kmeans_assign_reduce, kmeans_predict
 from ..ops.torch_ref import pairwise_sq_dists
-from ..utils import as_numpy, get_logger
+from ..utils import as_numpy
 
 
 class _KMeansParams(HasFeaturesCol, HasFeaturesCols, HasPredictionCol, HasWeightCol):

@@ -35,7 +35,7 @@ from ..params import (
     DictTypeConverters,
 )
 from ..ops.knn import knn_topk
-from ..utils import as_numpy, get_logger
+from ..utils import as_numpy
 
 
 class _NNParams(HasFeaturesCol, HasFeaturesCols, HasIDCol):

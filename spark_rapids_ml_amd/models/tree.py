@@ -36,7 +36,7 @@ from ..params import (
     TypeConverters,
     HasWeightCol,
 )
-from ..utils import as_numpy, get_logger
+from ..utils import as_numpy
 
 
 class _RandomForestParams(HasFeaturesCol, HasFeaturesCols, HasLabelCol, HasPredictionCol, HasWeightCol):

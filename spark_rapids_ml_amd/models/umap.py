@@ -28,7 +28,7 @@ from ..params import (
     TypeConverters,
 )
 from ..ops.knn import knn_topk
-from ..utils import as_numpy, get_logger
+from ..utils import as_numpy
 
 
 def find_ab_params(spread: float = 1.0, min_dist: float = 0.1) -> Tuple[float, float]:

@@ -10,7 +10,6 @@ from __future__ import annotations
 
 import importlib
 import os
-import sys
 from typing import Any, Optional
 
 import torch

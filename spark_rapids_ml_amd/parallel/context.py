@@ -23,12 +23,10 @@ Design notes vs the reference:
 from __future__ import annotations
 
 import datetime
-import json
 import os
 from dataclasses import dataclass
-from typing import Any, List, Optional, Sequence
+from typing import Any, List, Optional
 
-import numpy as np
 import torch
 import torch.distributed as dist
 
