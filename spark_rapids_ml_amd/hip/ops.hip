@@ -1286,27 +1286,30 @@ __global__ __launch_bounds__(512) void label_accumulate_lds_kernel(
     // pipelines.
     const int d2i = d >> 1;
     const bool act = lane < d2i;
-    const int64_t gs = 8 * (int64_t)nw;
-    int64_t r = rs + wave;
-    int labv = (lane < 8 && r + (int64_t)lane * nw < re)
-                   ? labels[r + (int64_t)lane * nw] : 0;
+    // each wave streams a CONTIGUOUS row slice: an 8-row group is one 4 KB
+    // burst (the interleaved wave-stride layout issued 8 scattered 512 B
+    // segments per group and capped at ~790 GB/s)
+    const int64_t span = (re - rs + nw - 1) / nw;
+    int64_t r = min(re, rs + (int64_t)wave * span);
+    const int64_t we = min(re, r + span);
+    int labv = (lane < 8 && r + lane < we) ? labels[r + lane] : 0;
     float2 va[8], vb[8];
 #pragma unroll
     for (int q = 0; q < 8; ++q) {
-      const int64_t rr = r + (int64_t)q * nw;
-      va[q] = (act && rr < re)
+      const int64_t rr = r + q;
+      va[q] = (act && rr < we)
                   ? reinterpret_cast<const float2*>(X + rr * (int64_t)d)[lane]
                   : float2{0.0f, 0.0f};
     }
-    for (; r + 7 * (int64_t)nw < re;) {
-      const int64_t rn = r + gs;
+    for (; r + 7 < we;) {
+      const int64_t rn = r + 8;
       int labn = 0;
-      if (rn + 7 * (int64_t)nw < re) {
-        labn = (lane < 8) ? labels[rn + (int64_t)lane * nw] : 0;
+      if (rn + 7 < we) {
+        labn = (lane < 8) ? labels[rn + lane] : 0;
 #pragma unroll
         for (int q = 0; q < 8; ++q)
           vb[q] = act ? reinterpret_cast<const float2*>(
-                            X + (rn + (int64_t)q * nw) * (int64_t)d)[lane]
+                            X + (rn + q) * (int64_t)d)[lane]
                       : float2{0.0f, 0.0f};
       }
 #pragma unroll
@@ -1324,7 +1327,7 @@ __global__ __launch_bounds__(512) void label_accumulate_lds_kernel(
 #pragma unroll
       for (int q = 0; q < 8; ++q) va[q] = vb[q];
     }
-    for (; r < re; r += nw) {
+    for (; r < we; ++r) {
       const float2* row = reinterpret_cast<const float2*>(X + r * (int64_t)d);
       float* dst = lacc + (size_t)labels[r] * d;
       for (int c2 = lane; c2 < d2i; c2 += 64) {
