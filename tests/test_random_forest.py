@@ -178,3 +178,34 @@ def test_rf_threshold_equality_consistent_with_training():
     pred = np.asarray(m.transform(df)[m.getOrDefault("predictionCol")])
     acc = float((pred == y).mean())
     assert acc == 1.0, f"train accuracy {acc} < 1.0 on separable integer data"
+
+
+def test_estimators_per_worker_split():
+    """Tree division across ranks (reference tree.py:330-341: floor division
+    with the remainder spread over the first ranks)."""
+    from spark_rapids_ml_amd.models.tree import _estimators_per_worker
+
+    assert _estimators_per_worker(10, 2) == [5, 5]
+    assert _estimators_per_worker(10, 3) == [4, 3, 3]
+    assert _estimators_per_worker(2, 4) == [1, 1, 0, 0]
+    assert _estimators_per_worker(7, 1) == [7]
+    assert sum(_estimators_per_worker(500, 8)) == 500
+
+
+def test_resolve_max_features_strategies():
+    """featureSubsetStrategy resolution (reference tree.py:93-135 mapping;
+    Spark semantics: auto = sqrt for classification, onethird for
+    regression)."""
+    from spark_rapids_ml_amd.models.tree import _resolve_max_features
+
+    d = 100
+    assert _resolve_max_features("all", d, "classification") == d
+    assert _resolve_max_features("sqrt", d, "classification") == 10
+    assert _resolve_max_features("log2", d, "classification") == 6
+    assert _resolve_max_features("onethird", d, "regression") == 33
+    assert _resolve_max_features("auto", d, "classification") == 10
+    assert _resolve_max_features("auto", d, "regression") == 33
+    assert _resolve_max_features("0.5", d, "regression") == 50
+    assert _resolve_max_features("25", d, "classification") == 25
+    # results are clamped to [1, d]
+    assert _resolve_max_features("0.0001", d, "regression") >= 1
